@@ -1,0 +1,515 @@
+"""GCS — the cluster control plane process.
+
+Feature counterpart of the reference's gcs_server (src/ray/gcs/gcs_server.h:99):
+node membership, actor lifecycle FSM + placement (gcs/actor/gcs_actor_manager.h:94),
+placement groups with 2-phase reserve/commit (gcs_placement_group_manager.h:51),
+namespaced KV (gcs_kv_manager.cc), job ids, and actor-state watch
+(pubsub). Re-designed as one asyncio process over the msgpack RPC layer.
+"""
+from __future__ import annotations
+
+import asyncio
+import time
+import traceback
+from typing import Any, Dict, List, Optional
+
+from . import ids
+from .protocol import RpcClient, RpcServer
+
+ACTOR_PENDING = "PENDING_CREATION"
+ACTOR_ALIVE = "ALIVE"
+ACTOR_RESTARTING = "RESTARTING"
+ACTOR_DEAD = "DEAD"
+
+
+class NodeInfo:
+    def __init__(self, node_id: bytes, addr: str, resources: Dict[str, float], labels):
+        self.node_id = node_id
+        self.addr = addr
+        self.resources_total = dict(resources)
+        self.resources_available = dict(resources)
+        self.labels = labels or {}
+        self.alive = True
+        self.client: Optional[RpcClient] = None
+        self.start_time = time.time()
+
+
+class ActorInfo:
+    def __init__(self, actor_id: bytes, spec: dict):
+        self.actor_id = actor_id
+        self.spec = spec
+        self.name = spec.get("name")
+        self.namespace = spec.get("namespace", "default")
+        self.state = ACTOR_PENDING
+        self.addr: Optional[str] = None
+        self.node_id: Optional[bytes] = None
+        self.num_restarts = 0
+        self.death_cause: Optional[str] = None
+        self.waiters: List[asyncio.Future] = []
+
+
+class PlacementGroupInfo:
+    def __init__(self, pg_id: bytes, bundles: List[dict], strategy: str, name: str):
+        self.pg_id = pg_id
+        self.bundles = bundles
+        self.strategy = strategy
+        self.name = name
+        self.state = "PENDING"
+        self.bundle_nodes: List[Optional[bytes]] = [None] * len(bundles)
+        self.waiters: List[asyncio.Future] = []
+
+
+class GcsServer:
+    def __init__(self, sock_path: str):
+        self.sock_path = sock_path
+        self.server = RpcServer()
+        self.kv: Dict[str, Dict[bytes, bytes]] = {}
+        self.nodes: Dict[bytes, NodeInfo] = {}
+        self.actors: Dict[bytes, ActorInfo] = {}
+        self.named_actors: Dict[tuple, bytes] = {}
+        self.pgs: Dict[bytes, PlacementGroupInfo] = {}
+        self.job_counter = 0
+        self._proto_node: Dict[int, bytes] = {}  # id(proto) -> node_id
+        for m in (
+            "kv_put kv_get kv_del kv_keys kv_exists register_node node_table "
+            "report_resources register_actor resolve_actor actor_exit "
+            "kill_actor list_actors next_job_id create_pg pg_wait_ready "
+            "remove_pg pg_table ping timeline_events drain_node"
+        ).split():
+            self.server.route(m, getattr(self, "h_" + m))
+        self.server.on_conn_lost = self._conn_lost
+        self._timeline: List[dict] = []
+
+    async def start(self):
+        await self.server.start_unix(self.sock_path)
+
+    # ---------- KV ----------
+    def h_kv_put(self, conn, p):
+        ns = self.kv.setdefault(p.get("ns", ""), {})
+        key = p["key"]
+        exists = key in ns
+        if p.get("overwrite", True) or not exists:
+            ns[key] = p["value"]
+            return not exists
+        return False
+
+    def h_kv_get(self, conn, p):
+        return self.kv.get(p.get("ns", ""), {}).get(p["key"])
+
+    def h_kv_del(self, conn, p):
+        return self.kv.get(p.get("ns", ""), {}).pop(p["key"], None) is not None
+
+    def h_kv_exists(self, conn, p):
+        return p["key"] in self.kv.get(p.get("ns", ""), {})
+
+    def h_kv_keys(self, conn, p):
+        prefix = p.get("prefix", b"")
+        return [k for k in self.kv.get(p.get("ns", ""), {}) if k.startswith(prefix)]
+
+    # ---------- nodes ----------
+    async def h_register_node(self, conn, p):
+        node_id = p["node_id"]
+        info = NodeInfo(node_id, p["addr"], p["resources"], p.get("labels"))
+        client = RpcClient()
+        await client.connect(p["addr"])
+        info.client = client
+        self.nodes[node_id] = info
+        self._proto_node[id(conn)] = node_id
+        return {"ok": True}
+
+    def h_node_table(self, conn, p):
+        return [
+            {
+                "node_id": n.node_id,
+                "addr": n.addr,
+                "alive": n.alive,
+                "resources_total": n.resources_total,
+                "resources_available": n.resources_available,
+                "labels": n.labels,
+            }
+            for n in self.nodes.values()
+        ]
+
+    def h_report_resources(self, conn, p):
+        n = self.nodes.get(p["node_id"])
+        if n is not None:
+            n.resources_available = p["available"]
+
+    def h_ping(self, conn, p):
+        return "pong"
+
+    def h_timeline_events(self, conn, p):
+        evs = p.get("events")
+        if evs:
+            self._timeline.extend(evs)
+            return len(self._timeline)
+        return self._timeline
+
+    def _conn_lost(self, proto, exc):
+        node_id = self._proto_node.pop(id(proto), None)
+        if node_id is not None and node_id in self.nodes:
+            asyncio.ensure_future(self._on_node_death(node_id))
+
+    async def h_drain_node(self, conn, p):
+        await self._on_node_death(p["node_id"])
+        return True
+
+    async def _on_node_death(self, node_id: bytes):
+        n = self.nodes.get(node_id)
+        if n is None or not n.alive:
+            return
+        n.alive = False
+        for a in list(self.actors.values()):
+            if a.node_id == node_id and a.state == ACTOR_ALIVE:
+                await self._on_actor_exit(a, "node died", expected=False)
+
+    # ---------- scheduling helpers ----------
+    def _alive_nodes(self) -> List[NodeInfo]:
+        return [n for n in self.nodes.values() if n.alive]
+
+    def _fits(self, node: NodeInfo, req: Dict[str, float]) -> bool:
+        for k, v in req.items():
+            if v > 0 and node.resources_available.get(k, 0.0) + 1e-9 < v:
+                return False
+        return True
+
+    def _pick_node(self, req: Dict[str, float], strategy: str = "hybrid",
+                   exclude=(), soft_affinity: Optional[bytes] = None) -> Optional[NodeInfo]:
+        """Hybrid policy (reference: policy/hybrid_scheduling_policy.h:28):
+        prefer the least-loaded feasible node; SPREAD picks round-robin."""
+        cands = [n for n in self._alive_nodes() if n.node_id not in exclude and self._fits(n, req)]
+        if not cands:
+            return None
+        if soft_affinity is not None:
+            for n in cands:
+                if n.node_id == soft_affinity:
+                    return n
+
+        def load(n: NodeInfo):
+            t = n.resources_total.get("CPU", 1.0) or 1.0
+            return 1.0 - n.resources_available.get("CPU", 0.0) / t
+
+        cands.sort(key=load)
+        return cands[0]
+
+    # ---------- actors ----------
+    async def h_register_actor(self, conn, p):
+        actor_id = p["actor_id"]
+        a = ActorInfo(actor_id, p)
+        if a.name:
+            key = (a.namespace, a.name)
+            if key in self.named_actors:
+                other = self.actors.get(self.named_actors[key])
+                if other is not None and other.state != ACTOR_DEAD:
+                    if p.get("get_if_exists"):
+                        return {"existing": self.named_actors[key]}
+                    raise ValueError(f"actor name {a.name!r} already taken")
+            self.named_actors[key] = actor_id
+        self.actors[actor_id] = a
+        asyncio.ensure_future(self._schedule_actor(a))
+        return {"existing": None}
+
+    async def _schedule_actor(self, a: ActorInfo):
+        req = dict(a.spec.get("resources", {}))
+        deadline = time.time() + 300.0
+        while a.state in (ACTOR_PENDING, ACTOR_RESTARTING):
+            node = self._pick_node(
+                req,
+                strategy=a.spec.get("scheduling_strategy", "hybrid"),
+                soft_affinity=a.spec.get("pg_node"),
+            )
+            if node is None:
+                if time.time() > deadline:
+                    await self._fail_actor(a, "resources unavailable for actor")
+                    return
+                await asyncio.sleep(0.05)
+                continue
+            try:
+                r = await node.client.call(
+                    "start_actor", {"actor_id": a.actor_id, "spec": a.spec}
+                )
+                a.node_id = node.node_id
+                a.addr = r["addr"]
+                a.state = ACTOR_ALIVE
+                for k, v in req.items():
+                    node.resources_available[k] = node.resources_available.get(k, 0) - v
+                for f in a.waiters:
+                    if not f.done():
+                        f.set_result(None)
+                a.waiters.clear()
+                return
+            except Exception:
+                traceback.print_exc()
+                if time.time() > deadline:
+                    await self._fail_actor(a, "actor start failed:\n" + traceback.format_exc())
+                    return
+                await asyncio.sleep(0.2)
+
+    async def _fail_actor(self, a: ActorInfo, cause: str):
+        a.state = ACTOR_DEAD
+        a.death_cause = cause
+        for f in a.waiters:
+            if not f.done():
+                f.set_result(None)
+        a.waiters.clear()
+
+    async def h_resolve_actor(self, conn, p):
+        aid = p.get("actor_id")
+        if aid is None:
+            key = (p.get("namespace", "default"), p["name"])
+            aid = self.named_actors.get(key)
+            if aid is None:
+                return {"state": "NOT_FOUND"}
+        a = self.actors.get(aid)
+        if a is None:
+            return {"state": "NOT_FOUND"}
+        if p.get("wait") and a.state in (ACTOR_PENDING, ACTOR_RESTARTING):
+            fut = asyncio.get_running_loop().create_future()
+            a.waiters.append(fut)
+            try:
+                await asyncio.wait_for(fut, p.get("timeout", 120.0))
+            except asyncio.TimeoutError:
+                pass
+        return {
+            "state": a.state,
+            "actor_id": aid,
+            "addr": a.addr,
+            "node_id": a.node_id,
+            "death_cause": a.death_cause,
+            "spec_kv_key": a.spec.get("spec_kv_key"),
+        }
+
+    async def h_actor_exit(self, conn, p):
+        a = self.actors.get(p["actor_id"])
+        if a is None:
+            return
+        await self._on_actor_exit(a, p.get("cause", "actor exited"), p.get("expected", True))
+
+    async def _on_actor_exit(self, a: ActorInfo, cause: str, expected: bool):
+        node = self.nodes.get(a.node_id) if a.node_id else None
+        if node is not None and a.state == ACTOR_ALIVE:
+            for k, v in a.spec.get("resources", {}).items():
+                node.resources_available[k] = node.resources_available.get(k, 0) + v
+        if (not expected) and a.num_restarts < a.spec.get("max_restarts", 0):
+            a.num_restarts += 1
+            a.state = ACTOR_RESTARTING
+            a.addr = None
+            asyncio.ensure_future(self._schedule_actor(a))
+        else:
+            a.state = ACTOR_DEAD
+            a.death_cause = cause
+            if a.name and self.named_actors.get((a.namespace, a.name)) == a.actor_id:
+                del self.named_actors[(a.namespace, a.name)]
+            for f in a.waiters:
+                if not f.done():
+                    f.set_result(None)
+            a.waiters.clear()
+
+    async def h_kill_actor(self, conn, p):
+        a = self.actors.get(p["actor_id"])
+        if a is None or a.state == ACTOR_DEAD:
+            return False
+        if a.addr and a.state == ACTOR_ALIVE:
+            node = self.nodes.get(a.node_id)
+            if node is not None and node.client is not None:
+                try:
+                    await node.client.call(
+                        "kill_worker", {"addr": a.addr, "no_restart": p.get("no_restart", True)}
+                    )
+                except Exception:
+                    pass
+        if p.get("no_restart", True):
+            a.spec["max_restarts"] = 0
+        await self._on_actor_exit(a, "ray.kill", expected=p.get("no_restart", True))
+        return True
+
+    def h_list_actors(self, conn, p):
+        out = []
+        for a in self.actors.values():
+            out.append(
+                {
+                    "actor_id": a.actor_id,
+                    "class_name": a.spec.get("class_name"),
+                    "name": a.name,
+                    "namespace": a.namespace,
+                    "state": a.state,
+                    "node_id": a.node_id,
+                    "pid": a.spec.get("pid"),
+                    "num_restarts": a.num_restarts,
+                }
+            )
+        return out
+
+    def h_next_job_id(self, conn, p):
+        self.job_counter += 1
+        return self.job_counter
+
+    # ---------- placement groups (2PC: reserve on each raylet, commit) ----------
+    async def h_create_pg(self, conn, p):
+        pg = PlacementGroupInfo(p["pg_id"], p["bundles"], p["strategy"], p.get("name", ""))
+        self.pgs[pg.pg_id] = pg
+        asyncio.ensure_future(self._schedule_pg(pg))
+        return {"pg_id": pg.pg_id}
+
+    async def _schedule_pg(self, pg: PlacementGroupInfo):
+        deadline = time.time() + 300.0
+        while pg.state == "PENDING":
+            plan = self._plan_pg(pg)
+            if plan is None:
+                if time.time() > deadline:
+                    pg.state = "INFEASIBLE"
+                    break
+                await asyncio.sleep(0.05)
+                continue
+            reserved = []
+            ok = True
+            for idx, node in enumerate(plan):
+                try:
+                    r = await node.client.call(
+                        "reserve_bundle",
+                        {"pg_id": pg.pg_id, "bundle_index": idx, "resources": pg.bundles[idx]},
+                    )
+                    if not r.get("ok"):
+                        ok = False
+                        break
+                    reserved.append((idx, node))
+                except Exception:
+                    ok = False
+                    break
+            if ok:
+                for idx, node in reserved:
+                    await node.client.call(
+                        "commit_bundle", {"pg_id": pg.pg_id, "bundle_index": idx}
+                    )
+                    pg.bundle_nodes[idx] = node.node_id
+                    for k, v in pg.bundles[idx].items():
+                        node.resources_available[k] = node.resources_available.get(k, 0) - v
+                pg.state = "CREATED"
+                break
+            for idx, node in reserved:
+                try:
+                    await node.client.call(
+                        "rollback_bundle", {"pg_id": pg.pg_id, "bundle_index": idx}
+                    )
+                except Exception:
+                    pass
+            await asyncio.sleep(0.05)
+            if time.time() > deadline:
+                pg.state = "INFEASIBLE"
+                break
+        for f in pg.waiters:
+            if not f.done():
+                f.set_result(None)
+        pg.waiters.clear()
+
+    def _plan_pg(self, pg: PlacementGroupInfo) -> Optional[List[NodeInfo]]:
+        nodes = self._alive_nodes()
+        if not nodes:
+            return None
+        avail = {n.node_id: dict(n.resources_available) for n in nodes}
+
+        def fits(nid, req):
+            a = avail[nid]
+            return all(a.get(k, 0) + 1e-9 >= v for k, v in req.items())
+
+        def take(nid, req):
+            for k, v in req.items():
+                avail[nid][k] = avail[nid].get(k, 0) - v
+
+        plan: List[NodeInfo] = []
+        strat = pg.strategy
+        if strat in ("STRICT_PACK", "PACK"):
+            # try to pack everything on one node first
+            for n in sorted(nodes, key=lambda n: -n.resources_available.get("CPU", 0)):
+                a = dict(avail[n.node_id])
+                if all(self._consume(a, b) for b in pg.bundles):
+                    return [n] * len(pg.bundles)
+            if strat == "STRICT_PACK":
+                return None
+        if strat in ("STRICT_SPREAD", "SPREAD"):
+            used = set()
+            for b in pg.bundles:
+                cand = [
+                    n for n in nodes
+                    if fits(n.node_id, b) and (n.node_id not in used or strat == "SPREAD")
+                ]
+                cand.sort(key=lambda n: (n.node_id in used, -avail[n.node_id].get("CPU", 0)))
+                if not cand:
+                    return None
+                n = cand[0]
+                take(n.node_id, b)
+                used.add(n.node_id)
+                plan.append(n)
+            return plan
+        # PACK fallback / default: greedy best-fit
+        plan = []
+        for b in pg.bundles:
+            cand = [n for n in nodes if fits(n.node_id, b)]
+            if not cand:
+                return None
+            cand.sort(key=lambda n: -avail[n.node_id].get("CPU", 0))
+            take(cand[0].node_id, b)
+            plan.append(cand[0])
+        return plan
+
+    @staticmethod
+    def _consume(avail: Dict[str, float], req: Dict[str, float]) -> bool:
+        if all(avail.get(k, 0) + 1e-9 >= v for k, v in req.items()):
+            for k, v in req.items():
+                avail[k] = avail.get(k, 0) - v
+            return True
+        return False
+
+    async def h_pg_wait_ready(self, conn, p):
+        pg = self.pgs.get(p["pg_id"])
+        if pg is None:
+            raise ValueError("no such placement group")
+        if pg.state == "PENDING":
+            fut = asyncio.get_running_loop().create_future()
+            pg.waiters.append(fut)
+            try:
+                await asyncio.wait_for(fut, p.get("timeout", 120.0))
+            except asyncio.TimeoutError:
+                pass
+        return {"state": pg.state, "bundle_nodes": pg.bundle_nodes}
+
+    async def h_remove_pg(self, conn, p):
+        pg = self.pgs.pop(p["pg_id"], None)
+        if pg is None:
+            return False
+        for idx, nid in enumerate(pg.bundle_nodes):
+            node = self.nodes.get(nid) if nid else None
+            if node is not None and node.alive:
+                try:
+                    await node.client.call(
+                        "remove_bundle", {"pg_id": pg.pg_id, "bundle_index": idx}
+                    )
+                    for k, v in pg.bundles[idx].items():
+                        node.resources_available[k] = node.resources_available.get(k, 0) + v
+                except Exception:
+                    pass
+        return True
+
+    def h_pg_table(self, conn, p):
+        return [
+            {"pg_id": g.pg_id, "name": g.name, "state": g.state, "strategy": g.strategy,
+             "bundles": g.bundles, "bundle_nodes": g.bundle_nodes}
+            for g in self.pgs.values()
+        ]
+
+
+def main():
+    import sys
+
+    sock = sys.argv[1]
+    gcs = GcsServer(sock)
+
+    async def run():
+        await gcs.start()
+        await asyncio.Event().wait()
+
+    asyncio.run(run())
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,603 @@
+"""Raylet — the per-node daemon.
+
+Feature counterpart of the reference NodeManager
+(src/ray/raylet/node_manager.h:146) + local lease manager
+(raylet/scheduling/cluster_lease_manager.h:41) + WorkerPool
+(raylet/worker_pool.h:284) + object-manager pull/push
+(object_manager/object_manager.h:137) + placement-group bundle 2PC
+participant (node_manager.h:621). Hosts the node's shm object store
+table and serves chunked remote pulls.
+"""
+from __future__ import annotations
+
+import asyncio
+import os
+import subprocess
+import sys
+import time
+import traceback
+from collections import deque
+from typing import Dict, List, Optional
+
+from . import ids, store
+from .protocol import RpcClient, RpcServer
+
+CHUNK = 8 * 1024 * 1024  # remote pull chunk (reference uses 5 MiB, ray_config_def.h:421)
+
+
+class WorkerProc:
+    def __init__(self, proc: subprocess.Popen):
+        self.proc = proc
+        self.addr: Optional[str] = None
+        self.proto = None
+        self.pid = proc.pid
+        self.kind = "task"
+        self.actor_id: Optional[bytes] = None
+        self.idle = True
+        self.lease_id: Optional[int] = None
+        self.gpu_ids: List[int] = []
+
+
+class Lease:
+    def __init__(self, lease_id, resources, worker, gpu_ids, pg=None):
+        self.lease_id = lease_id
+        self.resources = resources
+        self.worker: WorkerProc = worker
+        self.gpu_ids = gpu_ids
+        self.pg = pg
+
+
+class Raylet:
+    def __init__(
+        self,
+        session_dir: str,
+        gcs_addr: str,
+        resources: Dict[str, float],
+        node_name: str = "",
+        labels: Optional[dict] = None,
+        object_store_memory: Optional[int] = None,
+    ):
+        self.session_dir = session_dir
+        self.gcs_addr = gcs_addr
+        self.node_id = ids.new_node_id()
+        self.node_name = node_name or self.node_id.hex()[:8]
+        self.resources_total = dict(resources)
+        self.avail = dict(resources)
+        self.labels = labels or {}
+        self.server = RpcServer()
+        self.gcs = RpcClient()
+        self.addr: Optional[str] = None
+
+        shm_dir = os.environ.get("RAY_AMD_SHM_DIR") or os.path.join(session_dir, "shm")
+        spill_dir = os.path.join(session_dir, "spill")
+        cap = object_store_memory or int(
+            os.environ.get("RAY_AMD_OBJECT_STORE_MEMORY", 16 * 2**30)
+        )
+        self.store = store.LocalObjectStore(shm_dir, spill_dir, cap)
+
+        self.workers: Dict[int, WorkerProc] = {}  # pid -> worker
+        self._idle_task_workers: deque = deque()
+        self._starting = 0
+        self._lease_seq = 0
+        self.leases: Dict[int, Lease] = {}
+        self._pending: deque = deque()  # (req payload, future)
+        self._worker_ready: Dict[int, asyncio.Future] = {}  # pid -> fut
+        self._actor_start_futs: Dict[bytes, asyncio.Future] = {}
+        self._proto_worker: Dict[int, WorkerProc] = {}
+        # GPU instance pool
+        ngpus = int(resources.get("GPU", 0))
+        self._free_gpus = list(range(ngpus))
+        # placement-group bundles: (pg_id, idx) -> {"resources", "avail", "committed"}
+        self.bundles: Dict[tuple, dict] = {}
+        self._node_cache: List[dict] = []
+        self._node_cache_time = 0.0
+
+        for m in (
+            "register_worker request_lease return_lease seal_object wait_object "
+            "free_objects pull_object fetch_chunk object_stats start_actor "
+            "actor_ready actor_failed kill_worker reserve_bundle commit_bundle "
+            "rollback_bundle remove_bundle node_info ping prestart_workers "
+            "report_task_events"
+        ).split():
+            self.server.route(m, getattr(self, "h_" + m))
+        self.server.on_conn_lost = self._conn_lost
+        self._task_events: List[dict] = []
+
+    # ---------------- lifecycle ----------------
+
+    async def start(self):
+        sock = os.path.join(self.session_dir, "sock", f"raylet_{self.node_id.hex()[:8]}")
+        await self.server.start_unix(sock)
+        self.addr = "unix:" + sock
+        await self.gcs.connect(self.gcs_addr)
+        await self.gcs.call(
+            "register_node",
+            {
+                "node_id": self.node_id,
+                "addr": self.addr,
+                "resources": self.resources_total,
+                "labels": self.labels,
+            },
+        )
+        n_prestart = int(min(self.resources_total.get("CPU", 0), 8))
+        for _ in range(n_prestart):
+            self._spawn_worker()
+        asyncio.ensure_future(self._resource_reporter())
+
+    async def _resource_reporter(self):
+        last = None
+        while True:
+            snap = dict(self.avail)
+            if snap != last:
+                try:
+                    self.gcs.notify(
+                        "report_resources",
+                        {"node_id": self.node_id, "available": snap},
+                    )
+                    last = snap
+                except Exception:
+                    pass
+            await asyncio.sleep(0.2)
+
+    # ---------------- worker pool ----------------
+
+    def _spawn_worker(self, actor_spec: Optional[dict] = None) -> WorkerProc:
+        env = dict(os.environ)
+        env["RAY_AMD_SESSION_DIR"] = self.session_dir
+        env["RAY_AMD_GCS_ADDR"] = self.gcs_addr
+        env["RAY_AMD_RAYLET_ADDR"] = self.addr
+        env["RAY_AMD_NODE_ID"] = self.node_id.hex()
+        args = [sys.executable, "-m", "ray_amd._core.worker"]
+        if actor_spec is not None:
+            env["RAY_AMD_ACTOR_ID"] = actor_spec["actor_id"].hex()
+            for k, v in (actor_spec.get("env_vars") or {}).items():
+                env[str(k)] = str(v)
+        logdir = os.path.join(self.session_dir, "logs")
+        os.makedirs(logdir, exist_ok=True)
+        tag = (
+            f"actor_{actor_spec['actor_id'].hex()[:8]}"
+            if actor_spec
+            else f"worker_{len(self.workers)}_{os.urandom(2).hex()}"
+        )
+        out = open(os.path.join(logdir, tag + ".log"), "ab", buffering=0)
+        proc = subprocess.Popen(
+            args, env=env, stdout=out, stderr=subprocess.STDOUT,
+            start_new_session=True,
+        )
+        w = WorkerProc(proc)
+        if actor_spec is not None:
+            w.kind = "actor"
+            w.actor_id = actor_spec["actor_id"]
+            w.idle = False
+        else:
+            self._starting += 1
+        self.workers[w.pid] = w
+        return w
+
+    def h_register_worker(self, conn, p):
+        w = self.workers.get(p["pid"])
+        if w is None:
+            # worker we didn't spawn (shouldn't happen)
+            return {"ok": False}
+        w.addr = p["addr"]
+        w.proto = conn
+        self._proto_worker[id(conn)] = w
+        if w.kind == "task":
+            self._starting -= 1
+            self._idle_task_workers.append(w)
+            self._try_grant()
+        fut = self._worker_ready.pop(w.pid, None)
+        if fut is not None and not fut.done():
+            fut.set_result(w)
+        return {"ok": True, "node_id": self.node_id}
+
+    def _conn_lost(self, proto, exc_):
+        w = self._proto_worker.pop(id(proto), None)
+        if w is None:
+            return
+        self.workers.pop(w.pid, None)
+        if w in self._idle_task_workers:
+            try:
+                self._idle_task_workers.remove(w)
+            except ValueError:
+                pass
+        if w.lease_id is not None:
+            lease = self.leases.pop(w.lease_id, None)
+            if lease is not None:
+                self._release_resources(lease)
+        if w.kind == "actor" and w.actor_id is not None:
+            spec_res = getattr(w, "actor_resources", None) or {}
+            for k, v in spec_res.items():
+                if k == "GPU":
+                    continue
+                self.avail[k] = self.avail.get(k, 0) + v
+            for g in w.gpu_ids:
+                self._free_gpus.append(g)
+            if "GPU" in spec_res:
+                self.avail["GPU"] = self.avail.get("GPU", 0) + spec_res["GPU"]
+            rc = w.proc.poll()
+            asyncio.ensure_future(self._notify_actor_exit(w, rc))
+        self._try_grant()
+
+    async def _notify_actor_exit(self, w: WorkerProc, returncode):
+        try:
+            await self.gcs.call(
+                "actor_exit",
+                {
+                    "actor_id": w.actor_id,
+                    "expected": returncode == 0,
+                    "cause": f"actor process exited with code {returncode}",
+                },
+            )
+        except Exception:
+            pass
+
+    def h_prestart_workers(self, conn, p):
+        for _ in range(int(p.get("n", 1))):
+            self._spawn_worker()
+        return True
+
+    # ---------------- leases ----------------
+
+    def _fits(self, avail: Dict[str, float], req: Dict[str, float]) -> bool:
+        return all(avail.get(k, 0.0) + 1e-9 >= v for k, v in req.items() if v > 0)
+
+    def _feasible_total(self, req: Dict[str, float]) -> bool:
+        return all(
+            self.resources_total.get(k, 0.0) + 1e-9 >= v for k, v in req.items() if v > 0
+        )
+
+    async def h_request_lease(self, conn, p):
+        req = dict(p.get("resources") or {})
+        pg = None
+        if p.get("pg_id") is not None:
+            pg = (bytes(p["pg_id"]), p.get("bundle_index"))
+            bkey = self._bundle_key(pg)
+            if bkey is None:
+                return {"error": "placement group bundle not found"}
+        if pg is None and not self._feasible_total(req):
+            spill = await self._find_spill_target(req)
+            if spill:
+                return {"spill": spill}
+            return {"error": f"infeasible resource request {req} on this cluster"}
+        fut = asyncio.get_running_loop().create_future()
+        self._pending.append((req, pg, fut))
+        self._try_grant()
+        return await fut
+
+    def _bundle_key(self, pg):
+        pg_id, idx = pg
+        if idx is not None:
+            key = (pg_id, idx)
+            return key if key in self.bundles else None
+        # any bundle of this pg on this node
+        for key in self.bundles:
+            if key[0] == pg_id:
+                return key
+        return None
+
+    def _try_grant(self):
+        made_progress = True
+        while made_progress and self._pending:
+            made_progress = False
+            req, pg, fut = self._pending[0]
+            if fut.done():
+                self._pending.popleft()
+                made_progress = True
+                continue
+            pool_avail = self.avail
+            bundle = None
+            if pg is not None:
+                bkey = self._bundle_key(pg)
+                if bkey is None:
+                    self._pending.popleft()
+                    fut.set_result({"error": "placement group bundle lost"})
+                    continue
+                bundle = self.bundles[bkey]
+                pool_avail = bundle["avail"]
+            if not self._fits(pool_avail, req):
+                break
+            if not self._idle_task_workers:
+                if self._starting == 0 or len(self._pending) > self._starting:
+                    self._spawn_worker()
+                break
+            w = self._idle_task_workers.popleft()
+            if w.proc.poll() is not None or w.proto is None:
+                made_progress = True
+                continue
+            self._pending.popleft()
+            gpu_ids = []
+            ngpu = int(req.get("GPU", 0))
+            if bundle is None:
+                for k, v in req.items():
+                    self.avail[k] = self.avail.get(k, 0) - v
+                if ngpu:
+                    gpu_ids = [self._free_gpus.pop() for _ in range(min(ngpu, len(self._free_gpus)))]
+            else:
+                for k, v in req.items():
+                    bundle["avail"][k] = bundle["avail"].get(k, 0) - v
+                if ngpu:
+                    gpu_ids = [self._free_gpus.pop() for _ in range(min(ngpu, len(self._free_gpus)))]
+            self._lease_seq += 1
+            lease = Lease(self._lease_seq, req, w, gpu_ids, pg)
+            self.leases[lease.lease_id] = lease
+            w.idle = False
+            w.lease_id = lease.lease_id
+            w.gpu_ids = gpu_ids
+            fut.set_result(
+                {"addr": w.addr, "lease_id": lease.lease_id, "gpu_ids": gpu_ids,
+                 "raylet": self.addr}
+            )
+            made_progress = True
+
+    def _release_resources(self, lease: Lease):
+        if lease.pg is not None:
+            bkey = self._bundle_key(lease.pg)
+            if bkey is not None:
+                b = self.bundles[bkey]
+                for k, v in lease.resources.items():
+                    b["avail"][k] = b["avail"].get(k, 0) + v
+        else:
+            for k, v in lease.resources.items():
+                self.avail[k] = self.avail.get(k, 0) + v
+        for g in lease.gpu_ids:
+            self._free_gpus.append(g)
+
+    def h_return_lease(self, conn, p):
+        lease = self.leases.pop(p["lease_id"], None)
+        if lease is None:
+            return
+        self._release_resources(lease)
+        w = lease.worker
+        if not p.get("dead") and w.pid in self.workers and w.proc.poll() is None:
+            w.idle = True
+            w.lease_id = None
+            self._idle_task_workers.append(w)
+        self._try_grant()
+
+    async def _find_spill_target(self, req) -> Optional[str]:
+        now = time.time()
+        if now - self._node_cache_time > 0.5:
+            try:
+                self._node_cache = await self.gcs.call("node_table", {})
+                self._node_cache_time = now
+            except Exception:
+                return None
+        for n in self._node_cache:
+            if not n["alive"] or n["addr"] == self.addr:
+                continue
+            tot = n["resources_total"]
+            if all(tot.get(k, 0) + 1e-9 >= v for k, v in req.items() if v > 0):
+                return n["addr"]
+        return None
+
+    # ---------------- actors ----------------
+
+    async def h_start_actor(self, conn, p):
+        spec = p["spec"]
+        actor_id = p["actor_id"]
+        spec = dict(spec)
+        spec["actor_id"] = actor_id
+        res = spec.get("resources") or {}
+        if not self._fits(self.avail, res):
+            raise RuntimeError(f"node {self.node_name}: insufficient resources {res}")
+        ngpu = int(res.get("GPU", 0))
+        gpu_ids = [self._free_gpus.pop() for _ in range(min(ngpu, len(self._free_gpus)))]
+        for k, v in res.items():
+            self.avail[k] = self.avail.get(k, 0) - v
+        if gpu_ids:
+            spec.setdefault("env_vars", {})
+            spec["env_vars"] = dict(spec.get("env_vars") or {})
+            ids_str = ",".join(map(str, gpu_ids))
+            spec["env_vars"]["RAY_AMD_GPU_IDS"] = ids_str
+        w = self._spawn_worker(actor_spec=spec)
+        w.gpu_ids = gpu_ids
+        w.actor_resources = res
+        fut = asyncio.get_running_loop().create_future()
+        self._actor_start_futs[actor_id] = fut
+        try:
+            r = await asyncio.wait_for(fut, 120.0)
+        except asyncio.TimeoutError:
+            try:
+                w.proc.kill()
+            except Exception:
+                pass
+            raise RuntimeError("actor start timed out")
+        finally:
+            self._actor_start_futs.pop(actor_id, None)
+        if r.get("error"):
+            raise RuntimeError(r["error"])
+        return {"addr": r["addr"], "pid": w.pid}
+
+    def h_actor_ready(self, conn, p):
+        fut = self._actor_start_futs.get(bytes(p["actor_id"]))
+        if fut is not None and not fut.done():
+            fut.set_result({"addr": p["addr"]})
+        return True
+
+    def h_actor_failed(self, conn, p):
+        fut = self._actor_start_futs.get(bytes(p["actor_id"]))
+        if fut is not None and not fut.done():
+            fut.set_result({"error": p.get("error", "actor init failed")})
+        return True
+
+    async def h_kill_worker(self, conn, p):
+        addr = p["addr"]
+        for w in self.workers.values():
+            if w.addr == addr:
+                try:
+                    w.proc.kill()
+                except Exception:
+                    pass
+                return True
+        return False
+
+    # ---------------- placement-group bundles (2PC participant) ----------------
+
+    def h_reserve_bundle(self, conn, p):
+        res = p["resources"]
+        if not self._fits(self.avail, res):
+            return {"ok": False}
+        for k, v in res.items():
+            self.avail[k] = self.avail.get(k, 0) - v
+        self.bundles[(bytes(p["pg_id"]), p["bundle_index"])] = {
+            "resources": dict(res),
+            "avail": dict(res),
+            "committed": False,
+        }
+        return {"ok": True}
+
+    def h_commit_bundle(self, conn, p):
+        b = self.bundles.get((bytes(p["pg_id"]), p["bundle_index"]))
+        if b is not None:
+            b["committed"] = True
+        return {"ok": b is not None}
+
+    def h_rollback_bundle(self, conn, p):
+        b = self.bundles.pop((bytes(p["pg_id"]), p["bundle_index"]), None)
+        if b is not None:
+            for k, v in b["resources"].items():
+                self.avail[k] = self.avail.get(k, 0) + v
+        return True
+
+    def h_remove_bundle(self, conn, p):
+        return self.h_rollback_bundle(conn, p)
+
+    # ---------------- object store ----------------
+
+    def h_seal_object(self, conn, p):
+        self.store.seal(bytes(p["id"]), p["size"])
+        return {"ok": True}
+
+    async def h_wait_object(self, conn, p):
+        oid = bytes(p["id"])
+        ok = await self.store.wait_sealed(oid, p.get("timeout", 60.0))
+        if ok:
+            self.store.ensure_local(oid)
+        return {"ok": ok, "size": self.store.table.get(oid, [0])[0]}
+
+    def h_free_objects(self, conn, p):
+        self.store.free([bytes(i) for i in p["ids"]])
+
+    async def h_pull_object(self, conn, p):
+        """Pull an object from a remote node's store into ours, chunked
+        (reference: PullManager / object_manager gRPC chunks)."""
+        oid = bytes(p["id"])
+        if self.store.contains(oid):
+            self.store.ensure_local(oid)
+            return {"ok": True}
+        src = p["src"]
+        c = RpcClient()
+        try:
+            await c.connect(src, retries=5)
+            first = await c.call("fetch_chunk", {"id": oid, "off": 0, "len": CHUNK})
+            if not first.get("ok"):
+                return {"ok": False}
+            size = first["size"]
+            w = store.ObjectWriter(self.store.shm_dir, oid, size)
+            data = first["data"]
+            w.view[: len(data)] = data
+            off = len(data)
+            while off < size:
+                r = await c.call("fetch_chunk", {"id": oid, "off": off, "len": CHUNK})
+                if not r.get("ok"):
+                    return {"ok": False}
+                d = r["data"]
+                w.view[off : off + len(d)] = d
+                off += len(d)
+            w.seal()
+            self.store.seal(oid, size)
+            return {"ok": True}
+        except Exception:
+            traceback.print_exc()
+            return {"ok": False}
+        finally:
+            c.close()
+
+    async def h_fetch_chunk(self, conn, p):
+        oid = bytes(p["id"])
+        ok = await self.store.wait_sealed(oid, 30.0)
+        if not ok:
+            return {"ok": False}
+        self.store.ensure_local(oid)
+        path = store.shm_path(self.store.shm_dir, oid)
+        size = self.store.table[oid][0]
+        with open(path, "rb") as f:
+            f.seek(p["off"])
+            data = f.read(p["len"])
+        return {"ok": True, "size": size, "data": data}
+
+    def h_object_stats(self, conn, p):
+        n, used, cap = self.store.stats()
+        return {"num_objects": n, "used": used, "capacity": cap}
+
+    def h_node_info(self, conn, p):
+        return {
+            "node_id": self.node_id,
+            "addr": self.addr,
+            "resources_total": self.resources_total,
+            "resources_available": self.avail,
+            "num_workers": len(self.workers),
+        }
+
+    def h_ping(self, conn, p):
+        return "pong"
+
+    def h_report_task_events(self, conn, p):
+        self._task_events.extend(p.get("events", []))
+        if len(self._task_events) > 100000:
+            del self._task_events[:50000]
+        evs = self._task_events
+        if p.get("fetch"):
+            return evs
+        return len(evs)
+
+    def shutdown_workers(self):
+        for w in self.workers.values():
+            try:
+                w.proc.kill()
+            except Exception:
+                pass
+
+
+def main():
+    import argparse
+    import json
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--session-dir", required=True)
+    ap.add_argument("--gcs", required=True)
+    ap.add_argument("--resources", required=True, help="json dict")
+    ap.add_argument("--node-name", default="")
+    ap.add_argument("--labels", default="{}")
+    ap.add_argument("--object-store-memory", type=int, default=0)
+    ap.add_argument("--ready-file", default="")
+    args = ap.parse_args()
+
+    raylet = Raylet(
+        args.session_dir,
+        args.gcs,
+        json.loads(args.resources),
+        node_name=args.node_name,
+        labels=json.loads(args.labels),
+        object_store_memory=args.object_store_memory or None,
+    )
+
+    async def run():
+        await raylet.start()
+        if args.ready_file:
+            with open(args.ready_file, "w") as f:
+                f.write(raylet.addr + "\n" + raylet.node_id.hex())
+        try:
+            await asyncio.Event().wait()
+        finally:
+            raylet.shutdown_workers()
+
+    try:
+        asyncio.run(run())
+    finally:
+        raylet.shutdown_workers()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,967 @@
+"""CoreRuntime — the per-process runtime embedded in drivers and workers.
+
+Feature counterpart of the reference CoreWorker
+(src/ray/core_worker/core_worker.h:184): object put/get/wait, normal-task
+submission over cached worker leases (normal_task_submitter.cc:34),
+per-actor ordered submission (actor_task_submitter.cc), an in-process
+memory store for small results (memory_store.h:48), ownership-based
+object resolution (GetOwnershipInfo / locate-by-owner), and reference
+counting driving shm frees (reference_counter.h:44).
+
+Architecture here is MI355X-native/new: one asyncio loop per process
+(background thread in drivers, main thread in workers), msgpack RPC,
+shm-file object store, direct worker→worker connections for the actor
+hot path.
+"""
+from __future__ import annotations
+
+import asyncio
+import hashlib
+import os
+import threading
+import time
+import traceback
+from collections import deque
+from typing import Any, Dict, List, Optional, Tuple
+
+from .. import exceptions as exc
+from . import ids, serialization, store
+from .protocol import ConnectionLost, RpcClient, RpcError, RpcServer
+
+_runtime: Optional["CoreRuntime"] = None
+_runtime_lock = threading.Lock()
+
+
+def global_runtime() -> "CoreRuntime":
+    if _runtime is None:
+        raise RuntimeError("ray_amd.init() has not been called")
+    return _runtime
+
+
+def set_global_runtime(rt: Optional["CoreRuntime"]):
+    global _runtime
+    _runtime = rt
+
+
+def is_initialized() -> bool:
+    return _runtime is not None
+
+
+# --------------------------------------------------------------------------
+# ObjectRef
+# --------------------------------------------------------------------------
+
+_local_ref_ctx = threading.local()
+
+
+class ObjectRef:
+    """Handle to a (future) object. Owner-based, like the reference
+    (python/ray/includes/object_ref.pxi:50)."""
+
+    __slots__ = ("id", "owner_addr", "_rt", "__weakref__")
+
+    def __init__(self, oid: bytes, owner_addr: str, _register: bool = True):
+        self.id = oid
+        self.owner_addr = owner_addr
+        self._rt = _runtime
+        if _register and self._rt is not None:
+            self._rt._add_local_ref(oid, owner_addr)
+
+    def hex(self) -> str:
+        return self.id.hex()
+
+    def binary(self) -> bytes:
+        return self.id
+
+    def is_nil(self) -> bool:
+        return not self.id
+
+    def __hash__(self):
+        return hash(self.id)
+
+    def __eq__(self, other):
+        return isinstance(other, ObjectRef) and other.id == self.id
+
+    def __repr__(self):
+        return f"ObjectRef({self.id.hex()})"
+
+    def __reduce__(self):
+        cap = getattr(_local_ref_ctx, "captured", None)
+        if cap is not None:
+            cap.append(self)
+        return (_deserialize_ref, (self.id, self.owner_addr))
+
+    def __del__(self):
+        rt = self._rt
+        if rt is not None and not rt._closed:
+            try:
+                rt._remove_local_ref(self.id)
+            except Exception:
+                pass
+
+    def future(self):
+        import concurrent.futures
+
+        f = concurrent.futures.Future()
+
+        def _done():
+            try:
+                f.set_result(self._rt.get_sync([self], timeout=None)[0])
+            except BaseException as e:  # noqa
+                f.set_exception(e)
+
+        threading.Thread(target=_done, daemon=True).start()
+        return f
+
+
+def _deserialize_ref(oid: bytes, owner_addr: str) -> ObjectRef:
+    return ObjectRef(oid, owner_addr)
+
+
+# --------------------------------------------------------------------------
+# Lease pool (reference: normal_task_submitter.h:87 per-SchedulingKey pools)
+# --------------------------------------------------------------------------
+
+
+class _Lease:
+    __slots__ = ("lease_id", "addr", "client", "busy")
+
+    def __init__(self, lease_id, addr, client):
+        self.lease_id = lease_id
+        self.addr = addr
+        self.client = client
+        self.busy = False
+
+
+class _LeasePool:
+    def __init__(self, key, resources, pg):
+        self.key = key
+        self.resources = resources
+        self.pg = pg  # (pg_id, bundle_index) or None
+        self.leases: List[_Lease] = []
+        self.queue: deque = deque()  # pending task dispatch callables
+        self.requests_in_flight = 0
+
+
+# --------------------------------------------------------------------------
+
+
+class CoreRuntime:
+    def __init__(
+        self,
+        mode: str,
+        session_dir: str,
+        gcs_addr: str,
+        raylet_addr: str,
+        node_id: bytes,
+        loop: Optional[asyncio.AbstractEventLoop] = None,
+    ):
+        self.mode = mode
+        self.session_dir = session_dir
+        self.shm_dir = os.environ.get("RAY_AMD_SHM_DIR") or os.path.join(
+            session_dir, "shm"
+        )
+        self.gcs_addr = gcs_addr
+        self.raylet_addr = raylet_addr
+        self.node_id = node_id
+        self.job_id = 0
+        self._closed = False
+
+        self.loop = loop or asyncio.new_event_loop()
+        self._own_loop_thread: Optional[threading.Thread] = None
+        self._loop_thread_obj: Optional[threading.Thread] = None
+
+        self.server = RpcServer()
+        self.addr: Optional[str] = None
+        self.gcs = RpcClient()
+        self.raylet = RpcClient()
+        self._conns: Dict[str, RpcClient] = {}
+        self._conn_locks: Dict[str, asyncio.Lock] = {}
+
+        # memory store: oid -> tuple(kind, ...)
+        #   ("val", python_value)        deserialized / locally put value
+        #   ("val_ser", bytes)           serialized inline value
+        #   ("store", node_addr, size)   sealed in a node's shm store
+        #   ("err", bytes)               serialized exception
+        self.memory_store: Dict[bytes, tuple] = {}
+        self._events: Dict[bytes, asyncio.Event] = {}
+        self._mmaps: Dict[bytes, store.MappedObject] = {}
+
+        # refcounts: oid -> [local, submitted, owner_addr]
+        self._refs: Dict[bytes, list] = {}
+        self._refs_lock = threading.Lock()
+
+        self._pools: Dict[tuple, _LeasePool] = {}
+        self._fn_exported: Dict[bytes, asyncio.Future] = {}
+        self._fn_cache: Dict[bytes, Any] = {}
+        self._actors: Dict[bytes, dict] = {}
+        self._task_events: List[dict] = []
+        self.worker_id = os.urandom(8)
+
+        self.server.route("fetch_object", self._h_fetch_object)
+        self.server.route("locate_object", self._h_locate_object)
+        self.server.route("ping", lambda c, p: "pong")
+
+    # ------------- lifecycle -------------
+
+    def start_driver(self):
+        """Driver mode: run the loop in a background thread."""
+        ready = threading.Event()
+
+        def run():
+            asyncio.set_event_loop(self.loop)
+            self.loop.call_soon(ready.set)
+            self.loop.run_forever()
+
+        self._own_loop_thread = threading.Thread(
+            target=run, name="ray_amd_driver_loop", daemon=True
+        )
+        self._own_loop_thread.start()
+        self._loop_thread_obj = self._own_loop_thread
+        ready.wait()
+        self._run(self._async_start()).result()
+
+    def set_loop_thread(self, t: threading.Thread):
+        """Worker mode: record which thread runs the loop."""
+        self._loop_thread_obj = t
+
+    async def _async_start(self):
+        sock = os.path.join(
+            self.session_dir, "sock", f"rt_{os.getpid()}_{os.urandom(3).hex()}"
+        )
+        await self.server.start_unix(sock)
+        self.addr = "unix:" + sock
+        await self.gcs.connect(self.gcs_addr)
+        await self.raylet.connect(self.raylet_addr)
+
+    def _run(self, coro):
+        """Schedule a coroutine on the loop from any thread."""
+        return asyncio.run_coroutine_threadsafe(coro, self.loop)
+
+    def _call_sync(self, coro, timeout=None):
+        if threading.current_thread() is self._loop_thread():
+            raise RuntimeError("sync call from event-loop thread")
+        return self._run(coro).result(timeout)
+
+    def _loop_thread(self):
+        return self._loop_thread_obj
+
+    def shutdown(self):
+        if self._closed:
+            return
+        self._closed = True
+        set_global_runtime(None)
+
+        async def _close():
+            try:
+                await self.server.close()
+            except Exception:
+                pass
+            for c in [self.gcs, self.raylet, *self._conns.values()]:
+                try:
+                    c.close()
+                except Exception:
+                    pass
+
+        try:
+            self._run(_close()).result(2)
+        except Exception:
+            pass
+        if self._own_loop_thread is not None:
+            self.loop.call_soon_threadsafe(self.loop.stop)
+            self._own_loop_thread.join(2)
+
+    # ------------- connections -------------
+
+    async def _conn(self, addr: str) -> RpcClient:
+        c = self._conns.get(addr)
+        if c is not None and c.connected:
+            return c
+        lock = self._conn_locks.setdefault(addr, asyncio.Lock())
+        async with lock:
+            c = self._conns.get(addr)
+            if c is not None and c.connected:
+                return c
+            c = RpcClient()
+            await c.connect(addr, retries=10)
+            self._conns[addr] = c
+            return c
+
+    # ------------- reference counting -------------
+
+    def _add_local_ref(self, oid: bytes, owner_addr: str):
+        with self._refs_lock:
+            e = self._refs.get(oid)
+            if e is None:
+                self._refs[oid] = [1, 0, owner_addr]
+            else:
+                e[0] += 1
+
+    def _remove_local_ref(self, oid: bytes):
+        with self._refs_lock:
+            e = self._refs.get(oid)
+            if e is None:
+                return
+            e[0] -= 1
+            if e[0] > 0 or e[1] > 0:
+                return
+            del self._refs[oid]
+            owner = e[2]
+        if owner == self.addr:
+            self._free_owned(oid)
+
+    def _add_submitted_ref(self, oid: bytes):
+        with self._refs_lock:
+            e = self._refs.get(oid)
+            if e is not None:
+                e[1] += 1
+
+    def _remove_submitted_ref(self, oid: bytes):
+        with self._refs_lock:
+            e = self._refs.get(oid)
+            if e is None:
+                return
+            e[1] -= 1
+            if e[0] > 0 or e[1] > 0:
+                return
+            del self._refs[oid]
+            owner = e[2]
+        if owner == self.addr:
+            self._free_owned(oid)
+
+    def _free_owned(self, oid: bytes):
+        ent = self.memory_store.pop(oid, None)
+        self._events.pop(oid, None)
+        self._mmaps.pop(oid, None)
+        if ent is not None and ent[0] == "store" and not self._closed:
+
+            async def _free():
+                try:
+                    c = await self._conn(ent[1])
+                    c.notify("free_objects", {"ids": [oid]})
+                except Exception:
+                    pass
+
+            try:
+                self._run(_free())
+            except Exception:
+                pass
+
+    # ------------- memory store -------------
+
+    def _store_put(self, oid: bytes, entry: tuple):
+        self.memory_store[oid] = entry
+        ev = self._events.pop(oid, None)
+        if ev is not None:
+            ev.set()
+
+    def _store_put_threadsafe(self, oid: bytes, entry: tuple):
+        self.loop.call_soon_threadsafe(self._store_put, oid, entry)
+
+    async def _store_wait(self, oid: bytes, timeout=None) -> tuple:
+        ent = self.memory_store.get(oid)
+        if ent is not None:
+            return ent
+        ev = self._events.get(oid)
+        if ev is None:
+            ev = self._events[oid] = asyncio.Event()
+        try:
+            await asyncio.wait_for(ev.wait(), timeout)
+        except asyncio.TimeoutError:
+            raise exc.GetTimeoutError(
+                f"object {oid.hex()} not ready within {timeout}s"
+            )
+        ent = self.memory_store.get(oid)
+        if ent is None:
+            raise exc.ObjectLostError(oid.hex())
+        return ent
+
+    # ------------- serialization helpers -------------
+
+    def _serialize_capture(self, value) -> Tuple[bytes, list, List[ObjectRef]]:
+        _local_ref_ctx.captured = []
+        try:
+            meta, buffers = serialization.serialize(value)
+            return meta, buffers, _local_ref_ctx.captured
+        finally:
+            _local_ref_ctx.captured = None
+
+    # ------------- put / get / wait -------------
+
+    def put(self, value, _owner=None) -> ObjectRef:
+        oid = ids.new_object_id()
+        meta, buffers, captured = self._serialize_capture(value)
+        size = serialization.serialized_size(meta, buffers)
+        ref = ObjectRef(oid, self.addr)
+        if size <= serialization.INLINE_MAX:
+            blob = bytearray(size)
+            n = serialization.write_to(memoryview(blob), meta, buffers)
+            self._store_put_threadsafe(oid, ("val_ser", bytes(blob[:n])))
+        else:
+            store.put_serialized(self.shm_dir, oid, meta, buffers)
+            self.raylet_seal(oid, size)
+            self._store_put_threadsafe(oid, ("store", self.raylet_addr, size))
+        return ref
+
+    def raylet_seal(self, oid: bytes, size: int):
+        self._call_sync(self.raylet.call("seal_object", {"id": oid, "size": size}))
+
+    def get_sync(self, refs: List[ObjectRef], timeout=None) -> List[Any]:
+        return self._call_sync(self.get_async(refs, timeout))
+
+    async def get_async(self, refs: List[ObjectRef], timeout=None) -> List[Any]:
+        deadline = None if timeout is None else time.monotonic() + timeout
+        out = []
+        for r in refs:
+            rem = None if deadline is None else max(0.0, deadline - time.monotonic())
+            out.append(await self._get_one(r, rem))
+        return out
+
+    async def _get_one(self, ref: ObjectRef, timeout=None):
+        ent = self.memory_store.get(ref.id)
+        if ent is None:
+            if ref.owner_addr == self.addr:
+                ent = await self._store_wait(ref.id, timeout)
+            else:
+                ent = await self._fetch_from_owner(ref, timeout)
+        return await self._materialize(ref.id, ent, timeout)
+
+    async def _materialize(self, oid: bytes, ent: tuple, timeout=None):
+        kind = ent[0]
+        if kind == "val":
+            return ent[1]
+        if kind == "val_ser":
+            value = serialization.loads(ent[1])
+            self.memory_store[oid] = ("val", value)
+            return value
+        if kind == "err":
+            raise serialization.loads(ent[1])
+        if kind == "store":
+            node_addr, size = ent[1], ent[2]
+            path = store.shm_path(self.shm_dir, oid)
+            if not os.path.exists(path):
+                if node_addr == self.raylet_addr:
+                    r = await self.raylet.call(
+                        "wait_object", {"id": oid, "timeout": timeout or 60.0}
+                    )
+                    if not r.get("ok"):
+                        raise exc.ObjectLostError(oid.hex())
+                else:
+                    r = await self.raylet.call(
+                        "pull_object",
+                        {"id": oid, "src": node_addr, "timeout": timeout or 120.0},
+                    )
+                    if not r.get("ok"):
+                        raise exc.ObjectLostError(oid.hex())
+            mo = store.MappedObject(path)
+            self._mmaps[oid] = mo
+            value = serialization.loads_from(mo.view)
+            self.memory_store[oid] = ("val", value)
+            return value
+        raise exc.RaySystemError(f"bad store entry {kind}")
+
+    async def _fetch_from_owner(self, ref: ObjectRef, timeout=None) -> tuple:
+        try:
+            c = await self._conn(ref.owner_addr)
+            r = await asyncio.wait_for(
+                c.call("fetch_object", {"id": ref.id}), timeout
+            )
+        except asyncio.TimeoutError:
+            raise exc.GetTimeoutError(f"fetching {ref.id.hex()} timed out")
+        except (ConnectionLost, ConnectionError, RpcError) as e:
+            raise exc.ObjectLostError(
+                f"{ref.id.hex()} (owner {ref.owner_addr} unreachable: {e})"
+            )
+        if r is None:
+            raise exc.ObjectLostError(ref.id.hex())
+        kind = r["kind"]
+        if kind == "val_ser":
+            ent = ("val_ser", r["data"])
+        elif kind == "err":
+            ent = ("err", r["data"])
+        else:
+            ent = ("store", r["node_addr"], r["size"])
+        self._store_put(ref.id, ent)
+        return ent
+
+    async def _h_fetch_object(self, conn, p):
+        oid = p["id"]
+        ent = self.memory_store.get(oid)
+        if ent is None:
+            with self._refs_lock:
+                known = oid in self._refs
+            if not known and oid not in self._events:
+                return None
+            try:
+                ent = await self._store_wait(oid, 60.0)
+            except Exception:
+                return None
+        kind = ent[0]
+        if kind == "val":
+            return {"kind": "val_ser", "data": serialization.dumps(ent[1])}
+        if kind == "val_ser":
+            return {"kind": "val_ser", "data": ent[1]}
+        if kind == "err":
+            return {"kind": "err", "data": ent[1]}
+        return {"kind": "store", "node_addr": ent[1], "size": ent[2]}
+
+    def _h_locate_object(self, conn, p):
+        ent = self.memory_store.get(p["id"])
+        if ent is None:
+            return None
+        if ent[0] == "store":
+            return {"node_addr": ent[1], "size": ent[2]}
+        return {"inline": True}
+
+    def wait_sync(self, refs, num_returns=1, timeout=None, fetch_local=True):
+        return self._call_sync(
+            self._wait_async(refs, num_returns, timeout, fetch_local)
+        )
+
+    async def _wait_async(self, refs, num_returns, timeout, fetch_local):
+        pending = list(refs)
+        ready: List[ObjectRef] = []
+
+        async def _ready_one(r: ObjectRef):
+            ent = self.memory_store.get(r.id)
+            if ent is None:
+                if r.owner_addr == self.addr:
+                    await self._store_wait(r.id, None)
+                else:
+                    await self._fetch_from_owner(r, None)
+            return r
+
+        tasks = {asyncio.ensure_future(_ready_one(r)): r for r in pending}
+        deadline = None if timeout is None else time.monotonic() + timeout
+        try:
+            while len(ready) < num_returns and tasks:
+                rem = None if deadline is None else max(0, deadline - time.monotonic())
+                done, _ = await asyncio.wait(
+                    tasks.keys(), timeout=rem, return_when=asyncio.FIRST_COMPLETED
+                )
+                if not done:
+                    break
+                for t in done:
+                    r = tasks.pop(t)
+                    ready.append(r)
+                if deadline is not None and time.monotonic() >= deadline:
+                    break
+        finally:
+            for t in tasks:
+                t.cancel()
+        not_ready = [r for r in refs if r not in ready]
+        ready = ready[: max(num_returns, 0)]
+        not_ready = [r for r in refs if r not in ready]
+        return ready, not_ready
+
+    # ------------- function export -------------
+
+    async def _export_function(self, fn_id: bytes, pickled: bytes):
+        fut = self._fn_exported.get(fn_id)
+        if fut is not None:
+            await fut
+            return
+        fut = self._fn_exported[fn_id] = self.loop.create_future()
+        try:
+            await self.gcs.call(
+                "kv_put",
+                {"ns": "fn", "key": fn_id, "value": pickled, "overwrite": False},
+            )
+            fut.set_result(None)
+        except Exception as e:
+            fut.set_exception(e)
+            del self._fn_exported[fn_id]
+            raise
+
+    async def load_function(self, fn_id: bytes):
+        fn = self._fn_cache.get(fn_id)
+        if fn is None:
+            data = await self.gcs.call("kv_get", {"ns": "fn", "key": fn_id})
+            if data is None:
+                raise exc.RaySystemError(f"function {fn_id.hex()} not found in GCS")
+            import cloudpickle
+
+            fn = cloudpickle.loads(data)
+            self._fn_cache[fn_id] = fn
+        return fn
+
+    # ------------- normal task submission -------------
+
+    def submit_task(
+        self,
+        pickled_fn: bytes,
+        fn_id: bytes,
+        name: str,
+        args_tuple,
+        options: dict,
+    ) -> List[ObjectRef]:
+        num_returns = options.get("num_returns", 1)
+        returns = [ids.new_object_id() for _ in range(max(num_returns, 1))]
+        refs = [ObjectRef(oid, self.addr) for oid in returns]
+        meta, buffers, captured = self._serialize_capture(args_tuple)
+        size = serialization.serialized_size(meta, buffers)
+        for r in captured:
+            self._add_submitted_ref(r.id)
+        captured_ids = [(r.id, r.owner_addr) for r in captured]
+        spec = {
+            "task_id": ids.new_task_id(),
+            "fn_id": fn_id,
+            "name": name,
+            "returns": returns,
+            "caller": self.addr,
+            "num_returns": num_returns,
+            "env_vars": (options.get("runtime_env") or {}).get("env_vars"),
+        }
+        if size <= serialization.INLINE_MAX:
+            blob = bytearray(size)
+            n = serialization.write_to(memoryview(blob), meta, buffers)
+            spec["args"] = bytes(blob[:n])
+        else:
+            aid = ids.new_object_id()
+            store.put_serialized(self.shm_dir, aid, meta, buffers)
+            spec["args_store"] = (aid, self.addr, self.raylet_addr)
+            self._run(self._seal_async(aid, size))
+            self._store_put_threadsafe(aid, ("store", self.raylet_addr, size))
+            with self._refs_lock:
+                self._refs[aid] = [1, 0, self.addr]  # freed after task completes
+        retries = options.get("max_retries", 3)
+        self._run(
+            self._submit_with_retries(spec, options, retries, captured_ids)
+        )
+        return refs
+
+    async def _seal_async(self, oid, size):
+        await self.raylet.call("seal_object", {"id": oid, "size": size})
+
+    def _pool_key(self, options) -> tuple:
+        res = dict(options.get("resources") or {})
+        if options.get("num_cpus") is not None:
+            res["CPU"] = options["num_cpus"]
+        elif "CPU" not in res:
+            res["CPU"] = 1
+        if options.get("num_gpus"):
+            res["GPU"] = options["num_gpus"]
+        pg = options.get("placement_group")
+        pg_key = None
+        if pg is not None:
+            pg_key = (pg[0], pg[1])
+        return (tuple(sorted(res.items())), pg_key), res, pg_key
+
+    async def _submit_with_retries(self, spec, options, retries, captured_ids):
+        try:
+            while True:
+                try:
+                    reply = await self._dispatch_normal_task(spec, options)
+                    self._ingest_result(spec, reply)
+                    return
+                except (ConnectionLost, ConnectionError) as e:
+                    if retries > 0:
+                        retries -= 1
+                        continue
+                    err = serialization.dumps(
+                        exc.WorkerCrashedError(
+                            f"worker died running {spec['name']}: {e}"
+                        )
+                    )
+                    for oid in spec["returns"]:
+                        self._store_put(oid, ("err", err))
+                    return
+        except Exception:
+            err = serialization.dumps(
+                exc.RaySystemError(
+                    "task submission failed:\n" + traceback.format_exc()
+                )
+            )
+            for oid in spec["returns"]:
+                self._store_put(oid, ("err", err))
+        finally:
+            for oid, _owner in captured_ids:
+                self._remove_submitted_ref(oid)
+            a = spec.get("args_store")
+            if a is not None:
+                self._remove_local_ref(a[0])
+
+    async def _dispatch_normal_task(self, spec, options) -> dict:
+        key, res, pg_key = self._pool_key(options)
+        pool = self._pools.get(key)
+        if pool is None:
+            pool = self._pools[key] = _LeasePool(key, res, pg_key)
+        lease = await self._acquire_lease(pool, spec)
+        lease.busy = True
+        try:
+            reply = await lease.client.call("push_task", spec)
+            return reply
+        finally:
+            lease.busy = False
+            if lease.client.connected:
+                self._release_or_reuse(pool, lease)
+            else:
+                if lease in pool.leases:
+                    pool.leases.remove(lease)
+                self._run(self._return_lease(pool, lease, dead=True))
+
+    async def _acquire_lease(self, pool: _LeasePool, spec) -> _Lease:
+        for l in pool.leases:
+            if not l.busy and l.client.connected:
+                return l
+        fut = self.loop.create_future()
+        pool.queue.append(fut)
+        if pool.requests_in_flight < len(pool.queue):
+            pool.requests_in_flight += 1
+            asyncio.ensure_future(self._request_lease(pool))
+        return await fut
+
+    async def _request_lease(self, pool: _LeasePool):
+        try:
+            req = {
+                "resources": pool.resources,
+                "client": self.addr,
+            }
+            if pool.pg is not None:
+                req["pg_id"], req["bundle_index"] = pool.pg
+            raylet = self.raylet
+            for _hop in range(8):
+                r = await raylet.call("request_lease", req)
+                if r.get("spill"):
+                    raylet = await self._conn(r["spill"])
+                    continue
+                if r.get("error"):
+                    raise exc.RaySystemError(r["error"])
+                client = await self._conn(r["addr"])
+                lease = _Lease((r["lease_id"], r.get("raylet", raylet.addr)), r["addr"], client)
+                pool.leases.append(lease)
+                self._grant_to_queue(pool, lease)
+                return
+            raise exc.RaySystemError("lease spillback loop exceeded")
+        except Exception as e:
+            while pool.queue:
+                fut = pool.queue.popleft()
+                if not fut.done():
+                    fut.set_exception(e)
+        finally:
+            pool.requests_in_flight -= 1
+
+    def _grant_to_queue(self, pool: _LeasePool, lease: _Lease):
+        while pool.queue:
+            fut = pool.queue.popleft()
+            if not fut.done():
+                fut.set_result(lease)
+                return
+        # nobody waiting: keep lease idle; return after grace period
+        self.loop.call_later(0.25, self._maybe_return_idle, pool, lease)
+
+    def _release_or_reuse(self, pool: _LeasePool, lease: _Lease):
+        self._grant_to_queue(pool, lease)
+
+    def _maybe_return_idle(self, pool: _LeasePool, lease: _Lease):
+        if lease.busy or lease not in pool.leases:
+            return
+        if pool.queue:
+            self._grant_to_queue(pool, lease)
+            return
+        pool.leases.remove(lease)
+        asyncio.ensure_future(self._return_lease(pool, lease))
+
+    async def _return_lease(self, pool: _LeasePool, lease: _Lease, dead=False):
+        try:
+            lease_id, raylet_addr = lease.lease_id
+            c = self.raylet if raylet_addr == self.raylet_addr else await self._conn(raylet_addr)
+            c.notify("return_lease", {"lease_id": lease_id, "dead": dead})
+        except Exception:
+            pass
+
+    def _ingest_result(self, spec, reply):
+        status = reply.get("status")
+        if status == "error":
+            for oid in spec["returns"]:
+                self._store_put(oid, ("err", reply["error"]))
+            return
+        results = reply["results"]
+        for oid, r in zip(spec["returns"], results):
+            if r["kind"] == "inline":
+                self._store_put(oid, ("val_ser", r["data"]))
+            else:
+                self._store_put(oid, ("store", r["node_addr"], r["size"]))
+
+    # ------------- actor submission -------------
+
+    def create_actor(self, spec_kv_key: bytes, pickled_cls: bytes, options: dict,
+                     args_tuple) -> bytes:
+        actor_id = ids.new_actor_id()
+        meta, buffers, captured = self._serialize_capture(args_tuple)
+        args_blob = bytearray(serialization.serialized_size(meta, buffers))
+        n = serialization.write_to(memoryview(args_blob), meta, buffers)
+        for r in captured:
+            self._add_submitted_ref(r.id)
+
+        res = dict(options.get("resources") or {})
+        res["CPU"] = options.get("num_cpus", 1) or 0
+        if options.get("num_gpus"):
+            res["GPU"] = options["num_gpus"]
+        pg = options.get("placement_group")
+
+        async def _do():
+            await self.gcs.call(
+                "kv_put",
+                {"ns": "actorcls", "key": spec_kv_key, "value": pickled_cls,
+                 "overwrite": False},
+            )
+            await self.gcs.call(
+                "kv_put",
+                {"ns": "actorargs", "key": actor_id, "value": bytes(args_blob[:n])},
+            )
+            payload = {
+                "actor_id": actor_id,
+                "name": options.get("name"),
+                "namespace": options.get("namespace", "default"),
+                "get_if_exists": options.get("get_if_exists", False),
+                "class_name": options.get("class_name", "Actor"),
+                "spec_kv_key": spec_kv_key,
+                "resources": res,
+                "max_restarts": options.get("max_restarts", 0),
+                "max_concurrency": options.get("max_concurrency", 1),
+                "caller": self.addr,
+                "env_vars": (options.get("runtime_env") or {}).get("env_vars"),
+            }
+            if pg is not None:
+                payload["pg_id"], payload["bundle_index"] = pg[0], pg[1]
+                st = await self.gcs.call("pg_wait_ready", {"pg_id": pg[0]})
+                nodes = st.get("bundle_nodes") or []
+                if pg[1] is not None and pg[1] < len(nodes):
+                    payload["pg_node"] = nodes[pg[1]]
+            r = await self.gcs.call("register_actor", payload)
+            for cref in captured:
+                self._remove_submitted_ref(cref.id)
+            return r
+
+        r = self._call_sync(_do())
+        if r.get("existing"):
+            return r["existing"]
+        return actor_id
+
+    def submit_actor_task(self, actor_id: bytes, method: str, args_tuple,
+                          options: dict) -> List[ObjectRef]:
+        num_returns = options.get("num_returns", 1)
+        returns = [ids.new_object_id() for _ in range(max(num_returns, 1))]
+        refs = [ObjectRef(oid, self.addr) for oid in returns]
+        meta, buffers, captured = self._serialize_capture(args_tuple)
+        size = serialization.serialized_size(meta, buffers)
+        for r in captured:
+            self._add_submitted_ref(r.id)
+        captured_ids = [r.id for r in captured]
+        spec = {
+            "task_id": ids.new_task_id(),
+            "actor_id": actor_id,
+            "method": method,
+            "returns": returns,
+            "caller": self.addr,
+            "num_returns": num_returns,
+        }
+        if size <= serialization.INLINE_MAX:
+            blob = bytearray(size)
+            n = serialization.write_to(memoryview(blob), meta, buffers)
+            spec["args"] = bytes(blob[:n])
+        else:
+            aid = ids.new_object_id()
+            store.put_serialized(self.shm_dir, aid, meta, buffers)
+            spec["args_store"] = (aid, self.addr, self.raylet_addr)
+            self._run(self._seal_async(aid, size))
+            self._store_put_threadsafe(aid, ("store", self.raylet_addr, size))
+            with self._refs_lock:
+                self._refs[aid] = [1, 0, self.addr]
+        self._run(self._submit_actor_async(spec, captured_ids))
+        return refs
+
+    async def _actor_state(self, actor_id: bytes) -> dict:
+        st = self._actors.get(actor_id)
+        if st is None:
+            st = self._actors[actor_id] = {
+                "addr": None, "client": None, "resolving": None, "dead": None
+            }
+        return st
+
+    async def _resolve_actor(self, actor_id: bytes, st: dict):
+        if st["resolving"] is not None:
+            await st["resolving"]
+            return
+        fut = st["resolving"] = self.loop.create_future()
+        try:
+            r = await self.gcs.call(
+                "resolve_actor", {"actor_id": actor_id, "wait": True, "timeout": 120.0}
+            )
+            state = r.get("state")
+            if state == "ALIVE":
+                st["addr"] = r["addr"]
+                st["client"] = await self._conn(r["addr"])
+                st["dead"] = None
+            else:
+                st["dead"] = r.get("death_cause") or f"actor state {state}"
+            fut.set_result(None)
+        except Exception as e:
+            fut.set_exception(e)
+            raise
+        finally:
+            st["resolving"] = None
+
+    async def _submit_actor_async(self, spec, captured_ids):
+        actor_id = spec["actor_id"]
+        try:
+            for attempt in range(3):
+                st = await self._actor_state(actor_id)
+                if st["client"] is None or not st["client"].connected:
+                    st["client"] = None
+                    await self._resolve_actor(actor_id, st)
+                if st["dead"] is not None:
+                    err = serialization.dumps(
+                        exc.ActorDiedError(
+                            f"actor {actor_id.hex()} is dead: {st['dead']}",
+                        )
+                    )
+                    for oid in spec["returns"]:
+                        self._store_put(oid, ("err", err))
+                    return
+                try:
+                    reply = await st["client"].call("actor_call", spec)
+                    self._ingest_result(spec, reply)
+                    return
+                except (ConnectionLost, ConnectionError):
+                    st["client"] = None
+                    continue
+            err = serialization.dumps(
+                exc.ActorUnavailableError(f"actor {actor_id.hex()} unreachable")
+            )
+            for oid in spec["returns"]:
+                self._store_put(oid, ("err", err))
+        except Exception:
+            err = serialization.dumps(
+                exc.RaySystemError("actor call failed:\n" + traceback.format_exc())
+            )
+            for oid in spec["returns"]:
+                self._store_put(oid, ("err", err))
+        finally:
+            for oid in captured_ids:
+                self._remove_submitted_ref(oid)
+            a = spec.get("args_store")
+            if a is not None:
+                self._remove_local_ref(a[0])
+
+    def kill_actor(self, actor_id: bytes, no_restart=True):
+        self._call_sync(
+            self.gcs.call("kill_actor", {"actor_id": actor_id, "no_restart": no_restart})
+        )
+        st = self._actors.get(actor_id)
+        if st is not None:
+            st["client"] = None
+            st["dead"] = "ray.kill"
+
+    # ------------- misc -------------
+
+    def gcs_call(self, method, payload, timeout=None):
+        return self._call_sync(self.gcs.call(method, payload), timeout)
+
+    def raylet_call(self, method, payload, timeout=None):
+        return self._call_sync(self.raylet.call(method, payload), timeout)
+
+
+def fn_hash(pickled: bytes) -> bytes:
+    return hashlib.sha1(pickled).digest()

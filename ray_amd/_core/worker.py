@@ -1,0 +1,375 @@
+"""Worker process — executes normal tasks and hosts actors.
+
+Counterpart of the reference's worker side of CoreWorker: task receiver
+queues (task_execution/task_receiver.h:43), ordered actor execution,
+concurrency groups via a thread pool, async-actor coroutines on the
+event loop, and the Python execution callback
+(_raylet.pyx:2436 execute_task_with_cancellation_handler).
+
+One process == one worker == (optionally) one actor, one asyncio loop on
+the main thread, user code on executor thread(s) so the loop stays
+responsive for serving owned objects.
+"""
+from __future__ import annotations
+
+import asyncio
+import concurrent.futures
+import os
+import sys
+import threading
+import time
+import traceback
+
+# GPU pinning must happen before torch import anywhere in this process.
+_gpu_ids = os.environ.get("RAY_AMD_GPU_IDS")
+if _gpu_ids:
+    os.environ.setdefault("CUDA_VISIBLE_DEVICES", _gpu_ids)
+    os.environ.setdefault("HIP_VISIBLE_DEVICES", _gpu_ids)
+
+from .. import exceptions as exc  # noqa: E402
+from . import ids, serialization, store  # noqa: E402
+from .runtime import CoreRuntime, ObjectRef, set_global_runtime  # noqa: E402
+
+
+class TaskContext:
+    def __init__(self):
+        self.task_id = None
+        self.task_name = None
+        self.actor_id = None
+        self.gpu_ids = []
+
+
+_task_ctx = TaskContext()
+
+
+def current_task_context() -> TaskContext:
+    return _task_ctx
+
+
+class WorkerMain:
+    def __init__(self):
+        self.session_dir = os.environ["RAY_AMD_SESSION_DIR"]
+        self.gcs_addr = os.environ["RAY_AMD_GCS_ADDR"]
+        self.raylet_addr = os.environ["RAY_AMD_RAYLET_ADDR"]
+        self.node_id = bytes.fromhex(os.environ["RAY_AMD_NODE_ID"])
+        self.actor_id = (
+            bytes.fromhex(os.environ["RAY_AMD_ACTOR_ID"])
+            if os.environ.get("RAY_AMD_ACTOR_ID")
+            else None
+        )
+        self.rt: CoreRuntime = None
+        self.actor_instance = None
+        self.actor_spec = None
+        self.executor = concurrent.futures.ThreadPoolExecutor(
+            max_workers=1, thread_name_prefix="task_exec"
+        )
+        self._cancelled = set()
+        self._events = []
+        self._max_concurrency = 1
+
+    async def main(self):
+        loop = asyncio.get_running_loop()
+        rt = CoreRuntime(
+            "worker",
+            self.session_dir,
+            self.gcs_addr,
+            self.raylet_addr,
+            self.node_id,
+            loop=loop,
+        )
+        self.rt = rt
+        rt.set_loop_thread(threading.current_thread())
+        set_global_runtime(rt)
+        await rt._async_start()
+        rt.server.route("push_task", self.h_push_task)
+        rt.server.route("actor_call", self.h_actor_call)
+        rt.server.route("cancel_task", self.h_cancel_task)
+        rt.server.route("exit_worker", self.h_exit_worker)
+        r = await rt.raylet.call(
+            "register_worker", {"pid": os.getpid(), "addr": rt.addr}
+        )
+        if not r.get("ok"):
+            sys.exit(1)
+        if self.actor_id is not None:
+            ok = await self.init_actor()
+            if not ok:
+                await asyncio.sleep(0.2)
+                sys.exit(1)
+        # Stay alive until the raylet connection drops.
+        while rt.raylet.connected:
+            await asyncio.sleep(0.5)
+        sys.exit(0)
+
+    # ------------- actor init -------------
+
+    async def init_actor(self) -> bool:
+        try:
+            r = await self.rt.gcs.call(
+                "resolve_actor", {"actor_id": self.actor_id, "wait": False}
+            )
+            spec_key = r.get("spec_kv_key")
+            pickled_cls = await self.rt.gcs.call(
+                "kv_get", {"ns": "actorcls", "key": spec_key}
+            )
+            args_blob = await self.rt.gcs.call(
+                "kv_get", {"ns": "actorargs", "key": self.actor_id}
+            )
+            import cloudpickle
+
+            cls, self._max_concurrency = cloudpickle.loads(pickled_cls)
+            if self._max_concurrency > 1:
+                self.executor = concurrent.futures.ThreadPoolExecutor(
+                    max_workers=self._max_concurrency
+                )
+            args, kwargs = serialization.loads(args_blob)
+            args, kwargs = await self._resolve_args(args, kwargs)
+            _task_ctx.actor_id = self.actor_id
+            _task_ctx.gpu_ids = [
+                int(x) for x in (os.environ.get("RAY_AMD_GPU_IDS") or "").split(",") if x
+            ]
+
+            def _make():
+                return cls(*args, **kwargs)
+
+            self.actor_instance = await asyncio.get_running_loop().run_in_executor(
+                self.executor, _make
+            )
+            self.rt.raylet.notify(
+                "actor_ready", {"actor_id": self.actor_id, "addr": self.rt.addr}
+            )
+            return True
+        except Exception:
+            tb = traceback.format_exc()
+            try:
+                self.rt.raylet.notify(
+                    "actor_failed", {"actor_id": self.actor_id, "error": tb}
+                )
+            except Exception:
+                pass
+            sys.stderr.write(tb)
+            return False
+
+    # ------------- arg resolution -------------
+
+    async def _resolve_args(self, args, kwargs):
+        """Replace top-level ObjectRefs with their values (reference
+        semantics: only top-level args are resolved)."""
+        refs = [a for a in args if isinstance(a, ObjectRef)]
+        refs += [v for v in kwargs.values() if isinstance(v, ObjectRef)]
+        if refs:
+            vals = await self.rt.get_async(refs, timeout=600.0)
+            table = dict(zip([r.id for r in refs], vals))
+            args = tuple(
+                table[a.id] if isinstance(a, ObjectRef) else a for a in args
+            )
+            kwargs = {
+                k: table[v.id] if isinstance(v, ObjectRef) else v
+                for k, v in kwargs.items()
+            }
+        return args, kwargs
+
+    async def _load_args(self, spec):
+        if spec.get("args") is not None:
+            payload = serialization.loads(spec["args"])
+        else:
+            aid, owner, node_addr = spec["args_store"]
+            ref = ObjectRef(bytes(aid), owner)
+            self.rt.memory_store.setdefault(
+                ref.id, ("store", node_addr, 0)
+            )
+            # size unknown here; _materialize reads the file after wait
+            ent = self.rt.memory_store[ref.id]
+            payload = await self.rt._materialize(ref.id, ent)
+        return payload
+
+    # ------------- normal tasks -------------
+
+    async def h_push_task(self, conn, spec):
+        t0 = time.time()
+        task_id = bytes(spec["task_id"])
+        if task_id in self._cancelled:
+            self._cancelled.discard(task_id)
+            err = serialization.dumps(exc.TaskCancelledError(spec.get("name", "")))
+            return {"status": "error", "error": err}
+        try:
+            fn = await self.rt.load_function(bytes(spec["fn_id"]))
+            args, kwargs = await self._load_args(spec)
+            args, kwargs = await self._resolve_args(args, kwargs)
+        except Exception:
+            return self._error_reply(spec, traceback.format_exc())
+        if spec.get("env_vars"):
+            os.environ.update({str(k): str(v) for k, v in spec["env_vars"].items()})
+
+        loop = asyncio.get_running_loop()
+
+        def _exec():
+            _task_ctx.task_id = task_id
+            _task_ctx.task_name = spec.get("name")
+            try:
+                return True, fn(*args, **kwargs)
+            except BaseException as e:  # noqa
+                return False, e
+            finally:
+                _task_ctx.task_id = None
+
+        if asyncio.iscoroutinefunction(fn):
+            try:
+                result = await fn(*args, **kwargs)
+                ok = True
+            except BaseException as e:  # noqa
+                ok, result = False, e
+        else:
+            ok, result = await loop.run_in_executor(self.executor, _exec)
+        reply = self._build_reply(spec, ok, result)
+        self._record_event(spec, t0, ok)
+        return reply
+
+    # ------------- actor calls -------------
+
+    async def h_actor_call(self, conn, spec):
+        t0 = time.time()
+        method_name = spec["method"]
+        if self.actor_instance is None:
+            err = serialization.dumps(
+                exc.ActorDiedError("actor instance not initialized")
+            )
+            return {"status": "error", "error": err}
+        if method_name == "__ray_terminate__":
+            asyncio.get_running_loop().call_later(0.05, self._graceful_exit)
+            return {"status": "ok", "results": [
+                {"kind": "inline", "data": serialization.dumps(None)}]}
+        try:
+            args, kwargs = await self._load_args(spec)
+            args, kwargs = await self._resolve_args(args, kwargs)
+            method = getattr(self.actor_instance, method_name)
+        except Exception:
+            return self._error_reply(spec, traceback.format_exc())
+
+        loop = asyncio.get_running_loop()
+        exit_after = False
+        if asyncio.iscoroutinefunction(method):
+            try:
+                result = await method(*args, **kwargs)
+                ok = True
+            except SystemExit:
+                ok, result, exit_after = True, None, True
+            except BaseException as e:  # noqa
+                ok, result = False, e
+        else:
+
+            def _exec():
+                try:
+                    return True, method(*args, **kwargs), False
+                except SystemExit:
+                    return True, None, True
+                except BaseException as e:  # noqa
+                    return False, e, False
+
+            ok, result, exit_after = await loop.run_in_executor(self.executor, _exec)
+        reply = self._build_reply(spec, ok, result)
+        self._record_event(spec, t0, ok, method_name)
+        if exit_after or method_name == "__ray_terminate__":
+            loop.call_later(0.05, self._graceful_exit)
+        return reply
+
+    def _graceful_exit(self):
+        try:
+            self.rt.gcs.notify(
+                "actor_exit",
+                {"actor_id": self.actor_id, "expected": True, "cause": "exit_actor"},
+            )
+        except Exception:
+            pass
+        os._exit(0)
+
+    def h_cancel_task(self, conn, p):
+        self._cancelled.add(bytes(p["task_id"]))
+
+    def h_exit_worker(self, conn, p):
+        os._exit(0)
+
+    # ------------- replies -------------
+
+    def _error_reply(self, spec, tb: str, cause=None):
+        name = spec.get("name") or spec.get("method", "")
+        e = exc.RayTaskError(name, tb, None)
+        try:
+            data = serialization.dumps(e)
+        except Exception:
+            data = serialization.dumps(exc.RayTaskError(name, tb, None))
+        return {"status": "error", "error": data}
+
+    def _build_reply(self, spec, ok: bool, result):
+        name = spec.get("name") or spec.get("method", "")
+        if not ok:
+            tb = "".join(
+                traceback.format_exception(type(result), result, result.__traceback__)
+            )
+            try:
+                e = exc.RayTaskError(name, tb, result)
+                data = serialization.dumps(e)
+            except Exception:
+                data = serialization.dumps(exc.RayTaskError(name, tb, None))
+            return {"status": "error", "error": data}
+        num_returns = spec.get("num_returns", 1)
+        if num_returns == 1:
+            values = [result]
+        else:
+            if not isinstance(result, (tuple, list)) or len(result) != num_returns:
+                return self._error_reply(
+                    spec,
+                    f"task declared num_returns={num_returns} but returned "
+                    f"{type(result).__name__}",
+                )
+            values = list(result)
+        results = []
+        for oid, v in zip(spec["returns"], values):
+            try:
+                meta, buffers, _ = self.rt._serialize_capture(v)
+            except Exception:
+                return self._error_reply(spec, traceback.format_exc())
+            size = serialization.serialized_size(meta, buffers)
+            if size <= serialization.INLINE_MAX:
+                blob = bytearray(size)
+                n = serialization.write_to(memoryview(blob), meta, buffers)
+                results.append({"kind": "inline", "data": bytes(blob[:n])})
+            else:
+                store.put_serialized(self.rt.shm_dir, bytes(oid), meta, buffers)
+                # seal through raylet (async fire-and-forget then confirm)
+                self.rt.raylet.notify("seal_object", {"id": bytes(oid), "size": size})
+                results.append(
+                    {"kind": "store", "node_addr": self.raylet_addr, "size": size}
+                )
+        return {"status": "ok", "results": results}
+
+    def _record_event(self, spec, t0, ok, method=None):
+        self._events.append(
+            {
+                "task_id": bytes(spec["task_id"]).hex(),
+                "name": method or spec.get("name", ""),
+                "pid": os.getpid(),
+                "start": t0,
+                "end": time.time(),
+                "ok": ok,
+            }
+        )
+        if len(self._events) >= 100:
+            try:
+                self.rt.raylet.notify(
+                    "report_task_events", {"events": self._events}
+                )
+                self._events = []
+            except Exception:
+                self._events = []
+
+
+def main():
+    wm = WorkerMain()
+    try:
+        asyncio.run(wm.main())
+    except KeyboardInterrupt:
+        pass
+
+
+if __name__ == "__main__":
+    main()
