@@ -1,0 +1,183 @@
+"""Core task/object API tests (reference model:
+python/ray/tests/test_basic.py)."""
+import time
+
+import numpy as np
+import pytest
+
+import ray_amd as ray
+
+
+def test_put_get(ray_start_regular):
+    for v in [1, "x", None, [1, 2, {"a": (3, 4)}], b"bytes"]:
+        assert ray.get(ray.put(v)) == v
+
+
+def test_put_get_numpy_zero_copy(ray_start_regular):
+    a = np.random.rand(512, 512)
+    ref = ray.put(a)
+    b = ray.get(ref)
+    np.testing.assert_array_equal(a, b)
+    assert not b.flags.writeable  # zero-copy view onto shm
+    # repeated get returns cached value
+    c = ray.get(ref)
+    assert c is b
+
+
+def test_simple_task(ray_start_regular):
+    @ray.remote
+    def f(x):
+        return x * 2
+
+    assert ray.get(f.remote(21)) == 42
+
+
+def test_task_many(ray_start_regular):
+    @ray.remote
+    def f(i):
+        return i
+
+    assert ray.get([f.remote(i) for i in range(300)]) == list(range(300))
+
+
+def test_task_args_kwargs_defaults(ray_start_regular):
+    @ray.remote
+    def f(a, b=10, *args, **kwargs):
+        return a + b + sum(args) + kwargs.get("c", 0)
+
+    assert ray.get(f.remote(1)) == 11
+    assert ray.get(f.remote(1, 2)) == 3
+    assert ray.get(f.remote(1, 2, 3, c=4)) == 10
+
+
+def test_ref_as_arg_resolved(ray_start_regular):
+    @ray.remote
+    def plus1(x):
+        return x + 1
+
+    r = plus1.remote(plus1.remote(plus1.remote(0)))
+    assert ray.get(r) == 3
+
+
+def test_nested_refs_not_resolved(ray_start_regular):
+    @ray.remote
+    def check(lst):
+        return all(isinstance(x, ray.ObjectRef) for x in lst)
+
+    refs = [ray.put(i) for i in range(3)]
+    assert ray.get(check.remote(refs))
+
+
+def test_large_arg_and_return(ray_start_regular):
+    @ray.remote
+    def echo(a):
+        return a * 2
+
+    a = np.ones((1500, 1500), dtype=np.float32)
+    out = ray.get(echo.remote(a))
+    assert out.shape == (1500, 1500) and out[0, 0] == 2.0
+
+
+def test_num_returns(ray_start_regular):
+    @ray.remote(num_returns=3)
+    def three():
+        return 1, 2, 3
+
+    a, b, c = three.remote()
+    assert ray.get([a, b, c]) == [1, 2, 3]
+
+
+def test_task_exception(ray_start_regular):
+    @ray.remote
+    def boom():
+        raise ValueError("kaboom")
+
+    with pytest.raises(ray.exceptions.RayTaskError, match="kaboom"):
+        ray.get(boom.remote())
+
+
+def test_exception_in_dependency_propagates(ray_start_regular):
+    @ray.remote
+    def boom():
+        raise ValueError("kaboom")
+
+    @ray.remote
+    def consume(x):
+        return x
+
+    with pytest.raises(Exception, match="kaboom"):
+        ray.get(consume.remote(boom.remote()))
+
+
+def test_wait(ray_start_regular):
+    @ray.remote
+    def slow(t):
+        time.sleep(t)
+        return t
+
+    refs = [slow.remote(0.05), slow.remote(10)]
+    ready, not_ready = ray.wait(refs, num_returns=1, timeout=5)
+    assert ready == [refs[0]] and not_ready == [refs[1]]
+
+    ready, not_ready = ray.wait([slow.remote(0.01)], num_returns=1, timeout=5)
+    assert len(ready) == 1 and not not_ready
+
+
+def test_wait_timeout(ray_start_regular):
+    @ray.remote
+    def slow():
+        time.sleep(30)
+
+    ready, not_ready = ray.wait([slow.remote()], num_returns=1, timeout=0.2)
+    assert not ready and len(not_ready) == 1
+
+
+def test_get_timeout(ray_start_regular):
+    @ray.remote
+    def slow():
+        time.sleep(30)
+
+    with pytest.raises(ray.exceptions.GetTimeoutError):
+        ray.get(slow.remote(), timeout=0.2)
+
+
+def test_options_name(ray_start_regular):
+    @ray.remote
+    def f():
+        return 1
+
+    assert ray.get(f.options(name="custom").remote()) == 1
+
+
+def test_runtime_env_env_vars(ray_start_regular):
+    import os
+
+    @ray.remote
+    def read_env():
+        return os.environ.get("MY_TEST_VAR")
+
+    v = ray.get(
+        read_env.options(runtime_env={"env_vars": {"MY_TEST_VAR": "42"}}).remote()
+    )
+    assert v == "42"
+
+
+def test_cluster_resources(ray_start_regular):
+    res = ray.cluster_resources()
+    assert res["CPU"] == 4
+
+
+def test_put_of_object_ref_fails(ray_start_regular):
+    with pytest.raises(TypeError):
+        ray.put(ray.put(1))
+
+
+def test_async_task_function(ray_start_regular):
+    @ray.remote
+    async def afn(x):
+        import asyncio
+
+        await asyncio.sleep(0.01)
+        return x + 1
+
+    assert ray.get(afn.remote(1)) == 2
