@@ -1,0 +1,166 @@
+"""Flagship benchmark: Llama-3-8B DDP training step (BASELINE.json
+metric "samples/sec Ray Train Llama-3-8B DDP").
+
+Synthetic data (random tokens), random-init weights, bf16 compute,
+fp32 optimizer states (ray_amd.ops.FusedAdamW HIP kernel), DDP over
+RCCL with xGMI-tuned buckets. Launched by the driver as
+`torch.distributed.run --nproc-per-node N bench.py --gpus N ...` for
+N>1; reads RANK/LOCAL_RANK/WORLD_SIZE from the env.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", default="llama3-8b")
+    p.add_argument("--seq-len", type=int, default=4096)
+    p.add_argument("--micro-batch", type=int, default=2)
+    p.add_argument("--bucket-mb", type=int, default=128,
+                   help="DDP bucket size; xGMI rings favor large buckets")
+    p.add_argument("--grad-checkpoint", action="store_true")
+    p.add_argument("--device", default=None, help="cpu for local testing")
+    p.add_argument("--profile-steps", type=int, default=0)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    use_cpu = args.device == "cpu" or not torch.cuda.is_available()
+    if use_cpu:
+        device = torch.device("cpu")
+        dtype = torch.float32
+        if args.model == "llama3-8b":
+            args.model = "llama-tiny"
+            args.seq_len = min(args.seq_len, 128)
+    else:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+        dtype = torch.bfloat16
+
+    distributed = world_size > 1
+    if distributed:
+        import torch.distributed as dist
+
+        dist.init_process_group(
+            backend="gloo" if use_cpu else "nccl",
+            rank=rank,
+            world_size=world_size,
+        )
+
+    from ray_amd.models.llama import CONFIGS, LlamaModel
+    from ray_amd.ops import FusedAdamW
+
+    cfg = CONFIGS[args.model]
+    seq = min(args.seq_len, cfg.max_seq_len)
+    torch.manual_seed(1234 + rank)
+    model = LlamaModel(cfg, dtype=dtype,
+                       gradient_checkpointing=args.grad_checkpoint).to(device)
+    if not use_cpu:
+        model.cosT = model.cosT.to(device)
+        model.sinT = model.sinT.to(device)
+    n_params = model.num_params()
+
+    if distributed:
+        from torch.nn.parallel import DistributedDataParallel as DDP
+
+        model = DDP(
+            model,
+            device_ids=None if use_cpu else [local_rank],
+            bucket_cap_mb=args.bucket_mb,
+            gradient_as_bucket_view=True,
+        )
+
+    opt = FusedAdamW(model.parameters(), lr=1e-4, weight_decay=0.1)
+
+    B, T = args.micro_batch, seq
+    vocab = cfg.vocab_size
+    tokens = torch.randint(0, vocab, (B, T), device=device)
+    targets = torch.randint(0, vocab, (B, T), device=device)
+
+    def step():
+        loss = model(tokens, targets)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        return loss
+
+    def sync():
+        if distributed:
+            import torch.distributed as dist
+
+            dist.barrier()
+        if not use_cpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        loss = step()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if distributed:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device if not use_cpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    total_samples = args.steps * B * world_size
+    samples_per_sec = total_samples / elapsed
+    tokens_per_sec = samples_per_sec * T
+
+    if rank == 0:
+        result = {
+            "metric": "train_samples_per_sec",
+            "value": round(samples_per_sec, 3),
+            "unit": "samples/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32" if use_cpu else "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": B * world_size,
+                "micro_batch": B,
+                "seq_len": T,
+                "parallelism": f"dp{world_size}",
+                "n_params": n_params,
+                "tokens_per_sec": round(tokens_per_sec, 1),
+                "final_loss": round(float(loss.detach().float().cpu()), 4),
+                "bucket_mb": args.bucket_mb,
+                "grad_checkpoint": bool(args.grad_checkpoint),
+            },
+        }
+        print(json.dumps(result))
+
+    if distributed:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

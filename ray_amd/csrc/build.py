@@ -1,0 +1,81 @@
+"""Build the ray_amd HIP extension in-tree with hipcc for gfx950.
+
+Direct hipcc invocation (no hipify, no CUDA path): compiles
+csrc/ops.hip -> ray_amd/_hip_ops.so. The .so is git-ignored but travels
+to GPU boxes with the repo snapshot.
+"""
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+
+import torch
+from torch.utils import cpp_extension
+
+CSRC = os.path.dirname(os.path.abspath(__file__))
+PKG = os.path.dirname(CSRC)
+OUT = os.path.join(PKG, "_hip_ops.so")
+
+
+def _python_include() -> str:
+    import sysconfig
+
+    return sysconfig.get_paths()["include"]
+
+
+def needs_rebuild() -> bool:
+    if not os.path.exists(OUT):
+        return True
+    out_m = os.path.getmtime(OUT)
+    for root, _, files in os.walk(CSRC):
+        for f in files:
+            if f.endswith((".hip", ".h")) and os.path.getmtime(
+                os.path.join(root, f)
+            ) > out_m:
+                return True
+    return False
+
+
+def build(verbose: bool = True, force: bool = False) -> str:
+    if not force and not needs_rebuild():
+        return OUT
+    hipcc = os.path.join(
+        os.environ.get("ROCM_PATH", "/opt/rocm"), "bin", "hipcc"
+    )
+    includes = cpp_extension.include_paths() + [_python_include(), CSRC]
+    lib_dirs = cpp_extension.library_paths()
+    abi = int(torch.compiled_with_cxx11_abi())
+    cmd = [
+        hipcc,
+        "--offload-arch=gfx950",
+        "-O3",
+        "-std=c++17",
+        "-fPIC",
+        "-shared",
+        os.path.join(CSRC, "ops.hip"),
+        "-o",
+        OUT,
+        f"-D_GLIBCXX_USE_CXX11_ABI={abi}",
+        "-DTORCH_EXTENSION_NAME=_hip_ops",
+        "-DTORCH_API_INCLUDE_EXTENSION_H",
+        "-DUSE_ROCM=1",
+        "-fno-gpu-rdc",
+        "-Wno-unused-result",
+    ]
+    for i in includes:
+        cmd.append(f"-I{i}")
+    for l in lib_dirs:
+        cmd.append(f"-L{l}")
+        cmd.append(f"-Wl,-rpath,{l}")
+    cmd += ["-ltorch", "-ltorch_python", "-lc10", "-ltorch_hip", "-lc10_hip",
+            "-lamdhip64"]
+    if verbose:
+        print("[ray_amd build]", " ".join(cmd), file=sys.stderr)
+    subprocess.check_call(cmd)
+    return OUT
+
+
+if __name__ == "__main__":
+    build(force="--force" in sys.argv)
+    print(OUT)

@@ -1,0 +1,229 @@
+// ray_amd HIP ops extension: torch bindings for the CDNA4 kernels.
+// Built directly with hipcc for gfx950 (no hipify, no CUDA path) by
+// ray_amd/csrc/build.py -> ray_amd/_hip_ops.so (in-tree).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "hip/rmsnorm.hip"
+#include "hip/elementwise.hip"
+#include "hip/rl_scans.hip"
+#include "hip/cross_entropy.hip"
+
+#define CHECK_IN(x)                                                     \
+  TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                       \
+  TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+static inline int grid_for(long long n, int block = 256, int cap = 2048) {
+  long long g = (n + block - 1) / block;
+  return (int)std::min<long long>(g, cap);
+}
+
+// ---------------- RMSNorm ----------------
+
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
+  CHECK_IN(x);
+  CHECK_IN(w);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "x must be bf16");
+  int H = (int)x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "H must be a multiple of 8");
+  long long N = x.numel() / H;
+  auto y = at::empty_like(x);
+  auto inv = at::empty({N}, x.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(rmsnorm_fwd_bf16, dim3((unsigned)N), dim3(256), 0,
+                     cur_stream(), (const short*)x.data_ptr(),
+                     (const short*)w.data_ptr(), (short*)y.data_ptr(),
+                     inv.data_ptr<float>(), H, (float)eps);
+  return {y, inv};
+}
+
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                    at::Tensor inv) {
+  CHECK_IN(dy); CHECK_IN(x); CHECK_IN(w); CHECK_IN(inv);
+  int H = (int)x.size(-1);
+  long long N = x.numel() / H;
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(rmsnorm_bwd_dx_bf16, dim3((unsigned)N), dim3(256), 0,
+                     cur_stream(), (const short*)dy.data_ptr(),
+                     (const short*)x.data_ptr(), (const short*)w.data_ptr(),
+                     inv.data_ptr<float>(), (short*)dx.data_ptr(), H);
+  int row_splits = (int)std::min<long long>(256, std::max<long long>(1, N / 64));
+  hipLaunchKernelGGL(rmsnorm_bwd_dw_bf16,
+                     dim3((H + 255) / 256, row_splits), dim3(256), 0,
+                     cur_stream(), (const short*)dy.data_ptr(),
+                     (const short*)x.data_ptr(), inv.data_ptr<float>(),
+                     dw.data_ptr<float>(), N, H);
+  return {dx, dw};
+}
+
+// ---------------- SwiGLU ----------------
+
+at::Tensor swiglu_fwd(at::Tensor a, at::Tensor b) {
+  CHECK_IN(a); CHECK_IN(b);
+  TORCH_CHECK(a.numel() == b.numel() && a.numel() % 8 == 0);
+  auto y = at::empty_like(a);
+  long long n8 = a.numel() / 8;
+  hipLaunchKernelGGL(swiglu_fwd_bf16, dim3(grid_for(n8)), dim3(256), 0,
+                     cur_stream(), (const short*)a.data_ptr(),
+                     (const short*)b.data_ptr(), (short*)y.data_ptr(), n8);
+  return y;
+}
+
+std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor a, at::Tensor b) {
+  CHECK_IN(dy); CHECK_IN(a); CHECK_IN(b);
+  auto da = at::empty_like(a);
+  auto db = at::empty_like(b);
+  long long n8 = a.numel() / 8;
+  hipLaunchKernelGGL(swiglu_bwd_bf16, dim3(grid_for(n8)), dim3(256), 0,
+                     cur_stream(), (const short*)dy.data_ptr(),
+                     (const short*)a.data_ptr(), (const short*)b.data_ptr(),
+                     (short*)da.data_ptr(), (short*)db.data_ptr(), n8);
+  return {da, db};
+}
+
+// ---------------- RoPE ----------------
+
+at::Tensor rope_apply(at::Tensor x, at::Tensor cosT, at::Tensor sinT,
+                      int64_t n_heads, int64_t T, int64_t sign) {
+  CHECK_IN(x); CHECK_IN(cosT); CHECK_IN(sinT);
+  int D = (int)x.size(-1);
+  long long rows = x.numel() / D;
+  auto y = at::empty_like(x);
+  long long total = rows * (D / 2);
+  hipLaunchKernelGGL(rope_fwd_bf16, dim3(grid_for(total)), dim3(256), 0,
+                     cur_stream(), (const short*)x.data_ptr(),
+                     (short*)y.data_ptr(), cosT.data_ptr<float>(),
+                     sinT.data_ptr<float>(), rows, D, (int)n_heads, (int)T,
+                     (int)sign);
+  return y;
+}
+
+// ---------------- AdamW ----------------
+
+void adamw_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
+                at::Tensor master, double lr, double beta1, double beta2,
+                double eps, double wd, int64_t step, double grad_scale) {
+  CHECK_IN(p); CHECK_IN(g); CHECK_IN(m); CHECK_IN(v);
+  long long n = p.numel();
+  float bc1 = 1.f - powf((float)beta1, (float)step);
+  float bc2 = 1.f - powf((float)beta2, (float)step);
+  if (p.scalar_type() == at::kBFloat16) {
+    CHECK_IN(master);
+    hipLaunchKernelGGL(adamw_step_bf16, dim3(grid_for(n)), dim3(256), 0,
+                       cur_stream(), (short*)p.data_ptr(),
+                       (const short*)g.data_ptr(), m.data_ptr<float>(),
+                       v.data_ptr<float>(), master.data_ptr<float>(), n,
+                       (float)lr, (float)beta1, (float)beta2, (float)eps,
+                       (float)wd, bc1, bc2, (float)grad_scale);
+  } else {
+    hipLaunchKernelGGL(adamw_step_f32, dim3(grid_for(n)), dim3(256), 0,
+                       cur_stream(), p.data_ptr<float>(),
+                       g.data_ptr<float>(), m.data_ptr<float>(),
+                       v.data_ptr<float>(), n, (float)lr, (float)beta1,
+                       (float)beta2, (float)eps, (float)wd, bc1, bc2,
+                       (float)grad_scale);
+  }
+}
+
+// ---------------- RL scans ----------------
+
+std::vector<at::Tensor> gae(at::Tensor rewards, at::Tensor values,
+                            at::Tensor cont, double gamma, double lam) {
+  CHECK_IN(rewards); CHECK_IN(values); CHECK_IN(cont);
+  int T = (int)rewards.size(0);
+  int B = (int)rewards.size(1);
+  TORCH_CHECK(values.size(0) == T + 1, "values must be [T+1, B]");
+  auto adv = at::empty_like(rewards);
+  auto vt = at::empty_like(rewards);
+  hipLaunchKernelGGL(gae_scan_f32, dim3((B + 255) / 256), dim3(256), 0,
+                     cur_stream(), rewards.data_ptr<float>(),
+                     values.data_ptr<float>(), cont.data_ptr<float>(),
+                     adv.data_ptr<float>(), vt.data_ptr<float>(), T, B,
+                     (float)gamma, (float)lam);
+  return {adv, vt};
+}
+
+std::vector<at::Tensor> vtrace(at::Tensor log_rhos, at::Tensor rewards,
+                               at::Tensor values, at::Tensor cont,
+                               double gamma, double rho_clip, double c_clip,
+                               double rho_pg_clip) {
+  CHECK_IN(log_rhos); CHECK_IN(rewards); CHECK_IN(values); CHECK_IN(cont);
+  int T = (int)rewards.size(0);
+  int B = (int)rewards.size(1);
+  auto vs = at::empty_like(rewards);
+  auto pg = at::empty_like(rewards);
+  hipLaunchKernelGGL(vtrace_scan_f32, dim3((B + 255) / 256), dim3(256), 0,
+                     cur_stream(), log_rhos.data_ptr<float>(),
+                     rewards.data_ptr<float>(), values.data_ptr<float>(),
+                     cont.data_ptr<float>(), vs.data_ptr<float>(),
+                     pg.data_ptr<float>(), T, B, (float)gamma,
+                     (float)rho_clip, (float)c_clip, (float)rho_pg_clip);
+  return {vs, pg};
+}
+
+// ---------------- image normalize ----------------
+
+at::Tensor img_normalize(at::Tensor in, at::Tensor mean, at::Tensor inv_std) {
+  CHECK_IN(in); CHECK_IN(mean); CHECK_IN(inv_std);
+  TORCH_CHECK(in.scalar_type() == at::kByte && in.dim() == 4);
+  long long N = in.size(0);
+  int H = (int)in.size(1), W = (int)in.size(2), C = (int)in.size(3);
+  auto out = at::empty({N, C, H, W}, in.options().dtype(at::kBFloat16));
+  long long total = in.numel();
+  hipLaunchKernelGGL(img_norm_u8_bf16, dim3(grid_for(total)), dim3(256), 0,
+                     cur_stream(), in.data_ptr<unsigned char>(),
+                     (short*)out.data_ptr(), mean.data_ptr<float>(),
+                     inv_std.data_ptr<float>(), N, H, W, C);
+  return out;
+}
+
+// ---------------- cross entropy ----------------
+
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor target) {
+  CHECK_IN(logits); CHECK_IN(target);
+  TORCH_CHECK(logits.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(target.scalar_type() == at::kInt);
+  long long N = logits.size(0);
+  int V = (int)logits.size(1);
+  auto loss = at::empty({N}, logits.options().dtype(at::kFloat));
+  auto mx = at::empty({N}, logits.options().dtype(at::kFloat));
+  auto lse = at::empty({N}, logits.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(ce_fwd_bf16, dim3((unsigned)N), dim3(256), 0,
+                     cur_stream(), (const short*)logits.data_ptr(),
+                     target.data_ptr<int>(), loss.data_ptr<float>(),
+                     mx.data_ptr<float>(), lse.data_ptr<float>(), N, V);
+  return {loss, mx, lse};
+}
+
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor mx,
+                  at::Tensor lse, at::Tensor dloss) {
+  CHECK_IN(logits); CHECK_IN(target); CHECK_IN(dloss);
+  long long N = logits.size(0);
+  int V = (int)logits.size(1);
+  auto dl = at::empty_like(logits);
+  hipLaunchKernelGGL(ce_bwd_bf16, dim3((unsigned)N), dim3(256), 0,
+                     cur_stream(), (const short*)logits.data_ptr(),
+                     target.data_ptr<int>(), mx.data_ptr<float>(),
+                     lse.data_ptr<float>(), dloss.data_ptr<float>(),
+                     (short*)dl.data_ptr(), N, V);
+  return dl;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("rope_apply", &rope_apply);
+  m.def("adamw_step", &adamw_step);
+  m.def("gae", &gae);
+  m.def("vtrace", &vtrace);
+  m.def("img_normalize", &img_normalize);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
+  m.attr("gfx_arch") = "gfx950";
+}

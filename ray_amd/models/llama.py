@@ -1,0 +1,184 @@
+"""Llama-3 architecture, MI355X-first.
+
+Hand-written HIP kernels (ray_amd.ops) for RMSNorm / RoPE / SwiGLU /
+fused cross-entropy; GQA attention via torch SDPA (CK flash backend on
+ROCm); plain GEMMs via nn.Linear (hipBLASLt). bf16 weights, fp32
+optimizer states via ray_amd.ops.FusedAdamW.
+
+Config llama3-8b matches Meta-Llama-3-8B: hidden 4096, 32 layers,
+32 heads / 8 KV heads, ffn 14336, vocab 128256, rope base 500000.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ray_amd import ops
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    max_seq_len: int = 8192
+    rope_base: float = 500000.0
+    rms_eps: float = 1e-5
+    tie_embeddings: bool = False
+
+
+CONFIGS = {
+    "llama3-8b": LlamaConfig(),
+    "llama3-1b": LlamaConfig(
+        hidden_size=2048, intermediate_size=8192, num_layers=16,
+        num_heads=32, num_kv_heads=8
+    ),
+    # tiny config for CPU tests / smoke
+    "llama-tiny": LlamaConfig(
+        vocab_size=512, hidden_size=256, intermediate_size=512,
+        num_layers=2, num_heads=4, num_kv_heads=2, max_seq_len=512,
+    ),
+}
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype=torch.bfloat16):
+        super().__init__()
+        self.cfg = cfg
+        self.head_dim = cfg.hidden_size // cfg.num_heads
+        self.n_rep = cfg.num_heads // cfg.num_kv_heads
+        self.q_proj = nn.Linear(
+            cfg.hidden_size, cfg.num_heads * self.head_dim, bias=False, dtype=dtype
+        )
+        self.k_proj = nn.Linear(
+            cfg.hidden_size, cfg.num_kv_heads * self.head_dim, bias=False, dtype=dtype
+        )
+        self.v_proj = nn.Linear(
+            cfg.hidden_size, cfg.num_kv_heads * self.head_dim, bias=False, dtype=dtype
+        )
+        self.o_proj = nn.Linear(
+            cfg.num_heads * self.head_dim, cfg.hidden_size, bias=False, dtype=dtype
+        )
+
+    def forward(self, x, cosT, sinT, kv_cache=None, pos0: int = 0):
+        B, T, H = x.shape
+        cfg = self.cfg
+        q = self.q_proj(x).view(B, T, cfg.num_heads, self.head_dim)
+        k = self.k_proj(x).view(B, T, cfg.num_kv_heads, self.head_dim)
+        v = self.v_proj(x).view(B, T, cfg.num_kv_heads, self.head_dim)
+        q = ops.rope(q, cosT[pos0 : pos0 + T], sinT[pos0 : pos0 + T])
+        k = ops.rope(k, cosT[pos0 : pos0 + T], sinT[pos0 : pos0 + T])
+        q = q.transpose(1, 2)  # [B, Hq, T, D]
+        k = k.transpose(1, 2)
+        v = v.transpose(1, 2)
+        if kv_cache is not None:
+            k, v = kv_cache.update(k, v, pos0)
+        is_causal = kv_cache is None or T > 1
+        out = F.scaled_dot_product_attention(
+            q, k, v, is_causal=is_causal, enable_gqa=True
+        )
+        out = out.transpose(1, 2).reshape(B, T, -1)
+        return self.o_proj(out)
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype=torch.bfloat16):
+        super().__init__()
+        self.gate_proj = nn.Linear(
+            cfg.hidden_size, cfg.intermediate_size, bias=False, dtype=dtype
+        )
+        self.up_proj = nn.Linear(
+            cfg.hidden_size, cfg.intermediate_size, bias=False, dtype=dtype
+        )
+        self.down_proj = nn.Linear(
+            cfg.intermediate_size, cfg.hidden_size, bias=False, dtype=dtype
+        )
+
+    def forward(self, x):
+        return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class Block(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype=torch.bfloat16):
+        super().__init__()
+        self.attn_norm = ops.RMSNorm(cfg.hidden_size, cfg.rms_eps, dtype)
+        self.attn = Attention(cfg, dtype)
+        self.mlp_norm = ops.RMSNorm(cfg.hidden_size, cfg.rms_eps, dtype)
+        self.mlp = MLP(cfg, dtype)
+
+    def forward(self, x, cosT, sinT, kv_cache=None, pos0=0):
+        x = x + self.attn(self.attn_norm(x), cosT, sinT, kv_cache, pos0)
+        x = x + self.mlp(self.mlp_norm(x))
+        return x
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype=torch.bfloat16,
+                 gradient_checkpointing: bool = False):
+        super().__init__()
+        self.cfg = cfg
+        self.dtype = dtype
+        self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size, dtype=dtype)
+        self.layers = nn.ModuleList(Block(cfg, dtype) for _ in range(cfg.num_layers))
+        self.final_norm = ops.RMSNorm(cfg.hidden_size, cfg.rms_eps, dtype)
+        self.lm_head = nn.Linear(
+            cfg.hidden_size, cfg.vocab_size, bias=False, dtype=dtype
+        )
+        if cfg.tie_embeddings:
+            self.lm_head.weight = self.embed.weight
+        cosT, sinT = ops.rope_tables(
+            cfg.max_seq_len, cfg.hidden_size // cfg.num_heads, cfg.rope_base
+        )
+        self.register_buffer("cosT", cosT, persistent=False)
+        self.register_buffer("sinT", sinT, persistent=False)
+        self.gradient_checkpointing = gradient_checkpointing
+        self.apply(self._init_weights)
+
+    def _init_weights(self, m):
+        if isinstance(m, nn.Linear):
+            nn.init.normal_(m.weight, std=0.02)
+        elif isinstance(m, nn.Embedding):
+            nn.init.normal_(m.weight, std=0.02)
+
+    def forward(self, tokens, targets=None, kv_caches=None, pos0: int = 0):
+        x = self.embed(tokens)
+        for i, layer in enumerate(self.layers):
+            cache = kv_caches[i] if kv_caches is not None else None
+            if self.gradient_checkpointing and self.training:
+                x = torch.utils.checkpoint.checkpoint(
+                    layer, x, self.cosT, self.sinT, use_reentrant=False
+                )
+            else:
+                x = layer(x, self.cosT, self.sinT, cache, pos0)
+        x = self.final_norm(x)
+        if targets is not None:
+            logits = self.lm_head(x)
+            return ops.cross_entropy(
+                logits.view(-1, self.cfg.vocab_size), targets.reshape(-1)
+            )
+        return self.lm_head(x)
+
+    def num_params(self) -> int:
+        return sum(p.numel() for p in self.parameters())
+
+
+class KVCache:
+    """Per-layer KV cache for decode, preallocated in HBM."""
+
+    def __init__(self, B, max_T, n_kv, head_dim, device, dtype=torch.bfloat16):
+        self.k = torch.zeros(B, n_kv, max_T, head_dim, device=device, dtype=dtype)
+        self.v = torch.zeros(B, n_kv, max_T, head_dim, device=device, dtype=dtype)
+        self.len = 0
+
+    def update(self, k, v, pos0):
+        T = k.size(2)
+        self.k[:, :, pos0 : pos0 + T] = k
+        self.v[:, :, pos0 : pos0 + T] = v
+        self.len = pos0 + T
+        return self.k[:, :, : self.len], self.v[:, :, : self.len]

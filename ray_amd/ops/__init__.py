@@ -1,0 +1,348 @@
+"""ray_amd.ops — hand-written CDNA4 HIP kernels with torch autograd.
+
+Each op has a plain-PyTorch fp32 reference (`*_ref`) used on CPU and in
+numerics tests. On a GPU box the HIP extension is REQUIRED: if a CUDA
+tensor reaches an op and `ray_amd._hip_ops` is missing, we raise rather
+than silently falling back to eager torch.
+"""
+from __future__ import annotations
+
+import torch
+
+try:
+    from ray_amd import _hip_ops as _K
+
+    HAVE_HIP_OPS = True
+except ImportError:  # CPU-only environment without a built extension
+    _K = None
+    HAVE_HIP_OPS = False
+
+
+def _require_ext(name: str):
+    if _K is None:
+        raise RuntimeError(
+            f"ray_amd HIP extension not built but {name} was called on a GPU "
+            "tensor. Run `python ray_amd/csrc/build.py`."
+        )
+
+
+# --------------------------------------------------------------------------
+# RMSNorm
+# --------------------------------------------------------------------------
+
+
+def rmsnorm_ref(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5):
+    xf = x.float()
+    inv = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (xf * inv * w.float()).to(x.dtype)
+
+
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, eps):
+        _require_ext("rmsnorm")
+        y, inv = _K.rmsnorm_fwd(x.contiguous(), w.contiguous(), eps)
+        ctx.save_for_backward(x, w, inv)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, inv = ctx.saved_tensors
+        dx, dw = _K.rmsnorm_bwd(dy.contiguous(), x, w, inv)
+        return dx, dw.to(w.dtype), None
+
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5):
+    if not x.is_cuda:
+        # autograd-capable reference path
+        xf = x.float()
+        inv = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+        return (xf * inv * w.float()).to(x.dtype)
+    return _RMSNorm.apply(x, w, eps)
+
+
+class RMSNorm(torch.nn.Module):
+    def __init__(self, dim: int, eps: float = 1e-5, dtype=torch.bfloat16):
+        super().__init__()
+        self.weight = torch.nn.Parameter(torch.ones(dim, dtype=dtype))
+        self.eps = eps
+
+    def forward(self, x):
+        return rmsnorm(x, self.weight, self.eps)
+
+
+# --------------------------------------------------------------------------
+# SwiGLU
+# --------------------------------------------------------------------------
+
+
+def swiglu_ref(a: torch.Tensor, b: torch.Tensor):
+    return (torch.nn.functional.silu(a.float()) * b.float()).to(a.dtype)
+
+
+class _SwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        _require_ext("swiglu")
+        a = a.contiguous()
+        b = b.contiguous()
+        ctx.save_for_backward(a, b)
+        return _K.swiglu_fwd(a, b)
+
+    @staticmethod
+    def backward(ctx, dy):
+        a, b = ctx.saved_tensors
+        da, db = _K.swiglu_bwd(dy.contiguous(), a, b)
+        return da, db
+
+
+def swiglu(a: torch.Tensor, b: torch.Tensor):
+    if not a.is_cuda:
+        return torch.nn.functional.silu(a) * b
+    return _SwiGLU.apply(a, b)
+
+
+# --------------------------------------------------------------------------
+# RoPE
+# --------------------------------------------------------------------------
+
+
+def rope_tables(T: int, D: int, base: float = 500000.0, device="cpu"):
+    """Host-precomputed cos/sin [T, D/2] fp32 (guide §B: no device trig)."""
+    half = D // 2
+    inv_freq = 1.0 / (
+        base ** (torch.arange(0, half, dtype=torch.float64) / half)
+    )
+    t = torch.arange(T, dtype=torch.float64)
+    ang = torch.outer(t, inv_freq)
+    return (
+        ang.cos().float().contiguous().to(device),
+        ang.sin().float().contiguous().to(device),
+    )
+
+
+def rope_ref(x: torch.Tensor, cosT: torch.Tensor, sinT: torch.Tensor):
+    """x: [B, T, Hn, D]; rotate halves (LLaMA convention)."""
+    B, T, Hn, D = x.shape
+    half = D // 2
+    x1 = x[..., :half].float()
+    x2 = x[..., half:].float()
+    c = cosT[:T].view(1, T, 1, half)
+    s = sinT[:T].view(1, T, 1, half)
+    return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1).to(x.dtype)
+
+
+class _Rope(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cosT, sinT, n_heads, T):
+        _require_ext("rope")
+        ctx.meta = (cosT, sinT, n_heads, T)
+        return _K.rope_apply(x.contiguous(), cosT, sinT, n_heads, T, 1)
+
+    @staticmethod
+    def backward(ctx, dy):
+        cosT, sinT, n_heads, T = ctx.meta
+        dx = _K.rope_apply(dy.contiguous(), cosT, sinT, n_heads, T, -1)
+        return dx, None, None, None, None
+
+
+def rope(x: torch.Tensor, cosT: torch.Tensor, sinT: torch.Tensor):
+    """x: [B, T, Hn, D] bf16."""
+    B, T, Hn, D = x.shape
+    if not x.is_cuda:
+        return rope_ref(x, cosT, sinT)
+    return _Rope.apply(x, cosT[:T].contiguous(), sinT[:T].contiguous(), Hn, T)
+
+
+# --------------------------------------------------------------------------
+# Cross entropy (fused, bf16 logits)
+# --------------------------------------------------------------------------
+
+
+def cross_entropy_ref(logits: torch.Tensor, target: torch.Tensor):
+    return torch.nn.functional.cross_entropy(
+        logits.float(), target.long(), ignore_index=-100, reduction="mean"
+    )
+
+
+class _FusedCE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, target):
+        _require_ext("cross_entropy")
+        logits = logits.contiguous()
+        t32 = target.to(torch.int32).contiguous()
+        loss, mx, lse = _K.ce_fwd(logits, t32)
+        nvalid = (t32 >= 0).sum().clamp(min=1)
+        ctx.save_for_backward(logits, t32, mx, lse, nvalid)
+        return loss.sum() / nvalid.float()
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, t32, mx, lse, nvalid = ctx.saved_tensors
+        per_row = (dloss / nvalid.float()).expand(logits.size(0)).contiguous()
+        dl = _K.ce_bwd(logits, t32, mx, lse, per_row)
+        return dl, None
+
+
+def cross_entropy(logits: torch.Tensor, target: torch.Tensor):
+    """Mean CE over rows where target != -100. logits [N, V] bf16."""
+    if not logits.is_cuda:
+        return cross_entropy_ref(logits, target)
+    return _FusedCE.apply(logits, target)
+
+
+# --------------------------------------------------------------------------
+# Fused AdamW
+# --------------------------------------------------------------------------
+
+
+class FusedAdamW:
+    """AdamW with fp32 states (+ fp32 master weights for bf16 params),
+    one fused HIP kernel launch per tensor. CPU fallback uses torch ops."""
+
+    def __init__(self, params, lr=1e-4, betas=(0.9, 0.95), eps=1e-8,
+                 weight_decay=0.1):
+        self.params = [p for p in params if p.requires_grad]
+        self.lr = lr
+        self.beta1, self.beta2 = betas
+        self.eps = eps
+        self.wd = weight_decay
+        self.t = 0
+        self.state = {}
+        for p in self.params:
+            st = {"m": torch.zeros_like(p, dtype=torch.float32),
+                  "v": torch.zeros_like(p, dtype=torch.float32)}
+            if p.dtype == torch.bfloat16:
+                st["master"] = p.detach().float().clone()
+            self.state[id(p)] = st
+
+    @property
+    def param_groups(self):
+        return [{"params": self.params, "lr": self.lr}]
+
+    def zero_grad(self, set_to_none=True):
+        for p in self.params:
+            if set_to_none:
+                p.grad = None
+            elif p.grad is not None:
+                p.grad.zero_()
+
+    @torch.no_grad()
+    def step(self, grad_scale: float = 1.0):
+        self.t += 1
+        for p in self.params:
+            if p.grad is None:
+                continue
+            st = self.state[id(p)]
+            if p.is_cuda:
+                _require_ext("adamw")
+                _K.adamw_step(
+                    p.data, p.grad.contiguous(), st["m"], st["v"],
+                    st.get("master", st["m"]), self.lr, self.beta1,
+                    self.beta2, self.eps, self.wd, self.t, grad_scale,
+                )
+            else:
+                g = p.grad.float() * grad_scale
+                st["m"].mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+                st["v"].mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+                bc1 = 1 - self.beta1 ** self.t
+                bc2 = 1 - self.beta2 ** self.t
+                pf = st.get("master", None)
+                if pf is None:
+                    pf = p.data.float()
+                upd = (st["m"] / bc1) / ((st["v"] / bc2).sqrt() + self.eps)
+                pf -= self.lr * (upd + self.wd * pf)
+                if "master" in st:
+                    p.data.copy_(pf.to(p.dtype))
+                else:
+                    p.data.copy_(pf)
+
+    def state_dict(self):
+        return {"t": self.t, "lr": self.lr}
+
+    def load_state_dict(self, sd):
+        self.t = sd.get("t", 0)
+        self.lr = sd.get("lr", self.lr)
+
+
+# --------------------------------------------------------------------------
+# RL scans
+# --------------------------------------------------------------------------
+
+
+def gae_ref(rewards, values, cont, gamma, lam):
+    """rewards/cont [T,B] fp32, values [T+1,B]. Returns (adv, vtarg)."""
+    T, B = rewards.shape
+    adv = torch.zeros_like(rewards)
+    running = torch.zeros(B, dtype=rewards.dtype)
+    for t in reversed(range(T)):
+        delta = rewards[t] + gamma * cont[t] * values[t + 1] - values[t]
+        running = delta + gamma * lam * cont[t] * running
+        adv[t] = running
+    return adv, adv + values[:-1]
+
+
+def gae(rewards, values, cont, gamma=0.99, lam=0.95):
+    if not rewards.is_cuda:
+        return gae_ref(rewards, values, cont, gamma, lam)
+    _require_ext("gae")
+    out = _K.gae(rewards.contiguous(), values.contiguous(),
+                 cont.contiguous(), gamma, lam)
+    return out[0], out[1]
+
+
+def vtrace_ref(log_rhos, rewards, values, cont, gamma, rho_clip=1.0,
+               c_clip=1.0, rho_pg_clip=1.0):
+    """Reference recursion (cf. vtrace_torch_v2.py:73 semantics)."""
+    T, B = rewards.shape
+    rhos = log_rhos.exp()
+    rho = rhos.clamp(max=rho_clip)
+    cs = rhos.clamp(max=c_clip)
+    vs = torch.zeros_like(rewards)
+    diff = torch.zeros(B, dtype=rewards.dtype)
+    for t in reversed(range(T)):
+        delta = rho[t] * (rewards[t] + gamma * cont[t] * values[t + 1] - values[t])
+        cur = delta + gamma * cont[t] * cs[t] * diff
+        vs[t] = values[t] + cur
+        diff = cur
+    pg = torch.zeros_like(rewards)
+    for t in range(T):
+        vs_next = values[t + 1] if t == T - 1 else vs[t + 1]
+        pg[t] = rhos[t].clamp(max=rho_pg_clip) * (
+            rewards[t] + gamma * cont[t] * vs_next - values[t]
+        )
+    return vs, pg
+
+
+def vtrace(log_rhos, rewards, values, cont, gamma=0.99, rho_clip=1.0,
+           c_clip=1.0, rho_pg_clip=1.0):
+    if not rewards.is_cuda:
+        return vtrace_ref(log_rhos, rewards, values, cont, gamma, rho_clip,
+                          c_clip, rho_pg_clip)
+    _require_ext("vtrace")
+    out = _K.vtrace(log_rhos.contiguous(), rewards.contiguous(),
+                    values.contiguous(), cont.contiguous(), gamma,
+                    rho_clip, c_clip, rho_pg_clip)
+    return out[0], out[1]
+
+
+# --------------------------------------------------------------------------
+# Image normalize (Data preprocessing)
+# --------------------------------------------------------------------------
+
+
+def img_normalize_ref(x_u8, mean, std):
+    xf = x_u8.float() / 255.0
+    out = (xf - mean.view(1, 1, 1, -1)) / std.view(1, 1, 1, -1)
+    return out.permute(0, 3, 1, 2).contiguous().bfloat16()
+
+
+def img_normalize(x_u8: torch.Tensor, mean: torch.Tensor, std: torch.Tensor):
+    """uint8 NHWC -> bf16 NCHW normalized."""
+    if not x_u8.is_cuda:
+        return img_normalize_ref(x_u8, mean, std)
+    _require_ext("img_normalize")
+    return _K.img_normalize(
+        x_u8.contiguous(), mean.float().contiguous(),
+        (1.0 / std.float()).contiguous()
+    )
