@@ -24,7 +24,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", default="llama3-8b")
     p.add_argument("--seq-len", type=int, default=4096)
-    p.add_argument("--micro-batch", type=int, default=2)
+    p.add_argument("--micro-batch", type=int, default=8)
     p.add_argument("--bucket-mb", type=int, default=128,
                    help="DDP bucket size; xGMI rings favor large buckets")
     p.add_argument("--grad-checkpoint", action="store_true")
@@ -88,10 +88,12 @@ def main():
 
     B, T = args.micro_batch, seq
     vocab = cfg.vocab_size
-    tokens = torch.randint(0, vocab, (B, T), device=device)
-    targets = torch.randint(0, vocab, (B, T), device=device)
+    def make_batch():
+        return (torch.randint(0, vocab, (B, T), device=device),
+                torch.randint(0, vocab, (B, T), device=device))
 
     def step():
+        tokens, targets = make_batch()
         loss = model(tokens, targets)
         loss.backward()
         opt.step()
