@@ -203,6 +203,17 @@ class LocalCluster:
         shutil.rmtree(self.session_dir, ignore_errors=True)
 
 
+def export_driver_pythonpath():
+    """Propagate the driver's sys.path to workers (reference parity:
+    workers inherit the driver's import environment via PYTHONPATH /
+    working_dir runtime env)."""
+    import sys
+
+    os.environ["RAY_AMD_PYTHONPATH"] = os.pathsep.join(
+        p for p in sys.path if p
+    )
+
+
 def start_local_cluster(
     num_cpus: Optional[float] = None,
     num_gpus: Optional[float] = None,
@@ -212,6 +223,7 @@ def start_local_cluster(
 ) -> LocalCluster:
     session_dir = new_session_dir()
     os.environ["RAY_AMD_SHM_DIR"] = session_shm_dir(session_dir)
+    export_driver_pythonpath()
     gcs_proc, gcs_addr = start_gcs(session_dir)
     res = dict(resources or {})
     res.setdefault("CPU", num_cpus if num_cpus is not None else os.cpu_count())

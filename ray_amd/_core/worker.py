@@ -20,6 +20,15 @@ import threading
 import time
 import traceback
 
+# Driver sys.path propagation (so by-reference pickled functions import).
+_pp = os.environ.get("RAY_AMD_PYTHONPATH")
+if _pp:
+    import sys as _sys
+
+    for _p in reversed(_pp.split(os.pathsep)):
+        if _p and _p not in _sys.path:
+            _sys.path.insert(0, _p)
+
 # GPU pinning must happen before torch import anywhere in this process.
 _gpu_ids = os.environ.get("RAY_AMD_GPU_IDS")
 if _gpu_ids:

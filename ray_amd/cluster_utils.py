@@ -31,6 +31,7 @@ class Cluster:
                  connect: bool = False):
         self.session_dir = _node.new_session_dir()
         os.environ["RAY_AMD_SHM_DIR"] = _node.session_shm_dir(self.session_dir)
+        _node.export_driver_pythonpath()
         self.gcs_proc, self.gcs_addr = _node.start_gcs(self.session_dir)
         self.nodes = []
         self.head_node: Optional[NodeHandle] = None

@@ -1,0 +1,160 @@
+"""ray_amd.train tests: TorchTrainer DDP on CPU (gloo, world_size=2),
+report/checkpoint/resume, failure handling.
+Reference model: python/ray/train/tests/."""
+import os
+
+import pytest
+import torch
+
+import ray_amd as ray
+from ray_amd.train import (
+    Checkpoint,
+    FailureConfig,
+    RunConfig,
+    ScalingConfig,
+)
+from ray_amd.train.torch import TorchTrainer
+
+
+def train_fn_basic(config):
+    import ray_amd.train as train
+
+    ctx = train.get_context()
+    assert ctx.get_world_size() == 2
+    for i in range(3):
+        train.report({"loss": 1.0 / (i + 1), "rank": ctx.get_world_rank()})
+
+
+def test_torch_trainer_reports(ray_start_regular, tmp_path):
+    t = TorchTrainer(
+        train_fn_basic,
+        scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(name="t1", storage_path=str(tmp_path)),
+    )
+    res = t.fit()
+    assert res.error is None
+    assert res.metrics["loss"] == pytest.approx(1 / 3)
+    assert len(res.metrics_dataframe) == 3
+
+
+def train_fn_ddp(config):
+    import torch.distributed as dist
+
+    import ray_amd.train as train
+    from ray_amd.train.torch import prepare_model
+
+    model = torch.nn.Linear(4, 2)
+    model = prepare_model(model)
+    assert isinstance(model, torch.nn.parallel.DistributedDataParallel)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    x = torch.randn(8, 4)
+    y = torch.randn(8, 2)
+    for _ in range(2):
+        loss = ((model(x) - y) ** 2).mean()
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+    # all ranks see identical (averaged) grads -> identical weights
+    w = model.module.weight.detach().flatten()
+    gathered = [torch.zeros_like(w) for _ in range(2)]
+    dist.all_gather(gathered, w)
+    assert torch.allclose(gathered[0], gathered[1])
+    train.report({"final_loss": float(loss)})
+
+
+def test_torch_trainer_ddp_gloo(ray_start_regular, tmp_path):
+    t = TorchTrainer(
+        train_fn_ddp,
+        scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(name="ddp", storage_path=str(tmp_path)),
+    )
+    res = t.fit()
+    assert res.error is None
+    assert "final_loss" in res.metrics
+
+
+def train_fn_ckpt(config):
+    import tempfile
+
+    import ray_amd.train as train
+
+    ctx = train.get_context()
+    start = 0
+    ckpt = train.get_checkpoint()
+    if ckpt is not None:
+        with ckpt.as_directory() as d:
+            start = int(open(os.path.join(d, "it.txt")).read())
+    for i in range(start, start + 2):
+        with tempfile.TemporaryDirectory() as d:
+            with open(os.path.join(d, "it.txt"), "w") as f:
+                f.write(str(i + 1))
+            if ctx.get_world_rank() == 0:
+                train.report({"it": i + 1}, checkpoint=Checkpoint.from_directory(d))
+            else:
+                train.report({"it": i + 1})
+
+
+def test_checkpoint_and_resume(ray_start_regular, tmp_path):
+    t = TorchTrainer(
+        train_fn_ckpt,
+        scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(name="ck", storage_path=str(tmp_path)),
+    )
+    res = t.fit()
+    assert res.error is None
+    assert res.metrics["it"] == 2
+    assert res.checkpoint is not None
+    with res.checkpoint.as_directory() as d:
+        assert open(os.path.join(d, "it.txt")).read() == "2"
+
+    # resume
+    t2 = TorchTrainer(
+        train_fn_ckpt,
+        scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(name="ck2", storage_path=str(tmp_path)),
+        resume_from_checkpoint=res.checkpoint,
+    )
+    res2 = t2.fit()
+    assert res2.metrics["it"] == 4
+
+
+_fail_flag = os.path.join("/tmp", "ray_amd_train_fail_once")
+
+
+def train_fn_flaky(config):
+    import ray_amd.train as train
+
+    if not os.path.exists(config["flag"]):
+        open(config["flag"], "w").close()
+        raise RuntimeError("injected failure")
+    train.report({"ok": 1})
+
+
+def test_failure_retry(ray_start_regular, tmp_path):
+    flag = str(tmp_path / "flag")
+    t = TorchTrainer(
+        train_fn_flaky,
+        train_loop_config={"flag": flag},
+        scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(
+            name="flaky", storage_path=str(tmp_path),
+            failure_config=FailureConfig(max_failures=1),
+        ),
+    )
+    res = t.fit()
+    assert res.error is None
+    assert res.metrics["ok"] == 1
+
+
+def test_failure_no_retry(ray_start_regular, tmp_path):
+    def always_fail(config):
+        raise RuntimeError("always fails")
+
+    t = TorchTrainer(
+        always_fail,
+        scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(name="nf", storage_path=str(tmp_path)),
+    )
+    res = t.fit()
+    assert res.error is not None
+    assert "always fails" in str(res.error)
