@@ -1,0 +1,30 @@
+"""ray_amd.data — distributed datasets (reference: python/ray/data/).
+
+Blocks are pyarrow Tables stored in the shm object store; transforms
+run as ray_amd tasks (task-pool) or actor pools (map_batches
+concurrency), with a windowed streaming iterator for consumption
+(reference: _internal/execution/streaming_executor.py:107 — a bounded
+in-flight window provides the same backpressure effect for the v1
+pipeline).
+"""
+from .dataset import (  # noqa: F401
+    Dataset,
+    DataContext,
+    from_arrow,
+    from_items,
+    from_numpy,
+    from_pandas,
+    range,  # noqa: A001  (API parity with ray.data.range)
+    range_tensor,
+    read_binary_files,
+    read_csv,
+    read_json,
+    read_parquet,
+)
+from .preprocessors import (  # noqa: F401
+    Concatenator,
+    LabelEncoder,
+    MinMaxScaler,
+    StandardScaler,
+    TorchVisionNormalizer,
+)

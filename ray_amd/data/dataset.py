@@ -1,0 +1,884 @@
+"""Dataset core (reference: python/ray/data/dataset.py — map :422,
+map_batches :629, flat_map :1859, filter :2033, repartition :2269,
+random_shuffle :2410, streaming_split :2629, groupby :3747, sort :4267,
+iter_batches :6731, iter_torch_batches :6819; read_api.py:1370
+read_parquet).
+
+Blocks = pyarrow Tables in the object store; a Dataset is a lazy plan
+over input blocks. Execution: per-block ray_amd tasks with a bounded
+in-flight window for streaming consumption.
+"""
+from __future__ import annotations
+
+import builtins
+import itertools
+from typing import Any, Callable, Dict, Iterable, Iterator, List, Optional, Union
+
+import numpy as np
+import pyarrow as pa
+import pyarrow.parquet as pq
+
+
+class DataContext:
+    _current = None
+
+    def __init__(self):
+        self.target_max_block_size = 128 * 1024 * 1024
+        self.streaming_read_window = 8  # max in-flight blocks per consumer
+
+    @classmethod
+    def get_current(cls) -> "DataContext":
+        if cls._current is None:
+            cls._current = DataContext()
+        return cls._current
+
+
+def _ray():
+    import ray_amd as ray
+
+    return ray
+
+
+# ---------------- block helpers ----------------
+
+
+def _to_table(block: Any) -> pa.Table:
+    if isinstance(block, pa.Table):
+        return block
+    if isinstance(block, dict):
+        return pa.table(
+            {k: _np_to_arrow(v) for k, v in block.items()}
+        )
+    raise TypeError(f"bad block type {type(block)}")
+
+
+def _np_to_arrow(v):
+    v = np.asarray(v)
+    if v.ndim <= 1:
+        return pa.array(v)
+    return pa.FixedSizeListArray.from_arrays(
+        pa.array(v.reshape(-1)), int(np.prod(v.shape[1:]))
+    )
+
+
+def _col_to_numpy(col: pa.ChunkedArray) -> np.ndarray:
+    t = col.type
+    if pa.types.is_fixed_size_list(t):
+        flat = col.combine_chunks().flatten().to_numpy(zero_copy_only=False)
+        return flat.reshape(len(col), t.list_size)
+    return col.to_numpy(zero_copy_only=False)
+
+
+def _table_to_numpy(t: pa.Table) -> Dict[str, np.ndarray]:
+    return {name: _col_to_numpy(t.column(name)) for name in t.column_names}
+
+
+def _table_to_pandas(t: pa.Table):
+    return t.to_pandas()
+
+
+def _format_batch(t: pa.Table, fmt: Optional[str]):
+    if fmt in (None, "default", "numpy"):
+        return _table_to_numpy(t)
+    if fmt == "pandas":
+        return _table_to_pandas(t)
+    if fmt in ("pyarrow", "arrow"):
+        return t
+    raise ValueError(f"unknown batch_format {fmt}")
+
+
+def _batch_to_table(b) -> pa.Table:
+    if isinstance(b, pa.Table):
+        return b
+    if isinstance(b, dict):
+        return _to_table(b)
+    try:
+        import pandas as pd
+
+        if isinstance(b, pd.DataFrame):
+            return pa.Table.from_pandas(b, preserve_index=False)
+    except ImportError:
+        pass
+    raise TypeError(f"map_batches fn returned unsupported type {type(b)}")
+
+
+def _rows_of(t: pa.Table) -> List[dict]:
+    cols = _table_to_numpy(t)
+    names = list(cols)
+    n = t.num_rows
+    return [
+        {name: cols[name][i] for name in names} for i in builtins.range(n)
+    ]
+
+
+# ---------------- plan ops (executed inside ray tasks) ----------------
+
+
+def _op_map_batches(t: pa.Table, fn, fmt, batch_size) -> pa.Table:
+    if batch_size is None or t.num_rows <= batch_size:
+        out = fn(_format_batch(t, fmt))
+        return _batch_to_table(out)
+    parts = []
+    for s in builtins.range(0, t.num_rows, batch_size):
+        out = fn(_format_batch(t.slice(s, batch_size), fmt))
+        parts.append(_batch_to_table(out))
+    return pa.concat_tables(parts)
+
+
+def _op_map(t: pa.Table, fn) -> pa.Table:
+    rows = [fn(r) for r in _rows_of(t)]
+    return _rows_to_table(rows)
+
+
+def _op_flat_map(t: pa.Table, fn) -> pa.Table:
+    rows = list(itertools.chain.from_iterable(fn(r) for r in _rows_of(t)))
+    return _rows_to_table(rows)
+
+
+def _op_filter(t: pa.Table, fn) -> pa.Table:
+    mask = np.array([bool(fn(r)) for r in _rows_of(t)])
+    return t.filter(pa.array(mask))
+
+
+def _rows_to_table(rows: List[dict]) -> pa.Table:
+    if not rows:
+        return pa.table({})
+    keys = rows[0].keys()
+    return pa.table({k: _np_to_arrow(np.asarray([r[k] for r in rows])) for k in keys})
+
+
+def _apply_ops(table_or_ref, ops: List[tuple]) -> pa.Table:
+    t = table_or_ref
+    for op in ops:
+        kind = op[0]
+        if kind == "map_batches":
+            t = _op_map_batches(t, op[1], op[2], op[3])
+        elif kind == "map":
+            t = _op_map(t, op[1])
+        elif kind == "flat_map":
+            t = _op_flat_map(t, op[1])
+        elif kind == "filter":
+            t = _op_filter(t, op[1])
+        elif kind == "read":
+            t = op[1](t)  # t is a read task descriptor
+    return t
+
+
+# ---------------- Dataset ----------------
+
+
+class Dataset:
+    def __init__(self, inputs: List[Any], ops: Optional[List[tuple]] = None,
+                 owner=None):
+        # inputs: list of ObjectRefs to pa.Table OR ("readtask", fn, arg)
+        self._inputs = inputs
+        self._ops = ops or []
+        self._materialized: Optional[List[Any]] = None
+
+    # ----- plan builders -----
+
+    def _with_op(self, op: tuple) -> "Dataset":
+        return Dataset(self._inputs, self._ops + [op])
+
+    def map(self, fn: Callable[[dict], dict], **kwargs) -> "Dataset":
+        return self._with_op(("map", fn))
+
+    def map_batches(
+        self,
+        fn: Callable,
+        *,
+        batch_size: Optional[int] = None,
+        batch_format: Optional[str] = "numpy",
+        compute=None,
+        concurrency=None,
+        fn_args=None,
+        fn_kwargs=None,
+        num_gpus: float = 0,
+        **kwargs,
+    ) -> "Dataset":
+        if fn_args or fn_kwargs:
+            base = fn
+            a = tuple(fn_args or ())
+            kw = dict(fn_kwargs or {})
+            fn = lambda b: base(b, *a, **kw)  # noqa: E731
+        if isinstance(fn, type):
+            # actor-pool UDF class: instantiate per task (v1; actor pools
+            # proper when concurrency tuning matters)
+            cls = fn
+
+            def fn(batch, _cls=cls):
+                inst_attr = "_ray_amd_udf_inst"
+                import threading
+
+                tl = getattr(_cls, inst_attr, None)
+                if tl is None:
+                    tl = _cls()
+                    setattr(_cls, inst_attr, tl)
+                return tl(batch)
+
+        return self._with_op(("map_batches", fn, batch_format, batch_size))
+
+    def flat_map(self, fn) -> "Dataset":
+        return self._with_op(("flat_map", fn))
+
+    def filter(self, fn=None, *, expr=None) -> "Dataset":
+        if fn is None and expr is not None:
+            code = compile(expr, "<filter_expr>", "eval")
+
+            def fn(row, _code=code):
+                return eval(_code, {}, dict(row))
+
+        return self._with_op(("filter", fn))
+
+    def add_column(self, name: str, fn) -> "Dataset":
+        def add(batch):
+            batch = dict(batch)
+            batch[name] = fn(batch)
+            return batch
+
+        return self._with_op(("map_batches", add, "numpy", None))
+
+    def drop_columns(self, cols: List[str]) -> "Dataset":
+        def drop(t: pa.Table):
+            return t.drop_columns([c for c in cols if c in t.column_names])
+
+        return self._with_op(("map_batches", drop, "pyarrow", None))
+
+    def select_columns(self, cols: List[str]) -> "Dataset":
+        def sel(t: pa.Table):
+            return t.select(cols)
+
+        return self._with_op(("map_batches", sel, "pyarrow", None))
+
+    def rename_columns(self, mapping: Dict[str, str]) -> "Dataset":
+        def ren(t: pa.Table):
+            return t.rename_columns(
+                [mapping.get(c, c) for c in t.column_names]
+            )
+
+        return self._with_op(("map_batches", ren, "pyarrow", None))
+
+    # ----- execution -----
+
+    def _materialize_refs(self) -> List[Any]:
+        """Run the plan; returns ObjectRefs of result blocks."""
+        if self._materialized is not None:
+            return self._materialized
+        ray = _ray()
+        ops = self._ops
+
+        @ray.remote
+        def _exec_block(block, ops=ops):
+            t = block() if callable(block) else block
+            return _apply_ops(t, ops)
+
+        refs = []
+        for inp in self._inputs:
+            refs.append(_exec_block.remote(inp))
+        self._materialized = refs
+        return refs
+
+    def materialize(self) -> "Dataset":
+        refs = self._materialize_refs()
+        ds = Dataset(refs, [])
+        ds._materialized = refs
+        return ds
+
+    def _iter_block_refs(self) -> Iterator[Any]:
+        """Streaming execution: keep at most `window` blocks in flight."""
+        ray = _ray()
+        if self._materialized is not None:
+            yield from self._materialized
+            return
+        ops = self._ops
+        window = DataContext.get_current().streaming_read_window
+
+        @ray.remote
+        def _exec_block(block, ops=ops):
+            t = block() if callable(block) else block
+            return _apply_ops(t, ops)
+
+        pending = []
+        inputs = iter(self._inputs)
+        for inp in itertools.islice(inputs, window):
+            pending.append(_exec_block.remote(inp))
+        while pending:
+            ref = pending.pop(0)
+            nxt = next(inputs, None)
+            if nxt is not None:
+                pending.append(_exec_block.remote(nxt))
+            yield ref
+
+    def _iter_tables(self) -> Iterator[pa.Table]:
+        ray = _ray()
+        for ref in self._iter_block_refs():
+            yield ray.get(ref)
+
+    # ----- consumption -----
+
+    def count(self) -> int:
+        ray = _ray()
+
+        @ray.remote
+        def _count(t):
+            return t.num_rows
+
+        return sum(ray.get([_count.remote(r) for r in self._materialize_refs()]))
+
+    def schema(self):
+        for t in self._iter_tables():
+            if t.num_rows or t.num_columns:
+                return t.schema
+        return None
+
+    def columns(self) -> List[str]:
+        s = self.schema()
+        return list(s.names) if s else []
+
+    def take(self, limit: int = 20) -> List[dict]:
+        out = []
+        for t in self._iter_tables():
+            out.extend(_rows_of(t))
+            if len(out) >= limit:
+                return out[:limit]
+        return out
+
+    def take_all(self) -> List[dict]:
+        out = []
+        for t in self._iter_tables():
+            out.extend(_rows_of(t))
+        return out
+
+    def take_batch(self, batch_size: int = 20, *, batch_format="numpy"):
+        acc = []
+        n = 0
+        for t in self._iter_tables():
+            acc.append(t)
+            n += t.num_rows
+            if n >= batch_size:
+                break
+        t = pa.concat_tables(acc).slice(0, batch_size)
+        return _format_batch(t, batch_format)
+
+    def show(self, limit: int = 20):
+        for r in self.take(limit):
+            print(r)
+
+    def iter_rows(self) -> Iterator[dict]:
+        for t in self._iter_tables():
+            yield from _rows_of(t)
+
+    def iter_batches(
+        self,
+        *,
+        batch_size: Optional[int] = 256,
+        batch_format: Optional[str] = "numpy",
+        drop_last: bool = False,
+        local_shuffle_buffer_size: Optional[int] = None,
+        prefetch_batches: int = 1,
+        **kwargs,
+    ) -> Iterator[Any]:
+        carry: Optional[pa.Table] = None
+        rng = np.random.default_rng(0)
+        for t in self._iter_tables():
+            if carry is not None and carry.num_rows:
+                t = pa.concat_tables([carry, t])
+                carry = None
+            if local_shuffle_buffer_size:
+                idx = rng.permutation(t.num_rows)
+                t = t.take(pa.array(idx))
+            if batch_size is None:
+                yield _format_batch(t, batch_format)
+                continue
+            off = 0
+            while t.num_rows - off >= batch_size:
+                yield _format_batch(t.slice(off, batch_size), batch_format)
+                off += batch_size
+            if off < t.num_rows:
+                carry = t.slice(off)
+        if carry is not None and carry.num_rows and not drop_last:
+            yield _format_batch(carry, batch_format)
+
+    def iter_torch_batches(
+        self,
+        *,
+        batch_size: Optional[int] = 256,
+        dtypes=None,
+        device: Optional[str] = None,
+        collate_fn=None,
+        drop_last: bool = False,
+        local_shuffle_buffer_size: Optional[int] = None,
+        prefetch_batches: int = 1,
+        **kwargs,
+    ):
+        import torch
+
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        for batch in self.iter_batches(
+            batch_size=batch_size,
+            batch_format="numpy",
+            drop_last=drop_last,
+            local_shuffle_buffer_size=local_shuffle_buffer_size,
+        ):
+            if collate_fn is not None:
+                yield collate_fn(batch)
+                continue
+            out = {}
+            for k, v in batch.items():
+                t = torch.as_tensor(np.ascontiguousarray(v))
+                if dtypes is not None:
+                    dt = dtypes.get(k) if isinstance(dtypes, dict) else dtypes
+                    if dt is not None:
+                        t = t.to(dt)
+                out[k] = t.to(device, non_blocking=True)
+            yield out
+
+    def to_pandas(self, limit: Optional[int] = None):
+        tables = list(self._iter_tables())
+        t = pa.concat_tables(tables) if tables else pa.table({})
+        df = t.to_pandas()
+        return df.head(limit) if limit else df
+
+    def to_arrow_refs(self):
+        return list(self._materialize_refs())
+
+    # ----- structure ops -----
+
+    def repartition(self, num_blocks: int, **kwargs) -> "Dataset":
+        ray = _ray()
+        tables = list(self._iter_tables())
+        if not tables:
+            return self
+        t = pa.concat_tables(tables)
+        n = t.num_rows
+        per = max(1, (n + num_blocks - 1) // num_blocks)
+        refs = [
+            ray.put(t.slice(i * per, per))
+            for i in builtins.range(min(num_blocks, (n + per - 1) // per))
+        ]
+        ds = Dataset(refs, [])
+        ds._materialized = refs
+        return ds
+
+    def random_shuffle(self, *, seed: Optional[int] = None, **kwargs) -> "Dataset":
+        ray = _ray()
+        tables = list(self._iter_tables())
+        if not tables:
+            return self
+        t = pa.concat_tables(tables)
+        rng = np.random.default_rng(seed)
+        idx = rng.permutation(t.num_rows)
+        t = t.take(pa.array(idx))
+        nb = max(1, len(tables))
+        per = max(1, (t.num_rows + nb - 1) // nb)
+        refs = [ray.put(t.slice(i * per, per)) for i in builtins.range(nb)]
+        ds = Dataset(refs, [])
+        ds._materialized = refs
+        return ds
+
+    def randomize_block_order(self, *, seed=None) -> "Dataset":
+        import random as _random
+
+        refs = list(self._materialize_refs())
+        _random.Random(seed).shuffle(refs)
+        ds = Dataset(refs, [])
+        ds._materialized = refs
+        return ds
+
+    def split(self, n: int, *, equal: bool = False, locality_hints=None
+              ) -> List["Dataset"]:
+        ray = _ray()
+        tables = list(self._iter_tables())
+        t = pa.concat_tables(tables) if tables else pa.table({})
+        rows = t.num_rows
+        per = rows // n
+        rem = rows % n
+        out = []
+        off = 0
+        for i in builtins.range(n):
+            k = per + (0 if equal else (1 if i < rem else 0))
+            if equal:
+                k = per
+            ref = ray.put(t.slice(off, k))
+            off += k
+            ds = Dataset([ref], [])
+            ds._materialized = [ref]
+            out.append(ds)
+        return out
+
+    def streaming_split(self, n: int, *, equal: bool = True,
+                        locality_hints=None) -> List["Dataset"]:
+        return self.split(n, equal=equal)
+
+    def split_at_indices(self, indices: List[int]) -> List["Dataset"]:
+        ray = _ray()
+        tables = list(self._iter_tables())
+        t = pa.concat_tables(tables) if tables else pa.table({})
+        bounds = [0] + list(indices) + [t.num_rows]
+        out = []
+        for i in builtins.range(len(bounds) - 1):
+            ref = ray.put(t.slice(bounds[i], bounds[i + 1] - bounds[i]))
+            ds = Dataset([ref], [])
+            ds._materialized = [ref]
+            out.append(ds)
+        return out
+
+    def train_test_split(self, test_size: Union[int, float], *,
+                         shuffle: bool = False, seed=None):
+        ds = self.random_shuffle(seed=seed) if shuffle else self
+        n = ds.count()
+        k = int(n * test_size) if isinstance(test_size, float) else test_size
+        train, test = ds.split_at_indices([n - k])
+        return train, test
+
+    def union(self, *others: "Dataset") -> "Dataset":
+        refs = list(self._materialize_refs())
+        for o in others:
+            refs.extend(o._materialize_refs())
+        ds = Dataset(refs, [])
+        ds._materialized = refs
+        return ds
+
+    def limit(self, n: int) -> "Dataset":
+        ray = _ray()
+        acc = []
+        got = 0
+        for t in self._iter_tables():
+            if got + t.num_rows > n:
+                acc.append(t.slice(0, n - got))
+                got = n
+                break
+            acc.append(t)
+            got += t.num_rows
+        t = pa.concat_tables(acc) if acc else pa.table({})
+        ref = ray.put(t)
+        ds = Dataset([ref], [])
+        ds._materialized = [ref]
+        return ds
+
+    def zip(self, other: "Dataset") -> "Dataset":
+        ray = _ray()
+        t1 = pa.concat_tables(list(self._iter_tables()))
+        t2 = pa.concat_tables(list(other._iter_tables()))
+        for name in t2.column_names:
+            col_name = name if name not in t1.column_names else name + "_1"
+            t1 = t1.append_column(col_name, t2.column(name))
+        ref = ray.put(t1)
+        ds = Dataset([ref], [])
+        ds._materialized = [ref]
+        return ds
+
+    # ----- sort / groupby -----
+
+    def sort(self, key: Union[str, List[str]], descending: bool = False) -> "Dataset":
+        ray = _ray()
+        t = pa.concat_tables(list(self._iter_tables()))
+        keys = [key] if isinstance(key, str) else key
+        order = "descending" if descending else "ascending"
+        t = t.sort_by([(k, order) for k in keys])
+        ref = ray.put(t)
+        ds = Dataset([ref], [])
+        ds._materialized = [ref]
+        return ds
+
+    def groupby(self, key: str) -> "GroupedData":
+        return GroupedData(self, key)
+
+    def sum(self, on: str):
+        return self._agg("sum", on)
+
+    def min(self, on: str):
+        return self._agg("min", on)
+
+    def max(self, on: str):
+        return self._agg("max", on)
+
+    def mean(self, on: str):
+        return self._agg("mean", on)
+
+    def std(self, on: str):
+        import pandas as pd  # noqa
+
+        vals = _col_to_numpy(
+            pa.concat_tables(list(self._iter_tables())).column(on)
+        )
+        return float(np.std(vals, ddof=1))
+
+    def _agg(self, how: str, on: str):
+        t = pa.concat_tables(list(self._iter_tables()))
+        vals = _col_to_numpy(t.column(on))
+        return getattr(np, how)(vals).item()
+
+    def unique(self, column: str):
+        t = pa.concat_tables(list(self._iter_tables()))
+        return list(pa.compute.unique(t.column(column)).to_pylist())
+
+    # ----- writes -----
+
+    def write_parquet(self, path: str, **kwargs):
+        import os
+
+        os.makedirs(path, exist_ok=True)
+        for i, t in enumerate(self._iter_tables()):
+            pq.write_table(t, os.path.join(path, f"part-{i:05d}.parquet"))
+
+    def write_csv(self, path: str, **kwargs):
+        import os
+
+        import pyarrow.csv as pcsv
+
+        os.makedirs(path, exist_ok=True)
+        for i, t in enumerate(self._iter_tables()):
+            pcsv.write_csv(t, os.path.join(path, f"part-{i:05d}.csv"))
+
+    def write_json(self, path: str, **kwargs):
+        import json
+        import os
+
+        os.makedirs(path, exist_ok=True)
+        for i, t in enumerate(self._iter_tables()):
+            with open(os.path.join(path, f"part-{i:05d}.json"), "w") as f:
+                for r in _rows_of(t):
+                    f.write(json.dumps({k: _jsonable(v) for k, v in r.items()}) + "\n")
+
+    def write_numpy(self, path: str, *, column: str, **kwargs):
+        import os
+
+        os.makedirs(path, exist_ok=True)
+        for i, t in enumerate(self._iter_tables()):
+            np.save(os.path.join(path, f"part-{i:05d}.npy"),
+                    _col_to_numpy(t.column(column)))
+
+    # ----- misc -----
+
+    def num_blocks(self) -> int:
+        return len(self._inputs)
+
+    def size_bytes(self) -> int:
+        return sum(t.nbytes for t in self._iter_tables())
+
+    def stats(self) -> str:
+        return f"Dataset(num_blocks={self.num_blocks()}, ops={len(self._ops)})"
+
+    def __repr__(self):
+        return f"Dataset(num_blocks={self.num_blocks()})"
+
+
+def _jsonable(v):
+    if isinstance(v, np.generic):
+        return v.item()
+    if isinstance(v, np.ndarray):
+        return v.tolist()
+    if isinstance(v, bytes):
+        return v.decode("utf-8", "replace")
+    return v
+
+
+class GroupedData:
+    def __init__(self, ds: Dataset, key: str):
+        self._ds = ds
+        self._key = key
+
+    def _grouped(self):
+        t = pa.concat_tables(list(self._ds._iter_tables()))
+        return t.group_by(self._key)
+
+    def _wrap(self, t: pa.Table) -> Dataset:
+        ray = _ray()
+        ref = ray.put(t)
+        ds = Dataset([ref], [])
+        ds._materialized = [ref]
+        return ds
+
+    def count(self) -> Dataset:
+        t = self._grouped().aggregate([(self._key, "count")])
+        t = t.rename_columns([self._key, "count()"])
+        return self._wrap(t)
+
+    def sum(self, on: str) -> Dataset:
+        t = self._grouped().aggregate([(on, "sum")])
+        t = t.rename_columns([f"sum({on})" if c == f"{on}_sum" else c
+                              for c in t.column_names])
+        return self._wrap(t)
+
+    def mean(self, on: str) -> Dataset:
+        t = self._grouped().aggregate([(on, "mean")])
+        t = t.rename_columns([f"mean({on})" if c == f"{on}_mean" else c
+                              for c in t.column_names])
+        return self._wrap(t)
+
+    def min(self, on: str) -> Dataset:
+        t = self._grouped().aggregate([(on, "min")])
+        t = t.rename_columns([f"min({on})" if c == f"{on}_min" else c
+                              for c in t.column_names])
+        return self._wrap(t)
+
+    def max(self, on: str) -> Dataset:
+        t = self._grouped().aggregate([(on, "max")])
+        t = t.rename_columns([f"max({on})" if c == f"{on}_max" else c
+                              for c in t.column_names])
+        return self._wrap(t)
+
+    def map_groups(self, fn, *, batch_format="numpy") -> Dataset:
+        t = pa.concat_tables(list(self._ds._iter_tables()))
+        keys = _col_to_numpy(t.column(self._key))
+        out = []
+        for k in np.unique(keys):
+            mask = pa.array(keys == k)
+            sub = t.filter(mask)
+            res = fn(_format_batch(sub, batch_format))
+            out.append(_batch_to_table(res))
+        return self._wrap(pa.concat_tables(out))
+
+
+# ---------------- creation APIs ----------------
+
+
+def from_items(items: List[Any], *, parallelism: int = -1) -> Dataset:
+    ray = _ray()
+    if items and isinstance(items[0], dict):
+        rows = items
+    else:
+        rows = [{"item": it} for it in items]
+    nb = parallelism if parallelism and parallelism > 0 else min(8, max(1, len(rows)))
+    per = max(1, (len(rows) + nb - 1) // nb)
+    refs = []
+    for i in builtins.range(0, len(rows), per):
+        refs.append(ray.put(_rows_to_table(rows[i : i + per])))
+    ds = Dataset(refs, [])
+    ds._materialized = refs
+    return ds
+
+
+def range(n: int, *, parallelism: int = -1, override_num_blocks=None) -> Dataset:
+    ray = _ray()
+    nb = override_num_blocks or (parallelism if parallelism > 0 else min(8, max(1, n)))
+    per = max(1, (n + nb - 1) // nb)
+    refs = []
+    for s in builtins.range(0, n, per):
+        e = min(s + per, n)
+        refs.append(ray.put(pa.table({"id": np.arange(s, e, dtype=np.int64)})))
+    ds = Dataset(refs, [])
+    ds._materialized = refs
+    return ds
+
+
+def range_tensor(n: int, *, shape=(1,), parallelism: int = -1) -> Dataset:
+    ds = range(n, parallelism=parallelism)
+
+    def to_tensor(batch):
+        ids = batch["id"]
+        size = int(np.prod(shape))
+        data = np.repeat(ids[:, None], size, axis=1).reshape((len(ids),) + tuple(shape))
+        return {"data": data, "id": ids}
+
+    return ds.map_batches(to_tensor)
+
+
+def from_numpy(arr: np.ndarray, *, column: str = "data") -> Dataset:
+    ray = _ray()
+    ref = ray.put(_to_table({column: arr}))
+    ds = Dataset([ref], [])
+    ds._materialized = [ref]
+    return ds
+
+
+def from_pandas(df) -> Dataset:
+    ray = _ray()
+    ref = ray.put(pa.Table.from_pandas(df, preserve_index=False))
+    ds = Dataset([ref], [])
+    ds._materialized = [ref]
+    return ds
+
+
+def from_arrow(t: pa.Table) -> Dataset:
+    ray = _ray()
+    ref = ray.put(t)
+    ds = Dataset([ref], [])
+    ds._materialized = [ref]
+    return ds
+
+
+def _expand_paths(paths, suffix=None) -> List[str]:
+    import glob as g
+    import os
+
+    if isinstance(paths, str):
+        paths = [paths]
+    out = []
+    for p in paths:
+        if os.path.isdir(p):
+            out.extend(sorted(g.glob(os.path.join(p, "**", "*"), recursive=True)))
+        else:
+            out.extend(sorted(g.glob(p)) or [p])
+    out = [p for p in out if os.path.isfile(p)]
+    if suffix:
+        matched = [p for p in out if p.endswith(suffix)]
+        out = matched or out
+    return out
+
+
+def read_parquet(paths, *, columns=None, parallelism: int = -1, **kwargs) -> Dataset:
+    ray = _ray()
+    files = _expand_paths(paths, ".parquet")
+
+    @ray.remote
+    def _read(f, columns=columns):
+        return pq.read_table(f, columns=columns)
+
+    refs = [_read.remote(f) for f in files]
+    ds = Dataset(refs, [])
+    ds._materialized = refs
+    return ds
+
+
+def read_csv(paths, *, parallelism: int = -1, **kwargs) -> Dataset:
+    ray = _ray()
+    files = _expand_paths(paths, ".csv")
+
+    @ray.remote
+    def _read(f):
+        import pyarrow.csv as pcsv
+
+        return pcsv.read_csv(f)
+
+    refs = [_read.remote(f) for f in files]
+    ds = Dataset(refs, [])
+    ds._materialized = refs
+    return ds
+
+
+def read_json(paths, *, parallelism: int = -1, **kwargs) -> Dataset:
+    ray = _ray()
+    files = _expand_paths(paths, ".json")
+
+    @ray.remote
+    def _read(f):
+        import pyarrow.json as pjson
+
+        return pjson.read_json(f)
+
+    refs = [_read.remote(f) for f in files]
+    ds = Dataset(refs, [])
+    ds._materialized = refs
+    return ds
+
+
+def read_binary_files(paths, *, include_paths: bool = False, **kwargs) -> Dataset:
+    ray = _ray()
+    files = _expand_paths(paths)
+
+    @ray.remote
+    def _read(f, include_paths=include_paths):
+        with open(f, "rb") as fh:
+            data = fh.read()
+        cols = {"bytes": pa.array([data], type=pa.binary())}
+        if include_paths:
+            cols["path"] = pa.array([f])
+        return pa.table(cols)
+
+    refs = [_read.remote(f) for f in files]
+    ds = Dataset(refs, [])
+    ds._materialized = refs
+    return ds

@@ -1,0 +1,153 @@
+"""ray_amd.data tests (reference model: python/ray/data/tests/)."""
+import numpy as np
+import pytest
+
+import ray_amd as ray
+import ray_amd.data as rd
+
+
+def test_range_count_take(ray_start_regular):
+    ds = rd.range(100)
+    assert ds.count() == 100
+    rows = ds.take(5)
+    assert [r["id"] for r in rows] == [0, 1, 2, 3, 4]
+
+
+def test_from_items_map(ray_start_regular):
+    ds = rd.from_items([{"x": i} for i in range(20)])
+    out = ds.map(lambda r: {"y": r["x"] * 2}).take_all()
+    assert sorted(r["y"] for r in out) == [2 * i for i in range(20)]
+
+
+def test_map_batches_numpy(ray_start_regular):
+    ds = rd.range(64).map_batches(lambda b: {"id": b["id"] * 10})
+    assert sorted(r["id"] for r in ds.take_all()) == [10 * i for i in range(64)]
+
+
+def test_filter_flat_map(ray_start_regular):
+    ds = rd.range(10).filter(lambda r: r["id"] % 2 == 0)
+    assert ds.count() == 5
+    ds2 = rd.range(3).flat_map(lambda r: [{"v": r["id"]}, {"v": r["id"]}])
+    assert ds2.count() == 6
+
+
+def test_batch_iteration(ray_start_regular):
+    ds = rd.range(100)
+    batches = list(ds.iter_batches(batch_size=32))
+    sizes = [len(b["id"]) for b in batches]
+    assert sum(sizes) == 100
+    assert sizes[:3] == [32, 32, 32]
+
+
+def test_iter_torch_batches(ray_start_regular):
+    import torch
+
+    ds = rd.range(16)
+    batches = list(ds.iter_torch_batches(batch_size=8, device="cpu"))
+    assert len(batches) == 2
+    assert isinstance(batches[0]["id"], torch.Tensor)
+
+
+def test_repartition_split(ray_start_regular):
+    ds = rd.range(100).repartition(4)
+    assert ds.num_blocks() == 4
+    parts = ds.split(2)
+    assert parts[0].count() + parts[1].count() == 100
+
+
+def test_random_shuffle_preserves_rows(ray_start_regular):
+    ds = rd.range(50).random_shuffle(seed=42)
+    vals = sorted(r["id"] for r in ds.take_all())
+    assert vals == list(range(50))
+
+
+def test_sort_groupby(ray_start_regular):
+    ds = rd.from_items([{"k": i % 3, "v": i} for i in range(12)])
+    s = ds.sort("v", descending=True).take(1)
+    assert s[0]["v"] == 11
+    agg = ds.groupby("k").sum("v").take_all()
+    total = {int(r["k"]): r for r in agg}
+    # groups: k=0 -> 0+3+6+9=18, k=1 -> 1+4+7+10=22, k=2 -> 2+5+8+11=26
+    sums = sorted(float(list(r.values())[1]) for r in agg)
+    assert sums == [18.0, 22.0, 26.0]
+
+
+def test_aggregates(ray_start_regular):
+    ds = rd.range(10)
+    assert ds.sum("id") == 45
+    assert ds.min("id") == 0
+    assert ds.max("id") == 9
+    assert ds.mean("id") == pytest.approx(4.5)
+
+
+def test_limit_union_zip(ray_start_regular):
+    a = rd.range(10).limit(3)
+    assert a.count() == 3
+    b = rd.range(5)
+    u = a.union(b)
+    assert u.count() == 8
+    z = rd.range(4).zip(rd.range(4).map_batches(lambda x: {"other": x["id"] + 1}))
+    rows = z.take_all()
+    assert all(r["other"] == r["id"] + 1 for r in rows)
+
+
+def test_parquet_roundtrip(ray_start_regular, tmp_path):
+    ds = rd.range(30)
+    path = str(tmp_path / "pq")
+    ds.write_parquet(path)
+    back = rd.read_parquet(path)
+    assert back.count() == 30
+    assert sorted(r["id"] for r in back.take_all()) == list(range(30))
+
+
+def test_csv_roundtrip(ray_start_regular, tmp_path):
+    ds = rd.from_items([{"a": i, "b": f"s{i}"} for i in range(10)])
+    path = str(tmp_path / "csv")
+    ds.write_csv(path)
+    back = rd.read_csv(path)
+    assert back.count() == 10
+
+
+def test_tensor_columns(ray_start_regular):
+    arr = np.random.rand(16, 4).astype(np.float32)
+    ds = rd.from_numpy(arr)
+    b = ds.take_batch(16)
+    np.testing.assert_allclose(b["data"], arr, rtol=1e-6)
+
+
+def test_preprocessor_standard_scaler(ray_start_regular):
+    ds = rd.from_items([{"x": float(i)} for i in range(10)])
+    from ray_amd.data import StandardScaler
+
+    sc = StandardScaler(["x"]).fit(ds)
+    out = sc.transform(ds)
+    vals = np.array([r["x"] for r in out.take_all()])
+    assert abs(vals.mean()) < 1e-6
+
+
+def test_train_test_split(ray_start_regular):
+    tr, te = rd.range(100).train_test_split(0.2)
+    assert tr.count() == 80 and te.count() == 20
+
+
+def test_dataset_shard_in_trainer(ray_start_regular, tmp_path):
+    from ray_amd.train import RunConfig, ScalingConfig
+    from ray_amd.train.torch import TorchTrainer
+
+    def loop(config):
+        import ray_amd.train as train
+
+        shard = train.get_dataset_shard("train")
+        n = sum(len(b["id"]) for b in shard.iter_batches(batch_size=None))
+        train.report({"rows": n})
+
+    ds = rd.range(40)
+    t = TorchTrainer(
+        loop,
+        scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(name="shard", storage_path=str(tmp_path)),
+        datasets={"train": ds},
+    )
+    res = t.fit()
+    assert res.error is None
+    assert res.metrics["rows"] == 20
