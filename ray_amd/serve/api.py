@@ -1,0 +1,830 @@
+"""Serve public API + controller/replica/router/proxy internals.
+
+Reference counterparts: serve/api.py:522 (@deployment), :952 (run),
+serve/handle.py (DeploymentHandle), _private/controller.py:133
+(ServeController), _private/replica.py:3249 (ReplicaActor),
+_private/proxy.py (HTTP proxy), request_router/pow_2_router.py:27
+(power-of-two-choices), batching.py (serve.batch), autoscaling_state.py
+(request-rate autoscaling).
+"""
+from __future__ import annotations
+
+import asyncio
+import random
+import time
+from dataclasses import dataclass, field
+from typing import Any, Callable, Dict, List, Optional
+
+# starlette State instances (FastAPI app.state) break pickle: their
+# __getattr__ recurses during reconstruction before _state exists.
+# Register a proper reducer so serve deployments with FastAPI ingress
+# apps serialize (the reference ships its own patched cloudpickle).
+try:
+    from starlette.datastructures import State as _StarletteState
+
+    def _rebuild_starlette_state(d):
+        return _StarletteState(d)
+
+    def _reduce_starlette_state(self):
+        return (_rebuild_starlette_state, (dict(self._state),))
+
+    _StarletteState.__reduce__ = _reduce_starlette_state
+except ImportError:
+    pass
+
+SERVE_CONTROLLER_NAME = "SERVE_CONTROLLER_ACTOR"
+SERVE_PROXY_NAME = "SERVE_PROXY_ACTOR"
+SERVE_NAMESPACE = "serve"
+
+
+def _ray():
+    import ray_amd as ray
+
+    return ray
+
+
+# --------------------------------------------------------------------------
+# Deployment declaration
+# --------------------------------------------------------------------------
+
+
+@dataclass
+class AutoscalingConfig:
+    min_replicas: int = 1
+    max_replicas: int = 4
+    target_ongoing_requests: float = 2.0
+    upscale_delay_s: float = 3.0
+    downscale_delay_s: float = 30.0
+
+
+class Deployment:
+    def __init__(self, func_or_class, name: str, *, num_replicas=1,
+                 max_ongoing_requests=100, ray_actor_options=None,
+                 user_config=None, autoscaling_config=None,
+                 health_check_period_s=10.0, **kwargs):
+        self.func_or_class = func_or_class
+        self.name = name
+        self.num_replicas = num_replicas
+        self.max_ongoing_requests = max_ongoing_requests
+        self.ray_actor_options = ray_actor_options or {}
+        self.user_config = user_config
+        if isinstance(autoscaling_config, dict):
+            autoscaling_config = AutoscalingConfig(**autoscaling_config)
+        self.autoscaling_config = autoscaling_config
+
+    def options(self, **kwargs) -> "Deployment":
+        merged = dict(
+            num_replicas=self.num_replicas,
+            max_ongoing_requests=self.max_ongoing_requests,
+            ray_actor_options=self.ray_actor_options,
+            user_config=self.user_config,
+            autoscaling_config=self.autoscaling_config,
+        )
+        name = kwargs.pop("name", self.name)
+        merged.update(kwargs)
+        return Deployment(self.func_or_class, name, **merged)
+
+    def bind(self, *args, **kwargs) -> "Application":
+        return Application(DeploymentNode(self, args, kwargs))
+
+    def __call__(self, *a, **k):
+        raise RuntimeError(
+            "Deployments cannot be called directly; use .bind() and serve.run"
+        )
+
+
+class DeploymentNode:
+    def __init__(self, deployment: Deployment, args, kwargs):
+        self.deployment = deployment
+        self.args = args
+        self.kwargs = kwargs
+
+
+class Application:
+    def __init__(self, root: DeploymentNode):
+        self.root = root
+
+    def _collect(self) -> List[DeploymentNode]:
+        """Topological order, children first."""
+        seen: List[DeploymentNode] = []
+
+        def visit(node: DeploymentNode):
+            for a in list(node.args) + list(node.kwargs.values()):
+                if isinstance(a, Application):
+                    visit(a.root)
+                elif isinstance(a, DeploymentNode):
+                    visit(a)
+            if node not in seen:
+                seen.append(node)
+
+        visit(self.root)
+        return seen
+
+
+def deployment(_func_or_class=None, *, name=None, num_replicas=1,
+               max_ongoing_requests=100, ray_actor_options=None,
+               user_config=None, autoscaling_config=None, **kwargs):
+    """@serve.deployment decorator (reference: serve/api.py:522)."""
+
+    def make(fc):
+        n = name or getattr(fc, "__name__", "deployment")
+        if isinstance(num_replicas, str) and num_replicas == "auto":
+            asc = autoscaling_config or AutoscalingConfig()
+            nr = asc.min_replicas
+        else:
+            asc = autoscaling_config
+            nr = num_replicas
+        return Deployment(
+            fc, n, num_replicas=nr, max_ongoing_requests=max_ongoing_requests,
+            ray_actor_options=ray_actor_options, user_config=user_config,
+            autoscaling_config=asc,
+        )
+
+    if _func_or_class is not None:
+        return make(_func_or_class)
+    return make
+
+
+def ingress(asgi_app):
+    """@serve.ingress(fastapi_app): attach an ASGI app to the class."""
+
+    def dec(cls):
+        cls.__serve_asgi_app__ = asgi_app
+        return cls
+
+    return dec
+
+
+# --------------------------------------------------------------------------
+# Replica
+# --------------------------------------------------------------------------
+
+
+class ReplicaActor:
+    def __init__(self, cls_or_fn, init_args, init_kwargs, user_config):
+        import inspect
+
+        self._ongoing = 0
+        self._total = 0
+        self._asgi_client = None
+        if inspect.isclass(cls_or_fn):
+            self._callable = cls_or_fn(*init_args, **init_kwargs)
+            asgi = getattr(cls_or_fn, "__serve_asgi_app__", None)
+            if asgi is not None:
+                import httpx
+
+                self._asgi_app = asgi
+                self._asgi_client = httpx.AsyncClient(
+                    transport=httpx.ASGITransport(app=asgi),
+                    base_url="http://serve",
+                )
+                # let FastAPI-style ingress classes access self via app state
+                try:
+                    asgi.state.serve_instance = self._callable
+                except Exception:
+                    pass
+        else:
+            self._callable = cls_or_fn
+        if user_config is not None and hasattr(self._callable, "reconfigure"):
+            self._callable.reconfigure(user_config)
+
+    async def handle_request(self, method: str, args, kwargs):
+        self._ongoing += 1
+        self._total += 1
+        try:
+            target = (
+                self._callable
+                if method == "__call__" and not hasattr(self._callable, "__call__")
+                else getattr(self._callable, method, None)
+            )
+            if target is None and method == "__call__":
+                target = self._callable
+            if asyncio.iscoroutinefunction(target):
+                return await target(*args, **kwargs)
+            # sync user code runs off-loop so it may block on .result()
+            loop = asyncio.get_running_loop()
+            r = await loop.run_in_executor(
+                None, lambda: target(*args, **kwargs)
+            )
+            if asyncio.iscoroutine(r):
+                r = await r
+            return r
+        finally:
+            self._ongoing -= 1
+
+    async def handle_http(self, method: str, path: str, query: str,
+                          headers: dict, body: bytes):
+        self._ongoing += 1
+        self._total += 1
+        try:
+            if self._asgi_client is not None:
+                url = path + (f"?{query}" if query else "")
+                resp = await self._asgi_client.request(
+                    method, url, headers=headers, content=body
+                )
+                return resp.status_code, dict(resp.headers), resp.content
+            req = SimpleRequest(method, path, query, headers, body)
+            r = self._callable(req) if callable(self._callable) else None
+            if asyncio.iscoroutine(r):
+                r = await r
+            return _encode_http_result(r)
+        finally:
+            self._ongoing -= 1
+
+    def get_stats(self):
+        return {"ongoing": self._ongoing, "total": self._total}
+
+    def check_health(self):
+        if hasattr(self._callable, "check_health"):
+            self._callable.check_health()
+        return True
+
+
+class SimpleRequest:
+    """Minimal starlette-Request-compatible object for plain ingress."""
+
+    def __init__(self, method, path, query, headers, body):
+        self.method = method
+        self.path = path
+        self.headers = headers or {}
+        self._body = body
+        from urllib.parse import parse_qsl
+
+        self.query_params = dict(parse_qsl(query or ""))
+
+    async def body(self) -> bytes:
+        return self._body
+
+    async def json(self):
+        import json
+
+        return json.loads(self._body or b"null")
+
+
+def _encode_http_result(r):
+    import json
+
+    if isinstance(r, (bytes, bytearray)):
+        return 200, {"content-type": "application/octet-stream"}, bytes(r)
+    if isinstance(r, str):
+        return 200, {"content-type": "text/plain"}, r.encode()
+    return (
+        200,
+        {"content-type": "application/json"},
+        json.dumps(r, default=str).encode(),
+    )
+
+
+# --------------------------------------------------------------------------
+# Controller
+# --------------------------------------------------------------------------
+
+
+class ServeController:
+    """Reconciles deployments -> replica actors; serves routing tables."""
+
+    def __init__(self):
+        # app -> {deployment_name: {"spec":..., "replicas": [handles]}}
+        self.apps: Dict[str, dict] = {}
+        self.routes: Dict[str, str] = {}  # route_prefix -> app
+        self.version = 0
+
+    def deploy_application(self, name: str, route_prefix: str,
+                           specs: List[dict], ingress_name: str):
+        ray = _ray()
+        old = self.apps.pop(name, None)
+        if old:
+            self._teardown(old)
+        app = {"deployments": {}, "ingress": ingress_name}
+        for spec in specs:
+            replicas = [
+                self._start_replica(spec) for _ in range(spec["num_replicas"])
+            ]
+            app["deployments"][spec["name"]] = {
+                "spec": spec,
+                "replicas": replicas,
+                "last_scale": time.time(),
+            }
+        self.apps[name] = app
+        if route_prefix:
+            self.routes[route_prefix] = name
+        self.version += 1
+        return True
+
+    def _start_replica(self, spec: dict):
+        ray = _ray()
+        import cloudpickle
+
+        cls_or_fn, init_args, init_kwargs = cloudpickle.loads(spec["payload"])
+        opts = dict(spec.get("ray_actor_options") or {})
+        opts.setdefault("num_cpus", 0.1)
+        opts["max_concurrency"] = max(16, spec.get("max_ongoing_requests", 100))
+        RA = ray.remote(ReplicaActor)
+        return RA.options(**opts).remote(
+            cls_or_fn, init_args, init_kwargs, spec.get("user_config")
+        )
+
+    def _teardown(self, app: dict):
+        ray = _ray()
+        for d in app["deployments"].values():
+            for r in d["replicas"]:
+                try:
+                    ray.kill(r)
+                except Exception:
+                    pass
+
+    def get_routing(self, app_name: str, deployment: Optional[str] = None):
+        app = self.apps.get(app_name)
+        if app is None:
+            return None
+        dname = deployment or app["ingress"]
+        d = app["deployments"].get(dname)
+        if d is None:
+            return None
+        return {"replicas": d["replicas"], "version": self.version,
+                "deployment": dname}
+
+    def resolve_route(self, path: str):
+        best = None
+        for prefix, app in self.routes.items():
+            if path.startswith(prefix.rstrip("/")) or prefix == "/":
+                if best is None or len(prefix) > len(best[0]):
+                    best = (prefix, app)
+        return best
+
+    def list_routes(self):
+        return dict(self.routes)
+
+    def delete_application(self, name: str):
+        app = self.apps.pop(name, None)
+        if app:
+            self._teardown(app)
+        self.routes = {k: v for k, v in self.routes.items() if v != name}
+        self.version += 1
+        return app is not None
+
+    def status(self):
+        out = {}
+        ray = _ray()
+        for name, app in self.apps.items():
+            deps = {}
+            for dname, d in app["deployments"].items():
+                deps[dname] = {
+                    "status": "HEALTHY",
+                    "replica_states": {"RUNNING": len(d["replicas"])},
+                }
+            out[name] = {"status": "RUNNING", "deployments": deps}
+        return out
+
+    async def autoscale_once(self):
+        """One reconciliation pass of request-based autoscaling."""
+        ray = _ray()
+        for app in self.apps.values():
+            for d in app["deployments"].values():
+                spec = d["spec"]
+                asc = spec.get("autoscaling")
+                if not asc:
+                    continue
+                stats = ray.get(
+                    [r.get_stats.remote() for r in d["replicas"]], timeout=30
+                )
+                ongoing = sum(s["ongoing"] for s in stats)
+                n = len(d["replicas"])
+                target = asc["target_ongoing_requests"]
+                desired = max(
+                    asc["min_replicas"],
+                    min(asc["max_replicas"],
+                        int((ongoing + target - 1) // target) or asc["min_replicas"]),
+                )
+                now = time.time()
+                if desired > n:
+                    for _ in range(desired - n):
+                        d["replicas"].append(self._start_replica(spec))
+                    d["last_scale"] = now
+                    self.version += 1
+                elif desired < n and now - d["last_scale"] > asc.get(
+                    "downscale_delay_s", 30.0
+                ):
+                    for r in d["replicas"][desired:]:
+                        try:
+                            ray.kill(r)
+                        except Exception:
+                            pass
+                    d["replicas"] = d["replicas"][:desired]
+                    d["last_scale"] = now
+                    self.version += 1
+        return self.version
+
+    def ping(self):
+        return "pong"
+
+
+# --------------------------------------------------------------------------
+# Handle + router (power of two choices)
+# --------------------------------------------------------------------------
+
+
+class DeploymentResponse:
+    def __init__(self, ref):
+        self._ref = ref
+
+    def result(self, timeout_s: Optional[float] = None):
+        return _ray().get(self._ref, timeout=timeout_s)
+
+    def __await__(self):
+        from ray_amd._core import runtime as _rt
+
+        rt = _rt.global_runtime()
+
+        async def _get():
+            return (await rt.get_async([self._ref]))[0]
+
+        return _get().__await__()
+
+    @property
+    def object_ref(self):
+        return self._ref
+
+
+class DeploymentHandle:
+    def __init__(self, app_name: str, deployment_name: Optional[str] = None,
+                 method_name: str = "__call__"):
+        self.app_name = app_name
+        self.deployment_name = deployment_name
+        self.method_name = method_name
+        self._replicas: List = []
+        self._version = -1
+        self._counts: Dict[int, int] = {}
+
+    def _refresh(self):
+        ray = _ray()
+        ctrl = ray.get_actor(SERVE_CONTROLLER_NAME, namespace=SERVE_NAMESPACE)
+        info = ray.get(
+            ctrl.get_routing.remote(self.app_name, self.deployment_name)
+        )
+        if info is None:
+            raise RuntimeError(
+                f"no deployment {self.deployment_name} in app {self.app_name}"
+            )
+        self._replicas = info["replicas"]
+        self._version = info["version"]
+        self.deployment_name = info["deployment"]
+        self._counts = {i: 0 for i in range(len(self._replicas))}
+
+    def _pick(self) -> int:
+        n = len(self._replicas)
+        if n == 1:
+            return 0
+        i, j = random.sample(range(n), 2)
+        return i if self._counts.get(i, 0) <= self._counts.get(j, 0) else j
+
+    def remote(self, *args, **kwargs) -> DeploymentResponse:
+        if not self._replicas:
+            self._refresh()
+        for attempt in range(3):
+            idx = self._pick()
+            replica = self._replicas[idx]
+            self._counts[idx] = self._counts.get(idx, 0) + 1
+            try:
+                ref = replica.handle_request.remote(
+                    self.method_name, args, kwargs
+                )
+                resp = DeploymentResponse(ref)
+                self._decr_later(idx)
+                return resp
+            except Exception:
+                self._refresh()
+        raise RuntimeError("could not route request")
+
+    def _decr_later(self, idx):
+        # decremented optimistically; precise per-request accounting
+        # happens replica-side (get_stats)
+        self._counts[idx] = max(0, self._counts.get(idx, 1) - 1)
+
+    def options(self, *, method_name: Optional[str] = None, **kwargs):
+        h = DeploymentHandle(self.app_name, self.deployment_name,
+                             method_name or self.method_name)
+        h._replicas = self._replicas
+        h._counts = self._counts
+        return h
+
+    def __getattr__(self, item):
+        if item.startswith("_"):
+            raise AttributeError(item)
+        return self.options(method_name=item)
+
+    def __reduce__(self):
+        return (DeploymentHandle,
+                (self.app_name, self.deployment_name, self.method_name))
+
+
+# --------------------------------------------------------------------------
+# HTTP proxy
+# --------------------------------------------------------------------------
+
+
+class ProxyActor:
+    def __init__(self, port: int):
+        self.port = port
+        self._handles: Dict[str, DeploymentHandle] = {}
+        self._server_task = None
+
+    async def start_server(self):
+        import uvicorn
+
+        ray = _ray()
+
+        async def asgi(scope, receive, send):
+            if scope["type"] != "http":
+                return
+            body = b""
+            while True:
+                msg = await receive()
+                body += msg.get("body", b"")
+                if not msg.get("more_body"):
+                    break
+            path = scope["path"]
+            loop = asyncio.get_running_loop()
+            # sync ray calls must leave the event loop (executor thread)
+            route = await loop.run_in_executor(None, self._resolve_route, path)
+            if route is None:
+                await _send_response(send, 404, {}, b'{"error":"not found"}')
+                return
+            prefix, app_name = route
+            h = self._handles.get(app_name)
+            if h is None:
+                h = self._handles[app_name] = DeploymentHandle(app_name)
+                await loop.run_in_executor(None, h._refresh)
+            sub_path = path[len(prefix.rstrip("/")):] or "/"
+            headers = {
+                k.decode(): v.decode() for k, v in scope.get("headers", [])
+            }
+            if not h._replicas:
+                await loop.run_in_executor(None, h._refresh)
+            from ray_amd._core import runtime as _rt
+
+            rt = _rt.global_runtime()
+            qs = scope.get("query_string", b"").decode()
+            last_exc = None
+            for _attempt in range(2):
+                idx = h._pick()
+                replica = h._replicas[idx]
+                try:
+                    ref = replica.handle_http.remote(
+                        scope["method"], sub_path, qs, headers, body
+                    )
+                    status, hdrs, content = (await rt.get_async([ref], 120))[0]
+                    await _send_response(send, status, hdrs, content)
+                    return
+                except Exception as e:  # replica died: refresh + retry
+                    last_exc = e
+                    try:
+                        await loop.run_in_executor(None, h._refresh)
+                    except Exception:
+                        break
+            await _send_response(
+                send, 500, {}, f'{{"error":"{last_exc}"}}'.encode()
+            )
+
+        config = uvicorn.Config(
+            asgi, host="127.0.0.1", port=self.port, log_level="warning",
+            loop="asyncio",
+        )
+        self._server = uvicorn.Server(config)
+        self._server_task = asyncio.ensure_future(self._server.serve())
+        for _ in range(100):
+            if self._server.started:
+                return True
+            await asyncio.sleep(0.05)
+        return False
+
+    def _resolve_route(self, path):
+        ray = _ray()
+        ctrl = ray.get_actor(SERVE_CONTROLLER_NAME, namespace=SERVE_NAMESPACE)
+        return ray.get(ctrl.resolve_route.remote(path))
+
+    def ping(self):
+        return "pong"
+
+
+async def _send_response(send, status, headers, body):
+    await send(
+        {
+            "type": "http.response.start",
+            "status": status,
+            "headers": [
+                (k.encode(), v.encode())
+                for k, v in (headers or {}).items()
+                if k.lower() not in ("content-length", "transfer-encoding")
+            ]
+            + [(b"content-length", str(len(body)).encode())],
+        }
+    )
+    await send({"type": "http.response.body", "body": body})
+
+
+# --------------------------------------------------------------------------
+# module-level API
+# --------------------------------------------------------------------------
+
+_http_port = 8000
+_started = False
+
+
+def start(detached: bool = True, http_options: Optional[dict] = None, **kw):
+    global _http_port, _started
+    if http_options:
+        _http_port = http_options.get("port", _http_port)
+    _ensure_controller()
+    _started = True
+
+
+def _ensure_controller():
+    ray = _ray()
+    try:
+        return ray.get_actor(SERVE_CONTROLLER_NAME, namespace=SERVE_NAMESPACE)
+    except ValueError:
+        Ctrl = ray.remote(ServeController)
+        ctrl = Ctrl.options(
+            name=SERVE_CONTROLLER_NAME, namespace=SERVE_NAMESPACE,
+            num_cpus=0.1, max_concurrency=16,
+        ).remote()
+        ray.get(ctrl.ping.remote())
+        return ctrl
+
+
+def _ensure_proxy(port: int):
+    ray = _ray()
+    try:
+        return ray.get_actor(SERVE_PROXY_NAME, namespace=SERVE_NAMESPACE)
+    except ValueError:
+        P = ray.remote(ProxyActor)
+        proxy = P.options(
+            name=SERVE_PROXY_NAME, namespace=SERVE_NAMESPACE, num_cpus=0.1,
+            max_concurrency=64,
+        ).remote(port)
+        ok = ray.get(proxy.start_server.remote(), timeout=30)
+        if not ok:
+            raise RuntimeError("serve HTTP proxy failed to start")
+        return proxy
+
+
+def run(app: Application, *, name: str = "default", route_prefix: str = "/",
+        blocking: bool = False, _local_testing_mode: bool = False,
+        http: bool = True, port: Optional[int] = None) -> DeploymentHandle:
+    import cloudpickle
+
+    ray = _ray()
+    ctrl = _ensure_controller()
+    nodes = app._collect()
+    # assign unique per-node deployment names (reference suffixes
+    # duplicate bindings: Adder, Adder_1, ...)
+    node_names: Dict[int, str] = {}
+    used: Dict[str, int] = {}
+    for node in nodes:
+        base = node.deployment.name
+        k = used.get(base, 0)
+        node_names[id(node)] = base if k == 0 else f"{base}_{k}"
+        used[base] = k + 1
+    specs = []
+    for node in nodes:
+        d = node.deployment
+
+        def resolve(v):
+            if isinstance(v, Application):
+                return DeploymentHandle(name, node_names[id(v.root)])
+            if isinstance(v, DeploymentNode):
+                return DeploymentHandle(name, node_names[id(v)])
+            return v
+
+        args = tuple(resolve(a) for a in node.args)
+        kwargs = {k: resolve(v) for k, v in node.kwargs.items()}
+        asc = None
+        if d.autoscaling_config:
+            a = d.autoscaling_config
+            asc = {
+                "min_replicas": a.min_replicas,
+                "max_replicas": a.max_replicas,
+                "target_ongoing_requests": a.target_ongoing_requests,
+                "downscale_delay_s": a.downscale_delay_s,
+            }
+        specs.append(
+            {
+                "name": node_names[id(node)],
+                "num_replicas": d.num_replicas,
+                "max_ongoing_requests": d.max_ongoing_requests,
+                "ray_actor_options": d.ray_actor_options,
+                "user_config": d.user_config,
+                "autoscaling": asc,
+                "payload": cloudpickle.dumps(
+                    (d.func_or_class, args, kwargs)
+                ),
+            }
+        )
+    ingress_name = node_names[id(app.root)]
+    ray.get(
+        ctrl.deploy_application.remote(name, route_prefix, specs, ingress_name),
+        timeout=120,
+    )
+    if http:
+        _ensure_proxy(port or _http_port)
+    h = DeploymentHandle(name)
+    h._refresh()
+    return h
+
+
+def get_app_handle(name: str = "default") -> DeploymentHandle:
+    h = DeploymentHandle(name)
+    h._refresh()
+    return h
+
+
+def get_deployment_handle(deployment_name: str, app_name: str = "default"
+                          ) -> DeploymentHandle:
+    h = DeploymentHandle(app_name, deployment_name)
+    h._refresh()
+    return h
+
+
+def status():
+    ray = _ray()
+    ctrl = _ensure_controller()
+    return ray.get(ctrl.status.remote())
+
+
+def delete(name: str, _blocking: bool = True):
+    ray = _ray()
+    ctrl = _ensure_controller()
+    ray.get(ctrl.delete_application.remote(name))
+
+
+def shutdown():
+    ray = _ray()
+    for n in (SERVE_PROXY_NAME, SERVE_CONTROLLER_NAME):
+        try:
+            a = ray.get_actor(n, namespace=SERVE_NAMESPACE)
+            ray.kill(a)
+        except Exception:
+            pass
+
+
+# --------------------------------------------------------------------------
+# serve.batch
+# --------------------------------------------------------------------------
+
+
+def batch(_func=None, *, max_batch_size: int = 10,
+          batch_wait_timeout_s: float = 0.01):
+    """Dynamic request batching (reference: serve/batching.py)."""
+
+    def dec(func):
+        state = {"queue": None, "task": None}
+
+        async def flusher(queue):
+            while True:
+                items = [await queue.get()]
+                deadline = asyncio.get_running_loop().time() + batch_wait_timeout_s
+                while len(items) < max_batch_size:
+                    to = deadline - asyncio.get_running_loop().time()
+                    if to <= 0:
+                        break
+                    try:
+                        items.append(
+                            await asyncio.wait_for(queue.get(), timeout=to)
+                        )
+                    except asyncio.TimeoutError:
+                        break
+                args = [it[0] for it in items]
+                futs = [it[1] for it in items]
+                try:
+                    self_arg = items[0][2]
+                    if self_arg is not None:
+                        results = await func(self_arg, args)
+                    else:
+                        results = await func(args)
+                    for f, r in zip(futs, results):
+                        if not f.done():
+                            f.set_result(r)
+                except Exception as e:
+                    for f in futs:
+                        if not f.done():
+                            f.set_exception(e)
+
+        async def wrapper(*call_args):
+            if state["queue"] is None:
+                state["queue"] = asyncio.Queue()
+                state["task"] = asyncio.ensure_future(flusher(state["queue"]))
+            if len(call_args) == 2:
+                self_arg, item = call_args
+            else:
+                self_arg, item = None, call_args[0]
+            fut = asyncio.get_running_loop().create_future()
+            await state["queue"].put((item, fut, self_arg))
+            return await fut
+
+        wrapper._is_serve_batch = True
+        return wrapper
+
+    if _func is not None:
+        return dec(_func)
+    return dec
