@@ -125,16 +125,23 @@ class Raylet:
         asyncio.ensure_future(self._resource_reporter())
 
     async def _resource_reporter(self):
+        # Send on change AND at least once per second: the GCS deducts
+        # optimistically when placing actors, and a changed-only report
+        # can race that deduction and leave the GCS cache stale forever
+        # (reference: ray_syncer periodic RESOURCE_VIEW broadcasts).
         last = None
+        last_time = 0.0
         while True:
             snap = dict(self.avail)
-            if snap != last:
+            now = time.time()
+            if snap != last or now - last_time > 1.0:
                 try:
                     self.gcs.notify(
                         "report_resources",
                         {"node_id": self.node_id, "available": snap},
                     )
                     last = snap
+                    last_time = now
                 except Exception:
                     pass
             await asyncio.sleep(0.2)

@@ -1,0 +1,27 @@
+"""ray_amd.tune — hyperparameter tuning (reference: python/ray/tune/).
+
+Tuner.fit (tuner.py:332) over actor-based trial execution
+(execution/tune_controller.py), basic-variant search (grid + random),
+ASHA scheduler (schedulers/async_hyperband.py), ResultGrid.
+"""
+from .impl import (  # noqa: F401
+    ASHAScheduler,
+    Callback,
+    FIFOScheduler,
+    ResultGrid,
+    TuneConfig,
+    Tuner,
+    choice,
+    grid_search,
+    loguniform,
+    qrandint,
+    randint,
+    randn,
+    report,
+    run,
+    sample_from,
+    uniform,
+    with_parameters,
+    with_resources,
+)
+from ..train.session import get_checkpoint, get_context  # noqa: F401

@@ -1,0 +1,485 @@
+"""Tune implementation.
+
+Reference counterparts: tune/tuner.py:43 Tuner (.fit :332),
+tune/execution/tune_controller.py (actor-based trial loop),
+tune/search/basic_variant.py (grid/random variant generation),
+tune/schedulers/async_hyperband.py ASHA, tune/result_grid.py.
+"""
+from __future__ import annotations
+
+import itertools
+import math
+import os
+import random
+import threading
+import time
+import traceback
+from dataclasses import dataclass, field
+from typing import Any, Callable, Dict, List, Optional
+
+import numpy as np
+
+from ..train.checkpoint import Checkpoint
+from ..train.config import Result, RunConfig
+from ..train.session import TrainSession, _set_session
+
+# ---------------- search space primitives ----------------
+
+
+class Domain:
+    def sample(self, rng: random.Random):
+        raise NotImplementedError
+
+
+class _Uniform(Domain):
+    def __init__(self, lo, hi):
+        self.lo, self.hi = lo, hi
+
+    def sample(self, rng):
+        return rng.uniform(self.lo, self.hi)
+
+
+class _LogUniform(Domain):
+    def __init__(self, lo, hi):
+        self.lo, self.hi = lo, hi
+
+    def sample(self, rng):
+        return math.exp(rng.uniform(math.log(self.lo), math.log(self.hi)))
+
+
+class _Randn(Domain):
+    def __init__(self, mean, sd):
+        self.mean, self.sd = mean, sd
+
+    def sample(self, rng):
+        return rng.gauss(self.mean, self.sd)
+
+
+class _RandInt(Domain):
+    def __init__(self, lo, hi):
+        self.lo, self.hi = lo, hi
+
+    def sample(self, rng):
+        return rng.randrange(self.lo, self.hi)
+
+
+class _QRandInt(Domain):
+    def __init__(self, lo, hi, q):
+        self.lo, self.hi, self.q = lo, hi, q
+
+    def sample(self, rng):
+        return (rng.randrange(self.lo, self.hi + 1) // self.q) * self.q
+
+
+class _Choice(Domain):
+    def __init__(self, options):
+        self.options = list(options)
+
+    def sample(self, rng):
+        return rng.choice(self.options)
+
+
+class _SampleFrom(Domain):
+    def __init__(self, fn):
+        self.fn = fn
+
+    def sample(self, rng):
+        return self.fn(None)
+
+
+class _GridSearch:
+    def __init__(self, values):
+        self.values = list(values)
+
+
+def uniform(lo, hi):
+    return _Uniform(lo, hi)
+
+
+def loguniform(lo, hi):
+    return _LogUniform(lo, hi)
+
+
+def randn(mean=0.0, sd=1.0):
+    return _Randn(mean, sd)
+
+
+def randint(lo, hi):
+    return _RandInt(lo, hi)
+
+
+def qrandint(lo, hi, q=1):
+    return _QRandInt(lo, hi, q)
+
+
+def choice(options):
+    return _Choice(options)
+
+
+def sample_from(fn):
+    return _SampleFrom(fn)
+
+
+def grid_search(values):
+    return _GridSearch(values)
+
+
+def generate_variants(space: Dict[str, Any], num_samples: int, seed: int = 0
+                      ) -> List[Dict[str, Any]]:
+    """Basic variant generator: cartesian product of grid_search values,
+    each repeated num_samples times with random Domains resampled."""
+    rng = random.Random(seed)
+    grid_keys = [k for k, v in space.items() if isinstance(v, _GridSearch)]
+    grids = [space[k].values for k in grid_keys]
+    combos = list(itertools.product(*grids)) if grid_keys else [()]
+    variants = []
+    for _ in range(num_samples):
+        for combo in combos:
+            cfg = {}
+            for k, v in space.items():
+                if isinstance(v, _GridSearch):
+                    cfg[k] = combo[grid_keys.index(k)]
+                elif isinstance(v, Domain):
+                    cfg[k] = v.sample(rng)
+                elif isinstance(v, dict):
+                    cfg[k] = generate_variants(v, 1, rng.randrange(1 << 30))[0]
+                else:
+                    cfg[k] = v
+            variants.append(cfg)
+    return variants
+
+
+# ---------------- schedulers ----------------
+
+
+class FIFOScheduler:
+    def on_trial_result(self, trial_id: str, iteration: int, metric_value) -> str:
+        return "CONTINUE"
+
+
+class ASHAScheduler:
+    """Async successive halving (reference:
+    tune/schedulers/async_hyperband.py): rungs at grace_period *
+    reduction_factor^k; a trial stops at a rung if it is below the top
+    1/reduction_factor quantile of completed results at that rung."""
+
+    def __init__(self, metric: Optional[str] = None, mode: str = "max",
+                 max_t: int = 100, grace_period: int = 1,
+                 reduction_factor: float = 3, time_attr="training_iteration"):
+        self.metric = metric
+        self.mode = mode
+        self.max_t = max_t
+        self.grace = grace_period
+        self.rf = reduction_factor
+        self.rungs: Dict[int, List[float]] = {}
+        r = grace_period
+        while r < max_t:
+            self.rungs[int(r)] = []
+            r *= reduction_factor
+
+    def on_trial_result(self, trial_id, iteration, metric_value) -> str:
+        if metric_value is None:
+            return "CONTINUE"
+        v = float(metric_value) if self.mode == "max" else -float(metric_value)
+        if iteration >= self.max_t:
+            return "STOP"
+        rung = None
+        for r in sorted(self.rungs, reverse=True):
+            if iteration >= r:
+                rung = r
+                break
+        if rung is None:
+            return "CONTINUE"
+        recorded = self.rungs[rung]
+        recorded.append(v)
+        k = max(1, int(len(recorded) / self.rf))
+        cutoff = sorted(recorded, reverse=True)[k - 1]
+        return "CONTINUE" if v >= cutoff else "STOP"
+
+
+# ---------------- config ----------------
+
+
+@dataclass
+class TuneConfig:
+    metric: Optional[str] = None
+    mode: str = "max"
+    num_samples: int = 1
+    max_concurrent_trials: Optional[int] = None
+    scheduler: Any = None
+    search_alg: Any = None
+    time_budget_s: Optional[float] = None
+
+
+# ---------------- trial runner actor ----------------
+
+
+class _TrialActor:
+    def __init__(self):
+        self.session: Optional[TrainSession] = None
+        self.done = False
+        self.error: Optional[str] = None
+
+    def run(self, fn_bytes: bytes, config: dict, storage_dir: str,
+            trial_name: str):
+        import cloudpickle
+
+        fn = cloudpickle.loads(fn_bytes)
+        os.makedirs(storage_dir, exist_ok=True)
+        self.session = TrainSession(0, 1, 0, 1, storage_dir, trial_name)
+        self.done = False
+
+        def _run():
+            _set_session(self.session)
+            try:
+                if isinstance(fn, type):
+                    # class Trainable: step() loop until scheduler stops us
+                    inst = fn(config)
+                    while not self.session.stop_requested.is_set():
+                        res = inst.step()
+                        self.session.report(res)
+                        if res.get("done"):
+                            break
+                else:
+                    fn(config)
+            except BaseException:
+                self.error = traceback.format_exc()
+            finally:
+                self.done = True
+                _set_session(None)
+
+        threading.Thread(target=_run, daemon=True).start()
+        return True
+
+    def fetch(self):
+        out = []
+        if self.session:
+            while not self.session.results_queue.empty():
+                out.append(self.session.results_queue.get_nowait())
+        return {"results": out, "done": self.done, "error": self.error}
+
+    def request_stop(self):
+        if self.session:
+            self.session.stop_requested.set()
+        return True
+
+    def latest_checkpoint_path(self):
+        if self.session and self.session.latest_checkpoint:
+            return self.session.latest_checkpoint.path
+        return None
+
+
+# ---------------- Tuner ----------------
+
+
+class ResultGrid:
+    def __init__(self, results: List[Result], metric=None, mode="max"):
+        self._results = results
+        self._metric = metric
+        self._mode = mode
+
+    def __len__(self):
+        return len(self._results)
+
+    def __getitem__(self, i):
+        return self._results[i]
+
+    def __iter__(self):
+        return iter(self._results)
+
+    @property
+    def errors(self):
+        return [r.error for r in self._results if r.error]
+
+    def get_best_result(self, metric: Optional[str] = None,
+                        mode: Optional[str] = None) -> Result:
+        metric = metric or self._metric
+        mode = mode or self._mode
+        valid = [r for r in self._results
+                 if r.metrics and metric in r.metrics]
+        if not valid:
+            raise ValueError(f"no results with metric {metric!r}")
+        key = lambda r: r.metrics[metric]  # noqa: E731
+        return max(valid, key=key) if mode == "max" else min(valid, key=key)
+
+    def get_dataframe(self):
+        import pandas as pd
+
+        return pd.DataFrame([r.metrics for r in self._results if r.metrics])
+
+
+class Tuner:
+    def __init__(self, trainable, *, param_space: Optional[dict] = None,
+                 tune_config: Optional[TuneConfig] = None,
+                 run_config: Optional[RunConfig] = None):
+        self._trainable = trainable
+        self.param_space = param_space or {}
+        self.tune_config = tune_config or TuneConfig()
+        self.run_config = run_config or RunConfig(name=f"tune_{int(time.time())}")
+
+    def fit(self) -> ResultGrid:
+        import cloudpickle
+
+        import ray_amd as ray
+
+        tc = self.tune_config
+        storage = self.run_config.resolved_storage_path()
+        os.makedirs(storage, exist_ok=True)
+
+        # Trainer-as-trainable (reference: Train trainers are Trainables)
+        from ..train.trainer import DataParallelTrainer
+
+        if isinstance(self._trainable, DataParallelTrainer):
+            return self._fit_trainer_trials(storage)
+
+        variants = generate_variants(self.param_space, tc.num_samples)
+        scheduler = tc.scheduler or FIFOScheduler()
+        max_conc = tc.max_concurrent_trials or min(8, max(1, len(variants)))
+        fn_bytes = cloudpickle.dumps(self._trainable)
+        Actor = ray.remote(_TrialActor)
+
+        trials = []  # dicts: actor, config, rows, done, error, it
+        pending = list(enumerate(variants))
+        running: List[dict] = []
+        finished: List[dict] = []
+        t_start = time.time()
+
+        def launch(idx_cfg):
+            idx, cfg = idx_cfg
+            a = Actor.options(num_cpus=1).remote()
+            name = f"trial_{idx:05d}"
+            # fire-and-forget: the actor may stay PENDING until a CPU
+            # frees up; the fetch loop below is fully non-blocking.
+            a.run.remote(fn_bytes, cfg, os.path.join(storage, name), name)
+            t = {"actor": a, "config": cfg, "rows": [], "done": False,
+                 "error": None, "it": 0, "name": name, "stopped": False,
+                 "ckpt": None, "fetch_ref": None}
+            running.append(t)
+
+        while pending or running:
+            while pending and len(running) < max_conc:
+                launch(pending.pop(0))
+            # issue fetches and wait for any to become ready
+            ref_to_trial = {}
+            for t in running:
+                if t["fetch_ref"] is None:
+                    t["fetch_ref"] = t["actor"].fetch.remote()
+                ref_to_trial[t["fetch_ref"]] = t
+            if not ref_to_trial:
+                time.sleep(0.02)
+                continue
+            ready, _ = ray.wait(
+                list(ref_to_trial), num_returns=len(ref_to_trial), timeout=0.2
+            )
+            for ref in ready:
+                t = ref_to_trial[ref]
+                t["fetch_ref"] = None
+                try:
+                    st = ray.get(ref, timeout=30)
+                except (ray.exceptions.RayActorError, ray.exceptions.RayError):
+                    t["error"] = "trial actor died"
+                    t["done"] = True
+                    running.remove(t)
+                    finished.append(t)
+                    continue
+                for r in st["results"]:
+                    t["it"] += 1
+                    row = dict(r["metrics"])
+                    row["training_iteration"] = t["it"]
+                    row.update(
+                        {f"config/{k}": v for k, v in t["config"].items()}
+                    )
+                    t["rows"].append(row)
+                    if r.get("checkpoint_path"):
+                        t["ckpt"] = r["checkpoint_path"]
+                    mv = r["metrics"].get(tc.metric) if tc.metric else None
+                    decision = scheduler.on_trial_result(t["name"], t["it"], mv)
+                    if decision == "STOP" and not t["stopped"]:
+                        t["stopped"] = True
+                        t["actor"].request_stop.remote()
+                if st["done"] or (t["stopped"] and not st["results"]):
+                    t["error"] = t["error"] or st["error"]
+                    t["done"] = True
+                    running.remove(t)
+                    finished.append(t)
+                    try:
+                        ray.kill(t["actor"])
+                    except Exception:
+                        pass
+            if tc.time_budget_s and time.time() - t_start > tc.time_budget_s:
+                for t in running:
+                    t["actor"].request_stop.remote()
+
+        results = []
+        for t in finished:
+            metrics = t["rows"][-1] if t["rows"] else None
+            results.append(
+                Result(
+                    metrics=metrics,
+                    checkpoint=Checkpoint(t["ckpt"]) if t["ckpt"] else None,
+                    path=os.path.join(storage, t["name"]),
+                    error=RuntimeError(t["error"]) if t["error"] else None,
+                )
+            )
+        return ResultGrid(results, tc.metric, tc.mode)
+
+    def _fit_trainer_trials(self, storage: str) -> ResultGrid:
+        tc = self.tune_config
+        space = self.param_space.get("train_loop_config", {})
+        variants = generate_variants(space, tc.num_samples) if space else [
+            {} for _ in range(tc.num_samples)
+        ]
+        results = []
+        for i, cfg in enumerate(variants):
+            trainer = self._trainable
+            merged = dict(trainer._config)
+            merged.update(cfg)
+            t2 = type(trainer)(
+                trainer._fn,
+                train_loop_config=merged,
+                scaling_config=trainer.scaling_config,
+                run_config=RunConfig(
+                    name=f"{self.run_config.name}_t{i}", storage_path=storage
+                ),
+                datasets=trainer.datasets,
+            )
+            res = t2.fit()
+            if res.metrics is not None:
+                res.metrics.update({f"config/{k}": v for k, v in cfg.items()})
+            results.append(res)
+        return ResultGrid(results, tc.metric, tc.mode)
+
+
+def report(metrics: Dict[str, Any], checkpoint=None):
+    from ..train.session import report as _r
+
+    _r(metrics, checkpoint)
+
+
+def with_parameters(fn, **params):
+    def wrapped(config):
+        return fn(config, **params)
+
+    return wrapped
+
+
+def with_resources(fn, resources):
+    fn._tune_resources = resources
+    return fn
+
+
+class Callback:
+    def on_trial_result(self, *a, **k):
+        pass
+
+
+def run(trainable, config=None, num_samples=1, metric=None, mode="max",
+        scheduler=None, **kwargs):
+    """Legacy tune.run API shim over Tuner."""
+    tuner = Tuner(
+        trainable,
+        param_space=config or {},
+        tune_config=TuneConfig(metric=metric, mode=mode,
+                               num_samples=num_samples, scheduler=scheduler),
+    )
+    return tuner.fit()

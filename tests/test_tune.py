@@ -1,0 +1,126 @@
+"""Tune tests (reference model: python/ray/tune/tests/)."""
+import pytest
+
+import ray_amd as ray
+from ray_amd import tune
+from ray_amd.tune import ASHAScheduler, TuneConfig, Tuner
+
+
+def trainable_quadratic(config):
+    # minimize (x-3)^2 reported over a few iterations
+    x = config["x"]
+    for i in range(3):
+        tune.report({"score": -((x - 3.0) ** 2), "x": x})
+
+
+def test_grid_search(ray_start_regular, tmp_path):
+    from ray_amd.train import RunConfig
+
+    tuner = Tuner(
+        trainable_quadratic,
+        param_space={"x": tune.grid_search([0.0, 1.0, 3.0, 5.0])},
+        tune_config=TuneConfig(metric="score", mode="max"),
+        run_config=RunConfig(name="grid", storage_path=str(tmp_path)),
+    )
+    grid = tuner.fit()
+    assert len(grid) == 4
+    best = grid.get_best_result()
+    assert best.metrics["x"] == 3.0
+
+
+def test_random_search_uniform(ray_start_regular, tmp_path):
+    from ray_amd.train import RunConfig
+
+    tuner = Tuner(
+        trainable_quadratic,
+        param_space={"x": tune.uniform(0, 6)},
+        tune_config=TuneConfig(metric="score", mode="max", num_samples=6),
+        run_config=RunConfig(name="rand", storage_path=str(tmp_path)),
+    )
+    grid = tuner.fit()
+    assert len(grid) == 6
+    assert all(0 <= r.metrics["x"] <= 6 for r in grid if r.metrics)
+
+
+def test_variant_generation():
+    from ray_amd.tune.impl import generate_variants
+
+    vs = generate_variants(
+        {"a": tune.grid_search([1, 2]), "b": tune.choice([10]), "c": 5},
+        num_samples=2,
+    )
+    assert len(vs) == 4
+    assert all(v["c"] == 5 and v["b"] == 10 for v in vs)
+    assert sorted(v["a"] for v in vs) == [1, 1, 2, 2]
+
+
+def test_asha_stops_bad_trials(ray_start_regular, tmp_path):
+    from ray_amd.train import RunConfig
+
+    def slow_trainable(config):
+        import time
+
+        for i in range(10):
+            tune.report({"score": config["x"] * (i + 1)})
+            time.sleep(0.02)
+
+    sched = ASHAScheduler(metric="score", mode="max", max_t=10,
+                          grace_period=2, reduction_factor=2)
+    tuner = Tuner(
+        slow_trainable,
+        param_space={"x": tune.grid_search([0.1, 0.2, 1.0, 2.0])},
+        tune_config=TuneConfig(metric="score", mode="max", scheduler=sched,
+                               max_concurrent_trials=4),
+        run_config=RunConfig(name="asha", storage_path=str(tmp_path)),
+    )
+    grid = tuner.fit()
+    best = grid.get_best_result()
+    assert best.metrics["config/x"] == 2.0
+    # at least one weak trial should have been stopped early
+    iters = [r.metrics["training_iteration"] for r in grid if r.metrics]
+    assert min(iters) < 10
+
+
+def test_trial_error_captured(ray_start_regular, tmp_path):
+    from ray_amd.train import RunConfig
+
+    def failing(config):
+        if config["x"] == 1:
+            raise ValueError("bad trial")
+        tune.report({"ok": 1})
+
+    tuner = Tuner(
+        failing,
+        param_space={"x": tune.grid_search([0, 1])},
+        tune_config=TuneConfig(metric="ok", mode="max"),
+        run_config=RunConfig(name="err", storage_path=str(tmp_path)),
+    )
+    grid = tuner.fit()
+    assert len(grid.errors) == 1
+    best = grid.get_best_result()
+    assert best.metrics["ok"] == 1
+
+
+def test_tuner_with_torch_trainer(ray_start_regular, tmp_path):
+    from ray_amd.train import RunConfig, ScalingConfig
+    from ray_amd.train.torch import TorchTrainer
+
+    def loop(config):
+        import ray_amd.train as train
+
+        train.report({"lr_used": config["lr"]})
+
+    trainer = TorchTrainer(
+        loop,
+        scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(name="base", storage_path=str(tmp_path)),
+    )
+    tuner = Tuner(
+        trainer,
+        param_space={"train_loop_config": {"lr": tune.grid_search([0.1, 0.2])}},
+        tune_config=TuneConfig(metric="lr_used", mode="max"),
+        run_config=RunConfig(name="tt", storage_path=str(tmp_path)),
+    )
+    grid = tuner.fit()
+    assert len(grid) == 2
+    assert grid.get_best_result().metrics["lr_used"] == 0.2
