@@ -506,7 +506,18 @@ def main():
 
     async def run():
         await gcs.start()
-        await asyncio.Event().wait()
+        had_nodes = False
+        idle_since = None
+        while True:
+            await asyncio.sleep(1.0)
+            alive = any(n.alive for n in gcs.nodes.values())
+            if alive:
+                had_nodes = True
+                idle_since = None
+            elif had_nodes:
+                idle_since = idle_since or time.time()
+                if time.time() - idle_since > 10.0:
+                    return  # all raylets gone: session over
 
     asyncio.run(run())
 

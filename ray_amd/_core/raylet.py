@@ -305,7 +305,16 @@ class Raylet:
             if not self._fits(pool_avail, req):
                 break
             if not self._idle_task_workers:
-                if self._starting == 0 or len(self._pending) > self._starting:
+                # bounded pool (reference: worker_pool.h soft limit):
+                # never exceed what the CPU resource could run anyway
+                n_task_workers = sum(
+                    1 for w in self.workers.values() if w.kind == "task"
+                )
+                cap = int(max(self.resources_total.get("CPU", 1) * 2, 8))
+                if (
+                    self._starting < min(len(self._pending), 4)
+                    and n_task_workers + self._starting < cap
+                ):
                     self._spawn_worker()
                 break
             w = self._idle_task_workers.popleft()
@@ -596,7 +605,10 @@ def main():
             with open(args.ready_file, "w") as f:
                 f.write(raylet.addr + "\n" + raylet.node_id.hex())
         try:
-            await asyncio.Event().wait()
+            # exit (and kill workers) when the GCS goes away — prevents
+            # daemon leaks when the driver is killed hard
+            while raylet.gcs.connected:
+                await asyncio.sleep(0.5)
         finally:
             raylet.shutdown_workers()
 

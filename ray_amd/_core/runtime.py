@@ -332,8 +332,17 @@ class CoreRuntime:
     def _free_owned(self, oid: bytes):
         ent = self.memory_store.pop(oid, None)
         self._events.pop(oid, None)
-        self._mmaps.pop(oid, None)
+        mapped = self._mmaps.pop(oid, None)
         if ent is not None and ent[0] == "store" and not self._closed:
+            # recycle same-node segments into the hot pool (plasma-arena
+            # equivalent) unless we still hold a mapping of it
+            if ent[1] == self.raylet_addr and mapped is None:
+                path = store.shm_path(self.shm_dir, oid)
+                try:
+                    fsize = os.path.getsize(path)
+                    store._segment_pool.release(path, fsize)
+                except OSError:
+                    pass
 
             async def _free():
                 try:
@@ -356,6 +365,9 @@ class CoreRuntime:
             ev.set()
 
     def _store_put_threadsafe(self, oid: bytes, entry: tuple):
+        # immediate visibility for same-thread readers (e.g. a ref freed
+        # right after put); event wakeups still run on the loop
+        self.memory_store[oid] = entry
         self.loop.call_soon_threadsafe(self._store_put, oid, entry)
 
     async def _store_wait(self, oid: bytes, timeout=None) -> tuple:
@@ -407,6 +419,16 @@ class CoreRuntime:
         self._call_sync(self.raylet.call("seal_object", {"id": oid, "size": size}))
 
     def get_sync(self, refs: List[ObjectRef], timeout=None) -> List[Any]:
+        # fast path: everything already materialized in the memory store
+        ms = self.memory_store
+        out = []
+        for r in refs:
+            ent = ms.get(r.id)
+            if ent is None or ent[0] != "val":
+                break
+            out.append(ent[1])
+        else:
+            return out
         return self._call_sync(self.get_async(refs, timeout))
 
     async def get_async(self, refs: List[ObjectRef], timeout=None) -> List[Any]:
@@ -706,7 +728,10 @@ class CoreRuntime:
                 return l
         fut = self.loop.create_future()
         pool.queue.append(fut)
-        if pool.requests_in_flight < len(pool.queue):
+        # cap outstanding lease requests (reference:
+        # LeaseRequestRateLimiter) — unbounded requests pile up at the
+        # raylet and starve later submitters
+        if pool.requests_in_flight < min(len(pool.queue), 16):
             pool.requests_in_flight += 1
             asyncio.ensure_future(self._request_lease(pool))
         return await fut
@@ -747,8 +772,9 @@ class CoreRuntime:
             if not fut.done():
                 fut.set_result(lease)
                 return
-        # nobody waiting: keep lease idle; return after grace period
-        self.loop.call_later(0.25, self._maybe_return_idle, pool, lease)
+        # nobody waiting: keep lease idle; return after a short grace
+        # period (long enough for submit->get->submit reuse)
+        self.loop.call_later(0.05, self._maybe_return_idle, pool, lease)
 
     def _release_or_reuse(self, pool: _LeasePool, lease: _Lease):
         self._grant_to_queue(pool, lease)
@@ -795,7 +821,8 @@ class CoreRuntime:
             self._add_submitted_ref(r.id)
 
         res = dict(options.get("resources") or {})
-        res["CPU"] = options.get("num_cpus", 1) or 0
+        if options.get("num_cpus") is not None:
+            res["CPU"] = options["num_cpus"]
         if options.get("num_gpus"):
             res["GPU"] = options["num_gpus"]
         pg = options.get("placement_group")

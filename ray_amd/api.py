@@ -25,8 +25,12 @@ _cluster: Optional[_node.LocalCluster] = None
 _namespace = "default"
 
 DEFAULT_TASK_OPTIONS = dict(num_cpus=1, num_gpus=0, num_returns=1, max_retries=3)
+# Reference semantics (actor.py): an actor with no explicit resource
+# request reserves NOTHING while alive (scheduled by 1-CPU availability
+# but released after start); explicit num_cpus/num_gpus are reserved
+# for the actor's lifetime.
 DEFAULT_ACTOR_OPTIONS = dict(
-    num_cpus=1, num_gpus=0, max_restarts=0, max_concurrency=1, lifetime=None
+    num_cpus=None, num_gpus=0, max_restarts=0, max_concurrency=1, lifetime=None
 )
 
 
@@ -179,10 +183,23 @@ class RemoteFunction:
             "Remote functions cannot be called directly; use .remote()"
         )
 
+    def __reduce__(self):
+        # never pickle the runtime/export caches (they hold sockets)
+        return (_rebuild_remote_function,
+                (cloudpickle.dumps(self._function), self._options))
+
     def bind(self, *args, **kwargs):
         from .dag import FunctionNode
 
         return FunctionNode(self, args, kwargs)
+
+
+def _rebuild_remote_function(fn_bytes, options):
+    return RemoteFunction(cloudpickle.loads(fn_bytes), options)
+
+
+def _rebuild_actor_class(cls_bytes, options):
+    return ActorClass(cloudpickle.loads(cls_bytes), options)
 
 
 def _normalize_scheduling(opts: dict) -> dict:
@@ -317,6 +334,10 @@ class ActorClass:
         from .dag import ClassNode
 
         return ClassNode(self, args, kwargs)
+
+    def __reduce__(self):
+        return (_rebuild_actor_class,
+                (cloudpickle.dumps(self._cls), self._options))
 
     def __call__(self, *a, **k):
         raise TypeError("Actors cannot be instantiated directly; use .remote()")
