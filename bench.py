@@ -61,6 +61,10 @@ def main():
             world_size=world_size,
         )
 
+    if args.model == "resnet50":
+        run_resnet(args, world_size, rank, local_rank, device, use_cpu)
+        return
+
     from ray_amd.models.llama import CONFIGS, LlamaModel
     from ray_amd.ops import FusedAdamW
 
@@ -159,6 +163,88 @@ def main():
         print(json.dumps(result))
 
     if distributed:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+def run_resnet(args, world_size, rank, local_rank, device, use_cpu):
+    """North-star config 2: ResNet-50 DDP bf16 (autocast), synthetic
+    ImageNet-shaped data."""
+    import torch.nn.functional as F
+
+    from ray_amd.models.resnet import ResNet50
+
+    torch.manual_seed(1234 + rank)
+    model = ResNet50().to(device)
+    if world_size > 1:
+        from torch.nn.parallel import DistributedDataParallel as DDP
+
+        model = DDP(model, device_ids=None if use_cpu else [local_rank],
+                    bucket_cap_mb=args.bucket_mb,
+                    gradient_as_bucket_view=True)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
+                          weight_decay=1e-4)
+    B = args.micro_batch if args.micro_batch != 8 else 256
+    if use_cpu:
+        B = min(B, 8)
+    res = 224 if not use_cpu else 64
+    amp_dtype = torch.bfloat16
+
+    def step():
+        x = torch.randn(B, 3, res, res, device=device)
+        y = torch.randint(0, 1000, (B,), device=device)
+        with torch.autocast(device_type=device.type, dtype=amp_dtype,
+                            enabled=not use_cpu):
+            loss = F.cross_entropy(model(x), y)
+        loss.backward()
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+        return loss
+
+    def sync():
+        if world_size > 1:
+            import torch.distributed as dist
+
+            dist.barrier()
+        if not use_cpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        loss = step()
+    sync()
+    elapsed = time.perf_counter() - t0
+    if world_size > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if not use_cpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    total = args.steps * B * world_size
+    if rank == 0:
+        print(json.dumps({
+            "metric": "train_samples_per_sec",
+            "value": round(total / elapsed, 2),
+            "unit": "images/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if not use_cpu else "fp32",
+            "data": "synthetic",
+            "config": {"model": "resnet50", "global_batch": B * world_size,
+                       "resolution": res, "parallelism": f"dp{world_size}",
+                       "final_loss": round(float(loss.detach().float().cpu()), 4)},
+        }))
+    if world_size > 1:
         import torch.distributed as dist
 
         dist.destroy_process_group()

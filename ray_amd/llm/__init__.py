@@ -1,0 +1,7 @@
+"""ray_amd.llm — LLM serving on Serve (reference: python/ray/llm/,
+build_openai_app + vLLM engine integration). MI355X-native: the engine
+is our own bf16 Llama with hipGraph-captured decode
+(models/llama.py GraphedDecoder) — no external inference engine.
+"""
+from .engine import LLMConfig, LLMEngine  # noqa: F401
+from .serving import LLMServer, build_llm_deployment, build_openai_app  # noqa: F401

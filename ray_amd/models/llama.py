@@ -168,6 +168,109 @@ class LlamaModel(nn.Module):
         return sum(p.numel() for p in self.parameters())
 
 
+class GraphedDecoder:
+    """hipGraph-captured single-token decode (north-star config 5:
+    "hipGraph-captured decode"; HIP graphs instead of a tracing
+    compiler).
+
+    The decode step is made shape- and address-static so one capture
+    replays for every token: KV caches are preallocated at max_T, the
+    new K/V row is written via index_copy_ with a device position
+    tensor, and attention runs over the full cache with an additive
+    mask buffer updated in place. All tensors the graph reads are
+    static buffers; per-token updates are in-place copies."""
+
+    def __init__(self, model: "LlamaModel", batch_size: int, max_T: int,
+                 device):
+        self.m = model
+        cfg = model.cfg
+        self.B = batch_size
+        self.max_T = max_T
+        self.device = device
+        hd = cfg.hidden_size // cfg.num_heads
+        dt = model.dtype
+        self.cache_k = [
+            torch.zeros(batch_size, cfg.num_kv_heads, max_T, hd,
+                        device=device, dtype=dt)
+            for _ in range(cfg.num_layers)
+        ]
+        self.cache_v = [torch.zeros_like(k) for k in self.cache_k]
+        # static IO buffers
+        self.in_tok = torch.zeros(batch_size, 1, dtype=torch.long, device=device)
+        self.pos = torch.zeros(1, dtype=torch.long, device=device)
+        self.mask = torch.full((1, 1, 1, max_T), float("-inf"),
+                               device=device, dtype=torch.float32)
+        self.out_logits = None
+        self.graph = None
+
+    def _step(self):
+        m = self.m
+        cfg = m.cfg
+        B = self.B
+        hd = cfg.hidden_size // cfg.num_heads
+        x = m.embed(self.in_tok)  # [B,1,H]
+        cos = m.cosT.index_select(0, self.pos).view(1, 1, 1, hd // 2)
+        sin = m.sinT.index_select(0, self.pos).view(1, 1, 1, hd // 2)
+        for li, layer in enumerate(m.layers):
+            h = layer.attn_norm(x)
+            at = layer.attn
+            q = at.q_proj(h).view(B, 1, cfg.num_heads, hd)
+            k = at.k_proj(h).view(B, 1, cfg.num_kv_heads, hd)
+            v = at.v_proj(h).view(B, 1, cfg.num_kv_heads, hd)
+            q = _rope_one(q, cos, sin)
+            k = _rope_one(k, cos, sin)
+            # static-address cache update
+            self.cache_k[li].index_copy_(
+                2, self.pos, k.permute(0, 2, 1, 3).to(self.cache_k[li].dtype)
+            )
+            self.cache_v[li].index_copy_(
+                2, self.pos, v.permute(0, 2, 1, 3).to(self.cache_v[li].dtype)
+            )
+            attn = F.scaled_dot_product_attention(
+                q.permute(0, 2, 1, 3),
+                self.cache_k[li],
+                self.cache_v[li],
+                attn_mask=self.mask.to(q.dtype),
+                enable_gqa=True,
+            )
+            x = x + at.o_proj(attn.permute(0, 2, 1, 3).reshape(B, 1, -1))
+            x = x + layer.mlp(layer.mlp_norm(x))
+        x = m.final_norm(x)
+        return m.lm_head(x)[:, 0]
+
+    def capture(self):
+        assert self.device.type == "cuda"
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):  # warmup allocations
+                self.out_logits = self._step()
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.out_logits = self._step()
+
+    @torch.no_grad()
+    def decode(self, tok: torch.Tensor, position: int) -> torch.Tensor:
+        self.in_tok.copy_(tok.view(self.B, 1))
+        self.pos.fill_(position)
+        self.mask[..., : position + 1] = 0.0
+        if self.graph is not None:
+            self.graph.replay()
+            return self.out_logits
+        return self._step()
+
+
+def _rope_one(x, cos, sin):
+    """RoPE for a single position; x [B,1,Hn,D], cos/sin [1,1,1,D/2]."""
+    half = x.shape[-1] // 2
+    x1 = x[..., :half].float()
+    x2 = x[..., half:].float()
+    c = cos.to(torch.float32)
+    s = sin.to(torch.float32)
+    return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1).to(x.dtype)
+
+
 class KVCache:
     """Per-layer KV cache for decode, preallocated in HBM."""
 

@@ -1,0 +1,156 @@
+"""`ray_amd` CLI (reference: python/ray/scripts/scripts.py — start :800,
+stop :1341, status :2386, timeline :2289, memory :2346,
+microbenchmark :2275). Invoke as `python -m ray_amd <cmd>`."""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import signal
+import sys
+import time
+
+
+def cmd_start(args):
+    from ray_amd._core import node as _node
+
+    if not args.head:
+        print("only --head is supported in-round; worker nodes join via "
+              "cluster_utils.Cluster", file=sys.stderr)
+        return 1
+    cluster = _node.start_local_cluster(
+        num_cpus=args.num_cpus, num_gpus=args.num_gpus
+    )
+    with open(os.path.join(cluster.session_dir, "head_pids"), "w") as f:
+        json.dump(
+            {"gcs": cluster.gcs_proc.pid, "raylet": cluster.raylet_proc.pid}, f
+        )
+    print(f"started head; session_dir={cluster.session_dir}")
+    print(f"connect with: ray_amd.init(address='{cluster.session_dir}')")
+    return 0
+
+
+def cmd_stop(args):
+    from ray_amd._core.node import RAY_AMD_TMP
+
+    latest = os.path.join(RAY_AMD_TMP, "latest_session")
+    if not os.path.exists(latest):
+        print("no running session found")
+        return 0
+    with open(latest) as f:
+        session = f.read().strip()
+    pid_file = os.path.join(session, "head_pids")
+    if os.path.exists(pid_file):
+        with open(pid_file) as f:
+            pids = json.load(f)
+        for name, pid in pids.items():
+            try:
+                os.kill(pid, signal.SIGTERM)
+                print(f"stopped {name} (pid {pid})")
+            except OSError:
+                pass
+    return 0
+
+
+def cmd_status(args):
+    import ray_amd as ray
+
+    ray.init(address=args.address or "auto")
+    print("Nodes:")
+    for n in ray.nodes():
+        state = "ALIVE" if n["Alive"] else "DEAD"
+        print(f"  {n['NodeID'][:12]} {state} total={n['Resources']} "
+              f"avail={n['Available']}")
+    from ray_amd.util import state as state_api
+
+    actors = state_api.list_actors()
+    alive = sum(1 for a in actors if a["state"] == "ALIVE")
+    print(f"Actors: {alive} alive / {len(actors)} total")
+    st = state_api.list_objects()[0]
+    print(f"Object store: {st['num_objects_in_store']} objects, "
+          f"{st['store_used_bytes'] / 1e6:.1f} MB used")
+    ray.shutdown()
+    return 0
+
+
+def cmd_timeline(args):
+    import ray_amd as ray
+
+    ray.init(address=args.address or "auto")
+    from ray_amd.util import state as state_api
+
+    tasks = state_api.list_tasks()
+    events = []
+    for t in tasks:
+        events.append(
+            {
+                "cat": "task",
+                "name": t["name"],
+                "ph": "X",
+                "ts": t["start_time_ms"] * 1000,
+                "dur": (t["end_time_ms"] - t["start_time_ms"]) * 1000,
+                "pid": t["worker_pid"],
+                "tid": t["worker_pid"],
+            }
+        )
+    out = args.output or f"timeline-{time.strftime('%Y%m%d-%H%M%S')}.json"
+    with open(out, "w") as f:
+        json.dump(events, f)
+    print(f"wrote {len(events)} events to {out} (open in chrome://tracing)")
+    ray.shutdown()
+    return 0
+
+
+def cmd_memory(args):
+    import ray_amd as ray
+
+    ray.init(address=args.address or "auto")
+    rt = ray.api._rt.global_runtime()
+    st = rt.raylet_call("object_stats", {})
+    print(f"shm store: {st['num_objects']} objects, "
+          f"{st['used'] / 1e6:.1f}/{st['capacity'] / 1e6:.1f} MB")
+    ray.shutdown()
+    return 0
+
+
+def cmd_microbenchmark(args):
+    from ray_amd.microbenchmark import main as micro
+
+    micro(args.duration)
+    return 0
+
+
+def main(argv=None):
+    p = argparse.ArgumentParser(prog="ray_amd")
+    sub = p.add_subparsers(dest="cmd", required=True)
+
+    s = sub.add_parser("start")
+    s.add_argument("--head", action="store_true")
+    s.add_argument("--num-cpus", type=float, default=None)
+    s.add_argument("--num-gpus", type=float, default=None)
+    s.set_defaults(fn=cmd_start)
+
+    s = sub.add_parser("stop")
+    s.set_defaults(fn=cmd_stop)
+
+    for name, fn in (
+        ("status", cmd_status),
+        ("timeline", cmd_timeline),
+        ("memory", cmd_memory),
+    ):
+        s = sub.add_parser(name)
+        s.add_argument("--address", default=None)
+        if name == "timeline":
+            s.add_argument("--output", default=None)
+        s.set_defaults(fn=fn)
+
+    s = sub.add_parser("microbenchmark")
+    s.add_argument("--duration", type=float, default=2.0)
+    s.set_defaults(fn=cmd_microbenchmark)
+
+    args = p.parse_args(argv)
+    return args.fn(args)
+
+
+if __name__ == "__main__":
+    sys.exit(main())

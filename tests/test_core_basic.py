@@ -181,3 +181,30 @@ def test_async_task_function(ray_start_regular):
         return x + 1
 
     assert ray.get(afn.remote(1)) == 2
+
+
+def test_state_api_and_cli(ray_start_regular):
+    @ray.remote
+    def f():
+        return 1
+
+    ray.get([f.remote() for _ in range(10)])
+
+    @ray.remote
+    class A:
+        def ping(self):
+            return 1
+
+    a = A.remote()
+    ray.get(a.ping.remote())
+    import time as _t
+
+    _t.sleep(0.3)
+    from ray_amd.util import state as state_api
+
+    nodes = state_api.list_nodes()
+    assert nodes and nodes[0]["state"] == "ALIVE"
+    actors = state_api.list_actors()
+    assert any(x["class_name"] == "A" for x in actors)
+    objs = state_api.list_objects()
+    assert "num_objects_in_store" in objs[0]

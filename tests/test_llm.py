@@ -1,0 +1,73 @@
+"""LLM engine/serving tests. CPU: eager decode; GPU: hipGraph decode
+equivalence + Serve deployment (north-star config 5)."""
+import pytest
+import torch
+
+import ray_amd as ray
+from ray_amd.llm import LLMConfig, LLMEngine
+
+
+def test_engine_generate_cpu():
+    eng = LLMEngine(LLMConfig(model_id="llama-tiny", max_seq_len=128,
+                              use_hip_graph=False))
+    r = eng.generate([1, 2, 3, 4], max_new_tokens=8)
+    assert len(r["token_ids"]) == 8
+    assert all(0 <= t < 512 for t in r["token_ids"])
+    # greedy decode is deterministic
+    r2 = eng.generate([1, 2, 3, 4], max_new_tokens=8)
+    assert r["token_ids"] == r2["token_ids"]
+
+
+def test_engine_temperature_sampling_cpu():
+    eng = LLMEngine(LLMConfig(model_id="llama-tiny", max_seq_len=64,
+                              use_hip_graph=False))
+    torch.manual_seed(0)
+    r = eng.generate([5, 6], max_new_tokens=16, temperature=1.0)
+    assert len(r["token_ids"]) == 16
+
+
+def test_llm_serve_deployment_cpu(ray_start_regular):
+    from ray_amd import serve
+    from ray_amd.llm import build_llm_deployment
+
+    app = build_llm_deployment(
+        {"model_id": "llama-tiny", "max_seq_len": 64, "use_hip_graph": False}
+    )
+    h = serve.run(app, name="llm", http=False)
+    out = h.generate.remote([1, 2, 3], 4).result(timeout_s=120)
+    assert len(out["token_ids"]) == 4
+    serve.shutdown()
+
+
+@pytest.mark.gpu
+def test_hipgraph_decode_matches_eager():
+    """Graph-captured decode must produce identical greedy tokens to the
+    eager decode path."""
+    assert torch.cuda.is_available()
+    eager = LLMEngine(LLMConfig(model_id="llama-tiny", max_seq_len=128,
+                                use_hip_graph=False))
+    graphed = LLMEngine(LLMConfig(model_id="llama-tiny", max_seq_len=128,
+                                  use_hip_graph=True))
+    assert graphed.decoder is not None and graphed.decoder.graph is not None
+    prompt = [7, 11, 13, 17, 19]
+    r_e = eager.generate(prompt, max_new_tokens=16)
+    r_g = graphed.generate(prompt, max_new_tokens=16)
+    assert r_e["token_ids"] == r_g["token_ids"], (
+        r_e["token_ids"], r_g["token_ids"]
+    )
+
+
+@pytest.mark.gpu
+def test_hipgraph_decode_speedup():
+    eager = LLMEngine(LLMConfig(model_id="llama-tiny", max_seq_len=256,
+                                use_hip_graph=False))
+    graphed = LLMEngine(LLMConfig(model_id="llama-tiny", max_seq_len=256,
+                                  use_hip_graph=True))
+    prompt = list(range(8))
+    eager.generate(prompt, 16)
+    graphed.generate(prompt, 16)
+    r_e = eager.generate(prompt, 128)
+    r_g = graphed.generate(prompt, 128)
+    print(f"eager {r_e['decode_tok_s']:.0f} tok/s vs graph {r_g['decode_tok_s']:.0f} tok/s")
+    # launch-bound tiny model: graph replay must be significantly faster
+    assert r_g["decode_tok_s"] > r_e["decode_tok_s"] * 1.5
