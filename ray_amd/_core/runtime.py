@@ -118,6 +118,49 @@ def _deserialize_ref(oid: bytes, owner_addr: str) -> ObjectRef:
     return ObjectRef(oid, owner_addr)
 
 
+class ObjectRefGenerator:
+    """Iterator over a streaming task's return refs (reference:
+    python/ray/_raylet.pyx ObjectRefGenerator; num_returns="streaming").
+    next() blocks until the executor reports the next yielded item."""
+
+    def __init__(self, task_id: bytes, rt: "CoreRuntime"):
+        self._task_id = task_id
+        self._rt = rt
+        self._consumed = 0
+
+    def __iter__(self):
+        return self
+
+    def __next__(self) -> ObjectRef:
+        import threading as _th
+        import time as _time
+
+        st = self._rt._streams.get(self._task_id)
+        if st is None:
+            raise StopIteration
+        deadline = _time.monotonic() + 600.0
+        while True:
+            if self._consumed < len(st["items"]):
+                oid = st["items"][self._consumed]
+                self._consumed += 1
+                return ObjectRef(oid, self._rt.addr)
+            if st["done"]:
+                err = st.get("error")
+                del self._rt._streams[self._task_id]
+                if err is not None:
+                    raise serialization.loads(err)
+                raise StopIteration
+            ev = st["event"]
+            ev.wait(0.5)
+            ev.clear()
+            if _time.monotonic() > deadline:
+                raise exc.GetTimeoutError("streaming generator stalled")
+
+    def completed(self):
+        st = self._rt._streams.get(self._task_id)
+        return st is None or st["done"]
+
+
 # --------------------------------------------------------------------------
 # Lease pool (reference: normal_task_submitter.h:87 per-SchedulingKey pools)
 # --------------------------------------------------------------------------
@@ -191,6 +234,7 @@ class CoreRuntime:
         self._refs: Dict[bytes, list] = {}
         self._refs_lock = threading.Lock()
 
+        self._streams: Dict[bytes, dict] = {}  # task_id -> stream state
         self._pools: Dict[tuple, _LeasePool] = {}
         self._fn_exported: Dict[bytes, asyncio.Future] = {}
         self._fn_cache: Dict[bytes, Any] = {}
@@ -283,9 +327,41 @@ class CoreRuntime:
             if c is not None and c.connected:
                 return c
             c = RpcClient()
+            c.on_notify = self._on_conn_notify
             await c.connect(addr, retries=10)
             self._conns[addr] = c
             return c
+
+    def _on_conn_notify(self, method, payload):
+        if method == "stream_item":
+            self._h_stream_item(payload)
+
+    def _h_stream_item(self, p):
+        """Streaming-generator item pushed by the executing worker
+        (reference: ReportGeneratorItemReturns, core_worker.h:856)."""
+        task_id = bytes(p["task_id"])
+        st = self._streams.get(task_id)
+        if st is None:
+            return
+        oid = bytes(p["oid"])
+        if p["kind"] == "inline":
+            self._store_put(oid, ("val_ser", p["data"]))
+        else:
+            self._store_put(oid, ("store", p["node_addr"], p["size"]))
+        st["items"].append(oid)
+        ev = st.get("event")
+        if ev is not None:
+            self.loop.call_soon_threadsafe(ev.set)
+
+    def _stream_finish(self, task_id: bytes, error=None):
+        st = self._streams.get(task_id)
+        if st is None:
+            return
+        st["done"] = True
+        st["error"] = error
+        ev = st.get("event")
+        if ev is not None:
+            self.loop.call_soon_threadsafe(ev.set)
 
     # ------------- reference counting -------------
 
@@ -616,8 +692,11 @@ class CoreRuntime:
         name: str,
         args_tuple,
         options: dict,
-    ) -> List[ObjectRef]:
+    ):
         num_returns = options.get("num_returns", 1)
+        streaming = num_returns == "streaming"
+        if streaming:
+            num_returns = 0
         returns = [ids.new_object_id() for _ in range(max(num_returns, 1))]
         refs = [ObjectRef(oid, self.addr) for oid in returns]
         meta, buffers, captured = self._serialize_capture(args_tuple)
@@ -625,8 +704,9 @@ class CoreRuntime:
         for r in captured:
             self._add_submitted_ref(r.id)
         captured_ids = [(r.id, r.owner_addr) for r in captured]
+        task_id = ids.new_task_id()
         spec = {
-            "task_id": ids.new_task_id(),
+            "task_id": task_id,
             "fn_id": fn_id,
             "name": name,
             "returns": returns,
@@ -634,6 +714,14 @@ class CoreRuntime:
             "num_returns": num_returns,
             "env_vars": (options.get("runtime_env") or {}).get("env_vars"),
         }
+        if streaming:
+            import threading as _th
+
+            spec["streaming"] = True
+            self._streams[task_id] = {
+                "items": [], "done": False, "error": None,
+                "event": _th.Event(),
+            }
         if size <= serialization.INLINE_MAX:
             blob = bytearray(size)
             n = serialization.write_to(memoryview(blob), meta, buffers)
@@ -646,10 +734,12 @@ class CoreRuntime:
             self._store_put_threadsafe(aid, ("store", self.raylet_addr, size))
             with self._refs_lock:
                 self._refs[aid] = [1, 0, self.addr]  # freed after task completes
-        retries = options.get("max_retries", 3)
+        retries = 0 if streaming else options.get("max_retries", 3)
         self._run(
             self._submit_with_retries(spec, options, retries, captured_ids)
         )
+        if streaming:
+            return ObjectRefGenerator(task_id, self)
         return refs
 
     async def _seal_async(self, oid, size):
@@ -685,6 +775,9 @@ class CoreRuntime:
                             f"worker died running {spec['name']}: {e}"
                         )
                     )
+                    if spec.get("streaming"):
+                        self._stream_finish(spec["task_id"], err)
+                        return
                     for oid in spec["returns"]:
                         self._store_put(oid, ("err", err))
                     return
@@ -694,8 +787,11 @@ class CoreRuntime:
                     "task submission failed:\n" + traceback.format_exc()
                 )
             )
-            for oid in spec["returns"]:
-                self._store_put(oid, ("err", err))
+            if spec.get("streaming"):
+                self._stream_finish(spec["task_id"], err)
+            else:
+                for oid in spec["returns"]:
+                    self._store_put(oid, ("err", err))
         finally:
             for oid, _owner in captured_ids:
                 self._remove_submitted_ref(oid)
@@ -798,6 +894,12 @@ class CoreRuntime:
 
     def _ingest_result(self, spec, reply):
         status = reply.get("status")
+        if spec.get("streaming"):
+            self._stream_finish(
+                spec["task_id"],
+                reply.get("error") if status == "error" else None,
+            )
+            return
         if status == "error":
             for oid in spec["returns"]:
                 self._store_put(oid, ("err", reply["error"]))
@@ -867,8 +969,11 @@ class CoreRuntime:
         return actor_id
 
     def submit_actor_task(self, actor_id: bytes, method: str, args_tuple,
-                          options: dict) -> List[ObjectRef]:
+                          options: dict):
         num_returns = options.get("num_returns", 1)
+        streaming = num_returns == "streaming"
+        if streaming:
+            num_returns = 0
         returns = [ids.new_object_id() for _ in range(max(num_returns, 1))]
         refs = [ObjectRef(oid, self.addr) for oid in returns]
         meta, buffers, captured = self._serialize_capture(args_tuple)
@@ -876,14 +981,23 @@ class CoreRuntime:
         for r in captured:
             self._add_submitted_ref(r.id)
         captured_ids = [r.id for r in captured]
+        task_id = ids.new_task_id()
         spec = {
-            "task_id": ids.new_task_id(),
+            "task_id": task_id,
             "actor_id": actor_id,
             "method": method,
             "returns": returns,
             "caller": self.addr,
             "num_returns": num_returns,
         }
+        if streaming:
+            import threading as _th
+
+            spec["streaming"] = True
+            self._streams[task_id] = {
+                "items": [], "done": False, "error": None,
+                "event": _th.Event(),
+            }
         if size <= serialization.INLINE_MAX:
             blob = bytearray(size)
             n = serialization.write_to(memoryview(blob), meta, buffers)
@@ -897,6 +1011,8 @@ class CoreRuntime:
             with self._refs_lock:
                 self._refs[aid] = [1, 0, self.addr]
         self._run(self._submit_actor_async(spec, captured_ids))
+        if streaming:
+            return ObjectRefGenerator(task_id, self)
         return refs
 
     async def _actor_state(self, actor_id: bytes) -> dict:
@@ -944,6 +1060,9 @@ class CoreRuntime:
                             f"actor {actor_id.hex()} is dead: {st['dead']}",
                         )
                     )
+                    if spec.get("streaming"):
+                        self._stream_finish(spec["task_id"], err)
+                        return
                     for oid in spec["returns"]:
                         self._store_put(oid, ("err", err))
                     return

@@ -229,9 +229,84 @@ class WorkerMain:
                 ok, result = False, e
         else:
             ok, result = await loop.run_in_executor(self.executor, _exec)
-        reply = self._build_reply(spec, ok, result)
+        if ok and spec.get("streaming"):
+            reply = await self._run_streaming(conn, spec, result)
+        else:
+            reply = self._build_reply(spec, ok, result)
         self._record_event(spec, t0, ok)
         return reply
+
+    async def _run_streaming(self, conn, spec, gen):
+        """Drive a user generator; push each yielded value to the
+        caller as its own object (reference: streaming generators,
+        ReportGeneratorItemReturns + generator_waiter.h backpressure)."""
+        import struct
+
+        loop = asyncio.get_running_loop()
+        task_id = bytes(spec["task_id"])
+        idx = 0
+
+        def pull():
+            try:
+                return True, next(gen), None
+            except StopIteration:
+                return False, None, None
+            except BaseException as e:  # noqa
+                return False, None, e
+
+        async def apull():
+            try:
+                return True, await gen.__anext__(), None
+            except StopAsyncIteration:
+                return False, None, None
+            except BaseException as e:  # noqa
+                return False, None, e
+
+        is_async = hasattr(gen, "__anext__")
+        if not (is_async or hasattr(gen, "__next__")):
+            return self._error_reply(
+                spec, "num_returns='streaming' requires a generator"
+            )
+        while True:
+            if is_async:
+                more, item, err = await apull()
+            else:
+                more, item, err = await loop.run_in_executor(self.executor, pull)
+            if err is not None:
+                name = spec.get("name", "")
+                import traceback as tb
+
+                data = serialization.dumps(
+                    exc.RayTaskError(
+                        name,
+                        "".join(tb.format_exception(type(err), err,
+                                                    err.__traceback__)),
+                        None,
+                    )
+                )
+                return {"status": "error", "error": data}
+            if not more:
+                break
+            oid = task_id + struct.pack("<I", idx)
+            meta, buffers, _ = self.rt._serialize_capture(item)
+            size = serialization.serialized_size(meta, buffers)
+            payload = {"task_id": task_id, "oid": oid}
+            if size <= serialization.INLINE_MAX:
+                blob = bytearray(size)
+                n = serialization.write_to(memoryview(blob), meta, buffers)
+                payload.update({"kind": "inline", "data": bytes(blob[:n])})
+            else:
+                store.put_serialized(self.rt.shm_dir, oid, meta, buffers)
+                self.rt.raylet.notify("seal_object", {"id": oid, "size": size})
+                payload.update(
+                    {"kind": "store", "node_addr": self.raylet_addr,
+                     "size": size}
+                )
+            from .protocol import MSG_NOTIFY
+
+            conn.send([MSG_NOTIFY, 0, "stream_item", payload])
+            idx += 1
+        return {"status": "ok", "streaming_done": idx}
 
     # ------------- actor calls -------------
 
@@ -275,7 +350,10 @@ class WorkerMain:
                     return False, e, False
 
             ok, result, exit_after = await loop.run_in_executor(self.executor, _exec)
-        reply = self._build_reply(spec, ok, result)
+        if ok and spec.get("streaming"):
+            reply = await self._run_streaming(conn, spec, result)
+        else:
+            reply = self._build_reply(spec, ok, result)
         self._record_event(spec, t0, ok, method_name)
         if exit_after or method_name == "__ray_terminate__":
             loop.call_later(0.05, self._graceful_exit)

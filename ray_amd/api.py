@@ -254,6 +254,8 @@ class ActorMethod:
             (args, kwargs),
             {"num_returns": num_returns},
         )
+        if num_returns == "streaming":
+            return refs
         if num_returns == 1:
             return refs[0]
         return refs

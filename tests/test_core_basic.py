@@ -208,3 +208,51 @@ def test_state_api_and_cli(ray_start_regular):
     assert any(x["class_name"] == "A" for x in actors)
     objs = state_api.list_objects()
     assert "num_objects_in_store" in objs[0]
+
+
+def test_streaming_generator_task(ray_start_regular):
+    @ray.remote
+    def gen(n):
+        for i in range(n):
+            yield i * 10
+
+    g = gen.options(num_returns="streaming").remote(5)
+    vals = [ray.get(ref) for ref in g]
+    assert vals == [0, 10, 20, 30, 40]
+
+
+def test_streaming_generator_large_items(ray_start_regular):
+    @ray.remote
+    def gen():
+        for i in range(3):
+            yield np.full((600, 600), i, dtype=np.float64)  # ~2.9MB
+
+    g = gen.options(num_returns="streaming").remote()
+    arrs = [ray.get(r) for r in g]
+    assert [int(a[0, 0]) for a in arrs] == [0, 1, 2]
+
+
+def test_streaming_generator_error(ray_start_regular):
+    @ray.remote
+    def gen():
+        yield 1
+        raise ValueError("stream boom")
+
+    g = gen.options(num_returns="streaming").remote()
+    first = ray.get(next(g))
+    assert first == 1
+    with pytest.raises(Exception, match="stream boom"):
+        for r in g:
+            ray.get(r)
+
+
+def test_streaming_actor_method(ray_start_regular):
+    @ray.remote
+    class Gen:
+        def stream(self, n):
+            for i in range(n):
+                yield i
+
+    g = Gen.remote()
+    it = g.stream.options(num_returns="streaming").remote(4)
+    assert [ray.get(r) for r in it] == [0, 1, 2, 3]
