@@ -256,3 +256,21 @@ def test_streaming_actor_method(ray_start_regular):
     g = Gen.remote()
     it = g.stream.options(num_returns="streaming").remote(4)
     assert [ray.get(r) for r in it] == [0, 1, 2, 3]
+
+
+def test_metrics_api(ray_start_regular):
+    from ray_amd.util import metrics
+
+    c = metrics.Counter("test_requests", "requests", ("route",))
+    c.inc(1, {"route": "/a"})
+    c.inc(2, {"route": "/a"})
+    g = metrics.Gauge("test_inflight", "inflight")
+    g.set(5)
+    h = metrics.Histogram("test_lat", "latency", boundaries=[1, 10])
+    h.observe(0.5)
+    h.observe(5)
+    h.observe(50)
+    text = metrics.export_text()
+    assert "test_requests" in text and "3.0" in text
+    assert "test_inflight 5" in text
+    assert "test_lat_count 3" in text
