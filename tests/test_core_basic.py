@@ -274,3 +274,25 @@ def test_metrics_api(ray_start_regular):
     assert "test_requests" in text and "3.0" in text
     assert "test_inflight 5" in text
     assert "test_lat_count 3" in text
+
+
+def test_multiprocessing_pool(ray_start_regular):
+    from ray_amd.util.multiprocessing import Pool
+
+    with Pool(processes=2) as pool:
+        out = pool.map(lambda x: x * x, range(20))
+        assert out == [x * x for x in range(20)]
+        assert pool.apply(lambda a, b: a + b, (3, 4)) == 7
+        assert sorted(pool.imap_unordered(lambda x: -x, range(5))) == [-4, -3, -2, -1, 0]
+
+
+def test_internal_kv(ray_start_regular):
+    from ray_amd.experimental import internal_kv as kv
+
+    assert kv._internal_kv_initialized()
+    kv._internal_kv_put(b"k1", b"v1")
+    assert kv._internal_kv_get(b"k1") == b"v1"
+    assert kv._internal_kv_exists(b"k1")
+    assert b"k1" in kv._internal_kv_list(b"k")
+    kv._internal_kv_del(b"k1")
+    assert kv._internal_kv_get(b"k1") is None
