@@ -1,0 +1,126 @@
+"""DAG composition + shm channel tests (reference model:
+python/ray/dag/tests/, experimental/channel tests)."""
+import threading
+import time
+
+import pytest
+
+import ray_amd as ray
+from ray_amd.dag import InputNode, MultiOutputNode
+
+
+def test_function_dag(ray_start_regular):
+    @ray.remote
+    def a(x):
+        return x + 1
+
+    @ray.remote
+    def b(x):
+        return x * 2
+
+    with InputNode() as inp:
+        dag = b.bind(a.bind(inp))
+    assert ray.get(dag.execute(10)) == 22
+
+
+def test_actor_dag_and_compile(ray_start_regular):
+    @ray.remote
+    class Worker:
+        def __init__(self, k):
+            self.k = k
+
+        def scale(self, x):
+            return x * self.k
+
+    w = Worker.bind(3)
+    with InputNode() as inp:
+        dag = w.scale.bind(inp)
+    compiled = dag.experimental_compile()
+    assert ray.get(compiled.execute(5)) == 15
+    assert ray.get(compiled.execute(7)) == 21  # same actor reused
+
+
+def test_multi_output_shared_upstream(ray_start_regular):
+    calls = []
+
+    @ray.remote
+    def src(x):
+        return x + 100
+
+    @ray.remote
+    def left(x):
+        return ("L", x)
+
+    @ray.remote
+    def right(x):
+        return ("R", x)
+
+    with InputNode() as inp:
+        s = src.bind(inp)
+        dag = MultiOutputNode([left.bind(s), right.bind(s)])
+    compiled = dag.experimental_compile()
+    l, r = ray.get(compiled.execute(1))
+    assert l == ("L", 101) and r == ("R", 101)
+
+
+def test_channel_spsc(ray_start_regular, tmp_path):
+    from ray_amd.experimental.channel import Channel, ChannelReader
+
+    path = str(tmp_path / "chan1")
+    ch = Channel(path, capacity=1 << 16, num_readers=1, create=True)
+    reader = ChannelReader(Channel(path), slot=0)
+
+    got = []
+
+    def consume():
+        for _ in range(5):
+            got.append(reader.next_obj(timeout=10))
+
+    t = threading.Thread(target=consume)
+    t.start()
+    for i in range(5):
+        ch.write_obj({"i": i})
+    t.join(10)
+    assert [g["i"] for g in got] == list(range(5))
+
+
+def test_channel_backpressure(ray_start_regular, tmp_path):
+    from ray_amd.experimental.channel import Channel
+
+    path = str(tmp_path / "chan2")
+    ch = Channel(path, capacity=1 << 12, num_readers=1, create=True)
+    ch.write(b"first")
+    # second write must block until the reader acks
+    with pytest.raises(TimeoutError):
+        ch.write(b"second", timeout=0.3)
+
+
+def test_channel_between_actors(ray_start_regular):
+    from ray_amd.experimental.channel import channel_path
+
+    path = channel_path("t_actors")
+
+    @ray.remote
+    class Producer:
+        def run(self, path, n):
+            from ray_amd.experimental.channel import Channel
+
+            ch = Channel(path, capacity=1 << 16, num_readers=1, create=True)
+            for i in range(n):
+                ch.write_obj(i * 2)
+            return n
+
+    @ray.remote
+    class Consumer:
+        def run(self, path, n):
+            from ray_amd.experimental.channel import Channel, ChannelReader
+
+            r = ChannelReader(Channel(path))
+            return [r.next_obj(timeout=30) for _ in range(n)]
+
+    p = Producer.remote()
+    c = Consumer.remote()
+    pref = p.run.remote(path, 4)
+    cref = c.run.remote(path, 4)
+    assert ray.get(cref, timeout=60) == [0, 2, 4, 6]
+    assert ray.get(pref) == 4
