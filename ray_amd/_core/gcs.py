@@ -211,13 +211,31 @@ class GcsServer:
 
     async def _schedule_actor(self, a: ActorInfo):
         req = dict(a.spec.get("resources", {}))
+        in_pg = a.spec.get("pg_id") is not None
         deadline = time.time() + 300.0
         while a.state in (ACTOR_PENDING, ACTOR_RESTARTING):
-            node = self._pick_node(
-                req,
-                strategy=a.spec.get("scheduling_strategy", "hybrid"),
-                soft_affinity=a.spec.get("pg_node"),
-            )
+            if in_pg:
+                # bundle resources were reserved at PG creation; the
+                # actor MUST land on the bundle's node and must not be
+                # double-counted against node availability
+                pg_node = a.spec.get("pg_node")
+                node = None
+                if pg_node is not None:
+                    n = self.nodes.get(bytes(pg_node))
+                    node = n if n is not None and n.alive else None
+                else:
+                    pg = self.pgs.get(bytes(a.spec["pg_id"]))
+                    if pg is not None:
+                        for nid in pg.bundle_nodes:
+                            if nid and self.nodes.get(nid) and self.nodes[nid].alive:
+                                node = self.nodes[nid]
+                                break
+            else:
+                node = self._pick_node(
+                    req,
+                    strategy=a.spec.get("scheduling_strategy", "hybrid"),
+                    soft_affinity=a.spec.get("pg_node"),
+                )
             if node is None:
                 if time.time() > deadline:
                     await self._fail_actor(a, "resources unavailable for actor")
@@ -231,8 +249,11 @@ class GcsServer:
                 a.node_id = node.node_id
                 a.addr = r["addr"]
                 a.state = ACTOR_ALIVE
-                for k, v in req.items():
-                    node.resources_available[k] = node.resources_available.get(k, 0) - v
+                if not in_pg:
+                    for k, v in req.items():
+                        node.resources_available[k] = (
+                            node.resources_available.get(k, 0) - v
+                        )
                 for f in a.waiters:
                     if not f.done():
                         f.set_result(None)
@@ -287,7 +308,7 @@ class GcsServer:
 
     async def _on_actor_exit(self, a: ActorInfo, cause: str, expected: bool):
         node = self.nodes.get(a.node_id) if a.node_id else None
-        if node is not None and a.state == ACTOR_ALIVE:
+        if node is not None and a.state == ACTOR_ALIVE and not a.spec.get("pg_id"):
             for k, v in a.spec.get("resources", {}).items():
                 node.resources_available[k] = node.resources_available.get(k, 0) + v
         if (not expected) and a.num_restarts < a.spec.get("max_restarts", 0):

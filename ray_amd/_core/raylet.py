@@ -214,14 +214,16 @@ class Raylet:
                 self._release_resources(lease)
         if w.kind == "actor" and w.actor_id is not None:
             spec_res = getattr(w, "actor_resources", None) or {}
-            for k, v in spec_res.items():
-                if k == "GPU":
-                    continue
-                self.avail[k] = self.avail.get(k, 0) + v
+            bundle_key = getattr(w, "actor_bundle", None)
             for g in w.gpu_ids:
                 self._free_gpus.append(g)
-            if "GPU" in spec_res:
-                self.avail["GPU"] = self.avail.get("GPU", 0) + spec_res["GPU"]
+            if bundle_key is not None and bundle_key in self.bundles:
+                b = self.bundles[bundle_key]
+                for k, v in spec_res.items():
+                    b["avail"][k] = b["avail"].get(k, 0) + v
+            else:
+                for k, v in spec_res.items():
+                    self.avail[k] = self.avail.get(k, 0) + v
             rc = w.proc.poll()
             asyncio.ensure_future(self._notify_actor_exit(w, rc))
         self._try_grant()
@@ -395,12 +397,27 @@ class Raylet:
         spec = dict(spec)
         spec["actor_id"] = actor_id
         res = spec.get("resources") or {}
-        if not self._fits(self.avail, res):
+        bundle_key = None
+        if spec.get("pg_id") is not None:
+            bundle_key = self._bundle_key(
+                (bytes(spec["pg_id"]), spec.get("bundle_index"))
+            )
+            if bundle_key is None:
+                raise RuntimeError("placement group bundle not on this node")
+            bundle = self.bundles[bundle_key]
+            if not self._fits(bundle["avail"], res):
+                raise RuntimeError(
+                    f"bundle {bundle_key} lacks resources for actor {res}"
+                )
+            for k, v in res.items():
+                bundle["avail"][k] = bundle["avail"].get(k, 0) - v
+        elif not self._fits(self.avail, res):
             raise RuntimeError(f"node {self.node_name}: insufficient resources {res}")
         ngpu = int(res.get("GPU", 0))
         gpu_ids = [self._free_gpus.pop() for _ in range(min(ngpu, len(self._free_gpus)))]
-        for k, v in res.items():
-            self.avail[k] = self.avail.get(k, 0) - v
+        if bundle_key is None:
+            for k, v in res.items():
+                self.avail[k] = self.avail.get(k, 0) - v
         if gpu_ids:
             spec.setdefault("env_vars", {})
             spec["env_vars"] = dict(spec.get("env_vars") or {})
@@ -409,6 +426,7 @@ class Raylet:
         w = self._spawn_worker(actor_spec=spec)
         w.gpu_ids = gpu_ids
         w.actor_resources = res
+        w.actor_bundle = bundle_key
         fut = asyncio.get_running_loop().create_future()
         self._actor_start_futs[actor_id] = fut
         try:

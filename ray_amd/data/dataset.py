@@ -12,6 +12,7 @@ from __future__ import annotations
 
 import builtins
 import itertools
+import json
 from typing import Any, Callable, Dict, Iterable, Iterator, List, Optional, Union
 
 import numpy as np
@@ -46,31 +47,52 @@ def _to_table(block: Any) -> pa.Table:
     if isinstance(block, pa.Table):
         return block
     if isinstance(block, dict):
-        return pa.table(
-            {k: _np_to_arrow(v) for k, v in block.items()}
-        )
+        arrays, fields = [], []
+        for k, v in block.items():
+            arr, shape = _np_to_arrow(v)
+            meta = (
+                {b"tensor_shape": json.dumps(shape).encode()} if shape else None
+            )
+            fields.append(pa.field(k, arr.type, metadata=meta))
+            arrays.append(arr)
+        return pa.Table.from_arrays(arrays, schema=pa.schema(fields))
     raise TypeError(f"bad block type {type(block)}")
 
 
 def _np_to_arrow(v):
+    """Returns (arrow array, inner_shape or None). Multi-dim tensors
+    are stored as FixedSizeList with the inner shape kept in the field
+    metadata (the role of the reference's ArrowTensorArray extension,
+    data/extensions/)."""
     v = np.asarray(v)
     if v.ndim <= 1:
-        return pa.array(v)
-    return pa.FixedSizeListArray.from_arrays(
-        pa.array(v.reshape(-1)), int(np.prod(v.shape[1:]))
+        return pa.array(v), None
+    inner = list(v.shape[1:])
+    return (
+        pa.FixedSizeListArray.from_arrays(
+            pa.array(np.ascontiguousarray(v).reshape(-1)), int(np.prod(inner))
+        ),
+        inner,
     )
 
 
-def _col_to_numpy(col: pa.ChunkedArray) -> np.ndarray:
+def _col_to_numpy(col: pa.ChunkedArray, field: pa.Field = None) -> np.ndarray:
     t = col.type
     if pa.types.is_fixed_size_list(t):
         flat = col.combine_chunks().flatten().to_numpy(zero_copy_only=False)
+        shape = None
+        if field is not None and field.metadata and b"tensor_shape" in field.metadata:
+            shape = json.loads(field.metadata[b"tensor_shape"])
+        if shape:
+            return flat.reshape((len(col), *shape))
         return flat.reshape(len(col), t.list_size)
     return col.to_numpy(zero_copy_only=False)
 
 
 def _table_to_numpy(t: pa.Table) -> Dict[str, np.ndarray]:
-    return {name: _col_to_numpy(t.column(name)) for name in t.column_names}
+    return {
+        f.name: _col_to_numpy(t.column(f.name), f) for f in t.schema
+    }
 
 
 def _table_to_pandas(t: pa.Table):
@@ -144,7 +166,7 @@ def _rows_to_table(rows: List[dict]) -> pa.Table:
     if not rows:
         return pa.table({})
     keys = rows[0].keys()
-    return pa.table({k: _np_to_arrow(np.asarray([r[k] for r in rows])) for k in keys})
+    return _to_table({k: np.asarray([r[k] for r in rows]) for k in keys})
 
 
 def _apply_ops(table_or_ref, ops: List[tuple]) -> pa.Table:
