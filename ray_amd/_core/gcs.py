@@ -126,6 +126,7 @@ class GcsServer:
                 "resources_total": n.resources_total,
                 "resources_available": n.resources_available,
                 "labels": n.labels,
+                "pending": getattr(n, "pending", 0),
             }
             for n in self.nodes.values()
         ]
@@ -134,6 +135,7 @@ class GcsServer:
         n = self.nodes.get(p["node_id"])
         if n is not None:
             n.resources_available = p["available"]
+            n.pending = p.get("pending", 0)
 
     def h_ping(self, conn, p):
         return "pong"

@@ -134,11 +134,13 @@ class Raylet:
         while True:
             snap = dict(self.avail)
             now = time.time()
+            pending = len(self._pending)
             if snap != last or now - last_time > 1.0:
                 try:
                     self.gcs.notify(
                         "report_resources",
-                        {"node_id": self.node_id, "available": snap},
+                        {"node_id": self.node_id, "available": snap,
+                         "pending": pending},
                     )
                     last = snap
                     last_time = now

@@ -144,6 +144,20 @@ def main(argv=None):
             s.add_argument("--output", default=None)
         s.set_defaults(fn=fn)
 
+    s = sub.add_parser("dashboard")
+    s.add_argument("--address", default=None)
+    s.add_argument("--port", type=int, default=8265)
+    def _dash(args):
+        import ray_amd as ray
+
+        ray.init(address=args.address or "auto")
+        from ray_amd.dashboard import run_dashboard
+
+        print(f"dashboard on http://127.0.0.1:{args.port}")
+        run_dashboard(port=args.port)
+        return 0
+    s.set_defaults(fn=_dash)
+
     s = sub.add_parser("microbenchmark")
     s.add_argument("--duration", type=float, default=2.0)
     s.set_defaults(fn=cmd_microbenchmark)

@@ -296,3 +296,34 @@ def test_internal_kv(ray_start_regular):
     assert b"k1" in kv._internal_kv_list(b"k")
     kv._internal_kv_del(b"k1")
     assert kv._internal_kv_get(b"k1") is None
+
+
+def test_dashboard_api(ray_start_regular):
+    import anyio
+    import httpx
+
+    from ray_amd.dashboard import build_asgi_app
+
+    @ray.remote
+    class DashActor:
+        def ping(self):
+            return 1
+
+    a = DashActor.remote()
+    ray.get(a.ping.remote())
+
+    async def go():
+        transport = httpx.ASGITransport(app=build_asgi_app())
+        async with httpx.AsyncClient(transport=transport,
+                                     base_url="http://d") as client:
+            r = await client.get("/api/cluster_status")
+            assert r.status_code == 200
+            body = r.json()
+            assert body["nodes"] >= 1
+            assert body["actors_alive"] >= 1
+            r = await client.get("/api/nodes")
+            assert r.json()[0]["Alive"]
+            r = await client.get("/")
+            assert b"ray_amd" in r.content
+
+    anyio.run(go)
