@@ -8,6 +8,7 @@
 #include "hip/elementwise.hip"
 #include "hip/rl_scans.hip"
 #include "hip/cross_entropy.hip"
+#include "hip/flash_attn.hip"
 
 #define CHECK_IN(x)                                                     \
   TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                       \
@@ -213,7 +214,36 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor mx,
   return dl;
 }
 
+std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
+                                        at::Tensor v, bool causal,
+                                        int64_t q_offset, bool want_lse) {
+  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v);
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "q must be bf16");
+  TORCH_CHECK(q.size(3) == 128, "head_dim must be 128");
+  TORCH_CHECK(q.size(2) % 64 == 0, "T must be a multiple of 64 (pad)");
+  int B = (int)q.size(0), Hq = (int)q.size(1), T = (int)q.size(2);
+  int Hkv = (int)k.size(1), Tk = (int)k.size(2);
+  TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
+  auto o = at::empty_like(q);
+  at::Tensor lse;
+  float* lse_ptr = nullptr;
+  if (want_lse) {
+    lse = at::empty({B, Hq, T}, q.options().dtype(at::kFloat));
+    lse_ptr = lse.data_ptr<float>();
+  }
+  float scale = 1.0f / sqrtf((float)q.size(3));
+  dim3 grid(T / 64, B * Hq);
+  hipLaunchKernelGGL(flash_attn_fwd_bf16, grid, dim3(256), 0, cur_stream(),
+                     (const short*)q.data_ptr(), (const short*)k.data_ptr(),
+                     (const short*)v.data_ptr(), (short*)o.data_ptr(),
+                     lse_ptr, B, Hq, Hkv, T, Tk, causal ? 1 : 0,
+                     (int)q_offset, scale);
+  if (want_lse) return {o, lse};
+  return {o};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("swiglu_fwd", &swiglu_fwd);

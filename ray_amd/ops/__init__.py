@@ -346,3 +346,50 @@ def img_normalize(x_u8: torch.Tensor, mean: torch.Tensor, std: torch.Tensor):
         x_u8.contiguous(), mean.float().contiguous(),
         (1.0 / std.float()).contiguous()
     )
+
+
+# --------------------------------------------------------------------------
+# Flash attention (forward; fused CDNA4 kernel with LSE output)
+# --------------------------------------------------------------------------
+
+
+def flash_attention_ref(q, k, v, causal=True):
+    """fp32 reference; q [B,Hq,T,D], k/v [B,Hkv,Tk,D] (GQA)."""
+    import math
+
+    B, Hq, T, D = q.shape
+    Hkv = k.shape[1]
+    rep = Hq // Hkv
+    kf = k.float().repeat_interleave(rep, dim=1)
+    vf = v.float().repeat_interleave(rep, dim=1)
+    s = torch.matmul(q.float(), kf.transpose(-1, -2)) / math.sqrt(D)
+    if causal:
+        Tk = k.shape[2]
+        mask = torch.ones(T, Tk, dtype=torch.bool, device=q.device).tril_(
+            Tk - T
+        )
+        s = s.masked_fill(~mask, float("-inf"))
+    lse = torch.logsumexp(s, dim=-1)
+    out = torch.matmul(torch.softmax(s, -1), vf)
+    return out.to(q.dtype), lse
+
+
+def flash_attention(q, k, v, causal: bool = True, q_offset: int = 0,
+                    return_lse: bool = False):
+    """Fused CDNA4 flash-attention forward (inference / ring-attention
+    block op). q [B,Hq,T,128] bf16; T padded to 64 internally."""
+    if not q.is_cuda:
+        out, lse = flash_attention_ref(q, k, v, causal)
+        return (out, lse) if return_lse else out
+    _require_ext("flash_attention")
+    B, Hq, T, D = q.shape
+    pad = (-T) % 64
+    if pad:
+        q = torch.nn.functional.pad(q, (0, 0, 0, pad))
+    r = _K.flash_attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(),
+                          causal, q_offset, return_lse)
+    out = r[0][:, :, :T] if pad else r[0]
+    if return_lse:
+        lse = r[1][:, :, :T] if pad else r[1]
+        return out, lse
+    return out
