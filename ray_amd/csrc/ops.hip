@@ -9,6 +9,9 @@
 #include "hip/rl_scans.hip"
 #include "hip/cross_entropy.hip"
 #include "hip/flash_attn.hip"
+#include "hip/flash_attn_v2.hip"
+#include "hip/flash_attn_v3.hip"
+#include "hip/flash_attn_v4.hip"
 
 #define CHECK_IN(x)                                                     \
   TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                       \
@@ -220,7 +223,7 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
   CHECK_IN(q); CHECK_IN(k); CHECK_IN(v);
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "q must be bf16");
   TORCH_CHECK(q.size(3) == 128, "head_dim must be 128");
-  TORCH_CHECK(q.size(2) % 64 == 0, "T must be a multiple of 64 (pad)");
+  TORCH_CHECK(q.size(2) % 128 == 0, "T must be a multiple of 128 (pad)");
   int B = (int)q.size(0), Hq = (int)q.size(1), T = (int)q.size(2);
   int Hkv = (int)k.size(1), Tk = (int)k.size(2);
   TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
@@ -232,12 +235,32 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
     lse_ptr = lse.data_ptr<float>();
   }
   float scale = 1.0f / sqrtf((float)q.size(3));
-  dim3 grid(T / 64, B * Hq);
-  hipLaunchKernelGGL(flash_attn_fwd_bf16, grid, dim3(256), 0, cur_stream(),
-                     (const short*)q.data_ptr(), (const short*)k.data_ptr(),
-                     (const short*)v.data_ptr(), (short*)o.data_ptr(),
-                     lse_ptr, B, Hq, Hkv, T, Tk, causal ? 1 : 0,
-                     (int)q_offset, scale);
+  static const bool use_v1 = getenv("RAY_AMD_FA_V1") != nullptr;
+  static const bool use_v2 = getenv("RAY_AMD_FA_V2") != nullptr;
+  if (use_v1)
+    hipLaunchKernelGGL(flash_attn_fwd_bf16, dim3(T / 64, B * Hq), dim3(256),
+                       0, cur_stream(), (const short*)q.data_ptr(),
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
+                       causal ? 1 : 0, (int)q_offset, scale);
+  else if (use_v2)
+    hipLaunchKernelGGL(flash_attn_fwd_v2_bf16, dim3(T / 64, B * Hq),
+                       dim3(256), 0, cur_stream(), (const short*)q.data_ptr(),
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
+                       causal ? 1 : 0, (int)q_offset, scale);
+  else if (getenv("RAY_AMD_FA_V3") != nullptr)
+    hipLaunchKernelGGL(flash_attn_fwd_v3_bf16, dim3(T / 128, B * Hq),
+                       dim3(256), 0, cur_stream(), (const short*)q.data_ptr(),
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
+                       causal ? 1 : 0, (int)q_offset, scale);
+  else
+    hipLaunchKernelGGL(flash_attn_fwd_v4_bf16, dim3(T / 128, B * Hq),
+                       dim3(256), 0, cur_stream(), (const short*)q.data_ptr(),
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
+                       causal ? 1 : 0, (int)q_offset, scale);
   if (want_lse) return {o, lse};
   return {o};
 }

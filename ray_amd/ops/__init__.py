@@ -383,7 +383,7 @@ def flash_attention(q, k, v, causal: bool = True, q_offset: int = 0,
         return (out, lse) if return_lse else out
     _require_ext("flash_attention")
     B, Hq, T, D = q.shape
-    pad = (-T) % 64
+    pad = (-T) % 128
     if pad:
         q = torch.nn.functional.pad(q, (0, 0, 0, pad))
     r = _K.flash_attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(),
