@@ -27,7 +27,7 @@ def run(tag):
     flops = 4*B*Hq*T*T*D/2
     print(f"{tag}: err={err:.3f} lse_err={lse_err:.4f}  {dt*1000:.2f} ms  {flops/dt/1e12:.0f} TF")
 
-run("v3(default)")
+run("fa-default")
 os.environ["RAY_AMD_FA_V2"]="1"
 # env is read once (static); need subprocess for A/B — done by caller
 # sdpa reference
