@@ -22,7 +22,19 @@ def _block_attn(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
     q [B,H,Tq,D], k/v [B,H,Tk,D]. causal_mask: "full" (attend all),
     "causal" (Tq==Tk lower-triangular), "none" (skip — caller handles).
-    """
+    On GPU with head_dim 128 this runs the fused CDNA4 flash kernel
+    (LSE output was built for exactly this merge)."""
+    if (
+        q.is_cuda
+        and q.shape[-1] == 128
+        and q.dtype == torch.bfloat16
+    ):
+        from ray_amd import ops
+
+        out, lse = ops.flash_attention(
+            q, k, v, causal=(causal_mask == "causal"), return_lse=True
+        )
+        return out.float(), lse.unsqueeze(-1)
     scale = q.shape[-1] ** -0.5
     s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
     if causal_mask == "causal":
