@@ -196,11 +196,14 @@ class Dataset:
         self._inputs = inputs
         self._ops = ops or []
         self._materialized: Optional[List[Any]] = None
+        self._task_opts: Optional[dict] = None
 
     # ----- plan builders -----
 
     def _with_op(self, op: tuple) -> "Dataset":
-        return Dataset(self._inputs, self._ops + [op])
+        ds = Dataset(self._inputs, self._ops + [op])
+        ds._task_opts = self._task_opts
+        return ds
 
     def map(self, fn: Callable[[dict], dict], **kwargs) -> "Dataset":
         return self._with_op(("map", fn))
@@ -238,7 +241,15 @@ class Dataset:
                     setattr(_cls, inst_attr, tl)
                 return tl(batch)
 
-        return self._with_op(("map_batches", fn, batch_format, batch_size))
+        op = ("map_batches", fn, batch_format, batch_size)
+        if num_gpus or kwargs.get("num_cpus"):
+            ds = self._with_op(op)
+            ds._task_opts = {
+                "num_gpus": num_gpus,
+                "num_cpus": kwargs.get("num_cpus", 1),
+            }
+            return ds
+        return self._with_op(op)
 
     def flat_map(self, fn) -> "Dataset":
         return self._with_op(("flat_map", fn))
@@ -294,9 +305,14 @@ class Dataset:
             t = block() if callable(block) else block
             return _apply_ops(t, ops)
 
+        task = (
+            _exec_block.options(**self._task_opts)
+            if self._task_opts
+            else _exec_block
+        )
         refs = []
         for inp in self._inputs:
-            refs.append(_exec_block.remote(inp))
+            refs.append(task.remote(inp))
         self._materialized = refs
         return refs
 

@@ -376,7 +376,7 @@ class ServeController:
             out[name] = {"status": "RUNNING", "deployments": deps}
         return out
 
-    async def autoscale_once(self):
+    def autoscale_once(self):
         """One reconciliation pass of request-based autoscaling."""
         ray = _ray()
         for app in self.apps.values():

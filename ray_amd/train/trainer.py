@@ -207,6 +207,7 @@ class DataParallelTrainer:
             ], timeout=180)
 
             rows = []
+            kept_checkpoints = []
             last_metrics = None
             latest_checkpoint_path = latest_ckpt
             while True:
@@ -225,6 +226,8 @@ class DataParallelTrainer:
                     last_metrics = r["metrics"]
                     if r["checkpoint_path"]:
                         latest_checkpoint_path = r["checkpoint_path"]
+                        kept_checkpoints.append(r["checkpoint_path"])
+                        self._enforce_keep(kept_checkpoints)
                 if all(st["done"] for st in states):
                     break
                 time.sleep(0.05)
@@ -255,6 +258,18 @@ class DataParallelTrainer:
             except Exception:
                 pass
 
+    def _enforce_keep(self, kept: list):
+        """CheckpointConfig.num_to_keep (reference:
+        _internal/checkpoint_manager.py keep-top-k by recency here)."""
+        import shutil
+
+        cc = self.run_config.checkpoint_config
+        if not cc or not cc.num_to_keep:
+            return
+        while len(kept) > cc.num_to_keep:
+            victim = kept.pop(0)
+            shutil.rmtree(victim, ignore_errors=True)
+
     def _make_shards(self, n: int):
         """Split datasets across workers (streaming_split equivalent)."""
         payloads = [dict() for _ in range(n)]
@@ -267,6 +282,10 @@ class DataParallelTrainer:
                 for i in range(n):
                     payloads[i][name] = ds
         return [cloudpickle.dumps(p) if p else None for p in payloads]
+
+
+def _noop():
+    pass
 
 
 class _WorkerGroupError(Exception):

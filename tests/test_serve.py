@@ -191,3 +191,32 @@ def test_serve_batch(serve_session):
     assert vals == [i * 2 for i in range(8)]
     sizes = h.get_batch_sizes.remote().result(timeout_s=30)
     assert max(sizes) >= 2  # some batching happened
+
+
+def test_autoscaling_tick_scales_up(serve_session):
+    import time
+
+    @serve.deployment(
+        autoscaling_config={"min_replicas": 1, "max_replicas": 3,
+                            "target_ongoing_requests": 1},
+    )
+    class Slow:
+        async def __call__(self):
+            import asyncio
+
+            await asyncio.sleep(1.0)
+            return 1
+
+    h = serve.run(Slow.bind(), name="auto", http=False)
+    # pile up slow requests, then run one reconciliation pass
+    resps = [h.remote() for _ in range(6)]
+    time.sleep(0.2)
+    ctrl = ray.get_actor(
+        serve.api.SERVE_CONTROLLER_NAME, namespace=serve.api.SERVE_NAMESPACE
+    )
+    ray.get(ctrl.autoscale_once.remote(), timeout=60)
+    st = serve.status()
+    n = st["auto"]["deployments"]["Slow"]["replica_states"]["RUNNING"]
+    assert n >= 2, f"expected scale-up, have {n}"
+    for r in resps:
+        r.result(timeout_s=60)

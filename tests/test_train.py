@@ -158,3 +158,33 @@ def test_failure_no_retry(ray_start_regular, tmp_path):
     res = t.fit()
     assert res.error is not None
     assert "always fails" in str(res.error)
+
+
+def test_checkpoint_num_to_keep(ray_start_regular, tmp_path):
+    import os
+
+    from ray_amd.train import CheckpointConfig
+
+    def loop(config):
+        import tempfile
+
+        import ray_amd.train as train
+
+        for i in range(5):
+            with tempfile.TemporaryDirectory() as d:
+                open(os.path.join(d, "x.txt"), "w").write(str(i))
+                train.report({"i": i}, checkpoint=Checkpoint.from_directory(d))
+
+    t = TorchTrainer(
+        loop,
+        scaling_config=ScalingConfig(num_workers=1),
+        run_config=RunConfig(
+            name="keep", storage_path=str(tmp_path),
+            checkpoint_config=CheckpointConfig(num_to_keep=2),
+        ),
+    )
+    res = t.fit()
+    assert res.error is None
+    run_dir = os.path.join(str(tmp_path), "keep")
+    ckpts = [d for d in os.listdir(run_dir) if d.startswith("checkpoint_")]
+    assert len(ckpts) <= 2
