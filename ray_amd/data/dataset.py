@@ -186,6 +186,55 @@ def _apply_ops(table_or_ref, ops: List[tuple]) -> pa.Table:
     return t
 
 
+# ---------------- distributed shuffle (reference: hash_shuffle.py) ----------------
+
+
+def _hash_partition(t: pa.Table, key: Optional[str], num_parts: int,
+                    seed: int = 0) -> List[pa.Table]:
+    """Split a block into num_parts sub-tables by key hash (or random)."""
+    n = t.num_rows
+    if n == 0:
+        return [t] * num_parts
+    if key is None:
+        rng = np.random.default_rng(seed)
+        part = rng.integers(0, num_parts, n)
+    else:
+        vals = _col_to_numpy(t.column(key))
+        if vals.dtype.kind in "OUS":
+            part = np.array([hash(str(v)) % num_parts for v in vals])
+        else:
+            part = (vals.astype(np.int64, copy=False) * 2654435761 % 2**31
+                    ) % num_parts
+    return [t.filter(pa.array(part == p)) for p in builtins.range(num_parts)]
+
+
+def _shuffle_exchange(ray, block_refs: List, key: Optional[str],
+                      num_parts: int, seed: int = 0) -> List:
+    """Two-stage map/reduce exchange: every block is hash-partitioned in
+    parallel tasks, then each output partition concatenates its slice of
+    every block (the reference's shuffle_operators exchange)."""
+
+    @ray.remote
+    def _map(block, num_parts=num_parts, key=key, seed=seed):
+        return tuple(_hash_partition(block, key, num_parts, seed))
+
+    @ray.remote
+    def _reduce(*parts):
+        tables = [p for p in parts if p.num_rows > 0]
+        return pa.concat_tables(tables) if tables else pa.table({})
+
+    map_refs = [
+        _map.options(num_returns=num_parts).remote(b) for b in block_refs
+    ]
+    out = []
+    for p in builtins.range(num_parts):
+        cols = [mr[p] if isinstance(mr, list) else mr for mr in map_refs]
+        if num_parts == 1:
+            cols = map_refs
+        out.append(_reduce.remote(*cols))
+    return out
+
+
 # ---------------- Dataset ----------------
 
 
@@ -501,18 +550,29 @@ class Dataset:
 
     def random_shuffle(self, *, seed: Optional[int] = None, **kwargs) -> "Dataset":
         ray = _ray()
-        tables = list(self._iter_tables())
-        if not tables:
-            return self
-        t = pa.concat_tables(tables)
-        rng = np.random.default_rng(seed)
-        idx = rng.permutation(t.num_rows)
-        t = t.take(pa.array(idx))
-        nb = max(1, len(tables))
-        per = max(1, (t.num_rows + nb - 1) // nb)
-        refs = [ray.put(t.slice(i * per, per)) for i in builtins.range(nb)]
-        ds = Dataset(refs, [])
-        ds._materialized = refs
+        refs = self._materialize_refs()
+        if len(refs) <= 1:
+            tables = list(self._iter_tables())
+            if not tables:
+                return self
+            t = pa.concat_tables(tables)
+            rng = np.random.default_rng(seed)
+            t = t.take(pa.array(rng.permutation(t.num_rows)))
+            ref = ray.put(t)
+            ds = Dataset([ref], [])
+            ds._materialized = [ref]
+            return ds
+        # distributed: hash-exchange rows randomly, then permute locally
+        parts = _shuffle_exchange(ray, refs, None, len(refs), seed or 0)
+
+        @ray.remote
+        def _perm(t, s):
+            rng = np.random.default_rng(s)
+            return t.take(pa.array(rng.permutation(t.num_rows)))
+
+        out = [_perm.remote(p, (seed or 0) + i) for i, p in enumerate(parts)]
+        ds = Dataset(out, [])
+        ds._materialized = out
         return ds
 
     def randomize_block_order(self, *, seed=None) -> "Dataset":
@@ -722,6 +782,31 @@ class GroupedData:
         t = pa.concat_tables(list(self._ds._iter_tables()))
         return t.group_by(self._key)
 
+    def _agg_distributed(self, on: str, how: str) -> Optional[Dataset]:
+        """Multi-block: hash-exchange by key, per-partition aggregate
+        tasks (all rows of a key land in one partition), concat."""
+        ray = _ray()
+        refs = self._ds._materialize_refs()
+        if len(refs) <= 1:
+            return None
+        key = self._key
+        parts = _shuffle_exchange(ray, refs, key, len(refs))
+
+        @ray.remote
+        def _agg(t, on=on, how=how, key=key):
+            if t.num_rows == 0:
+                return t
+            out = t.group_by(key).aggregate([(on, how)])
+            return out.rename_columns(
+                [f"{how}({on})" if c == f"{on}_{how}" else c
+                 for c in out.column_names]
+            )
+
+        out_refs = [_agg.remote(p) for p in parts]
+        ds = Dataset(out_refs, [])
+        ds._materialized = out_refs
+        return ds
+
     def _wrap(self, t: pa.Table) -> Dataset:
         ray = _ray()
         ref = ray.put(t)
@@ -735,24 +820,36 @@ class GroupedData:
         return self._wrap(t)
 
     def sum(self, on: str) -> Dataset:
+        d = self._agg_distributed(on, "sum")
+        if d is not None:
+            return d
         t = self._grouped().aggregate([(on, "sum")])
         t = t.rename_columns([f"sum({on})" if c == f"{on}_sum" else c
                               for c in t.column_names])
         return self._wrap(t)
 
     def mean(self, on: str) -> Dataset:
+        d = self._agg_distributed(on, "mean")
+        if d is not None:
+            return d
         t = self._grouped().aggregate([(on, "mean")])
         t = t.rename_columns([f"mean({on})" if c == f"{on}_mean" else c
                               for c in t.column_names])
         return self._wrap(t)
 
     def min(self, on: str) -> Dataset:
+        d = self._agg_distributed(on, "min")
+        if d is not None:
+            return d
         t = self._grouped().aggregate([(on, "min")])
         t = t.rename_columns([f"min({on})" if c == f"{on}_min" else c
                               for c in t.column_names])
         return self._wrap(t)
 
     def max(self, on: str) -> Dataset:
+        d = self._agg_distributed(on, "max")
+        if d is not None:
+            return d
         t = self._grouped().aggregate([(on, "max")])
         t = t.rename_columns([f"max({on})" if c == f"{on}_max" else c
                               for c in t.column_names])

@@ -151,3 +151,23 @@ def test_dataset_shard_in_trainer(ray_start_regular, tmp_path):
     res = t.fit()
     assert res.error is None
     assert res.metrics["rows"] == 20
+
+
+def test_distributed_shuffle_exchange(ray_start_regular):
+    # many blocks -> hash-exchange path
+    ds = rd.range(200, parallelism=8)
+    sh = ds.random_shuffle(seed=7)
+    vals = sorted(r["id"] for r in sh.take_all())
+    assert vals == list(range(200))
+    assert sh.num_blocks() == 8
+
+
+def test_distributed_groupby(ray_start_regular):
+    ds = rd.from_items([{"k": i % 5, "v": i} for i in range(100)],
+                       parallelism=8)
+    agg = ds.groupby("k").sum("v").take_all()
+    got = {int(r["k"]): float(r["sum(v)"]) for r in agg}
+    expect = {}
+    for i in range(100):
+        expect[i % 5] = expect.get(i % 5, 0) + i
+    assert got == expect
