@@ -300,6 +300,8 @@ class GcsServer:
             "node_id": a.node_id,
             "death_cause": a.death_cause,
             "spec_kv_key": a.spec.get("spec_kv_key"),
+            "working_dir": a.spec.get("working_dir"),
+            "py_modules": a.spec.get("py_modules"),
         }
 
     async def h_actor_exit(self, conn, p):

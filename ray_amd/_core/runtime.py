@@ -713,6 +713,8 @@ class CoreRuntime:
             "caller": self.addr,
             "num_returns": num_returns,
             "env_vars": (options.get("runtime_env") or {}).get("env_vars"),
+            "working_dir": (options.get("runtime_env") or {}).get("working_dir"),
+            "py_modules": (options.get("runtime_env") or {}).get("py_modules"),
         }
         if streaming:
             import threading as _th
@@ -951,6 +953,8 @@ class CoreRuntime:
                 "max_concurrency": options.get("max_concurrency", 1),
                 "caller": self.addr,
                 "env_vars": (options.get("runtime_env") or {}).get("env_vars"),
+                "working_dir": (options.get("runtime_env") or {}).get("working_dir"),
+                "py_modules": (options.get("runtime_env") or {}).get("py_modules"),
             }
             if pg is not None:
                 payload["pg_id"], payload["bundle_index"] = pg[0], pg[1]

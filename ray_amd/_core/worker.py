@@ -40,6 +40,21 @@ from . import ids, serialization, store  # noqa: E402
 from .runtime import CoreRuntime, ObjectRef, set_global_runtime  # noqa: E402
 
 
+def _apply_code_env(spec: dict):
+    """working_dir / py_modules (already staged into the session dir by
+    the driver): chdir + sys.path injection."""
+    import sys as _sys
+
+    wd = spec.get("working_dir")
+    if wd and os.path.isdir(wd):
+        os.chdir(wd)
+        if wd not in _sys.path:
+            _sys.path.insert(0, wd)
+    for m in spec.get("py_modules") or []:
+        if m and os.path.isdir(m) and m not in _sys.path:
+            _sys.path.insert(0, m)
+
+
 class TaskContext:
     def __init__(self):
         self.task_id = None
@@ -132,6 +147,7 @@ class WorkerMain:
                 )
             args, kwargs = serialization.loads(args_blob)
             args, kwargs = await self._resolve_args(args, kwargs)
+            _apply_code_env(r)
             _task_ctx.actor_id = self.actor_id
             _task_ctx.gpu_ids = [
                 int(x) for x in (os.environ.get("RAY_AMD_GPU_IDS") or "").split(",") if x
@@ -208,6 +224,7 @@ class WorkerMain:
             return self._error_reply(spec, traceback.format_exc())
         if spec.get("env_vars"):
             os.environ.update({str(k): str(v) for k, v in spec["env_vars"].items()})
+        _apply_code_env(spec)
 
         loop = asyncio.get_running_loop()
 

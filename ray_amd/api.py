@@ -172,6 +172,9 @@ class RemoteFunction:
         rt = _rt.global_runtime()
         self._ensure_exported(rt)
         opts = _normalize_scheduling(opts)
+        if opts.get("runtime_env"):
+            opts = dict(opts)
+            opts["runtime_env"] = _stage_runtime_env(opts["runtime_env"])
         name = opts.get("name") or getattr(self._function, "__name__", "fn")
         refs = rt.submit_task(self._pickled, self._fn_id, name, (args, kwargs), opts)
         if opts.get("num_returns", 1) == 1:
@@ -200,6 +203,38 @@ def _rebuild_remote_function(fn_bytes, options):
 
 def _rebuild_actor_class(cls_bytes, options):
     return ActorClass(cloudpickle.loads(cls_bytes), options)
+
+
+def _stage_runtime_env(runtime_env: Optional[dict]) -> Optional[dict]:
+    """Copy working_dir / py_modules into the session dir so every
+    worker on the node can import them (reference: runtime_env
+    working_dir upload; conda/pip are N/A offline)."""
+    if not runtime_env:
+        return runtime_env
+    needs = runtime_env.get("working_dir") or runtime_env.get("py_modules")
+    if not needs:
+        return runtime_env
+    import hashlib as _h
+    import shutil as _sh
+
+    rt = _rt.global_runtime()
+    out = dict(runtime_env)
+    base = os.path.join(rt.session_dir, "runtime_env")
+    os.makedirs(base, exist_ok=True)
+
+    def stage(path):
+        path = os.path.abspath(path)
+        tag = _h.sha1(path.encode()).hexdigest()[:12]
+        dest = os.path.join(base, tag)
+        if not os.path.exists(dest):
+            _sh.copytree(path, dest)
+        return dest
+
+    if out.get("working_dir"):
+        out["working_dir"] = stage(out["working_dir"])
+    if out.get("py_modules"):
+        out["py_modules"] = [stage(p) for p in out["py_modules"]]
+    return out
 
 
 def _normalize_scheduling(opts: dict) -> dict:
@@ -325,6 +360,8 @@ class ActorClass:
         rt = _rt.global_runtime()
         opts = _normalize_scheduling(opts)
         opts = dict(opts)
+        if opts.get("runtime_env"):
+            opts["runtime_env"] = _stage_runtime_env(opts["runtime_env"])
         opts.setdefault("namespace", _namespace)
         opts["class_name"] = self._cls.__name__
         mc = opts.get("max_concurrency", 1)

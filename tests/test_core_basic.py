@@ -327,3 +327,45 @@ def test_dashboard_api(ray_start_regular):
             assert b"ray_amd" in r.content
 
     anyio.run(go)
+
+
+def test_runtime_env_working_dir(ray_start_regular, tmp_path):
+    import os
+
+    mod_dir = tmp_path / "my_wd"
+    mod_dir.mkdir()
+    (mod_dir / "wd_module.py").write_text("MAGIC = 'from-working-dir'\n")
+    (mod_dir / "data.txt").write_text("hello-wd")
+
+    @ray.remote
+    def use_wd():
+        import wd_module
+
+        return wd_module.MAGIC, open("data.txt").read()
+
+    magic, data = ray.get(
+        use_wd.options(
+            runtime_env={"working_dir": str(mod_dir)}
+        ).remote(),
+        timeout=60,
+    )
+    assert magic == "from-working-dir"
+    assert data == "hello-wd"
+
+
+def test_runtime_env_py_modules(ray_start_regular, tmp_path):
+    pkg = tmp_path / "mypkg"
+    pkg.mkdir()
+    (pkg / "extra_mod.py").write_text("VALUE = 42\n")
+
+    @ray.remote
+    class UsesModule:
+        def get(self):
+            import extra_mod
+
+            return extra_mod.VALUE
+
+    a = UsesModule.options(
+        runtime_env={"py_modules": [str(pkg)]}
+    ).remote()
+    assert ray.get(a.get.remote(), timeout=60) == 42
