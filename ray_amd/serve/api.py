@@ -446,12 +446,28 @@ class DeploymentResponse:
         return self._ref
 
 
+class DeploymentResponseGenerator:
+    """Streaming response: iterates the replica's generator output
+    (reference: handle.options(stream=True) -> generator of results)."""
+
+    def __init__(self, ref_gen):
+        self._gen = ref_gen
+
+    def __iter__(self):
+        return self
+
+    def __next__(self):
+        ref = next(self._gen)
+        return _ray().get(ref)
+
+
 class DeploymentHandle:
     def __init__(self, app_name: str, deployment_name: Optional[str] = None,
-                 method_name: str = "__call__"):
+                 method_name: str = "__call__", stream: bool = False):
         self.app_name = app_name
         self.deployment_name = deployment_name
         self.method_name = method_name
+        self._stream = stream
         self._replicas: List = []
         self._version = -1
         self._counts: Dict[int, int] = {}
@@ -486,6 +502,12 @@ class DeploymentHandle:
             replica = self._replicas[idx]
             self._counts[idx] = self._counts.get(idx, 0) + 1
             try:
+                if self._stream:
+                    gen = replica.handle_request.options(
+                        num_returns="streaming"
+                    ).remote(self.method_name, args, kwargs)
+                    self._decr_later(idx)
+                    return DeploymentResponseGenerator(gen)
                 ref = replica.handle_request.remote(
                     self.method_name, args, kwargs
                 )
@@ -501,9 +523,13 @@ class DeploymentHandle:
         # happens replica-side (get_stats)
         self._counts[idx] = max(0, self._counts.get(idx, 1) - 1)
 
-    def options(self, *, method_name: Optional[str] = None, **kwargs):
-        h = DeploymentHandle(self.app_name, self.deployment_name,
-                             method_name or self.method_name)
+    def options(self, *, method_name: Optional[str] = None,
+                stream: Optional[bool] = None, **kwargs):
+        h = DeploymentHandle(
+            self.app_name, self.deployment_name,
+            method_name or self.method_name,
+            self._stream if stream is None else stream,
+        )
         h._replicas = self._replicas
         h._counts = self._counts
         return h

@@ -220,3 +220,15 @@ def test_autoscaling_tick_scales_up(serve_session):
     assert n >= 2, f"expected scale-up, have {n}"
     for r in resps:
         r.result(timeout_s=60)
+
+
+def test_streaming_handle(serve_session):
+    @serve.deployment
+    class Streamer:
+        def stream_nums(self, n):
+            for i in range(n):
+                yield i * 3
+
+    h = serve.run(Streamer.bind(), name="streamy", http=False)
+    gen = h.options(method_name="stream_nums", stream=True).remote(4)
+    assert list(gen) == [0, 3, 6, 9]
