@@ -206,7 +206,12 @@ class ReplicaActor:
             r = await loop.run_in_executor(
                 None, lambda: target(*args, **kwargs)
             )
-            if asyncio.iscoroutine(r):
+            import inspect as _insp
+
+            # NB: asyncio.iscoroutine() is True for plain generators
+            # (legacy coroutines) — generators must pass through intact
+            # for streaming responses.
+            if _insp.iscoroutine(r) and not _insp.isgenerator(r):
                 r = await r
             return r
         finally:
