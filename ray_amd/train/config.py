@@ -16,6 +16,11 @@ class ScalingConfig:
     resources_per_worker: Optional[Dict[str, float]] = None
     placement_strategy: str = "PACK"
     trainer_resources: Optional[Dict[str, float]] = None
+    # elastic training (reference: v2 scaling_policy/elastic.py:27):
+    # on failure-retry the group restarts with as many workers as the
+    # cluster can currently place, down to min_workers.
+    elastic: bool = False
+    min_workers: int = 1
 
     def worker_resources(self) -> Dict[str, float]:
         res = dict(self.resources_per_worker or {})

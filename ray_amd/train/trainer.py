@@ -151,9 +151,10 @@ class DataParallelTrainer:
         os.makedirs(storage_dir, exist_ok=True)
         latest_ckpt = self._resume.path if self._resume else None
         last_err: Optional[str] = None
+        n_workers = self.scaling_config.num_workers
         while True:
             try:
-                return self._fit_once(ray, storage_dir, latest_ckpt)
+                return self._fit_once(ray, storage_dir, latest_ckpt, n_workers)
             except _WorkerGroupError as e:
                 last_err = e.error
                 latest_ckpt = e.latest_ckpt or latest_ckpt
@@ -166,10 +167,29 @@ class DataParallelTrainer:
                     )
                 if retries > 0:
                     retries -= 1
+                if self.scaling_config.elastic:
+                    n_workers = self._elastic_world_size(ray, n_workers)
 
-    def _fit_once(self, ray, storage_dir: str, latest_ckpt: Optional[str]) -> Result:
+    def _elastic_world_size(self, ray, current: int) -> int:
+        """Shrink to what the cluster can place right now (elastic
+        restart-from-checkpoint; reference: ElasticScalingPolicy)."""
+        import time as _t
+
         sc = self.scaling_config
-        n = sc.num_workers
+        res = sc.worker_resources()
+        _t.sleep(1.0)  # let node death propagate
+        avail = ray.available_resources()
+        fit = current
+        for k, v in res.items():
+            if v > 0:
+                fit = min(fit, int(avail.get(k, 0) // v))
+        n = max(sc.min_workers, min(current, fit))
+        return max(n, 1)
+
+    def _fit_once(self, ray, storage_dir: str, latest_ckpt: Optional[str],
+                  n_override: Optional[int] = None) -> Result:
+        sc = self.scaling_config
+        n = n_override or sc.num_workers
         res = sc.worker_resources()
         from ..util import PlacementGroupSchedulingStrategy, placement_group, remove_placement_group
 
@@ -189,22 +209,27 @@ class DataParallelTrainer:
             for i in range(n)
         ]
         try:
-            port = _free_port()
-            ray.get([
-                w.setup_dist.remote("127.0.0.1", port, self._backend())
-                for w in workers
-            ], timeout=180)
+            try:
+                port = _free_port()
+                ray.get([
+                    w.setup_dist.remote("127.0.0.1", port, self._backend())
+                    for w in workers
+                ], timeout=180)
 
-            shard_payloads = self._make_shards(n)
-            fn_bytes = cloudpickle.dumps(self._fn)
-            ray.get([
-                w.start_training.remote(
-                    fn_bytes, self._config, storage_dir,
-                    self.run_config.name or "run", latest_ckpt,
-                    shard_payloads[i],
+                shard_payloads = self._make_shards(n)
+                fn_bytes = cloudpickle.dumps(self._fn)
+                ray.get([
+                    w.start_training.remote(
+                        fn_bytes, self._config, storage_dir,
+                        self.run_config.name or "run", latest_ckpt,
+                        shard_payloads[i],
+                    )
+                    for i, w in enumerate(workers)
+                ], timeout=180)
+            except (ray.exceptions.RayActorError, ray.exceptions.RayError) as e:
+                raise _WorkerGroupError(
+                    f"worker group setup failed: {e}", latest_ckpt
                 )
-                for i, w in enumerate(workers)
-            ], timeout=180)
 
             rows = []
             kept_checkpoints = []
@@ -214,9 +239,9 @@ class DataParallelTrainer:
                 try:
                     states = ray.get([w.fetch.remote() for w in workers],
                                      timeout=120)
-                except ray.exceptions.RayActorError:
+                except ray.exceptions.RayError as e:
                     raise _WorkerGroupError(
-                        "a training worker died", latest_checkpoint_path
+                        f"a training worker died: {e}", latest_checkpoint_path
                     )
                 for st in states:
                     if st["error"]:

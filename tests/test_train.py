@@ -188,3 +188,55 @@ def test_checkpoint_num_to_keep(ray_start_regular, tmp_path):
     run_dir = os.path.join(str(tmp_path), "keep")
     ckpts = [d for d in os.listdir(run_dir) if d.startswith("checkpoint_")]
     assert len(ckpts) <= 2
+
+
+def test_elastic_restart_with_fewer_workers(ray_start_cluster, tmp_path):
+    """Elastic policy: after losing a node, the retry restarts with the
+    workers that still fit (reference: v2 elastic scaling policy)."""
+    import time
+
+    cluster = ray_start_cluster  # head 4 CPUs
+    extra = cluster.add_node(num_cpus=2, resources={"extra": 2})
+    cluster.connect()
+    cluster.wait_for_nodes()
+    from ray_amd.train import FailureConfig, RunConfig, ScalingConfig
+    from ray_amd.train.torch import TorchTrainer
+
+    flag = str(tmp_path / "fail_once")
+
+    def loop(config):
+        import os
+
+        import ray_amd.train as train
+
+        import time as _t
+
+        ctx = train.get_context()
+        if not os.path.exists(config["flag"]) and ctx.get_world_rank() == 0:
+            open(config["flag"], "w").close()
+            _t.sleep(1.5)  # let the chopper remove the extra node first
+            raise RuntimeError("injected failure while node shrinks")
+        train.report({"world": ctx.get_world_size()})
+
+    # 6 single-CPU workers fit across both nodes initially
+    t = TorchTrainer(
+        loop,
+        train_loop_config={"flag": flag},
+        scaling_config=ScalingConfig(num_workers=5, elastic=True,
+                                     min_workers=1),
+        run_config=RunConfig(
+            name="elastic", storage_path=str(tmp_path),
+            failure_config=FailureConfig(max_failures=2),
+        ),
+    )
+    # remove the extra node while the first attempt fails
+    import threading
+
+    def chopper():
+        time.sleep(0.5)
+        cluster.remove_node(extra)
+
+    threading.Thread(target=chopper, daemon=True).start()
+    res = t.fit()
+    assert res.error is None
+    assert res.metrics["world"] <= 4  # shrunk below the original 5
