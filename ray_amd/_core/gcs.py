@@ -33,6 +33,14 @@ class NodeInfo:
         self.client: Optional[RpcClient] = None
         self.start_time = time.time()
 
+    def adjust(self, delta: Dict[str, float], sign: float = 1.0):
+        """Optimistic cache adjustment, clamped to [0, total] — raylet
+        reports are the truth and overwrite this within a second."""
+        for k, v in delta.items():
+            cur = self.resources_available.get(k, 0.0) + sign * v
+            hi = self.resources_total.get(k, cur)
+            self.resources_available[k] = min(max(cur, 0.0), hi)
+
 
 class ActorInfo:
     def __init__(self, actor_id: bytes, spec: dict):
@@ -252,10 +260,7 @@ class GcsServer:
                 a.addr = r["addr"]
                 a.state = ACTOR_ALIVE
                 if not in_pg:
-                    for k, v in req.items():
-                        node.resources_available[k] = (
-                            node.resources_available.get(k, 0) - v
-                        )
+                    node.adjust(req, -1.0)
                 for f in a.waiters:
                     if not f.done():
                         f.set_result(None)
@@ -313,8 +318,7 @@ class GcsServer:
     async def _on_actor_exit(self, a: ActorInfo, cause: str, expected: bool):
         node = self.nodes.get(a.node_id) if a.node_id else None
         if node is not None and a.state == ACTOR_ALIVE and not a.spec.get("pg_id"):
-            for k, v in a.spec.get("resources", {}).items():
-                node.resources_available[k] = node.resources_available.get(k, 0) + v
+            node.adjust(a.spec.get("resources", {}), +1.0)
         if (not expected) and a.num_restarts < a.spec.get("max_restarts", 0):
             a.num_restarts += 1
             a.state = ACTOR_RESTARTING
@@ -407,8 +411,7 @@ class GcsServer:
                         "commit_bundle", {"pg_id": pg.pg_id, "bundle_index": idx}
                     )
                     pg.bundle_nodes[idx] = node.node_id
-                    for k, v in pg.bundles[idx].items():
-                        node.resources_available[k] = node.resources_available.get(k, 0) - v
+                    node.adjust(pg.bundles[idx], -1.0)
                 pg.state = "CREATED"
                 break
             for idx, node in reserved:
@@ -509,8 +512,7 @@ class GcsServer:
                     await node.client.call(
                         "remove_bundle", {"pg_id": pg.pg_id, "bundle_index": idx}
                     )
-                    for k, v in pg.bundles[idx].items():
-                        node.resources_available[k] = node.resources_available.get(k, 0) + v
+                    node.adjust(pg.bundles[idx], +1.0)
                 except Exception:
                     pass
         return True

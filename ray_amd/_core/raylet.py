@@ -219,13 +219,20 @@ class Raylet:
             bundle_key = getattr(w, "actor_bundle", None)
             for g in w.gpu_ids:
                 self._free_gpus.append(g)
-            if bundle_key is not None and bundle_key in self.bundles:
-                b = self.bundles[bundle_key]
-                for k, v in spec_res.items():
-                    b["avail"][k] = b["avail"].get(k, 0) + v
+            if bundle_key is not None:
+                b = self.bundles.get(bundle_key)
+                if b is not None:
+                    for k, v in spec_res.items():
+                        b["avail"][k] = b["avail"].get(k, 0) + v
+                # else: the bundle was already rolled back — its
+                # resources went back to the node then; adding here
+                # would double-release (observed: avail > total)
             else:
                 for k, v in spec_res.items():
-                    self.avail[k] = self.avail.get(k, 0) + v
+                    self.avail[k] = min(
+                        self.avail.get(k, 0) + v,
+                        self.resources_total.get(k, v),
+                    )
             rc = w.proc.poll()
             asyncio.ensure_future(self._notify_actor_exit(w, rc))
         self._try_grant()
