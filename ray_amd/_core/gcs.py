@@ -68,8 +68,14 @@ class PlacementGroupInfo:
 
 
 class GcsServer:
-    def __init__(self, sock_path: str):
+    """Set persist_path to journal KV + actor + PG state; a restarted
+    GCS on the same socket replays it (reference: gcs/store_client Redis
+    persistence + gcs_init_data replay)."""
+
+    def __init__(self, sock_path: str, persist_path: str = ""):
         self.sock_path = sock_path
+        self.persist_path = persist_path
+        self._dirty = False
         self.server = RpcServer()
         self.kv: Dict[str, Dict[bytes, bytes]] = {}
         self.nodes: Dict[bytes, NodeInfo] = {}
@@ -89,7 +95,69 @@ class GcsServer:
         self._timeline: List[dict] = []
 
     async def start(self):
+        if self.persist_path:
+            self._restore()
+            asyncio.ensure_future(self._persist_loop())
+        import os as _os
+
+        try:
+            _os.unlink(self.sock_path)
+        except OSError:
+            pass
         await self.server.start_unix(self.sock_path)
+
+    # ---------- persistence ----------
+    def _snapshot(self) -> bytes:
+        import pickle
+
+        return pickle.dumps(
+            {
+                "kv": self.kv,
+                "named_actors": self.named_actors,
+                "actor_specs": {
+                    aid: a.spec for aid, a in self.actors.items()
+                    if a.state != ACTOR_DEAD
+                },
+                "job_counter": self.job_counter,
+            }
+        )
+
+    def _restore(self):
+        import os as _os
+        import pickle
+
+        if not _os.path.exists(self.persist_path):
+            return
+        try:
+            with open(self.persist_path, "rb") as f:
+                snap = pickle.load(f)
+        except Exception:
+            return
+        self.kv = snap.get("kv", {})
+        self.named_actors = snap.get("named_actors", {})
+        self.job_counter = snap.get("job_counter", 0)
+        # actors come back PENDING and are rescheduled once a raylet
+        # re-registers (restart-based recovery)
+        for aid, spec in snap.get("actor_specs", {}).items():
+            a = ActorInfo(aid, spec)
+            self.actors[aid] = a
+            asyncio.ensure_future(self._schedule_actor(a))
+
+    async def _persist_loop(self):
+        while True:
+            await asyncio.sleep(0.5)
+            if not self._dirty:
+                continue
+            self._dirty = False
+            try:
+                tmp = self.persist_path + ".tmp"
+                with open(tmp, "wb") as f:
+                    f.write(self._snapshot())
+                import os as _os
+
+                _os.replace(tmp, self.persist_path)
+            except Exception:
+                pass
 
     # ---------- KV ----------
     def h_kv_put(self, conn, p):
@@ -98,6 +166,7 @@ class GcsServer:
         exists = key in ns
         if p.get("overwrite", True) or not exists:
             ns[key] = p["value"]
+            self._dirty = True
             return not exists
         return False
 
@@ -216,6 +285,7 @@ class GcsServer:
                     raise ValueError(f"actor name {a.name!r} already taken")
             self.named_actors[key] = actor_id
         self.actors[actor_id] = a
+        self._dirty = True
         asyncio.ensure_future(self._schedule_actor(a))
         return {"existing": None}
 
@@ -529,7 +599,8 @@ def main():
     import sys
 
     sock = sys.argv[1]
-    gcs = GcsServer(sock)
+    persist = sys.argv[2] if len(sys.argv) > 2 else ""
+    gcs = GcsServer(sock, persist)
 
     async def run():
         await gcs.start()

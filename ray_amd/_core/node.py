@@ -68,8 +68,9 @@ def detect_gpus() -> int:
 def start_gcs(session_dir: str, env=None) -> tuple:
     sock = os.path.join(session_dir, "sock", "gcs")
     logf = open(os.path.join(session_dir, "logs", "gcs.log"), "ab", buffering=0)
+    persist = os.path.join(session_dir, "gcs_state.bin")
     proc = subprocess.Popen(
-        [sys.executable, "-m", "ray_amd._core.gcs", sock],
+        [sys.executable, "-m", "ray_amd._core.gcs", sock, persist],
         stdout=logf,
         stderr=subprocess.STDOUT,
         env=env or os.environ.copy(),

@@ -632,10 +632,34 @@ def main():
             with open(args.ready_file, "w") as f:
                 f.write(raylet.addr + "\n" + raylet.node_id.hex())
         try:
-            # exit (and kill workers) when the GCS goes away — prevents
-            # daemon leaks when the driver is killed hard
-            while raylet.gcs.connected:
-                await asyncio.sleep(0.5)
+            # on GCS loss: try to reconnect + re-register for 30s (GCS
+            # restart with persisted state); exit if it stays gone
+            while True:
+                if raylet.gcs.connected:
+                    await asyncio.sleep(0.5)
+                    continue
+                recovered = False
+                deadline = time.time() + 30.0
+                while time.time() < deadline:
+                    try:
+                        c = RpcClient()
+                        await c.connect(raylet.gcs_addr, retries=5)
+                        raylet.gcs = c
+                        await raylet.gcs.call(
+                            "register_node",
+                            {
+                                "node_id": raylet.node_id,
+                                "addr": raylet.addr,
+                                "resources": raylet.resources_total,
+                                "labels": raylet.labels,
+                            },
+                        )
+                        recovered = True
+                        break
+                    except Exception:
+                        await asyncio.sleep(0.5)
+                if not recovered:
+                    return
         finally:
             raylet.shutdown_workers()
 

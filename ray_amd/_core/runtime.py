@@ -269,6 +269,30 @@ class CoreRuntime:
         """Worker mode: record which thread runs the loop."""
         self._loop_thread_obj = t
 
+    async def _gcs_rpc(self, method, payload):
+        """GCS call with reconnect (reference: retryable_grpc_client —
+        clients survive a GCS restart when state is persisted)."""
+        try:
+            return await self.gcs.call(method, payload)
+        except ConnectionLost:
+            await self._reconnect_gcs()
+            return await self.gcs.call(method, payload)
+
+    async def _reconnect_gcs(self):
+        deadline = time.monotonic() + 30.0
+        last = None
+        while time.monotonic() < deadline:
+            try:
+                c = RpcClient()
+                await c.connect(self.gcs_addr, retries=5)
+                if await asyncio.wait_for(c.call("ping", {}), 5) == "pong":
+                    self.gcs = c
+                    return
+            except Exception as e:
+                last = e
+            await asyncio.sleep(0.25)
+        raise ConnectionLost(f"GCS unreachable after restart wait: {last}")
+
     async def _async_start(self):
         sock = os.path.join(
             self.session_dir, "sock", f"rt_{os.getpid()}_{os.urandom(3).hex()}"
@@ -661,7 +685,7 @@ class CoreRuntime:
             return
         fut = self._fn_exported[fn_id] = self.loop.create_future()
         try:
-            await self.gcs.call(
+            await self._gcs_rpc(
                 "kv_put",
                 {"ns": "fn", "key": fn_id, "value": pickled, "overwrite": False},
             )
@@ -674,7 +698,7 @@ class CoreRuntime:
     async def load_function(self, fn_id: bytes):
         fn = self._fn_cache.get(fn_id)
         if fn is None:
-            data = await self.gcs.call("kv_get", {"ns": "fn", "key": fn_id})
+            data = await self._gcs_rpc("kv_get", {"ns": "fn", "key": fn_id})
             if data is None:
                 raise exc.RaySystemError(f"function {fn_id.hex()} not found in GCS")
             import cloudpickle
@@ -932,12 +956,12 @@ class CoreRuntime:
         pg = options.get("placement_group")
 
         async def _do():
-            await self.gcs.call(
+            await self._gcs_rpc(
                 "kv_put",
                 {"ns": "actorcls", "key": spec_kv_key, "value": pickled_cls,
                  "overwrite": False},
             )
-            await self.gcs.call(
+            await self._gcs_rpc(
                 "kv_put",
                 {"ns": "actorargs", "key": actor_id, "value": bytes(args_blob[:n])},
             )
@@ -958,11 +982,11 @@ class CoreRuntime:
             }
             if pg is not None:
                 payload["pg_id"], payload["bundle_index"] = pg[0], pg[1]
-                st = await self.gcs.call("pg_wait_ready", {"pg_id": pg[0]})
+                st = await self._gcs_rpc("pg_wait_ready", {"pg_id": pg[0]})
                 nodes = st.get("bundle_nodes") or []
                 if pg[1] is not None and pg[1] < len(nodes):
                     payload["pg_node"] = nodes[pg[1]]
-            r = await self.gcs.call("register_actor", payload)
+            r = await self._gcs_rpc("register_actor", payload)
             for cref in captured:
                 self._remove_submitted_ref(cref.id)
             return r
@@ -1033,7 +1057,7 @@ class CoreRuntime:
             return
         fut = st["resolving"] = self.loop.create_future()
         try:
-            r = await self.gcs.call(
+            r = await self._gcs_rpc(
                 "resolve_actor", {"actor_id": actor_id, "wait": True, "timeout": 120.0}
             )
             state = r.get("state")
@@ -1097,7 +1121,7 @@ class CoreRuntime:
 
     def kill_actor(self, actor_id: bytes, no_restart=True):
         self._call_sync(
-            self.gcs.call("kill_actor", {"actor_id": actor_id, "no_restart": no_restart})
+            self._gcs_rpc("kill_actor", {"actor_id": actor_id, "no_restart": no_restart})
         )
         st = self._actors.get(actor_id)
         if st is not None:
@@ -1107,7 +1131,7 @@ class CoreRuntime:
     # ------------- misc -------------
 
     def gcs_call(self, method, payload, timeout=None):
-        return self._call_sync(self.gcs.call(method, payload), timeout)
+        return self._call_sync(self._gcs_rpc(method, payload), timeout)
 
     def raylet_call(self, method, payload, timeout=None):
         return self._call_sync(self.raylet.call(method, payload), timeout)
