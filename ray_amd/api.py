@@ -124,6 +124,8 @@ def shutdown(_exiting_interpreter: bool = False):
         if _rt.is_initialized():
             rt = _rt.global_runtime()
             rt.shutdown()
+        # only the driver that STARTED the cluster tears it down;
+        # connected drivers just disconnect
         if _cluster is not None:
             _cluster.shutdown()
             _cluster = None

@@ -369,3 +369,37 @@ def test_runtime_env_py_modules(ray_start_regular, tmp_path):
         runtime_env={"py_modules": [str(pkg)]}
     ).remote()
     assert ray.get(a.get.remote(), timeout=60) == 42
+
+
+def test_second_driver_connects(ray_start_regular):
+    """A separate driver process connects to the running cluster by
+    address and shares named actors (reference: multi-driver clusters)."""
+    import subprocess
+    import sys
+
+    @ray.remote
+    class Shared:
+        def __init__(self):
+            self.v = 41
+
+        def bump(self):
+            self.v += 1
+            return self.v
+
+    Shared.options(name="shared_counter", lifetime="detached").remote()
+    rt = ray.api._rt.global_runtime()
+    script = f"""
+import ray_amd as ray
+ray.init(address={rt.session_dir!r})
+h = ray.get_actor("shared_counter")
+print("RESULT", ray.get(h.bump.remote(), timeout=30))
+ray.shutdown(_exiting_interpreter=True)
+"""
+    out = subprocess.run(
+        [sys.executable, "-c", script], capture_output=True, text=True,
+        timeout=120,
+    )
+    assert "RESULT 42" in out.stdout, out.stdout + out.stderr
+    # first driver still sees the state
+    h = ray.get_actor("shared_counter")
+    assert ray.get(h.bump.remote()) == 43
