@@ -121,13 +121,19 @@ def init(
 def shutdown(_exiting_interpreter: bool = False):
     global _cluster
     with _init_lock:
-        if _rt.is_initialized():
-            rt = _rt.global_runtime()
-            rt.shutdown()
+        try:
+            if _rt.is_initialized():
+                rt = _rt.global_runtime()
+                rt.shutdown()
+        except Exception:
+            pass
         # only the driver that STARTED the cluster tears it down;
         # connected drivers just disconnect
         if _cluster is not None:
-            _cluster.shutdown()
+            try:
+                _cluster.shutdown()
+            except Exception:
+                pass
             _cluster = None
 
 
