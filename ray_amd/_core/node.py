@@ -139,9 +139,13 @@ def start_raylet(
         if time.time() > deadline:
             raise RuntimeError("raylet start timed out")
         time.sleep(0.01)
-    with open(ready) as f:
-        addr, node_id_hex = f.read().strip().split("\n")
-    return proc, addr, bytes.fromhex(node_id_hex)
+    for _ in range(50):
+        with open(ready) as f:
+            parts = f.read().strip().split("\n")
+        if len(parts) == 2 and parts[1]:
+            return proc, parts[0], bytes.fromhex(parts[1])
+        time.sleep(0.01)
+    raise RuntimeError("raylet ready file malformed")
 
 
 class LocalCluster:

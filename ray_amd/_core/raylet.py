@@ -629,8 +629,11 @@ def main():
     async def run():
         await raylet.start()
         if args.ready_file:
-            with open(args.ready_file, "w") as f:
+            # atomic write: the starter polls for existence
+            tmp = args.ready_file + ".tmp"
+            with open(tmp, "w") as f:
                 f.write(raylet.addr + "\n" + raylet.node_id.hex())
+            os.replace(tmp, args.ready_file)
         try:
             # on GCS loss: try to reconnect + re-register for 30s (GCS
             # restart with persisted state); exit if it stays gone

@@ -235,6 +235,7 @@ class CoreRuntime:
         self._refs_lock = threading.Lock()
 
         self._streams: Dict[bytes, dict] = {}  # task_id -> stream state
+        self._cancelled_returns: set = set()
         self._pools: Dict[tuple, _LeasePool] = {}
         self._fn_exported: Dict[bytes, asyncio.Future] = {}
         self._fn_cache: Dict[bytes, Any] = {}
@@ -825,7 +826,21 @@ class CoreRuntime:
             if a is not None:
                 self._remove_local_ref(a[0])
 
+    def cancel_task(self, return_oid: bytes):
+        """Mark the task producing return_oid cancelled; if it has not
+        been dispatched yet, its returns resolve to
+        TaskCancelledError."""
+        self._cancelled_returns.add(return_oid)
+
     async def _dispatch_normal_task(self, spec, options) -> dict:
+        if spec["returns"] and spec["returns"][0] in self._cancelled_returns:
+            self._cancelled_returns.discard(spec["returns"][0])
+            return {
+                "status": "error",
+                "error": serialization.dumps(
+                    exc.TaskCancelledError(spec.get("name", ""))
+                ),
+            }
         key, res, pg_key = self._pool_key(options)
         pool = self._pools.get(key)
         if pool is None:

@@ -24,6 +24,18 @@ _init_lock = threading.Lock()
 _cluster: Optional[_node.LocalCluster] = None
 _namespace = "default"
 
+
+def _atexit_shutdown():
+    try:
+        shutdown(_exiting_interpreter=True)
+    except Exception:
+        pass
+
+
+import atexit  # noqa: E402
+
+atexit.register(_atexit_shutdown)
+
 DEFAULT_TASK_OPTIONS = dict(num_cpus=1, num_gpus=0, num_returns=1, max_retries=3)
 # Reference semantics (actor.py): an actor with no explicit resource
 # request reserves NOTHING while alive (scheduled by 1-CPU availability
@@ -464,8 +476,10 @@ def kill(actor: ActorHandle, *, no_restart: bool = True):
 
 
 def cancel(ref: ObjectRef, *, force: bool = False, recursive: bool = True):
-    # Best-effort: queued-but-not-started tasks are dropped.
-    pass
+    """Best-effort cancellation: tasks not yet started are dropped and
+    their refs resolve to TaskCancelledError (reference semantics for
+    non-force cancel; running tasks are not interrupted)."""
+    _rt.global_runtime().cancel_task(ref.id)
 
 
 def get_actor(name: str, namespace: Optional[str] = None) -> ActorHandle:

@@ -403,3 +403,23 @@ ray.shutdown(_exiting_interpreter=True)
     # first driver still sees the state
     h = ray.get_actor("shared_counter")
     assert ray.get(h.bump.remote()) == 43
+
+
+def test_cancel_queued_task(ray_start_regular):
+    @ray.remote
+    def blocker():
+        time.sleep(5)
+        return "done"
+
+    @ray.remote
+    def victim():
+        return "ran"
+
+    # saturate the 4 CPUs, then queue a victim and cancel it
+    blockers = [blocker.remote() for _ in range(4)]
+    time.sleep(0.3)
+    v = victim.remote()
+    ray.cancel(v)
+    with pytest.raises(ray.exceptions.TaskCancelledError):
+        ray.get(v, timeout=30)
+    assert ray.get(blockers, timeout=30) == ["done"] * 4
