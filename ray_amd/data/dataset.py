@@ -498,19 +498,23 @@ class Dataset:
         prefetch_batches: int = 1,
         **kwargs,
     ):
+        """Batches as torch tensors on `device`. With a CUDA target the
+        CPU-side work (format, pin) runs in a background thread
+        `prefetch_batches` ahead and H2D copies are non_blocking on
+        pinned memory, overlapping the consumer's compute (SURVEY §2.9
+        #7: async H2D staging)."""
+        import queue as _q
+        import threading as _th
+
         import torch
 
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
-        for batch in self.iter_batches(
-            batch_size=batch_size,
-            batch_format="numpy",
-            drop_last=drop_last,
-            local_shuffle_buffer_size=local_shuffle_buffer_size,
-        ):
+        use_cuda = str(device).startswith("cuda") and torch.cuda.is_available()
+
+        def to_cpu_tensors(batch):
             if collate_fn is not None:
-                yield collate_fn(batch)
-                continue
+                return ("collated", collate_fn(batch))
             out = {}
             for k, v in batch.items():
                 t = torch.as_tensor(np.ascontiguousarray(v))
@@ -518,8 +522,52 @@ class Dataset:
                     dt = dtypes.get(k) if isinstance(dtypes, dict) else dtypes
                     if dt is not None:
                         t = t.to(dt)
-                out[k] = t.to(device, non_blocking=True)
-            yield out
+                if use_cuda:
+                    t = t.pin_memory()
+                out[k] = t
+            return ("dict", out)
+
+        src = self.iter_batches(
+            batch_size=batch_size,
+            batch_format="numpy",
+            drop_last=drop_last,
+            local_shuffle_buffer_size=local_shuffle_buffer_size,
+        )
+
+        def move(item):
+            kind, val = item
+            if kind == "collated":
+                return val
+            return {
+                k: (t.to(device, non_blocking=True) if use_cuda
+                    else t.to(device))
+                for k, t in val.items()
+            }
+
+        if prefetch_batches and prefetch_batches > 0:
+            q: "_q.Queue" = _q.Queue(maxsize=prefetch_batches)
+            DONE = object()
+
+            def producer():
+                try:
+                    for b in src:
+                        q.put(to_cpu_tensors(b))
+                except BaseException as e:  # noqa
+                    q.put(("error", e))
+                finally:
+                    q.put(DONE)
+
+            _th.Thread(target=producer, daemon=True).start()
+            while True:
+                item = q.get()
+                if item is DONE:
+                    return
+                if item[0] == "error":
+                    raise item[1]
+                yield move(item)
+        else:
+            for b in src:
+                yield move(to_cpu_tensors(b))
 
     def to_pandas(self, limit: Optional[int] = None):
         tables = list(self._iter_tables())
