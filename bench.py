@@ -64,6 +64,9 @@ def main():
     if args.model == "resnet50":
         run_resnet(args, world_size, rank, local_rank, device, use_cpu)
         return
+    if args.model == "ppo":
+        run_ppo(args, world_size, rank, use_cpu)
+        return
 
     from ray_amd.models.llama import CONFIGS, LlamaModel
     from ray_amd.ops import FusedAdamW
@@ -163,6 +166,70 @@ def main():
         print(json.dumps(result))
 
     if distributed:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+def run_ppo(args, world_size, rank, use_cpu):
+    """RLlib PPO CartPole env-steps/sec (BASELINE metric 2; config #1 is
+    the CPU plumbing case). Each rank runs an independent PPO learner
+    (GPU if available) with its own rollout workers — weak scaling."""
+    import ray_amd as ray
+
+    ray.init(num_cpus=max(4, (os.cpu_count() or 8) // max(world_size, 1)),
+             num_gpus=0 if use_cpu else 1)
+    from ray_amd.rllib.algorithms.ppo import PPOConfig
+
+    config = (
+        PPOConfig()
+        .environment("CartPole-v1")
+        .env_runners(num_env_runners=2, num_envs_per_env_runner=8)
+        .training(train_batch_size=4000, minibatch_size=512, num_epochs=8)
+        .learners(num_gpus_per_learner=0 if use_cpu else 1)
+    )
+    algo = config.build()
+    for _ in range(max(args.warmup, 1)):
+        algo.train()
+    if world_size > 1:
+        import torch.distributed as dist
+
+        dist.barrier()
+    t0 = time.perf_counter()
+    steps = 0
+    for _ in range(args.steps):
+        r = algo.train()
+        steps += r["num_env_steps_sampled"]
+    elapsed = time.perf_counter() - t0
+    if world_size > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    total = steps * world_size
+    if rank == 0:
+        print(json.dumps({
+            "metric": "ppo_env_steps_per_sec",
+            "value": round(total / elapsed, 2),
+            "unit": "env_steps/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {"model": "ppo-cartpole", "train_batch_size": 4000,
+                       "env_runners_per_learner": 2,
+                       "parallelism": f"dp{world_size}",
+                       "reward_mean": r.get("episode_reward_mean")},
+        }))
+    algo.stop()
+    ray.shutdown()
+    if world_size > 1:
         import torch.distributed as dist
 
         dist.destroy_process_group()

@@ -240,3 +240,43 @@ def test_elastic_restart_with_fewer_workers(ray_start_cluster, tmp_path):
     res = t.fit()
     assert res.error is None
     assert res.metrics["world"] <= 4  # shrunk below the original 5
+
+
+def test_huggingface_model_in_trainer(ray_start_regular, tmp_path):
+    """HF Transformers models train inside TorchTrainer (reference:
+    train/huggingface integration). Random-init config (offline image)."""
+
+    def loop(config):
+        import torch
+        from transformers import BertConfig, BertForSequenceClassification
+
+        import ray_amd.train as train
+        from ray_amd.train.torch import prepare_model
+
+        cfg = BertConfig(
+            vocab_size=128, hidden_size=32, num_hidden_layers=2,
+            num_attention_heads=2, intermediate_size=64,
+            max_position_embeddings=64, num_labels=2,
+        )
+        model = prepare_model(BertForSequenceClassification(cfg))
+        opt = torch.optim.AdamW(model.parameters(), lr=5e-4)
+        x = torch.randint(0, 128, (8, 16))
+        y = torch.randint(0, 2, (8,))
+        first = None
+        for _ in range(4):
+            out = model(input_ids=x, labels=y)
+            out.loss.backward()
+            opt.step()
+            opt.zero_grad()
+            if first is None:
+                first = float(out.loss)
+        train.report({"first": first, "last": float(out.loss)})
+
+    t = TorchTrainer(
+        loop,
+        scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(name="hf", storage_path=str(tmp_path)),
+    )
+    res = t.fit()
+    assert res.error is None
+    assert res.metrics["last"] <= res.metrics["first"] + 0.1
