@@ -1114,8 +1114,19 @@ class CoreRuntime:
                     self._ingest_result(spec, reply)
                     return
                 except (ConnectionLost, ConnectionError):
+                    # at-most-once (reference default, max_task_retries=0):
+                    # the call may have executed before the actor died,
+                    # so it must NOT be resent to a restarted instance
                     st["client"] = None
-                    continue
+                    err = serialization.dumps(
+                        exc.ActorUnavailableError(
+                            f"actor {actor_id.hex()} died while this call "
+                            "was in flight"
+                        )
+                    )
+                    for oid in spec["returns"]:
+                        self._store_put(oid, ("err", err))
+                    return
             err = serialization.dumps(
                 exc.ActorUnavailableError(f"actor {actor_id.hex()} unreachable")
             )

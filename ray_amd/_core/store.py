@@ -22,6 +22,11 @@ from typing import Dict, Optional, Tuple
 
 from . import serialization
 
+try:
+    from ray_amd import _shm_native  # native C++ data path (csrc/shm_store.cpp)
+except ImportError:
+    _shm_native = None
+
 
 def shm_path(shm_dir: str, object_id: bytes) -> str:
     return os.path.join(shm_dir, object_id.hex())
@@ -125,11 +130,18 @@ class ObjectWriter:
 
 class MappedObject:
     """A read-only mmap of a sealed object; keeps the map alive while
-    deserialized zero-copy views reference it."""
+    deserialized zero-copy views reference it. Native path when built."""
 
-    __slots__ = ("_mm", "view")
+    __slots__ = ("_mm", "_owner", "view")
 
     def __init__(self, path: str):
+        if _shm_native is not None:
+            self._owner = _shm_native.map_object(path)
+            # memoryview(exporter) keeps the mapping alive through any
+            # numpy views derived from it
+            self.view = memoryview(self._owner)
+            self._mm = None
+            return
         fd = os.open(path, os.O_RDONLY)
         try:
             size = os.fstat(fd).st_size
@@ -140,8 +152,33 @@ class MappedObject:
 
 
 def put_serialized(shm_dir: str, object_id: bytes, meta: bytes, buffers) -> int:
-    """Write a serialized value to shm; returns the sealed size."""
+    """Write a serialized value to shm; returns the sealed size.
+
+    Uses the native C++ path (GIL-released multithreaded memcpy) when
+    the _shm_native extension is built; pure-python fallback otherwise."""
     size = serialization.serialized_size(meta, buffers)
+    if _shm_native is not None:
+        import struct as _struct
+
+        path, pooled_cls = _segment_pool.acquire(shm_dir, object_id, size)
+        file_size = pooled_cls or SegmentPool.size_class(size)
+        nbufs = len(buffers)
+        header = bytearray(16 + 8 * nbufs + len(meta))
+        _struct.pack_into("<IIQ", header, 0, serialization.MAGIC, nbufs,
+                          len(meta))
+        off = 16
+        raws = []
+        for b in buffers:
+            raw = b.raw()
+            if raw.format != "B" or raw.ndim != 1:
+                raw = raw.cast("B")
+            raws.append(raw)
+            _struct.pack_into("<Q", header, off, raw.nbytes)
+            off += 8
+        header[off:] = meta
+        _shm_native.write_object(path + ".tmp", path, file_size,
+                                 bytes(header), raws, 64)
+        return size
     w = ObjectWriter(shm_dir, object_id, size)
     serialization.write_to(w.view, meta, buffers)
     w.seal()

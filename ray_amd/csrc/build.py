@@ -16,6 +16,7 @@ from torch.utils import cpp_extension
 CSRC = os.path.dirname(os.path.abspath(__file__))
 PKG = os.path.dirname(CSRC)
 OUT = os.path.join(PKG, "_hip_ops.so")
+OUT_SHM = os.path.join(PKG, "_shm_native.so")
 
 
 def _python_include() -> str:
@@ -37,7 +38,27 @@ def needs_rebuild() -> bool:
     return False
 
 
+def build_shm(verbose: bool = True, force: bool = False) -> str:
+    """Native shm store data path: plain C++ (g++), no HIP needed."""
+    src = os.path.join(CSRC, "shm_store.cpp")
+    if (not force and os.path.exists(OUT_SHM)
+            and os.path.getmtime(OUT_SHM) >= os.path.getmtime(src)):
+        return OUT_SHM
+    import pybind11
+
+    cmd = [
+        "g++", "-O3", "-std=c++17", "-fPIC", "-shared", src, "-o", OUT_SHM,
+        f"-I{pybind11.get_include()}", f"-I{_python_include()}",
+        "-pthread",
+    ]
+    if verbose:
+        print("[ray_amd build]", " ".join(cmd), file=sys.stderr)
+    subprocess.check_call(cmd)
+    return OUT_SHM
+
+
 def build(verbose: bool = True, force: bool = False) -> str:
+    build_shm(verbose, force)
     if not force and not needs_rebuild():
         return OUT
     hipcc = os.path.join(

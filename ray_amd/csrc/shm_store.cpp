@@ -1,0 +1,147 @@
+// Native shm object-store data path (pybind11).
+//
+// The role of the reference's C++ plasma client/store data plane
+// (object_manager/plasma/): creating, writing and mapping shared-memory
+// objects without Python byte-shuffling overhead. The control plane
+// (tables, pulls, refcounts) stays in ray_amd/_core/store.py; this
+// module owns the hot memcpy path:
+//   write_object(tmp_path, final_path, file_size, header, buffers)
+//       create+ftruncate+mmap, copy header + out-of-band buffers with
+//       GIL released and multithreaded memcpy for large payloads,
+//       fsync-free rename-to-seal.
+//   map_object(path) -> read-only memoryview over a persistent mmap.
+#include <fcntl.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
+
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <thread>
+#include <vector>
+
+#include <pybind11/pybind11.h>
+
+namespace py = pybind11;
+
+static void parallel_memcpy(char* dst, const char* src, size_t n) {
+  constexpr size_t kParMin = 32u << 20;
+  if (n < kParMin) {
+    std::memcpy(dst, src, n);
+    return;
+  }
+  const int nt = 4;
+  size_t chunk = (n + nt - 1) / nt;
+  std::vector<std::thread> ts;
+  for (int i = 0; i < nt; ++i) {
+    size_t s = i * chunk;
+    if (s >= n) break;
+    size_t e = std::min(s + chunk, n);
+    ts.emplace_back(
+        [=] { std::memcpy(dst + s, src + s, e - s); });
+  }
+  for (auto& t : ts) t.join();
+}
+
+static size_t write_object(const std::string& tmp_path,
+                           const std::string& final_path, size_t file_size,
+                           py::buffer header, py::list buffers,
+                           size_t align) {
+  py::buffer_info hdr = header.request();
+  // collect source views while holding the GIL
+  struct Src {
+    const char* p;
+    size_t n;
+  };
+  std::vector<Src> srcs;
+  srcs.reserve(buffers.size());
+  std::vector<py::buffer_info> infos;
+  infos.reserve(buffers.size());
+  for (auto h : buffers) {
+    infos.emplace_back(py::reinterpret_borrow<py::buffer>(h).request());
+    auto& b = infos.back();
+    srcs.push_back({static_cast<const char*>(b.ptr),
+                    static_cast<size_t>(b.size * b.itemsize)});
+  }
+  const char* hp = static_cast<const char*>(hdr.ptr);
+  size_t hn = static_cast<size_t>(hdr.size * hdr.itemsize);
+
+  size_t written;
+  {
+    py::gil_scoped_release release;
+    int fd = ::open(tmp_path.c_str(), O_RDWR, 0600);
+    bool fresh = false;
+    if (fd < 0) {
+      fd = ::open(tmp_path.c_str(), O_CREAT | O_RDWR | O_EXCL, 0600);
+      fresh = true;
+    }
+    if (fd < 0) throw std::runtime_error("shm open failed: " + tmp_path);
+    if (fresh && ::ftruncate(fd, (off_t)file_size) != 0) {
+      ::close(fd);
+      throw std::runtime_error("ftruncate failed");
+    }
+    struct stat st;
+    ::fstat(fd, &st);
+    size_t map_size = (size_t)st.st_size;
+    char* base = static_cast<char*>(
+        ::mmap(nullptr, map_size, PROT_READ | PROT_WRITE, MAP_SHARED, fd, 0));
+    ::close(fd);
+    if (base == MAP_FAILED) throw std::runtime_error("mmap failed");
+    size_t off = 0;
+    std::memcpy(base, hp, hn);
+    off = (hn + align - 1) & ~(align - 1);
+    for (auto& s : srcs) {
+      parallel_memcpy(base + off, s.p, s.n);
+      off = (off + s.n + align - 1) & ~(align - 1);
+    }
+    written = off;
+    ::munmap(base, map_size);
+    if (::rename(tmp_path.c_str(), final_path.c_str()) != 0)
+      throw std::runtime_error("seal rename failed");
+  }
+  return written;
+}
+
+struct Mapping {
+  // Buffer-protocol exporter: memoryview(Mapping) keeps the Mapping
+  // alive, so the munmap happens only after every view (and every
+  // numpy array built on one) is gone — same lifetime contract as
+  // CPython's mmap objects.
+  void* base = nullptr;
+  size_t size = 0;
+  Mapping(const std::string& path) {
+    int fd = ::open(path.c_str(), O_RDONLY);
+    if (fd < 0) throw std::runtime_error("open failed: " + path);
+    struct stat st;
+    ::fstat(fd, &st);
+    size = (size_t)st.st_size;
+    base = ::mmap(nullptr, size, PROT_READ, MAP_SHARED, fd, 0);
+    ::close(fd);
+    if (base == MAP_FAILED) {
+      base = nullptr;
+      throw std::runtime_error("mmap failed: " + path);
+    }
+  }
+  ~Mapping() {
+    if (base) ::munmap(base, size);
+  }
+};
+
+PYBIND11_MODULE(_shm_native, m) {
+  m.doc() = "ray_amd native shm object-store data path";
+  m.def("write_object", &write_object, py::arg("tmp_path"),
+        py::arg("final_path"), py::arg("file_size"), py::arg("header"),
+        py::arg("buffers"), py::arg("align") = 64);
+  py::class_<Mapping>(m, "Mapping", py::buffer_protocol())
+      .def(py::init<const std::string&>())
+      .def_buffer([](Mapping& mp) {
+        return py::buffer_info(mp.base, 1,
+                               py::format_descriptor<unsigned char>::format(),
+                               1, {(ssize_t)mp.size}, {(ssize_t)1},
+                               /*readonly=*/true);
+      });
+  m.def("map_object",
+        [](const std::string& path) { return new Mapping(path); },
+        py::return_value_policy::take_ownership);
+}
