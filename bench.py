@@ -34,6 +34,12 @@ def parse_args():
 
 
 def main():
+    if os.environ.get("BENCH_TRACE"):
+        import faulthandler
+
+        faulthandler.dump_traceback_later(
+            int(os.environ["BENCH_TRACE"]), exit=True
+        )
     args = parse_args()
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))

@@ -152,6 +152,10 @@ class Raylet:
 
     def _spawn_worker(self, actor_spec: Optional[dict] = None) -> WorkerProc:
         env = dict(os.environ)
+        # reference parity (worker defaults): cap BLAS/torch threads so
+        # many workers on one box don't thrash (Ray sets this too);
+        # user runtime_env env_vars can override
+        env.setdefault("OMP_NUM_THREADS", "1")
         env["RAY_AMD_SESSION_DIR"] = self.session_dir
         env["RAY_AMD_GCS_ADDR"] = self.gcs_addr
         env["RAY_AMD_RAYLET_ADDR"] = self.addr
