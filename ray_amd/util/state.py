@@ -94,8 +94,31 @@ def summarize_tasks(**kwargs) -> dict:
     return by_name
 
 
-def get_log(*a, **k):
-    raise NotImplementedError("per-worker logs live under <session>/logs/")
+def get_log(filename: str = None, node_id: str = None, tail: int = 1000,
+            **kwargs):
+    """Yield log lines from the session's log directory (reference:
+    ray.util.state.get_log)."""
+    import os
+
+    rt = _rt()
+    logdir = os.path.join(rt.session_dir, "logs")
+    if filename is None:
+        raise ValueError(
+            f"filename required; available: {sorted(os.listdir(logdir))}"
+        )
+    path = os.path.join(logdir, filename)
+    with open(path, "r", errors="replace") as f:
+        lines = f.readlines()
+    for line in lines[-tail:]:
+        yield line.rstrip("\n")
+
+
+def list_logs(node_id: str = None, **kwargs):
+    import os
+
+    rt = _rt()
+    logdir = os.path.join(rt.session_dir, "logs")
+    return {"worker_out": sorted(os.listdir(logdir))}
 
 
 def _match(row: dict, filters) -> bool:

@@ -423,3 +423,53 @@ def test_cancel_queued_task(ray_start_regular):
     with pytest.raises(ray.exceptions.TaskCancelledError):
         ray.get(v, timeout=30)
     assert ray.get(blockers, timeout=30) == ["done"] * 4
+
+
+def test_joblib_backend(ray_start_regular):
+    import joblib
+
+    from ray_amd.util.joblib import register_ray
+
+    register_ray()
+    with joblib.parallel_backend("ray_amd", n_jobs=2):
+        out = joblib.Parallel()(joblib.delayed(lambda x: x * x)(i)
+                                for i in range(12))
+    assert out == [i * i for i in range(12)]
+
+
+def test_joblib_sklearn(ray_start_regular):
+    import joblib
+    import numpy as np
+    from sklearn.ensemble import RandomForestClassifier
+
+    from ray_amd.util.joblib import register_ray
+
+    register_ray()
+    X = np.random.rand(80, 5)
+    y = (X[:, 0] > 0.5).astype(int)
+    with joblib.parallel_backend("ray_amd", n_jobs=2):
+        clf = RandomForestClassifier(n_estimators=8, n_jobs=2).fit(X, y)
+    assert clf.score(X, y) > 0.8
+
+
+def test_state_get_log(ray_start_regular):
+    @ray.remote
+    def noisy():
+        print("hello-from-worker-log")
+        return 1
+
+    ray.get([noisy.remote() for _ in range(3)])
+    import time as _t
+
+    _t.sleep(0.3)
+    from ray_amd.util import state as state_api
+
+    logs = state_api.list_logs()["worker_out"]
+    assert any(f.startswith("worker_") for f in logs)
+    found = False
+    for f in logs:
+        if f.startswith("worker_"):
+            for line in state_api.get_log(filename=f):
+                if "hello-from-worker-log" in line:
+                    found = True
+    assert found
