@@ -156,6 +156,7 @@ class Raylet:
         # many workers on one box don't thrash (Ray sets this too);
         # user runtime_env env_vars can override
         env.setdefault("OMP_NUM_THREADS", "1")
+        env["PYTHONUNBUFFERED"] = "1"  # live worker logs
         env["RAY_AMD_SESSION_DIR"] = self.session_dir
         env["RAY_AMD_GCS_ADDR"] = self.gcs_addr
         env["RAY_AMD_RAYLET_ADDR"] = self.addr
