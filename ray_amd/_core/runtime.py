@@ -791,6 +791,13 @@ class CoreRuntime:
             while True:
                 try:
                     reply = await self._dispatch_normal_task(spec, options)
+                    if (
+                        reply.get("status") == "error"
+                        and options.get("retry_exceptions")
+                        and retries > 0
+                    ):
+                        retries -= 1
+                        continue
                     self._ingest_result(spec, reply)
                     return
                 except (ConnectionLost, ConnectionError) as e:

@@ -473,3 +473,28 @@ def test_state_get_log(ray_start_regular):
                 if "hello-from-worker-log" in line:
                     found = True
     assert found
+
+
+def test_retry_exceptions(ray_start_regular, tmp_path):
+    marker = str(tmp_path / "attempts")
+
+    @ray.remote(max_retries=3, retry_exceptions=True)
+    def flaky(marker):
+        import os
+
+        n = int(open(marker).read()) if os.path.exists(marker) else 0
+        open(marker, "w").write(str(n + 1))
+        if n < 2:
+            raise RuntimeError(f"attempt {n} fails")
+        return n
+
+    assert ray.get(flaky.remote(marker), timeout=60) == 2
+
+
+def test_no_retry_exceptions_by_default(ray_start_regular):
+    @ray.remote
+    def boom():
+        raise RuntimeError("once")
+
+    with pytest.raises(ray.exceptions.RayTaskError):
+        ray.get(boom.remote(), timeout=30)
