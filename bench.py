@@ -80,9 +80,16 @@ def main():
     cfg = CONFIGS[args.model]
     seq = min(args.seq_len, cfg.max_seq_len)
     torch.manual_seed(1234 + rank)
-    model = LlamaModel(cfg, dtype=dtype,
-                       gradient_checkpointing=args.grad_checkpoint).to(device)
-    if not use_cpu:
+    if use_cpu:
+        model = LlamaModel(cfg, dtype=dtype,
+                           gradient_checkpointing=args.grad_checkpoint)
+    else:
+        # build directly on the GPU: skips 16 GB of CPU init + H2D per
+        # rank (8-rank scale runs would otherwise serialize minutes of
+        # host-side random init)
+        with torch.device(device):
+            model = LlamaModel(cfg, dtype=dtype,
+                               gradient_checkpointing=args.grad_checkpoint)
         model.cosT = model.cosT.to(device)
         model.sinT = model.sinT.to(device)
     n_params = model.num_params()
