@@ -12,6 +12,7 @@
 #include "hip/flash_attn_v2.hip"
 #include "hip/flash_attn_v3.hip"
 #include "hip/flash_attn_v4.hip"
+#include "hip/flash_attn_v5.hip"
 
 #define CHECK_IN(x)                                                     \
   TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                       \
@@ -255,8 +256,14 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
                        (const short*)k.data_ptr(), (const short*)v.data_ptr(),
                        (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
                        causal ? 1 : 0, (int)q_offset, scale);
-  else
+  else if (getenv("RAY_AMD_FA_V4") != nullptr)
     hipLaunchKernelGGL(flash_attn_fwd_v4_bf16, dim3(T / 128, B * Hq),
+                       dim3(256), 0, cur_stream(), (const short*)q.data_ptr(),
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
+                       causal ? 1 : 0, (int)q_offset, scale);
+  else
+    hipLaunchKernelGGL(flash_attn_fwd_v5_bf16, dim3(T / 128, B * Hq),
                        dim3(256), 0, cur_stream(), (const short*)q.data_ptr(),
                        (const short*)k.data_ptr(), (const short*)v.data_ptr(),
                        (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,

@@ -1,4 +1,5 @@
 """Quick flash-attention kernel iteration bench (GPU box)."""
+import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import os, time, torch
 import torch.nn.functional as F
 from ray_amd import ops
