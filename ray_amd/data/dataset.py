@@ -731,6 +731,29 @@ class Dataset:
     def groupby(self, key: str) -> "GroupedData":
         return GroupedData(self, key)
 
+    def join(self, other: "Dataset", *, on: str, how: str = "inner",
+             num_partitions: Optional[int] = None) -> "Dataset":
+        """Hash join on a key column (reference: data join operator).
+        Both sides are hash-partitioned on `on`; per-partition joins run
+        as parallel tasks."""
+        ray = _ray()
+        left_refs = self._materialize_refs()
+        right_refs = other._materialize_refs()
+        P = num_partitions or max(len(left_refs), len(right_refs), 1)
+        lp = _shuffle_exchange(ray, left_refs, on, P)
+        rp = _shuffle_exchange(ray, right_refs, on, P)
+
+        @ray.remote
+        def _join(lt, rt, on=on, how=how):
+            if lt.num_rows == 0 and rt.num_rows == 0:
+                return pa.table({})
+            return lt.join(rt, keys=on, join_type=how)
+
+        out = [_join.remote(lp[i], rp[i]) for i in builtins.range(P)]
+        ds = Dataset(out, [])
+        ds._materialized = out
+        return ds
+
     def sum(self, on: str):
         return self._agg("sum", on)
 

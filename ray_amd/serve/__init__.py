@@ -7,6 +7,8 @@ ServeController actor reconciling replica sets
 _private/proxy.py:1046), serve.batch, and request-rate autoscaling.
 """
 from .api import (  # noqa: F401
+    get_multiplexed_model_id,
+    multiplexed,
     Application,
     Deployment,
     DeploymentHandle,

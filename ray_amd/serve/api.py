@@ -800,6 +800,55 @@ def shutdown():
 
 
 # --------------------------------------------------------------------------
+# serve.multiplexed (reference: serve/multiplex.py — per-replica LRU of
+# loaded models keyed by model id)
+# --------------------------------------------------------------------------
+
+_mux_ctx_model_id: Optional[str] = None
+
+
+def get_multiplexed_model_id() -> str:
+    return _mux_ctx_model_id or ""
+
+
+def multiplexed(_func=None, *, max_num_models_per_replica: int = 3):
+    """Decorator for an async model-loader method; calls are routed with
+    handle.options(multiplexed_model_id=...) and the replica keeps an
+    LRU of loaded models."""
+
+    def dec(loader):
+        import collections
+
+        cache: "collections.OrderedDict" = collections.OrderedDict()
+
+        async def wrapper(self_arg, model_id: str):
+            global _mux_ctx_model_id
+            if model_id in cache:
+                cache.move_to_end(model_id)
+                return cache[model_id]
+            _mux_ctx_model_id = model_id
+            try:
+                model = loader(self_arg, model_id)
+                if asyncio.iscoroutine(model):
+                    model = await model
+            finally:
+                _mux_ctx_model_id = None
+            cache[model_id] = model
+            while len(cache) > max_num_models_per_replica:
+                evicted_id, evicted = cache.popitem(last=False)
+                dtor = getattr(evicted, "__del__", None)
+                del evicted
+            return model
+
+        wrapper._serve_multiplexed = True
+        return wrapper
+
+    if _func is not None:
+        return dec(_func)
+    return dec
+
+
+# --------------------------------------------------------------------------
 # serve.batch
 # --------------------------------------------------------------------------
 

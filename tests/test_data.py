@@ -171,3 +171,16 @@ def test_distributed_groupby(ray_start_regular):
     for i in range(100):
         expect[i % 5] = expect.get(i % 5, 0) + i
     assert got == expect
+
+
+def test_hash_join(ray_start_regular):
+    left = rd.from_items(
+        [{"k": i, "a": i * 10} for i in range(20)], parallelism=4
+    )
+    right = rd.from_items(
+        [{"k": i, "b": i * 100} for i in range(10, 30)], parallelism=4
+    )
+    joined = left.join(right, on="k").take_all()
+    assert len(joined) == 10  # keys 10..19
+    for r in joined:
+        assert r["a"] == r["k"] * 10 and r["b"] == r["k"] * 100

@@ -232,3 +232,31 @@ def test_streaming_handle(serve_session):
     h = serve.run(Streamer.bind(), name="streamy", http=False)
     gen = h.options(method_name="stream_nums", stream=True).remote(4)
     assert list(gen) == [0, 3, 6, 9]
+
+
+def test_multiplexed_models(serve_session):
+    @serve.deployment
+    class MuxModel:
+        def __init__(self):
+            self.loads = []
+
+        @serve.multiplexed(max_num_models_per_replica=2)
+        async def get_model(self, model_id: str):
+            self.loads.append(model_id)
+            return {"id": model_id, "scale": len(model_id)}
+
+        async def __call__(self, model_id, x):
+            m = await self.get_model(model_id)
+            return x * m["scale"]
+
+        def load_count(self):
+            return len(self.loads)
+
+    h = serve.run(MuxModel.bind(), name="mux", http=False)
+    assert h.remote("ab", 10).result(timeout_s=30) == 20
+    assert h.remote("abc", 10).result(timeout_s=30) == 30
+    assert h.remote("ab", 5).result(timeout_s=30) == 10  # cached
+    assert h.load_count.remote().result(timeout_s=30) == 2
+    # third model evicts LRU ("abc" is most recent? "ab" most recent)
+    assert h.remote("abcd", 1).result(timeout_s=30) == 4
+    assert h.load_count.remote().result(timeout_s=30) == 3
