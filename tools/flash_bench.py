@@ -3,8 +3,11 @@
 Measures ray_amd.ops.flash_attention (HIP v5 kernel) vs torch SDPA on
 the Llama-3-8B prefill shape, plus max error vs an fp32 reference.
 """
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
