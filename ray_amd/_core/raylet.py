@@ -519,6 +519,11 @@ class Raylet:
 
     async def h_wait_object(self, conn, p):
         oid = bytes(p["id"])
+        if p.get("known_sealed") and not self.store.contains(oid):
+            # caller knows the object was sealed here once: absence from
+            # the table means it was freed/lost — fail fast so the owner
+            # can reconstruct from lineage instead of stalling
+            return {"ok": False}
         ok = await self.store.wait_sealed(oid, p.get("timeout", 60.0))
         if ok:
             self.store.ensure_local(oid)

@@ -243,8 +243,18 @@ class CoreRuntime:
         self._task_events: List[dict] = []
         self.worker_id = os.urandom(8)
 
+        # lineage: return oid -> shared record {spec, options, retries,
+        # live (undropped returns), pins} enabling re-execution of the
+        # producing task when a stored result is lost (reference:
+        # core_worker/task_manager.cc lineage reconstruction). Pins keep
+        # the serialized args + captured arg refs alive (lineage pinning)
+        # until every return is freed or the results turned out inline.
+        self._lineage: Dict[bytes, dict] = {}
+        self._reconstructing: Dict[bytes, asyncio.Future] = {}
+
         self.server.route("fetch_object", self._h_fetch_object)
         self.server.route("locate_object", self._h_locate_object)
+        self.server.route("reconstruct_object", self._h_reconstruct_object)
         self.server.route("ping", lambda c, p: "pong")
 
     # ------------- lifecycle -------------
@@ -434,6 +444,7 @@ class CoreRuntime:
         ent = self.memory_store.pop(oid, None)
         self._events.pop(oid, None)
         mapped = self._mmaps.pop(oid, None)
+        self._drop_lineage_for(oid)
         if ent is not None and ent[0] == "store" and not self._closed:
             # recycle same-node segments into the hot pool (plasma-arena
             # equivalent) unless we still hold a mapping of it
@@ -547,9 +558,10 @@ class CoreRuntime:
                 ent = await self._store_wait(ref.id, timeout)
             else:
                 ent = await self._fetch_from_owner(ref, timeout)
-        return await self._materialize(ref.id, ent, timeout)
+        return await self._materialize(ref.id, ent, timeout, ref.owner_addr)
 
-    async def _materialize(self, oid: bytes, ent: tuple, timeout=None):
+    async def _materialize(self, oid: bytes, ent: tuple, timeout=None,
+                           owner_addr=None):
         kind = ent[0]
         if kind == "val":
             return ent[1]
@@ -560,28 +572,186 @@ class CoreRuntime:
         if kind == "err":
             raise serialization.loads(ent[1])
         if kind == "store":
-            node_addr, size = ent[1], ent[2]
-            path = store.shm_path(self.shm_dir, oid)
-            if not os.path.exists(path):
-                if node_addr == self.raylet_addr:
-                    r = await self.raylet.call(
-                        "wait_object", {"id": oid, "timeout": timeout or 60.0}
-                    )
-                    if not r.get("ok"):
-                        raise exc.ObjectLostError(oid.hex())
-                else:
-                    r = await self.raylet.call(
-                        "pull_object",
-                        {"id": oid, "src": node_addr, "timeout": timeout or 120.0},
-                    )
-                    if not r.get("ok"):
-                        raise exc.ObjectLostError(oid.hex())
-            mo = store.MappedObject(path)
-            self._mmaps[oid] = mo
-            value = serialization.loads_from(mo.view)
-            self.memory_store[oid] = ("val", value)
-            return value
+            while True:
+                try:
+                    return await self._materialize_store(oid, ent, timeout)
+                except exc.ObjectLostError:
+                    # re-execute the producing task (lineage) or ask the
+                    # owner to; _recover_entry raises ObjectLostError when
+                    # no lineage / retries exhausted, bounding this loop.
+                    ent = await self._recover_entry(oid, owner_addr)
+                    if ent[0] != "store":
+                        return await self._materialize(
+                            oid, ent, timeout, owner_addr
+                        )
         raise exc.RaySystemError(f"bad store entry {kind}")
+
+    async def _materialize_store(self, oid: bytes, ent: tuple, timeout=None):
+        node_addr, size = ent[1], ent[2]
+        path = store.shm_path(self.shm_dir, oid)
+        if not os.path.exists(path):
+            if node_addr == self.raylet_addr:
+                r = await self.raylet.call(
+                    "wait_object",
+                    {"id": oid, "timeout": timeout or 60.0,
+                     "known_sealed": True},
+                )
+                if not r.get("ok"):
+                    raise exc.ObjectLostError(oid.hex())
+            else:
+                r = await self.raylet.call(
+                    "pull_object",
+                    {"id": oid, "src": node_addr, "timeout": timeout or 120.0},
+                )
+                if not r.get("ok"):
+                    raise exc.ObjectLostError(oid.hex())
+        try:
+            mo = store.MappedObject(path)
+        except OSError:
+            raise exc.ObjectLostError(oid.hex())
+        self._mmaps[oid] = mo
+        value = serialization.loads_from(mo.view)
+        self.memory_store[oid] = ("val", value)
+        return value
+
+    # ------------- lineage reconstruction -------------
+
+    async def _recover_entry(self, oid: bytes, owner_addr=None) -> tuple:
+        """The stored copy of `oid` is gone: recover a fresh memory-store
+        entry, either by re-executing the producing task locally (we are
+        the owner) or by asking the owner to (we borrowed the ref)."""
+        if owner_addr is None or owner_addr == self.addr:
+            await self._reconstruct(oid)
+            ent = self.memory_store.get(oid)
+            if ent is None:
+                raise exc.ObjectLostError(oid.hex())
+            return ent
+        try:
+            c = await self._conn(owner_addr)
+            r = await c.call("reconstruct_object", {"id": oid})
+        except (ConnectionLost, ConnectionError, RpcError):
+            raise exc.ObjectLostError(
+                f"{oid.hex()} (owner {owner_addr} unreachable)"
+            )
+        if not r or not r.get("ok"):
+            raise exc.ObjectLostError(oid.hex())
+        if r["kind"] == "val_ser":
+            ent = ("val_ser", r["data"])
+        elif r["kind"] == "err":
+            ent = ("err", r["data"])
+        else:
+            ent = ("store", r["node_addr"], r["size"])
+        self._store_put(oid, ent)
+        return ent
+
+    async def _reconstruct(self, oid: bytes):
+        lin = self._lineage.get(oid)
+        if lin is None or lin["retries"] <= 0:
+            raise exc.ObjectLostError(
+                f"{oid.hex()} (no lineage / reconstruction retries left)"
+            )
+        tid = lin["spec"]["task_id"]
+        fut = self._reconstructing.get(tid)
+        if fut is None:
+            fut = self.loop.create_future()
+            self._reconstructing[tid] = fut
+            try:
+                await self._reexecute(lin)
+                fut.set_result(True)
+            except Exception as e:
+                fut.set_exception(e)
+                fut.exception()  # mark retrieved; we re-raise our copy
+                raise
+            finally:
+                self._reconstructing.pop(tid, None)
+        else:
+            await asyncio.shield(fut)
+
+    async def _reexecute(self, lin: dict):
+        spec = lin["spec"]
+        last = None
+        while lin["retries"] > 0:
+            lin["retries"] -= 1
+            try:
+                reply = await self._dispatch_normal_task(spec, lin["options"])
+            except (ConnectionLost, ConnectionError) as e:
+                last = e
+                continue
+            if reply.get("status") == "error":
+                last = serialization.loads(reply["error"])
+                continue
+            for oid2, r in zip(spec["returns"], reply["results"]):
+                self._mmaps.pop(oid2, None)
+                if r["kind"] == "inline":
+                    ent = ("val_ser", r["data"])
+                else:
+                    ent = ("store", r["node_addr"], r["size"])
+                self._store_put(oid2, ent)
+            return
+        raise exc.ObjectLostError(
+            f"reconstruction of {spec.get('name')} failed: {last!r}"
+        )
+
+    def _record_lineage(self, spec, options, captured, returns):
+        retries = options.get("max_retries", 3)
+        if retries <= 0 or spec.get("streaming"):
+            return
+        for r in captured:
+            self._add_submitted_ref(r.id)
+        a = spec.get("args_store")
+        if a is not None:
+            self._add_local_ref(a[0], self.addr)
+        lin = {
+            "spec": spec,
+            "options": options,
+            "retries": retries,
+            "live": set(returns),
+            "captured": [r.id for r in captured],
+            "args_pin": a[0] if a is not None else None,
+            "pinned": True,
+        }
+        for oid in returns:
+            self._lineage[oid] = lin
+
+    def _drop_lineage_pins(self, lin: dict):
+        if not lin.get("pinned"):
+            return
+        lin["pinned"] = False
+        for oid in lin["captured"]:
+            self._remove_submitted_ref(oid)
+        if lin["args_pin"] is not None:
+            self._remove_local_ref(lin["args_pin"])
+
+    def _drop_lineage_for(self, oid: bytes):
+        lin = self._lineage.pop(oid, None)
+        if lin is None:
+            return
+        lin["live"].discard(oid)
+        if not lin["live"]:
+            self._drop_lineage_pins(lin)
+
+    async def _h_reconstruct_object(self, conn, p):
+        oid = p["id"]
+        ent = self.memory_store.get(oid)
+        if ent is not None and ent[0] == "val":
+            # we still hold the value in memory: re-serialize instead of
+            # re-executing
+            return {"ok": True, "kind": "val_ser",
+                    "data": serialization.dumps(ent[1])}
+        try:
+            await self._reconstruct(oid)
+        except Exception:
+            return {"ok": False}
+        ent = self.memory_store.get(oid)
+        if ent is None:
+            return {"ok": False}
+        if ent[0] == "val":
+            return {"ok": True, "kind": "val_ser",
+                    "data": serialization.dumps(ent[1])}
+        if ent[0] in ("val_ser", "err"):
+            return {"ok": True, "kind": ent[0], "data": ent[1]}
+        return {"ok": True, "kind": "store", "node_addr": ent[1],
+                "size": ent[2]}
 
     async def _fetch_from_owner(self, ref: ObjectRef, timeout=None) -> tuple:
         try:
@@ -762,6 +932,8 @@ class CoreRuntime:
             with self._refs_lock:
                 self._refs[aid] = [1, 0, self.addr]  # freed after task completes
         retries = 0 if streaming else options.get("max_retries", 3)
+        if not streaming:
+            self._record_lineage(spec, options, captured, returns)
         self._run(
             self._submit_with_retries(spec, options, retries, captured_ids)
         )
@@ -813,6 +985,7 @@ class CoreRuntime:
                         self._stream_finish(spec["task_id"], err)
                         return
                     for oid in spec["returns"]:
+                        self._drop_lineage_for(oid)
                         self._store_put(oid, ("err", err))
                     return
         except Exception:
@@ -825,6 +998,7 @@ class CoreRuntime:
                 self._stream_finish(spec["task_id"], err)
             else:
                 for oid in spec["returns"]:
+                    self._drop_lineage_for(oid)
                     self._store_put(oid, ("err", err))
         finally:
             for oid, _owner in captured_ids:
@@ -950,6 +1124,7 @@ class CoreRuntime:
             return
         if status == "error":
             for oid in spec["returns"]:
+                self._drop_lineage_for(oid)
                 self._store_put(oid, ("err", reply["error"]))
             return
         results = reply["results"]
@@ -958,6 +1133,11 @@ class CoreRuntime:
                 self._store_put(oid, ("val_ser", r["data"]))
             else:
                 self._store_put(oid, ("store", r["node_addr"], r["size"]))
+        if all(r["kind"] == "inline" for r in results):
+            # inline results live in the owner's memory and cannot be
+            # lost; release the lineage pins now
+            for oid in spec["returns"]:
+                self._drop_lineage_for(oid)
 
     # ------------- actor submission -------------
 

@@ -157,3 +157,101 @@ def test_gcs_restart_with_persistence(ray_start_cluster):
         return 7
 
     assert ray.get(f.remote(), timeout=60) == 7
+
+
+def test_lineage_reconstruction_local_loss(ray_start_regular, tmp_path):
+    """A stored (non-inline) result whose shm copy is lost is recovered
+    by re-executing the producing task from lineage (reference:
+    core_worker/task_manager.cc lineage reconstruction)."""
+    from ray_amd._core import runtime as rtmod
+    from ray_amd._core import store as storemod
+
+    marker = str(tmp_path / "exec_count")
+
+    @ray.remote
+    def make(tag, marker=marker):
+        with open(marker, "a") as f:
+            f.write("x")
+        return np.full(300_000, tag, dtype=np.uint8)
+
+    ref = make.remote(7)
+    ready, _ = ray.wait([ref], timeout=30, fetch_local=False)
+    assert ready
+
+    rt = rtmod.global_runtime()
+    ent = rt.memory_store[ref.id]
+    assert ent[0] == "store"  # big enough to live in the shm store
+
+    # simulate loss of the only copy: free it from the raylet table and
+    # remove the shm file
+    rt._call_sync(rt.raylet.call("free_objects", {"ids": [ref.id]}))
+    path = storemod.shm_path(rt.shm_dir, ref.id)
+    if os.path.exists(path):
+        os.remove(path)
+
+    val = ray.get(ref, timeout=60)
+    assert val.shape == (300_000,) and val[0] == 7
+    with open(marker) as f:
+        assert len(f.read()) == 2  # original execution + reconstruction
+
+
+def test_lineage_reconstruction_node_death(ray_start_cluster, tmp_path):
+    """Result stored on a node that dies is reconstructed on another
+    node that can satisfy the task's resources."""
+    cluster = ray_start_cluster
+    n2 = cluster.add_node(num_cpus=2, resources={"b": 1})
+    cluster.connect()
+    cluster.wait_for_nodes()
+
+    marker = str(tmp_path / "exec_count")
+
+    @ray.remote(resources={"b": 1})
+    def make(marker=marker):
+        with open(marker, "a") as f:
+            f.write("x")
+        return np.arange(200_000, dtype=np.int32)
+
+    ref = make.remote()
+    ready, _ = ray.wait([ref], timeout=60, fetch_local=False)
+    assert ready
+
+    cluster.remove_node(n2)
+    # this localhost harness shares one shm dir across "nodes", so node
+    # death alone does not lose the bytes — delete the file as a real
+    # remote-node death would
+    from ray_amd._core import runtime as rtmod
+    from ray_amd._core import store as storemod
+
+    rt = rtmod.global_runtime()
+    path = storemod.shm_path(rt.shm_dir, ref.id)
+    if os.path.exists(path):
+        os.remove(path)
+    cluster.add_node(num_cpus=2, resources={"b": 1})
+    cluster.wait_for_nodes()
+    time.sleep(1.0)
+
+    val = ray.get(ref, timeout=90)
+    assert val[123456] == 123456
+    with open(marker) as f:
+        assert len(f.read()) == 2
+
+
+def test_lineage_retries_exhausted(ray_start_regular):
+    """With max_retries=0 a lost object raises ObjectLostError."""
+    from ray_amd._core import runtime as rtmod
+    from ray_amd._core import store as storemod
+
+    @ray.remote(max_retries=0)
+    def make():
+        return np.zeros(300_000, dtype=np.uint8)
+
+    ref = make.remote()
+    ready, _ = ray.wait([ref], timeout=30, fetch_local=False)
+    assert ready
+    rt = rtmod.global_runtime()
+    rt._call_sync(rt.raylet.call("free_objects", {"ids": [ref.id]}))
+    path = storemod.shm_path(rt.shm_dir, ref.id)
+    if os.path.exists(path):
+        os.remove(path)
+    with pytest.raises(ray.exceptions.ObjectLostError):
+        ray.get(ref, timeout=30)
