@@ -8,6 +8,7 @@ from .impl import (  # noqa: F401
     ASHAScheduler,
     Callback,
     FIFOScheduler,
+    PopulationBasedTraining,
     ResultGrid,
     TuneConfig,
     Tuner,
@@ -24,4 +25,5 @@ from .impl import (  # noqa: F401
     with_parameters,
     with_resources,
 )
+from ..train.checkpoint import Checkpoint  # noqa: F401
 from ..train.session import get_checkpoint, get_context  # noqa: F401

@@ -200,6 +200,77 @@ class ASHAScheduler:
 # ---------------- config ----------------
 
 
+class PopulationBasedTraining:
+    """PBT (reference: tune/schedulers/pbt.py PopulationBasedTraining).
+
+    Every `perturbation_interval` iterations a bottom-quantile trial is
+    restarted from a top-quantile trial's latest checkpoint with a
+    perturbed clone of its config (exploit + explore)."""
+
+    def __init__(self, *, time_attr="training_iteration", metric=None,
+                 mode="max", perturbation_interval=4,
+                 hyperparam_mutations=None, quantile_fraction=0.25,
+                 resample_probability=0.25, seed=0):
+        self.metric = metric
+        self.mode = mode
+        self.perturbation_interval = perturbation_interval
+        self.hyperparam_mutations = hyperparam_mutations or {}
+        self.quantile_fraction = quantile_fraction
+        self.resample_probability = resample_probability
+        self._rng = random.Random(seed)
+        self._scores: Dict[str, tuple] = {}       # trial -> (it, score)
+        self._last_perturb: Dict[str, int] = {}
+        self.num_perturbations = 0
+
+    def on_trial_result(self, trial_id, iteration, metric_value) -> str:
+        if metric_value is not None:
+            self._scores[trial_id] = (iteration, metric_value)
+        return "CONTINUE"
+
+    def should_perturb(self, trial_id, iteration) -> bool:
+        return (
+            iteration - self._last_perturb.get(trial_id, 0)
+            >= self.perturbation_interval
+        )
+
+    def exploit_target(self, trial_id) -> Optional[str]:
+        """Donor trial if `trial_id` is in the bottom quantile, else None."""
+        if len(self._scores) < 2:
+            return None
+        items = sorted(
+            self._scores.items(), key=lambda kv: kv[1][1],
+            reverse=(self.mode == "max"),
+        )
+        q = max(1, int(len(items) * self.quantile_fraction))
+        top = [k for k, _ in items[:q]]
+        bottom = {k for k, _ in items[-q:]}
+        if trial_id not in bottom or trial_id in top:
+            return None
+        return self._rng.choice(top)
+
+    def explore(self, config: dict) -> dict:
+        new = dict(config)
+        for k, dom in self.hyperparam_mutations.items():
+            if self._rng.random() < self.resample_probability:
+                if isinstance(dom, Domain):
+                    new[k] = dom.sample(self._rng)
+                elif isinstance(dom, list):
+                    new[k] = self._rng.choice(dom)
+                elif callable(dom):
+                    new[k] = dom()
+                continue
+            v = new.get(k)
+            if isinstance(dom, list) and v in dom:
+                i = dom.index(v) + self._rng.choice([-1, 1])
+                new[k] = dom[max(0, min(len(dom) - 1, i))]
+            elif isinstance(v, bool):
+                new[k] = not v if self._rng.random() < 0.5 else v
+            elif isinstance(v, (int, float)):
+                f = 1.2 if self._rng.random() < 0.5 else 0.8
+                new[k] = type(v)(v * f) if isinstance(v, int) else v * f
+        return new
+
+
 @dataclass
 class TuneConfig:
     metric: Optional[str] = None
@@ -221,12 +292,14 @@ class _TrialActor:
         self.error: Optional[str] = None
 
     def run(self, fn_bytes: bytes, config: dict, storage_dir: str,
-            trial_name: str):
+            trial_name: str, restore_path: Optional[str] = None):
         import cloudpickle
 
         fn = cloudpickle.loads(fn_bytes)
         os.makedirs(storage_dir, exist_ok=True)
-        self.session = TrainSession(0, 1, 0, 1, storage_dir, trial_name)
+        ckpt = Checkpoint(restore_path) if restore_path else None
+        self.session = TrainSession(0, 1, 0, 1, storage_dir, trial_name,
+                                    latest_checkpoint=ckpt)
         self.done = False
 
         def _run():
@@ -397,6 +470,21 @@ class Tuner:
                     if decision == "STOP" and not t["stopped"]:
                         t["stopped"] = True
                         t["actor"].request_stop.remote()
+                if (
+                    isinstance(scheduler, PopulationBasedTraining)
+                    and not t["stopped"]
+                    and t["rows"]
+                    and scheduler.should_perturb(t["name"], t["it"])
+                ):
+                    scheduler._last_perturb[t["name"]] = t["it"]
+                    donor_name = scheduler.exploit_target(t["name"])
+                    donor = next(
+                        (x for x in running if x["name"] == donor_name), None
+                    )
+                    if donor is not None and donor is not t:
+                        self._pbt_exploit(
+                            ray, scheduler, t, donor, fn_bytes, storage
+                        )
                 if st["done"] or (t["stopped"] and not st["results"]):
                     t["error"] = t["error"] or st["error"]
                     t["done"] = True
@@ -422,6 +510,34 @@ class Tuner:
                 )
             )
         return ResultGrid(results, tc.metric, tc.mode)
+
+    def _pbt_exploit(self, ray, scheduler, t, donor, fn_bytes, storage):
+        """Restart trial `t` from `donor`'s latest checkpoint with a
+        perturbed clone of the donor's config."""
+        try:
+            ckpt = ray.get(
+                donor["actor"].latest_checkpoint_path.remote(), timeout=15
+            )
+        except ray.exceptions.RayError:
+            ckpt = donor.get("ckpt")
+        if ckpt is None:
+            ckpt = donor.get("ckpt")
+        new_cfg = scheduler.explore(donor["config"])
+        try:
+            ray.kill(t["actor"])
+        except Exception:
+            pass
+        a = ray.remote(_TrialActor).options(num_cpus=1).remote()
+        a.run.remote(
+            fn_bytes, new_cfg, os.path.join(storage, t["name"]), t["name"],
+            restore_path=ckpt,
+        )
+        t["actor"] = a
+        t["config"] = new_cfg
+        t["fetch_ref"] = None
+        if ckpt:
+            t["ckpt"] = ckpt
+        scheduler.num_perturbations += 1
 
     def _fit_trainer_trials(self, storage: str) -> ResultGrid:
         tc = self.tune_config

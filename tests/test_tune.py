@@ -124,3 +124,45 @@ def test_tuner_with_torch_trainer(ray_start_regular, tmp_path):
     grid = tuner.fit()
     assert len(grid) == 2
     assert grid.get_best_result().metrics["lr_used"] == 0.2
+
+
+def test_pbt_exploit_and_explore(ray_start_regular, tmp_path):
+    """Bottom-quantile trials adopt a top trial's checkpoint + perturbed
+    config (reference: tune/schedulers/pbt.py)."""
+    from ray_amd import tune
+
+    def trainable(config):
+        import json
+        import os as _os
+        import time as _tm
+
+        ckpt = tune.get_checkpoint()
+        score = 0.0
+        if ckpt is not None:
+            with open(_os.path.join(ckpt.path, "state.json")) as f:
+                score = json.load(f)["score"]
+        for _ in range(30):
+            score += config["lr"]  # higher lr -> faster "learning"
+            d = str(tmp_path / f"ck_{_os.getpid()}_{_tm.time_ns()}")
+            _os.makedirs(d, exist_ok=True)
+            with open(_os.path.join(d, "state.json"), "w") as f:
+                json.dump({"score": score}, f)
+            tune.report({"score": score},
+                        checkpoint=tune.Checkpoint(d))
+            _tm.sleep(0.05)
+
+    sched = tune.PopulationBasedTraining(
+        metric="score", mode="max", perturbation_interval=5,
+        hyperparam_mutations={"lr": tune.uniform(0.1, 1.0)}, seed=1,
+    )
+    tuner = tune.Tuner(
+        trainable,
+        param_space={"lr": tune.grid_search([0.01, 0.02, 0.9, 1.0])},
+        tune_config=tune.TuneConfig(metric="score", mode="max",
+                                    scheduler=sched,
+                                    max_concurrent_trials=4),
+    )
+    rg = tuner.fit()
+    assert sched.num_perturbations >= 1
+    best = rg.get_best_result()
+    assert best.metrics["score"] > 5.0
