@@ -236,6 +236,48 @@ class ReplicaActor:
         finally:
             self._ongoing -= 1
 
+    def is_http_streaming(self) -> bool:
+        """True when the deployment's __call__ is a (sync or async)
+        generator function — the proxy then uses handle_http_stream and
+        chunked transfer (reference: _private/proxy.py streaming)."""
+        import inspect
+
+        if self._asgi_client is not None:
+            return False
+        target = getattr(self._callable, "__call__", self._callable)
+        return inspect.isgeneratorfunction(
+            target
+        ) or inspect.isasyncgenfunction(target)
+
+    async def handle_http_stream(self, method: str, path: str, query: str,
+                                 headers: dict, body: bytes):
+        """Async generator: first item is (status, headers), the rest are
+        body chunks — driven by the streaming-generator task path."""
+        import inspect
+
+        self._ongoing += 1
+        self._total += 1
+        try:
+            req = SimpleRequest(method, path, query, headers, body)
+            r = self._callable(req)
+            yield (200, {"content-type": "text/plain; charset=utf-8"})
+            if inspect.isasyncgen(r):
+                async for chunk in r:
+                    yield _chunk_bytes(chunk)
+            else:
+                it = iter(r)
+                loop = asyncio.get_running_loop()
+                sentinel = object()
+                while True:
+                    chunk = await loop.run_in_executor(
+                        None, next, it, sentinel
+                    )
+                    if chunk is sentinel:
+                        break
+                    yield _chunk_bytes(chunk)
+        finally:
+            self._ongoing -= 1
+
     def get_stats(self):
         return {"ongoing": self._ongoing, "total": self._total}
 
@@ -264,6 +306,16 @@ class SimpleRequest:
         import json
 
         return json.loads(self._body or b"null")
+
+
+def _chunk_bytes(chunk) -> bytes:
+    import json
+
+    if isinstance(chunk, (bytes, bytearray)):
+        return bytes(chunk)
+    if isinstance(chunk, str):
+        return chunk.encode()
+    return (json.dumps(chunk, default=str) + "\n").encode()
 
 
 def _encode_http_result(r):
@@ -558,6 +610,7 @@ class ProxyActor:
     def __init__(self, port: int):
         self.port = port
         self._handles: Dict[str, DeploymentHandle] = {}
+        self._streaming: Dict[str, bool] = {}
         self._server_task = None
 
     async def start_server(self):
@@ -601,6 +654,17 @@ class ProxyActor:
                 idx = h._pick()
                 replica = h._replicas[idx]
                 try:
+                    st_flag = self._streaming.get(app_name)
+                    if st_flag is None:
+                        fref = replica.is_http_streaming.remote()
+                        st_flag = (await rt.get_async([fref], 30))[0]
+                        self._streaming[app_name] = st_flag
+                    if st_flag:
+                        await self._proxy_stream(
+                            send, replica, scope["method"], sub_path, qs,
+                            headers, body, rt, loop,
+                        )
+                        return
                     ref = replica.handle_http.remote(
                         scope["method"], sub_path, qs, headers, body
                     )
@@ -628,6 +692,47 @@ class ProxyActor:
                 return True
             await asyncio.sleep(0.05)
         return False
+
+    async def _proxy_stream(self, send, replica, method, path, qs, headers,
+                            body, rt, loop):
+        """Forward a streaming replica response as chunked HTTP body
+        parts — each yielded item flushes to the client immediately."""
+        gen = replica.handle_http_stream.options(
+            num_returns="streaming"
+        ).remote(method, path, qs, headers, body)
+        started = False
+        try:
+            while True:
+                ref = await loop.run_in_executor(None, next, gen, None)
+                if ref is None:
+                    break
+                item = (await rt.get_async([ref], 120))[0]
+                if not started:
+                    status, hdrs = item
+                    await send({
+                        "type": "http.response.start",
+                        "status": status,
+                        "headers": [
+                            (k.encode(), v.encode())
+                            for k, v in (hdrs or {}).items()
+                            if k.lower() not in
+                            ("content-length", "transfer-encoding")
+                        ],
+                    })
+                    started = True
+                else:
+                    await send({
+                        "type": "http.response.body", "body": item,
+                        "more_body": True,
+                    })
+        except Exception as e:
+            if not started:
+                await _send_response(
+                    send, 500, {}, f'{{"error":"{e}"}}'.encode()
+                )
+                return
+        await send({"type": "http.response.body", "body": b"",
+                    "more_body": False})
 
     def _resolve_route(self, path):
         ray = _ray()

@@ -260,3 +260,34 @@ def test_multiplexed_models(serve_session):
     # third model evicts LRU ("abc" is most recent? "ab" most recent)
     assert h.remote("abcd", 1).result(timeout_s=30) == 4
     assert h.load_count.remote().result(timeout_s=30) == 3
+
+
+def test_http_chunked_streaming(serve_session):
+    """Generator deployments stream chunked HTTP bodies through the
+    proxy; the first chunk arrives before the generator finishes."""
+    import httpx
+
+    @serve.deployment
+    class Streamer:
+        def __call__(self, request):
+            import time as _tm
+
+            for i in range(5):
+                yield f"chunk{i};"
+                _tm.sleep(0.15)
+
+    port = 18433
+    serve.run(Streamer.bind(), name="stream_app", route_prefix="/",
+              port=port)
+    t0 = time.time()
+    first_at = None
+    parts = []
+    with httpx.stream("GET", f"http://127.0.0.1:{port}/", timeout=30) as r:
+        assert r.status_code == 200
+        for chunk in r.iter_raw():
+            if first_at is None and chunk:
+                first_at = time.time() - t0
+            parts.append(chunk)
+    bodytext = b"".join(parts).decode()
+    assert bodytext == "chunk0;chunk1;chunk2;chunk3;chunk4;"
+    assert first_at is not None and first_at < 0.45  # streamed, not buffered
