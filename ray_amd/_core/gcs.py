@@ -84,11 +84,15 @@ class GcsServer:
         self.pgs: Dict[bytes, PlacementGroupInfo] = {}
         self.job_counter = 0
         self._proto_node: Dict[int, bytes] = {}  # id(proto) -> node_id
+        # pub/sub topic bus (reference: gcs_publisher/gcs_subscriber —
+        # log/error/node/actor channels): channel -> subscribed conns
+        self._subs: Dict[str, set] = {}
         for m in (
             "kv_put kv_get kv_del kv_keys kv_exists register_node node_table "
             "report_resources register_actor resolve_actor actor_exit "
             "kill_actor list_actors next_job_id create_pg pg_wait_ready "
-            "remove_pg pg_table ping timeline_events drain_node"
+            "remove_pg pg_table ping timeline_events drain_node "
+            "subscribe unsubscribe publish"
         ).split():
             self.server.route(m, getattr(self, "h_" + m))
         self.server.on_conn_lost = self._conn_lost
@@ -225,9 +229,37 @@ class GcsServer:
         return self._timeline
 
     def _conn_lost(self, proto, exc):
+        for subs in self._subs.values():
+            subs.discard(proto)
         node_id = self._proto_node.pop(id(proto), None)
         if node_id is not None and node_id in self.nodes:
             asyncio.ensure_future(self._on_node_death(node_id))
+
+    # ---------- pub/sub ----------
+
+    def h_subscribe(self, conn, p):
+        self._subs.setdefault(p["channel"], set()).add(conn)
+        return True
+
+    def h_unsubscribe(self, conn, p):
+        subs = self._subs.get(p["channel"])
+        if subs is not None:
+            subs.discard(conn)
+        return True
+
+    def h_publish(self, conn, p):
+        from .protocol import MSG_NOTIFY
+
+        ch = p["channel"]
+        n = 0
+        for proto in list(self._subs.get(ch, ())):
+            if proto.transport is None or proto.transport.is_closing():
+                self._subs[ch].discard(proto)
+                continue
+            proto.send([MSG_NOTIFY, 0, "pubsub",
+                        {"channel": ch, "data": p["data"]}])
+            n += 1
+        return n
 
     async def h_drain_node(self, conn, p):
         await self._on_node_death(p["node_id"])

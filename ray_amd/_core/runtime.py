@@ -252,6 +252,9 @@ class CoreRuntime:
         self._lineage: Dict[bytes, dict] = {}
         self._reconstructing: Dict[bytes, asyncio.Future] = {}
 
+        # pub/sub: channel -> local callbacks (invoked on the loop thread)
+        self._pubsub_cbs: Dict[str, list] = {}
+
         self.server.route("fetch_object", self._h_fetch_object)
         self.server.route("locate_object", self._h_locate_object)
         self.server.route("reconstruct_object", self._h_reconstruct_object)
@@ -297,7 +300,10 @@ class CoreRuntime:
                 c = RpcClient()
                 await c.connect(self.gcs_addr, retries=5)
                 if await asyncio.wait_for(c.call("ping", {}), 5) == "pong":
+                    c.on_notify = self._on_conn_notify
                     self.gcs = c
+                    for ch in self._pubsub_cbs:
+                        await c.call("subscribe", {"channel": ch})
                     return
             except Exception as e:
                 last = e
@@ -311,6 +317,7 @@ class CoreRuntime:
         await self.server.start_unix(sock)
         self.addr = "unix:" + sock
         await self.gcs.connect(self.gcs_addr)
+        self.gcs.on_notify = self._on_conn_notify
         await self.raylet.connect(self.raylet_addr)
 
     def _run(self, coro):
@@ -370,6 +377,42 @@ class CoreRuntime:
     def _on_conn_notify(self, method, payload):
         if method == "stream_item":
             self._h_stream_item(payload)
+        elif method == "pubsub":
+            for cb in self._pubsub_cbs.get(payload["channel"], []):
+                try:
+                    cb(payload["data"])
+                except Exception:
+                    traceback.print_exc()
+
+    # ------------- pub/sub (GCS topic bus) -------------
+
+    def pubsub_publish(self, channel: str, data) -> int:
+        """Publish to a GCS channel; returns the subscriber count
+        reached."""
+        return self._call_sync(
+            self._gcs_rpc("publish", {"channel": channel, "data": data})
+        )
+
+    def pubsub_subscribe(self, channel: str, cb):
+        first = channel not in self._pubsub_cbs
+        self._pubsub_cbs.setdefault(channel, []).append(cb)
+        if first:
+            self._call_sync(self._gcs_rpc("subscribe", {"channel": channel}))
+
+    def pubsub_unsubscribe(self, channel: str, cb):
+        cbs = self._pubsub_cbs.get(channel)
+        if cbs is None:
+            return
+        if cb in cbs:
+            cbs.remove(cb)
+        if not cbs:
+            del self._pubsub_cbs[channel]
+            try:
+                self._call_sync(
+                    self._gcs_rpc("unsubscribe", {"channel": channel})
+                )
+            except Exception:
+                pass
 
     def _h_stream_item(self, p):
         """Streaming-generator item pushed by the executing worker

@@ -498,3 +498,33 @@ def test_no_retry_exceptions_by_default(ray_start_regular):
 
     with pytest.raises(ray.exceptions.RayTaskError):
         ray.get(boom.remote(), timeout=30)
+
+
+def test_pubsub_bus(ray_start_regular):
+    """GCS pub/sub: cross-process fan-out, no replay for late subs."""
+    import queue
+
+    from ray_amd.util import pubsub
+
+    with pubsub.Subscriber("events") as sub:
+        # another process (a task) publishes
+        @ray.remote
+        def pub(msg):
+            from ray_amd.util import pubsub as ps
+
+            return ps.publish("events", msg)
+
+        reached = ray.get(pub.remote({"k": 1}), timeout=30)
+        assert reached >= 1
+        assert sub.poll(timeout=10) == {"k": 1}
+
+        # driver-side publish also delivers
+        pubsub.publish("events", "plain-string")
+        assert sub.poll(timeout=10) == "plain-string"
+
+    # closed subscriber no longer receives
+    assert pubsub.publish("events", "late") == 0
+
+    with pubsub.Subscriber("events") as sub2:
+        with pytest.raises(queue.Empty):
+            sub2.poll(timeout=0.2)  # no replay of earlier messages
