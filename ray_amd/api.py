@@ -97,6 +97,18 @@ def init(
             os.environ.update(
                 {str(k): str(v) for k, v in runtime_env["env_vars"].items()}
             )
+        if address and (
+            address.startswith("ray_amd://") or address.startswith("ray://")
+        ):
+            # Ray-Client mode: proxy every op to a ClientServer on the
+            # cluster (reference: ray.init("ray://...") util/client/)
+            from .client import ClientRuntime
+
+            crt = ClientRuntime(address.split("://", 1)[1],
+                                namespace=_namespace)
+            crt.connect()
+            _rt.set_global_runtime(crt)
+            return RayContext(crt.session_dir, crt.node_id)
         if address in (None, "local"):
             cluster = _node.start_local_cluster(
                 num_cpus=num_cpus,
