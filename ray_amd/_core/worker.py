@@ -342,7 +342,16 @@ class WorkerMain:
         try:
             args, kwargs = await self._load_args(spec)
             args, kwargs = await self._resolve_args(args, kwargs)
-            method = getattr(self.actor_instance, method_name)
+            if method_name == "__ray_apply__":
+                # generic in-actor execution (reference: __ray_call__):
+                # first arg is fn(instance, *rest) — used by compiled
+                # DAG loops and debugging helpers
+                fn, args = args[0], args[1:]
+                import functools
+
+                method = functools.partial(fn, self.actor_instance)
+            else:
+                method = getattr(self.actor_instance, method_name)
         except Exception:
             return self._error_reply(spec, traceback.format_exc())
 

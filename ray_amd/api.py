@@ -334,7 +334,9 @@ class ActorHandle:
         self._class_name = class_name
 
     def __getattr__(self, item):
-        if item.startswith("_"):
+        if item.startswith("_") and item not in (
+            "__ray_apply__", "__ray_terminate__"
+        ):
             raise AttributeError(item)
         return ActorMethod(self, item)
 
@@ -445,6 +447,10 @@ def method(**kwargs):
 
 
 def get(refs: Union[ObjectRef, Sequence[ObjectRef]], *, timeout: Optional[float] = None):
+    from .dag import DAGFuture
+
+    if isinstance(refs, DAGFuture):  # compiled-DAG result handle
+        return refs.get(timeout if timeout is not None else 60.0)
     rt = _rt.global_runtime()
     if isinstance(refs, ObjectRef):
         return rt.get_sync([refs], timeout)[0]
