@@ -217,6 +217,8 @@ class GcsServer:
         if n is not None:
             n.resources_available = p["available"]
             n.pending = p.get("pending", 0)
+            if p.get("total") is not None:  # dynamic resource change
+                n.resources_total = p["total"]
 
     def h_ping(self, conn, p):
         return "pong"

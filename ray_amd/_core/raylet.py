@@ -93,7 +93,7 @@ class Raylet:
         self._node_cache_time = 0.0
 
         for m in (
-            "register_worker request_lease return_lease seal_object wait_object "
+            "register_worker request_lease return_lease seal_object wait_object set_resource "
             "free_objects pull_object fetch_chunk object_stats start_actor "
             "actor_ready actor_failed kill_worker reserve_bundle commit_bundle "
             "rollback_bundle remove_bundle node_info ping prestart_workers "
@@ -512,6 +512,30 @@ class Raylet:
         return self.h_rollback_bundle(conn, p)
 
     # ---------------- object store ----------------
+
+    def h_set_resource(self, conn, p):
+        """Dynamic custom resources (reference:
+        experimental/dynamic_resources.py ray.experimental.set_resource):
+        adjust this node's capacity for one resource at runtime."""
+        name = p["resource"]
+        cap = float(p["capacity"])
+        used = self.resources_total.get(name, 0.0) - self.avail.get(name, 0.0)
+        if cap <= 0:
+            self.resources_total.pop(name, None)
+            self.avail.pop(name, None)
+        else:
+            self.resources_total[name] = cap
+            self.avail[name] = max(0.0, cap - used)
+        try:
+            self.gcs.notify(
+                "report_resources",
+                {"node_id": self.node_id, "available": dict(self.avail),
+                 "pending": len(self._pending),
+                 "total": dict(self.resources_total)},
+            )
+        except Exception:
+            pass
+        return {"ok": True, "total": self.resources_total.get(name, 0.0)}
 
     def h_seal_object(self, conn, p):
         self.store.seal(bytes(p["id"]), p["size"])

@@ -528,3 +528,27 @@ def test_pubsub_bus(ray_start_regular):
     with pubsub.Subscriber("events") as sub2:
         with pytest.raises(queue.Empty):
             sub2.poll(timeout=0.2)  # no replay of earlier messages
+
+
+def test_object_locations_and_dynamic_resources(ray_start_regular):
+    from ray_amd import experimental as exp
+
+    # locations: a big object lives in this node's store
+    big = ray.put(np.zeros(500_000, dtype=np.uint8))
+    loc = exp.get_object_locations([big])[big]
+    assert loc["object_size"] and loc["object_size"] >= 500_000
+    assert len(loc["node_ids"]) == 1
+
+    # dynamic resource: create at runtime, schedule on it, delete
+    exp.set_resource("tokens", 2)
+    time.sleep(0.3)
+    assert ray.cluster_resources().get("tokens") == 2
+
+    @ray.remote(resources={"tokens": 1})
+    def use():
+        return "ok"
+
+    assert ray.get(use.remote(), timeout=30) == "ok"
+    exp.set_resource("tokens", 0)
+    time.sleep(0.3)
+    assert "tokens" not in ray.cluster_resources()
