@@ -21,8 +21,9 @@ from typing import Dict, List, Optional
 
 from . import ids, store
 from .protocol import RpcClient, RpcServer
+from .._config import config as _cfg
 
-CHUNK = 8 * 1024 * 1024  # remote pull chunk (reference uses 5 MiB, ray_config_def.h:421)
+CHUNK = _cfg.pull_chunk_bytes  # remote pull chunk (ray_config_def.h:421)
 
 
 class WorkerProc:
@@ -70,9 +71,7 @@ class Raylet:
 
         shm_dir = os.environ.get("RAY_AMD_SHM_DIR") or os.path.join(session_dir, "shm")
         spill_dir = os.path.join(session_dir, "spill")
-        cap = object_store_memory or int(
-            os.environ.get("RAY_AMD_OBJECT_STORE_MEMORY", 16 * 2**30)
-        )
+        cap = object_store_memory or _cfg.object_store_memory
         self.store = store.LocalObjectStore(shm_dir, spill_dir, cap)
 
         self.workers: Dict[int, WorkerProc] = {}  # pid -> worker
@@ -119,7 +118,8 @@ class Raylet:
                 "labels": self.labels,
             },
         )
-        n_prestart = int(min(self.resources_total.get("CPU", 0), 8))
+        n_prestart = int(min(self.resources_total.get("CPU", 0),
+                            _cfg.worker_prestart))
         for _ in range(n_prestart):
             self._spawn_worker()
         asyncio.ensure_future(self._resource_reporter())
@@ -146,7 +146,7 @@ class Raylet:
                     last_time = now
                 except Exception:
                     pass
-            await asyncio.sleep(0.2)
+            await asyncio.sleep(_cfg.resource_report_period_s)
 
     # ---------------- worker pool ----------------
 
@@ -326,7 +326,8 @@ class Raylet:
                 n_task_workers = sum(
                     1 for w in self.workers.values() if w.kind == "task"
                 )
-                cap = int(max(self.resources_total.get("CPU", 1) * 2, 8))
+                cap = int(max(self.resources_total.get("CPU", 1)
+                              * _cfg.worker_cap_factor, 8))
                 if (
                     self._starting < min(len(self._pending), 4)
                     and n_task_workers + self._starting < cap

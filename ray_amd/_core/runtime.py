@@ -25,6 +25,7 @@ from collections import deque
 from typing import Any, Dict, List, Optional, Tuple
 
 from .. import exceptions as exc
+from .._config import config as _cfg
 from . import ids, serialization, store
 from .protocol import ConnectionLost, RpcClient, RpcError, RpcServer
 
@@ -736,7 +737,7 @@ class CoreRuntime:
         )
 
     def _record_lineage(self, spec, options, captured, returns):
-        retries = options.get("max_retries", 3)
+        retries = options.get("max_retries", _cfg.default_max_retries)
         if retries <= 0 or spec.get("streaming"):
             return
         for r in captured:
@@ -974,7 +975,8 @@ class CoreRuntime:
             self._store_put_threadsafe(aid, ("store", self.raylet_addr, size))
             with self._refs_lock:
                 self._refs[aid] = [1, 0, self.addr]  # freed after task completes
-        retries = 0 if streaming else options.get("max_retries", 3)
+        retries = (0 if streaming else
+                   options.get("max_retries", _cfg.default_max_retries))
         if not streaming:
             self._record_lineage(spec, options, captured, returns)
         self._run(
@@ -1092,7 +1094,8 @@ class CoreRuntime:
         # cap outstanding lease requests (reference:
         # LeaseRequestRateLimiter) — unbounded requests pile up at the
         # raylet and starve later submitters
-        if pool.requests_in_flight < min(len(pool.queue), 16):
+        if pool.requests_in_flight < min(len(pool.queue),
+                                         _cfg.lease_request_cap):
             pool.requests_in_flight += 1
             asyncio.ensure_future(self._request_lease(pool))
         return await fut
@@ -1135,7 +1138,8 @@ class CoreRuntime:
                 return
         # nobody waiting: keep lease idle; return after a short grace
         # period (long enough for submit->get->submit reuse)
-        self.loop.call_later(0.05, self._maybe_return_idle, pool, lease)
+        self.loop.call_later(_cfg.lease_idle_grace_s,
+                             self._maybe_return_idle, pool, lease)
 
     def _release_or_reuse(self, pool: _LeasePool, lease: _Lease):
         self._grant_to_queue(pool, lease)

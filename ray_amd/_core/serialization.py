@@ -58,7 +58,9 @@ def _copy_range(dst, src, s, e):
 # Threshold below which values are inlined into RPC replies instead of
 # the object store (reference: max_direct_call_object_size=100KiB,
 # ray_config_def.h:274).
-INLINE_MAX = 100 * 1024
+from .._config import config as _cfg
+
+INLINE_MAX = _cfg.inline_max_bytes  # reference: max_direct_call_object_size
 
 
 def _pad(n: int) -> int:

@@ -17,7 +17,9 @@ import torch
 from .session import get_context
 from .trainer import DataParallelTrainer
 
-XGMI_BUCKET_CAP_MB = 128
+from .._config import config as _cfg
+
+XGMI_BUCKET_CAP_MB = _cfg.ddp_bucket_cap_mb
 
 
 class TorchTrainer(DataParallelTrainer):

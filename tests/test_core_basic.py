@@ -552,3 +552,20 @@ def test_object_locations_and_dynamic_resources(ray_start_regular):
     exp.set_resource("tokens", 0)
     time.sleep(0.3)
     assert "tokens" not in ray.cluster_resources()
+
+
+def test_ray_config_flags(monkeypatch):
+    """Central flag table (reference: ray_config_def.h RayConfig)."""
+    from ray_amd._config import RayConfig, config
+
+    desc = RayConfig.describe()
+    assert len(desc) >= 10
+    assert all("env" in v and "doc" in v for v in desc.values())
+    assert config.lease_request_cap == 16
+    monkeypatch.setenv("RAY_AMD_LEASE_REQUEST_CAP", "4")
+    config.reload()
+    assert config.lease_request_cap == 4
+    monkeypatch.delenv("RAY_AMD_LEASE_REQUEST_CAP")
+    config.reload()
+    with pytest.raises(AttributeError):
+        config.no_such_flag
