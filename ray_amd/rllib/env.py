@@ -84,7 +84,61 @@ class CartPoleEnv:
         return self.state.copy(), 1.0, terminated, truncated, {}
 
 
-_REGISTRY = {"CartPole-v1": CartPoleEnv}
+class PendulumEnv:
+    """Pendulum-v1: standard classic-control swing-up dynamics
+    (continuous torque in [-2, 2]); the built-in continuous-action env
+    for SAC (gymnasium is not in the image)."""
+
+    max_episode_steps = 200
+
+    def __init__(self, seed: Optional[int] = None):
+        self.max_speed = 8.0
+        self.max_torque = 2.0
+        self.dt = 0.05
+        self.g = 10.0
+        self.m = 1.0
+        self.l = 1.0
+        self.rng = np.random.default_rng(seed)
+        self.state = None
+        self.steps = 0
+        self.observation_space = Box(
+            np.array([-1.0, -1.0, -self.max_speed]),
+            np.array([1.0, 1.0, self.max_speed]), (3,))
+        self.action_space = Box(-self.max_torque, self.max_torque, (1,))
+
+    def _obs(self):
+        th, thdot = self.state
+        return np.array([np.cos(th), np.sin(th), thdot], dtype=np.float32)
+
+    def reset(self, *, seed: Optional[int] = None):
+        if seed is not None:
+            self.rng = np.random.default_rng(seed)
+        self.state = np.array([
+            self.rng.uniform(-np.pi, np.pi),
+            self.rng.uniform(-1.0, 1.0),
+        ])
+        self.steps = 0
+        return self._obs(), {}
+
+    def step(self, action):
+        th, thdot = self.state
+        u = float(np.clip(np.asarray(action).reshape(-1)[0],
+                          -self.max_torque, self.max_torque))
+        angle = ((th + np.pi) % (2 * np.pi)) - np.pi
+        cost = angle**2 + 0.1 * thdot**2 + 0.001 * u**2
+        thdot = thdot + (
+            3 * self.g / (2 * self.l) * np.sin(th)
+            + 3.0 / (self.m * self.l**2) * u
+        ) * self.dt
+        thdot = np.clip(thdot, -self.max_speed, self.max_speed)
+        th = th + thdot * self.dt
+        self.state = np.array([th, thdot])
+        self.steps += 1
+        truncated = self.steps >= self.max_episode_steps
+        return self._obs(), -float(cost), False, truncated, {}
+
+
+_REGISTRY = {"CartPole-v1": CartPoleEnv, "Pendulum-v1": PendulumEnv}
 _USER_REGISTRY = {}
 
 
@@ -133,8 +187,9 @@ class VectorEnv:
 
     def step(self, actions):
         obs, rews, terms, truncs = [], [], [], []
+        discrete = isinstance(self.action_space, Discrete)
         for i, (e, a) in enumerate(zip(self.envs, actions)):
-            o, r, term, trunc, _ = e.step(int(a))
+            o, r, term, trunc, _ = e.step(int(a) if discrete else a)
             self._episode_returns[i] += r
             self._episode_lens[i] += 1
             if term or trunc:
