@@ -65,6 +65,15 @@ _FLAGS: Dict[str, _Flag] = {
     "resource_report_period_s": _Flag(
         "RAY_AMD_RESOURCE_REPORT_PERIOD_S", 0.2, float,
         "raylet resource reporter loop period"),
+    # scheduling
+    "scheduler_top_k_fraction": _Flag(
+        "RAY_AMD_SCHEDULER_TOP_K_FRACTION", 0.2, float,
+        "hybrid policy picks randomly among the best k = "
+        "max(absolute, fraction*n) feasible nodes "
+        "(reference: scheduler_top_k_fraction)"),
+    "scheduler_top_k_absolute": _Flag(
+        "RAY_AMD_SCHEDULER_TOP_K_ABSOLUTE", 1, int,
+        "lower bound for the hybrid policy's top-k candidate pool"),
     # GPU data plane
     "ddp_bucket_cap_mb": _Flag(
         "RAY_AMD_DDP_BUCKET_MB", 128, int,

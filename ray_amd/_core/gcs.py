@@ -289,7 +289,9 @@ class GcsServer:
     def _pick_node(self, req: Dict[str, float], strategy: str = "hybrid",
                    exclude=(), soft_affinity: Optional[bytes] = None) -> Optional[NodeInfo]:
         """Hybrid policy (reference: policy/hybrid_scheduling_policy.h:28):
-        prefer the least-loaded feasible node; SPREAD picks round-robin."""
+        sort feasible nodes by load and pick uniformly among the top k
+        (top-k randomization avoids herding every scheduler decision
+        onto one node; reference scheduler_top_k_fraction)."""
         cands = [n for n in self._alive_nodes() if n.node_id not in exclude and self._fits(n, req)]
         if not cands:
             return None
@@ -303,7 +305,16 @@ class GcsServer:
             return 1.0 - n.resources_available.get("CPU", 0.0) / t
 
         cands.sort(key=load)
-        return cands[0]
+        from .._config import config as _cfgmod
+
+        k = max(
+            _cfgmod.scheduler_top_k_absolute,
+            int(len(cands) * _cfgmod.scheduler_top_k_fraction),
+        )
+        k = min(k, len(cands))
+        import random as _random
+
+        return _random.choice(cands[:k])
 
     # ---------- actors ----------
     async def h_register_actor(self, conn, p):

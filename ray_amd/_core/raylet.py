@@ -396,13 +396,18 @@ class Raylet:
                 self._node_cache_time = now
             except Exception:
                 return None
+        feasible = []
         for n in self._node_cache:
             if not n["alive"] or n["addr"] == self.addr:
                 continue
             tot = n["resources_total"]
             if all(tot.get(k, 0) + 1e-9 >= v for k, v in req.items() if v > 0):
-                return n["addr"]
-        return None
+                feasible.append(n["addr"])
+        if not feasible:
+            return None
+        import random as _random
+
+        return _random.choice(feasible)  # spread spillback, avoid herding
 
     # ---------------- actors ----------------
 
