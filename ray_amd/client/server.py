@@ -30,11 +30,21 @@ class ClientServer:
         # id(proto) -> {"refs": {oid: ObjectRef}, "actors": set[bytes]}
         self._sessions: Dict[int, dict] = {}
         self._lock = threading.Lock()
+        # msgpack-native task registry for non-Python clients (C++):
+        # name -> RemoteFunction, registered by the hosting driver
+        self._tasks: Dict[str, object] = {}
         for m in ("c_init c_put c_get c_wait c_export c_task c_cancel "
                   "c_actor_create c_actor_call c_kill c_gcs "
-                  "c_release").split():
+                  "c_release c_actor_msgpack c_task_msgpack").split():
             self.server.route(m, getattr(self, "h_" + m))
         self.server.on_conn_lost = self._conn_lost
+
+    def register_task(self, name: str, remote_fn):
+        """Expose a @ray.remote function to msgpack-native (e.g. C++)
+        clients under `name` (reference: C++ worker task registry,
+        src/ray/internal/internal.h)."""
+        self._tasks[name] = remote_fn
+        return self
 
     def start(self):
         """Start serving (requires ray_amd.init() in this process)."""
@@ -189,6 +199,54 @@ class ClientServer:
 
     async def h_c_gcs(self, conn, p):
         return await self.rt._gcs_rpc(p["method"], p["payload"])
+
+    # ---- msgpack-native surface (C++ / non-Python clients) ----
+
+    async def h_c_actor_msgpack(self, conn, p):
+        """Call a NAMED actor with plain-data (msgpack) args; the result
+        must be plain data too. No pickle anywhere — usable from C++."""
+
+        def do():
+            import ray_amd as ray
+
+            try:
+                h = ray.get_actor(p["name"], namespace=p.get("namespace"))
+                ref = getattr(h, p["method"]).remote(
+                    *(p.get("args") or []), **(p.get("kwargs") or {})
+                )
+                v = ray.get(ref, timeout=p.get("timeout", 60.0))
+                from .._core.protocol import pack
+
+                pack(v)  # plain-data check: fail here, not in the reply
+                return {"ok": True, "value": v}
+            except BaseException as e:  # noqa
+                return {"ok": False, "error": f"{type(e).__name__}: {e}"}
+
+        return await self._off(do)
+
+    async def h_c_task_msgpack(self, conn, p):
+        """Run a registered task with plain-data args and return its
+        plain-data result."""
+
+        def do():
+            import ray_amd as ray
+
+            fn = self._tasks.get(p["name"])
+            if fn is None:
+                return {"ok": False,
+                        "error": f"no registered task {p['name']!r}"}
+            try:
+                ref = fn.remote(*(p.get("args") or []),
+                                **(p.get("kwargs") or {}))
+                v = ray.get(ref, timeout=p.get("timeout", 60.0))
+                from .._core.protocol import pack
+
+                pack(v)
+                return {"ok": True, "value": v}
+            except BaseException as e:  # noqa
+                return {"ok": False, "error": f"{type(e).__name__}: {e}"}
+
+        return await self._off(do)
 
     async def h_c_release(self, conn, p):
         s = self._session(conn)
