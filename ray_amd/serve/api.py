@@ -401,6 +401,20 @@ class ServeController:
         return {"replicas": d["replicas"], "version": self.version,
                 "deployment": dname}
 
+    async def listen_for_change(self, app_name: str, deployment,
+                                known_version: int, timeout_s: float = 30.0):
+        """Long-poll (reference: _private/long_poll.py LongPollHost):
+        returns fresh routing once the config version moves past
+        `known_version`, or None at timeout (unchanged)."""
+        import time as _t
+
+        deadline = _t.monotonic() + timeout_s
+        while _t.monotonic() < deadline:
+            if self.version != known_version:
+                return self.get_routing(app_name, deployment)
+            await asyncio.sleep(0.1)
+        return None
+
     def resolve_route(self, path: str):
         best = None
         for prefix, app in self.routes.items():
@@ -518,6 +532,20 @@ class DeploymentResponseGenerator:
         return _ray().get(ref)
 
 
+class _RouterState:
+    """Routing state shared by every handle clone of one deployment
+    (reference: one LongPollClient per router, not per handle)."""
+
+    def __init__(self):
+        self.replicas: List = []
+        self.version = -1
+        self.counts: Dict[int, int] = {}
+        self.watch = None
+
+
+_router_states: Dict[tuple, _RouterState] = {}
+
+
 class DeploymentHandle:
     def __init__(self, app_name: str, deployment_name: Optional[str] = None,
                  method_name: str = "__call__", stream: bool = False):
@@ -525,9 +553,71 @@ class DeploymentHandle:
         self.deployment_name = deployment_name
         self.method_name = method_name
         self._stream = stream
-        self._replicas: List = []
-        self._version = -1
-        self._counts: Dict[int, int] = {}
+        self._rs = _router_states.setdefault(
+            (app_name, deployment_name or ""), _RouterState()
+        )
+
+    def __getstate__(self):
+        d = dict(self.__dict__)
+        d.pop("_rs", None)  # router state never crosses the wire
+        return d
+
+    def __setstate__(self, d):
+        self.__dict__.update(d)
+        self._rs = _router_states.setdefault(
+            (self.app_name, self.deployment_name or ""), _RouterState()
+        )
+
+    @property
+    def _replicas(self):
+        return self._rs.replicas
+
+    @property
+    def _version(self):
+        return self._rs.version
+
+    @property
+    def _counts(self):
+        return self._rs.counts
+
+    def _start_watch(self):
+        """Push-based config updates (reference: LongPollClient in every
+        router/proxy): one daemon thread per deployment long-polls the
+        controller and swaps the shared replica set when the version
+        moves."""
+        rs = self._rs
+        if rs.watch is not None:
+            return
+        import threading
+        import time as _t
+
+        app, dep = self.app_name, self.deployment_name
+
+        def watch():
+            ray = _ray()
+            failures = 0
+            while failures < 30:
+                try:
+                    ctrl = ray.get_actor(SERVE_CONTROLLER_NAME,
+                                         namespace=SERVE_NAMESPACE)
+                    info = ray.get(
+                        ctrl.listen_for_change.remote(app, dep, rs.version),
+                        timeout=45,
+                    )
+                    failures = 0
+                except Exception:
+                    failures += 1
+                    _t.sleep(1.0)
+                    continue
+                if info is not None:
+                    rs.replicas = info["replicas"]
+                    rs.version = info["version"]
+                    rs.counts = {i: 0 for i in range(len(info["replicas"]))}
+
+        t = threading.Thread(target=watch, daemon=True,
+                             name="serve-longpoll")
+        t.start()
+        rs.watch = t
 
     def _refresh(self):
         ray = _ray()
@@ -539,10 +629,14 @@ class DeploymentHandle:
             raise RuntimeError(
                 f"no deployment {self.deployment_name} in app {self.app_name}"
             )
-        self._replicas = info["replicas"]
-        self._version = info["version"]
-        self.deployment_name = info["deployment"]
-        self._counts = {i: 0 for i in range(len(self._replicas))}
+        rs = self._rs
+        rs.replicas = info["replicas"]
+        rs.version = info["version"]
+        rs.counts = {i: 0 for i in range(len(rs.replicas))}
+        if self.deployment_name != info["deployment"]:
+            self.deployment_name = info["deployment"]
+            # alias the resolved name so clones share this state
+            _router_states[(self.app_name, self.deployment_name)] = rs
 
     def _pick(self) -> int:
         n = len(self._replicas)
@@ -554,6 +648,7 @@ class DeploymentHandle:
     def remote(self, *args, **kwargs) -> DeploymentResponse:
         if not self._replicas:
             self._refresh()
+        self._start_watch()
         for attempt in range(3):
             idx = self._pick()
             replica = self._replicas[idx]
@@ -582,14 +677,11 @@ class DeploymentHandle:
 
     def options(self, *, method_name: Optional[str] = None,
                 stream: Optional[bool] = None, **kwargs):
-        h = DeploymentHandle(
+        return DeploymentHandle(
             self.app_name, self.deployment_name,
             method_name or self.method_name,
             self._stream if stream is None else stream,
         )
-        h._replicas = self._replicas
-        h._counts = self._counts
-        return h
 
     def __getattr__(self, item):
         if item.startswith("_"):
@@ -902,6 +994,7 @@ def shutdown():
             ray.kill(a)
         except Exception:
             pass
+    _router_states.clear()  # drop stale routing (+ dead watch threads)
 
 
 # --------------------------------------------------------------------------

@@ -291,3 +291,28 @@ def test_http_chunked_streaming(serve_session):
     bodytext = b"".join(parts).decode()
     assert bodytext == "chunk0;chunk1;chunk2;chunk3;chunk4;"
     assert first_at is not None and first_at < 0.45  # streamed, not buffered
+
+
+def test_long_poll_config_push(serve_session):
+    """Handles learn about redeployments via controller long-poll
+    without a failing request forcing a refresh."""
+    @serve.deployment(num_replicas=1)
+    class V:
+        def version(self):
+            return "v1"
+
+    h = serve.run(V.bind(), name="lp_app", http=False)
+    assert h.version.remote().result(timeout_s=30) == "v1"
+    assert len(h._replicas) == 1
+
+    @serve.deployment(num_replicas=3, name="V")
+    class V2:
+        def version(self):
+            return "v2"
+
+    serve.run(V2.bind(), name="lp_app", http=False)
+    deadline = time.time() + 15
+    while time.time() < deadline and len(h._replicas) != 3:
+        time.sleep(0.2)
+    assert len(h._replicas) == 3  # pushed, not pulled on failure
+    assert h.version.remote().result(timeout_s=30) == "v2"
