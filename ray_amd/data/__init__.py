@@ -16,7 +16,9 @@ from .dataset import (  # noqa: F401
     from_pandas,
     range,  # noqa: A001  (API parity with ray.data.range)
     range_tensor,
+    from_huggingface,
     read_binary_files,
+    read_text,
     read_csv,
     read_json,
     read_parquet,

@@ -26,6 +26,11 @@ class DataContext:
     def __init__(self):
         self.target_max_block_size = 128 * 1024 * 1024
         self.streaming_read_window = 8  # max in-flight blocks per consumer
+        # streaming-executor resource budget (reference:
+        # _internal/execution/streaming_executor.py resource manager):
+        # stop launching new block tasks while the local object store is
+        # above this fraction of capacity — in-flight blocks drain first
+        self.object_store_memory_fraction = 0.8
 
     @classmethod
     def get_current(cls) -> "DataContext":
@@ -38,6 +43,28 @@ def _ray():
     import ray_amd as ray
 
     return ray
+
+
+def _store_pressure(ray) -> bool:
+    """True while the local object store exceeds the streaming budget
+    (cheap: cached 0.2 s)."""
+    import time as _t
+
+    now = _t.time()
+    t0, val = getattr(_store_pressure, "_cache", (0.0, False))
+    if now - t0 < 0.2:
+        return val
+    try:
+        from ray_amd._core import runtime as _rtmod
+
+        rt = _rtmod.global_runtime()
+        r = rt._call_sync(rt.raylet.call("object_stats", {}), timeout=2)
+        frac = DataContext.get_current().object_store_memory_fraction
+        val = r["capacity"] > 0 and r["used"] / r["capacity"] > frac
+    except Exception:
+        val = False
+    _store_pressure._cache = (now, val)
+    return val
 
 
 # ---------------- block helpers ----------------
@@ -246,12 +273,14 @@ class Dataset:
         self._ops = ops or []
         self._materialized: Optional[List[Any]] = None
         self._task_opts: Optional[dict] = None
+        self._concurrency: Optional[int] = None  # per-op in-flight cap
 
     # ----- plan builders -----
 
     def _with_op(self, op: tuple) -> "Dataset":
         ds = Dataset(self._inputs, self._ops + [op])
         ds._task_opts = self._task_opts
+        ds._concurrency = self._concurrency
         return ds
 
     def map(self, fn: Callable[[dict], dict], **kwargs) -> "Dataset":
@@ -291,12 +320,15 @@ class Dataset:
                 return tl(batch)
 
         op = ("map_batches", fn, batch_format, batch_size)
-        if num_gpus or kwargs.get("num_cpus"):
+        if num_gpus or kwargs.get("num_cpus") or concurrency:
             ds = self._with_op(op)
-            ds._task_opts = {
-                "num_gpus": num_gpus,
-                "num_cpus": kwargs.get("num_cpus", 1),
-            }
+            if num_gpus or kwargs.get("num_cpus"):
+                ds._task_opts = {
+                    "num_gpus": num_gpus,
+                    "num_cpus": kwargs.get("num_cpus", 1),
+                }
+            if concurrency:
+                ds._concurrency = int(concurrency)
             return ds
         return self._with_op(op)
 
@@ -379,6 +411,8 @@ class Dataset:
             return
         ops = self._ops
         window = DataContext.get_current().streaming_read_window
+        if self._concurrency:
+            window = min(window, self._concurrency)
 
         @ray.remote
         def _exec_block(block, ops=ops):
@@ -387,13 +421,28 @@ class Dataset:
 
         pending = []
         inputs = iter(self._inputs)
-        for inp in itertools.islice(inputs, window):
-            pending.append(_exec_block.remote(inp))
+        sentinel = object()
+        held = None
+
+        def refill():
+            # top the window up; under store pressure stop producing and
+            # let in-flight blocks drain — but never starve (always
+            # launch when nothing is in flight)
+            nonlocal held
+            while len(pending) < window:
+                if held is None:
+                    held = next(inputs, sentinel)
+                if held is sentinel:
+                    return
+                if pending and _store_pressure(ray):
+                    return
+                pending.append(_exec_block.remote(held))
+                held = None
+
+        refill()
         while pending:
             ref = pending.pop(0)
-            nxt = next(inputs, None)
-            if nxt is not None:
-                pending.append(_exec_block.remote(nxt))
+            refill()
             yield ref
 
     def _iter_tables(self) -> Iterator[pa.Table]:
@@ -1066,6 +1115,51 @@ def read_json(paths, *, parallelism: int = -1, **kwargs) -> Dataset:
         return pjson.read_json(f)
 
     refs = [_read.remote(f) for f in files]
+    ds = Dataset(refs, [])
+    ds._materialized = refs
+    return ds
+
+
+def read_text(paths, *, encoding: str = "utf-8", drop_empty_lines=True,
+              **kwargs) -> Dataset:
+    """One row per line (reference: data/read_api.py read_text)."""
+    ray = _ray()
+    files = _expand_paths(paths)
+
+    @ray.remote
+    def _read(f, encoding=encoding, drop=drop_empty_lines):
+        with open(f, "r", encoding=encoding) as fh:
+            lines = fh.read().splitlines()
+        if drop:
+            lines = [l for l in lines if l.strip()]
+        return pa.table({"text": pa.array(lines)})
+
+    refs = [_read.remote(f) for f in files]
+    ds = Dataset(refs, [])
+    ds._materialized = refs
+    return ds
+
+
+def from_huggingface(hf_dataset, *, parallelism: int = -1) -> Dataset:
+    """Zero-copy-ish ingestion of a HuggingFace `datasets.Dataset`
+    (reference: data/read_api.py from_huggingface) — the HF dataset's
+    arrow table is sliced into blocks and put into the object store."""
+    ray = _ray()
+    try:
+        t = hf_dataset.data.table  # datasets.Dataset -> pyarrow Table
+    except AttributeError:
+        t = hf_dataset.with_format("arrow")[:]
+    t = t.combine_chunks()
+    n = t.num_rows
+    if parallelism <= 0:
+        parallelism = max(1, min(64, n // 10_000 or 1))
+    step = (n + parallelism - 1) // parallelism
+    import builtins
+
+    refs = [
+        ray.put(t.slice(i, min(step, n - i)))
+        for i in builtins.range(0, n, step)  # module fn `range` shadows
+    ]
     ds = Dataset(refs, [])
     ds._materialized = refs
     return ds

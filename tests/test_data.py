@@ -184,3 +184,29 @@ def test_hash_join(ray_start_regular):
     assert len(joined) == 10  # keys 10..19
     for r in joined:
         assert r["a"] == r["k"] * 10 and r["b"] == r["k"] * 100
+
+
+def test_read_text_and_from_huggingface(ray_start_regular, tmp_path):
+    from ray_amd import data as ray_data
+
+    f = tmp_path / "lines.txt"
+    f.write_text("alpha\n\nbeta\ngamma\n")
+    ds = ray_data.read_text(str(f))
+    assert [r["text"] for r in ds.take_all()] == ["alpha", "beta", "gamma"]
+
+    import datasets as hf
+
+    hfd = hf.Dataset.from_dict({"x": list(range(100)), "y": ["a"] * 100})
+    ds2 = ray_data.from_huggingface(hfd)
+    assert ds2.count() == 100
+    out = ds2.map_batches(lambda b: {"x2": b["x"] * 2}).take(3)
+    assert out[1]["x2"] == 2
+
+
+def test_map_batches_concurrency_cap(ray_start_regular):
+    from ray_amd import data as ray_data
+
+    ds = ray_data.range(64, override_num_blocks=16).map_batches(
+        lambda b: b, concurrency=2
+    )
+    assert ds.count() == 64  # capped window still processes everything
