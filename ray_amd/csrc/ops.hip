@@ -13,6 +13,7 @@
 #include "hip/flash_attn_v3.hip"
 #include "hip/flash_attn_v4.hip"
 #include "hip/flash_attn_v5.hip"
+#include "hip/fa_bwd.hip"
 
 #define CHECK_IN(x)                                                     \
   TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                       \
@@ -272,8 +273,48 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
   return {o};
 }
 
+std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
+                                        at::Tensor v, at::Tensor o,
+                                        at::Tensor d_o, at::Tensor lse,
+                                        bool causal) {
+  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v); CHECK_IN(o); CHECK_IN(d_o);
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "q must be bf16");
+  TORCH_CHECK(q.size(3) == 128, "head_dim must be 128");
+  TORCH_CHECK(q.size(2) % 128 == 0, "T must be a multiple of 128");
+  TORCH_CHECK(q.size(2) == k.size(2),
+              "bwd requires T == Tk (training self-attention)");
+  int B = (int)q.size(0), Hq = (int)q.size(1), T = (int)q.size(2);
+  int Hkv = (int)k.size(1);
+  TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto dsum = at::empty({B, Hq, T}, q.options().dtype(at::kFloat));
+  float scale = 1.0f / sqrtf((float)q.size(3));
+  long long rows = (long long)B * Hq * T;
+  hipLaunchKernelGGL(fa_bwd_prep_bf16, dim3((rows + 3) / 4), dim3(256), 0,
+                     cur_stream(), (const short*)d_o.data_ptr(),
+                     (const short*)o.data_ptr(), dsum.data_ptr<float>(),
+                     rows);
+  hipLaunchKernelGGL(fa_bwd_dq_bf16, dim3(T / 128, B * Hq), dim3(256), 0,
+                     cur_stream(), (const short*)q.data_ptr(),
+                     (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                     (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
+                     dsum.data_ptr<float>(), (short*)dq.data_ptr(), B, Hq,
+                     Hkv, T, causal ? 1 : 0, scale);
+  hipLaunchKernelGGL(fa_bwd_dkv_bf16, dim3(T / 128, B * Hkv), dim3(256), 0,
+                     cur_stream(), (const short*)q.data_ptr(),
+                     (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                     (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
+                     dsum.data_ptr<float>(), (short*)dk.data_ptr(),
+                     (short*)dv.data_ptr(), B, Hq, Hkv, T, causal ? 1 : 0,
+                     scale);
+  return {dq, dk, dv};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_fwd", &flash_attn_fwd);
+  m.def("flash_attn_bwd", &flash_attn_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("swiglu_fwd", &swiglu_fwd);

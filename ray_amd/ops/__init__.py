@@ -374,15 +374,55 @@ def flash_attention_ref(q, k, v, causal=True):
     return out.to(q.dtype), lse
 
 
+class _FlashAttnFn(torch.autograd.Function):
+    """Training flash attention: v5 forward + FA2-style two-kernel
+    backward (csrc/hip/fa_bwd.hip). Requires T == Tk, T % 128 == 0,
+    D == 128 (the Llama training shape)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal):
+        qc, kc, vc = q.contiguous(), k.contiguous(), v.contiguous()
+        out, lse = _K.flash_attn_fwd(qc, kc, vc, causal, 0, True)
+        ctx.save_for_backward(qc, kc, vc, out, lse)
+        ctx.causal = causal
+        return out
+
+    @staticmethod
+    def backward(ctx, d_out):
+        q, k, v, out, lse = ctx.saved_tensors
+        dq, dk, dv = _K.flash_attn_bwd(
+            q, k, v, out, d_out.contiguous(), lse, ctx.causal
+        )
+        return dq, dk, dv, None
+
+
 def flash_attention(q, k, v, causal: bool = True, q_offset: int = 0,
                     return_lse: bool = False):
-    """Fused CDNA4 flash-attention forward (inference / ring-attention
-    block op). q [B,Hq,T,128] bf16; T padded to 64 internally."""
+    """Fused CDNA4 flash attention. q [B,Hq,T,128] bf16 (GQA k/v
+    [B,Hkv,Tk,128]). Inference pads T to 128 internally; the
+    differentiable path (any input requires_grad) additionally needs
+    T == Tk and T % 128 == 0 and supports backward via the HIP
+    fa_bwd kernels."""
     if not q.is_cuda:
         out, lse = flash_attention_ref(q, k, v, causal)
         return (out, lse) if return_lse else out
     _require_ext("flash_attention")
     B, Hq, T, D = q.shape
+    needs_grad = torch.is_grad_enabled() and (
+        q.requires_grad or k.requires_grad or v.requires_grad
+    )
+    if needs_grad:
+        if return_lse or q_offset:
+            raise NotImplementedError(
+                "flash_attention backward does not support return_lse/"
+                "q_offset (ring-attention bwd lands next round)"
+            )
+        if T % 128 != 0 or T != k.shape[2]:
+            raise NotImplementedError(
+                "flash_attention backward requires T == Tk and "
+                "T % 128 == 0"
+            )
+        return _FlashAttnFn.apply(q, k, v, causal)
     pad = (-T) % 128
     if pad:
         q = torch.nn.functional.pad(q, (0, 0, 0, pad))

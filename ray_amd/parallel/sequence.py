@@ -24,10 +24,14 @@ def _block_attn(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     "causal" (Tq==Tk lower-triangular), "none" (skip — caller handles).
     On GPU with head_dim 128 this runs the fused CDNA4 flash kernel
     (LSE output was built for exactly this merge)."""
+    needs_grad = torch.is_grad_enabled() and (
+        q.requires_grad or k.requires_grad or v.requires_grad
+    )
     if (
         q.is_cuda
         and q.shape[-1] == 128
         and q.dtype == torch.bfloat16
+        and not needs_grad  # lse-merging bwd lands next round
     ):
         from ray_amd import ops
 
