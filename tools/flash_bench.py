@@ -58,6 +58,34 @@ def main():
         f"{flops/t_sdpa/1e12:.1f} TFLOP/s\n"
         f"max |err| vs fp32 ref: {err:.4f}"
     )
+
+    # ---- backward (fwd+bwd round trip, grads on) ----
+    bwd_flops = flops * 3.5  # fwd (1x) + bwd (2.5x)
+    qg = q.clone().requires_grad_()
+    kg = k.clone().requires_grad_()
+    vg = v.clone().requires_grad_()
+    g = torch.randn_like(q)
+
+    def ours_fb():
+        out = ops.flash_attention(qg, kg, vg, causal=causal)
+        out.backward(g)
+        qg.grad = kg.grad = vg.grad = None
+
+    def sdpa_fb():
+        out = torch.nn.functional.scaled_dot_product_attention(
+            qg, kg, vg, is_causal=causal
+        )
+        out.backward(g)
+        qg.grad = kg.grad = vg.grad = None
+
+    t_ours_fb = bench(ours_fb, iters=20, warmup=5)
+    t_sdpa_fb = bench(sdpa_fb, iters=20, warmup=5)
+    print(
+        f"fwd+bwd ray_amd : {t_ours_fb*1e3:.3f} ms  "
+        f"{bwd_flops/t_ours_fb/1e12:.1f} TFLOP/s\n"
+        f"fwd+bwd SDPA    : {t_sdpa_fb*1e3:.3f} ms  "
+        f"{bwd_flops/t_sdpa_fb/1e12:.1f} TFLOP/s"
+    )
     return 0
 
 
