@@ -363,3 +363,245 @@ extern "C" __global__ __launch_bounds__(256) void fa_bwd_dkv_bf16(
           f2bf(dv_acc[dt][r]);
     }
 }
+
+
+// ---------------------------------------------------------------------
+// split variant: dv-only and dk-only kernels. Each drops one 64-reg
+// accumulator pair (and dv drops the dP mfmas), landing ~230 regs →
+// 2 waves/SIMD, at +25% total mfma work vs the fused kernel.
+// ---------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(256) void fa_bwd_dv_bf16(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, const short* __restrict__ dO,
+    const float* __restrict__ LSE, short* __restrict__ dV, int B,
+    int Hq, int Hkv, int T, int causal, float scale) {
+  __shared__ short k_lds[128][FAB_LDK];
+  __shared__ short do_lds[32][FAB_LDK];
+  __shared__ short q_lds[32][FAB_LDK];
+  __shared__ float lse_lds[32];
+  __shared__ short scratch[4][32][40];
+
+  const int k0 = blockIdx.x * 128;
+  const int bh = blockIdx.y;
+  const int b = bh / Hkv;
+  const int hkv = bh % Hkv;
+  const int rep = Hq / Hkv;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int hi = lane >> 5;
+  const int a_off = 8 * hi;
+
+  const long long kbase = (((long long)b * Hkv + hkv) * T + k0) * FAB_D;
+  for (int i = threadIdx.x; i < 128 * FAB_D / 8; i += 256) {
+    int r = i / (FAB_D / 8);
+    int c = (i % (FAB_D / 8)) * 8;
+    *reinterpret_cast<short8*>(&k_lds[r][c]) =
+        *reinterpret_cast<const short8*>(K + kbase +
+                                         (long long)r * FAB_D + c);
+  }
+
+  fab_f32x16 dv_acc[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) dv_acc[t] = fab_f32x16{};
+  const float L2E = 1.4426950408889634f;
+
+  for (int g = 0; g < rep; ++g) {
+    const int hq = hkv * rep + g;
+    const long long qbase0 = (((long long)b * Hq + hq) * T) * FAB_D;
+    const long long lbase = ((long long)b * Hq + hq) * T;
+    const int q_start = causal ? k0 : 0;
+    for (int q0s = q_start; q0s < T; q0s += 32) {
+      __syncthreads();
+      for (int i = threadIdx.x; i < 32 * FAB_D / 8; i += 256) {
+        int r = i / (FAB_D / 8);
+        int c = (i % (FAB_D / 8)) * 8;
+        *reinterpret_cast<short8*>(&q_lds[r][c]) =
+            *reinterpret_cast<const short8*>(
+                Q + qbase0 + (long long)(q0s + r) * FAB_D + c);
+        *reinterpret_cast<short8*>(&do_lds[r][c]) =
+            *reinterpret_cast<const short8*>(
+                dO + qbase0 + (long long)(q0s + r) * FAB_D + c);
+      }
+      if (threadIdx.x < 32)
+        lse_lds[threadIdx.x] = LSE[lbase + q0s + threadIdx.x];
+      __syncthreads();
+      if (causal && k0 + 32 * wave > q0s + 31) continue;
+
+      fab_f32x16 s_acc{};
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        fab_bf16x8 kf =
+            fab_ld8(&k_lds[32 * wave + (lane & 31)][16 * c + a_off]);
+        fab_bf16x8 qf = fab_ld8(&q_lds[lane & 31][16 * c + a_off]);
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf, s_acc,
+                                                        0, 0, 0);
+      }
+      const int gq = q0s + (lane & 31);
+      const float lse_q = lse_lds[lane & 31];
+      const bool full_tile = !causal || (k0 + 32 * wave + 31 <= q0s);
+      short(*scr)[40] = scratch[wave];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int gk = k0 + 32 * wave + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float pv = __builtin_amdgcn_exp2f(
+            __builtin_fmaf(s_acc[r] * scale, L2E, -lse_q * L2E));
+        if (!full_tile && gk > gq) pv = 0.f;
+        scr[(r & 3) + 8 * (r >> 2) + 4 * hi][lane & 31] = f2bf(pv);
+      }
+      __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+      for (int qc = 0; qc < 2; ++qc) {
+        fab_bf16x8 af = __builtin_bit_cast(
+            fab_bf16x8, *reinterpret_cast<const short8*>(
+                            &scr[lane & 31][16 * qc + a_off]));
+#pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          short dtmp[8];
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            dtmp[i] = do_lds[16 * qc + a_off + i][32 * dt + (lane & 31)];
+          fab_bf16x8 bf = __builtin_bit_cast(
+              fab_bf16x8, *reinterpret_cast<short8*>(dtmp));
+          dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af, bf, dv_acc[dt], 0, 0, 0);
+        }
+      }
+    }
+  }
+  short* outv = dV + kbase + (long long)wave * 32 * FAB_D;
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int krow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      outv[(long long)krow * FAB_D + 32 * dt + (lane & 31)] =
+          f2bf(dv_acc[dt][r]);
+    }
+}
+
+extern "C" __global__ __launch_bounds__(256) void fa_bwd_dk_bf16(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, const short* __restrict__ dO,
+    const float* __restrict__ LSE, const float* __restrict__ Dsum,
+    short* __restrict__ dK, int B, int Hq, int Hkv, int T, int causal,
+    float scale) {
+  __shared__ short k_lds[128][FAB_LDK];
+  __shared__ short q_lds[32][FAB_LDK];
+  __shared__ short do_lds[32][FAB_LDK];
+  __shared__ float lse_lds[32];
+  __shared__ float dsum_lds[32];
+  __shared__ short scratch[4][32][40];
+
+  const int k0 = blockIdx.x * 128;
+  const int bh = blockIdx.y;
+  const int b = bh / Hkv;
+  const int hkv = bh % Hkv;
+  const int rep = Hq / Hkv;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int hi = lane >> 5;
+  const int a_off = 8 * hi;
+
+  const long long kbase = (((long long)b * Hkv + hkv) * T + k0) * FAB_D;
+  for (int i = threadIdx.x; i < 128 * FAB_D / 8; i += 256) {
+    int r = i / (FAB_D / 8);
+    int c = (i % (FAB_D / 8)) * 8;
+    *reinterpret_cast<short8*>(&k_lds[r][c]) =
+        *reinterpret_cast<const short8*>(K + kbase +
+                                         (long long)r * FAB_D + c);
+  }
+  // V is only ever read as this wave's own 32 rows: keep it in
+  // registers (A-operand layout) and off the LDS budget
+  fab_bf16x8 v_frag[8];
+  {
+    const short* vp =
+        V + kbase + ((long long)32 * wave + (lane & 31)) * FAB_D;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) v_frag[c] = fab_ld8(vp + 16 * c + a_off);
+  }
+
+  fab_f32x16 dk_acc[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) dk_acc[t] = fab_f32x16{};
+  const float L2E = 1.4426950408889634f;
+
+  for (int g = 0; g < rep; ++g) {
+    const int hq = hkv * rep + g;
+    const long long qbase0 = (((long long)b * Hq + hq) * T) * FAB_D;
+    const long long lbase = ((long long)b * Hq + hq) * T;
+    const int q_start = causal ? k0 : 0;
+    for (int q0s = q_start; q0s < T; q0s += 32) {
+      __syncthreads();
+      for (int i = threadIdx.x; i < 32 * FAB_D / 8; i += 256) {
+        int r = i / (FAB_D / 8);
+        int c = (i % (FAB_D / 8)) * 8;
+        *reinterpret_cast<short8*>(&q_lds[r][c]) =
+            *reinterpret_cast<const short8*>(
+                Q + qbase0 + (long long)(q0s + r) * FAB_D + c);
+        *reinterpret_cast<short8*>(&do_lds[r][c]) =
+            *reinterpret_cast<const short8*>(
+                dO + qbase0 + (long long)(q0s + r) * FAB_D + c);
+      }
+      if (threadIdx.x < 32) {
+        lse_lds[threadIdx.x] = LSE[lbase + q0s + threadIdx.x];
+        dsum_lds[threadIdx.x] = Dsum[lbase + q0s + threadIdx.x];
+      }
+      __syncthreads();
+      if (causal && k0 + 32 * wave > q0s + 31) continue;
+
+      fab_f32x16 s_acc{}, dp_acc{};
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        fab_bf16x8 kf =
+            fab_ld8(&k_lds[32 * wave + (lane & 31)][16 * c + a_off]);
+        fab_bf16x8 qf = fab_ld8(&q_lds[lane & 31][16 * c + a_off]);
+        fab_bf16x8 dof = fab_ld8(&do_lds[lane & 31][16 * c + a_off]);
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf, s_acc,
+                                                        0, 0, 0);
+        dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(v_frag[c], dof,
+                                                         dp_acc, 0, 0, 0);
+      }
+      const int gq = q0s + (lane & 31);
+      const float lse_q = lse_lds[lane & 31];
+      const float d_q = dsum_lds[lane & 31];
+      const bool full_tile = !causal || (k0 + 32 * wave + 31 <= q0s);
+      short(*scr)[40] = scratch[wave];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int gk = k0 + 32 * wave + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float pv = __builtin_amdgcn_exp2f(
+            __builtin_fmaf(s_acc[r] * scale, L2E, -lse_q * L2E));
+        if (!full_tile && gk > gq) pv = 0.f;
+        float dsv = pv * (dp_acc[r] - d_q) * scale;
+        scr[(r & 3) + 8 * (r >> 2) + 4 * hi][lane & 31] = f2bf(dsv);
+      }
+      __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+      for (int qc = 0; qc < 2; ++qc) {
+        fab_bf16x8 af = __builtin_bit_cast(
+            fab_bf16x8, *reinterpret_cast<const short8*>(
+                            &scr[lane & 31][16 * qc + a_off]));
+#pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          short qtmp[8];
+#pragma unroll
+          for (int i = 0; i < 8; ++i)
+            qtmp[i] = q_lds[16 * qc + a_off + i][32 * dt + (lane & 31)];
+          fab_bf16x8 bf = __builtin_bit_cast(
+              fab_bf16x8, *reinterpret_cast<short8*>(qtmp));
+          dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af, bf, dk_acc[dt], 0, 0, 0);
+        }
+      }
+    }
+  }
+  short* outk = dK + kbase + (long long)wave * 32 * FAB_D;
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int krow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      outk[(long long)krow * FAB_D + 32 * dt + (lane & 31)] =
+          f2bf(dk_acc[dt][r]);
+    }
+}
