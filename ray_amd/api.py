@@ -83,6 +83,10 @@ def init(
     **kwargs,
 ) -> RayContext:
     global _cluster, _namespace
+    if address is None:
+        # job drivers / tooling inherit the cluster address from the
+        # environment (reference: RAY_ADDRESS)
+        address = os.environ.get("RAY_AMD_ADDRESS") or None
     with _init_lock:
         if _rt.is_initialized():
             if ignore_reinit_error:

@@ -162,6 +162,76 @@ def main(argv=None):
     s.add_argument("--duration", type=float, default=2.0)
     s.set_defaults(fn=cmd_microbenchmark)
 
+    # ray_amd job submit|status|logs|stop|list (reference: `ray job ...`)
+    j = sub.add_parser("job")
+    jsub = j.add_subparsers(dest="job_cmd", required=True)
+
+    def _client(args):
+        from ray_amd.job_submission import JobSubmissionClient
+
+        return JobSubmissionClient(address=args.address)
+
+    js = jsub.add_parser("submit")
+    js.add_argument("--address", default=None)
+    js.add_argument("--submission-id", default=None)
+    js.add_argument("--working-dir", default=None)
+    js.add_argument("--no-wait", action="store_true")
+    js.add_argument("entrypoint", nargs=argparse.REMAINDER)
+
+    def _submit(args):
+        from ray_amd.job_submission import JobStatus
+
+        import shlex
+
+        c = _client(args)
+        parts = list(args.entrypoint)
+        if parts and parts[0] == "--":
+            parts = parts[1:]
+        ep = shlex.join(parts)
+        renv = {}
+        if args.working_dir:
+            renv["working_dir"] = args.working_dir
+        jid = c.submit_job(entrypoint=ep, runtime_env=renv or None,
+                           submission_id=args.submission_id)
+        print(jid)
+        if args.no_wait:
+            return 0
+        for chunk in c.tail_job_logs(jid):
+            print(chunk, end="")
+        return 0 if c.get_job_status(jid) == JobStatus.SUCCEEDED else 1
+
+    js.set_defaults(fn=_submit)
+
+    for jname in ("status", "logs", "stop"):
+        js = jsub.add_parser(jname)
+        js.add_argument("--address", default=None)
+        js.add_argument("submission_id")
+
+        def _mk(jname):
+            def run(args):
+                c = _client(args)
+                if jname == "status":
+                    print(c.get_job_status(args.submission_id))
+                elif jname == "logs":
+                    print(c.get_job_logs(args.submission_id), end="")
+                else:
+                    print(c.stop_job(args.submission_id))
+                return 0
+
+            return run
+
+        js.set_defaults(fn=_mk(jname))
+
+    js = jsub.add_parser("list")
+    js.add_argument("--address", default=None)
+
+    def _list(args):
+        for j in _client(args).list_jobs():
+            print(f"{j.submission_id}  {j.status}  {j.entrypoint[:60]}")
+        return 0
+
+    js.set_defaults(fn=_list)
+
     args = p.parse_args(argv)
     return args.fn(args)
 
