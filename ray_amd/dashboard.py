@@ -31,6 +31,8 @@ def build_asgi_app():
                 body = _json(st.list_tasks(limit=1000))
             elif path == "/api/cluster_status":
                 body = _json(_cluster_status())
+            elif path == "/api/jobs":
+                body = _json(_jobs())
             elif path == "/api/placement_groups":
                 from ray_amd.util import state as st
 
@@ -68,6 +70,21 @@ def _nodes():
     return ray.nodes()
 
 
+def _jobs():
+    """Job manager listing (reference: dashboard/modules/job)."""
+    import ray_amd as ray
+    from ray_amd.job_submission import (
+        JOB_MANAGER_NAME,
+        JOB_NAMESPACE,
+    )
+
+    try:
+        mgr = ray.get_actor(JOB_MANAGER_NAME, namespace=JOB_NAMESPACE)
+        return ray.get(mgr.list.remote(), timeout=10)
+    except Exception:
+        return []
+
+
 def _cluster_status():
     import ray_amd as ray
     from ray_amd.util import state as st
@@ -98,7 +115,7 @@ async function refresh(){
 }
 refresh(); setInterval(refresh, 2000);
 </script>
-<p>APIs: /api/nodes /api/actors /api/tasks /api/placement_groups /metrics</p>
+<p>APIs: /api/nodes /api/actors /api/tasks /api/jobs /api/placement_groups /metrics</p>
 </body></html>"""
 
 
