@@ -65,6 +65,10 @@ _FLAGS: Dict[str, _Flag] = {
     "resource_report_period_s": _Flag(
         "RAY_AMD_RESOURCE_REPORT_PERIOD_S", 0.2, float,
         "raylet resource reporter loop period"),
+    "streaming_backpressure_items": _Flag(
+        "RAY_AMD_STREAMING_BACKPRESSURE_ITEMS", 64, int,
+        "a streaming generator pauses when this many yielded items are "
+        "unconsumed (reference: generator_backpressure_num_objects)"),
     # scheduling
     "scheduler_top_k_fraction": _Flag(
         "RAY_AMD_SCHEDULER_TOP_K_FRACTION", 0.2, float,

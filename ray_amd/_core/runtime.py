@@ -144,6 +144,9 @@ class ObjectRefGenerator:
             if self._consumed < len(st["items"]):
                 oid = st["items"][self._consumed]
                 self._consumed += 1
+                prod = st.get("producer")
+                if prod:
+                    self._rt._stream_ack(self._task_id, self._consumed, prod)
                 return ObjectRef(oid, self._rt.addr)
             if st["done"]:
                 err = st.get("error")
@@ -415,6 +418,20 @@ class CoreRuntime:
             except Exception:
                 pass
 
+    def _stream_ack(self, task_id: bytes, consumed: int, addr: str):
+        async def _send():
+            try:
+                c = await self._conn(addr)
+                c.notify("stream_ack",
+                         {"task_id": task_id, "consumed": consumed})
+            except Exception:
+                pass
+
+        try:
+            self._run(_send())
+        except Exception:
+            pass
+
     def _h_stream_item(self, p):
         """Streaming-generator item pushed by the executing worker
         (reference: ReportGeneratorItemReturns, core_worker.h:856)."""
@@ -428,6 +445,8 @@ class CoreRuntime:
         else:
             self._store_put(oid, ("store", p["node_addr"], p["size"]))
         st["items"].append(oid)
+        if p.get("addr"):
+            st["producer"] = p["addr"]
         ev = st.get("event")
         if ev is not None:
             self.loop.call_soon_threadsafe(ev.set)

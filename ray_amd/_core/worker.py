@@ -90,6 +90,8 @@ class WorkerMain:
         self._cancelled = set()
         self._events = []
         self._max_concurrency = 1
+        # producer-side streaming backpressure: task_id -> {acked, event}
+        self._stream_prod = {}
 
     async def main(self):
         loop = asyncio.get_running_loop()
@@ -109,6 +111,7 @@ class WorkerMain:
         rt.server.route("actor_call", self.h_actor_call)
         rt.server.route("cancel_task", self.h_cancel_task)
         rt.server.route("exit_worker", self.h_exit_worker)
+        rt.server.route("stream_ack", self.h_stream_ack)
         r = await rt.raylet.call(
             "register_worker", {"pid": os.getpid(), "addr": rt.addr}
         )
@@ -123,6 +126,14 @@ class WorkerMain:
         while rt.raylet.connected:
             await asyncio.sleep(0.5)
         sys.exit(0)
+
+    def h_stream_ack(self, conn, p):
+        """Consumer progress for a streaming generator — wakes the
+        paused producer (reference: ReportGeneratorItemReturns acks)."""
+        st = self._stream_prod.get(bytes(p["task_id"]))
+        if st is not None:
+            st["acked"] = max(st["acked"], p["consumed"])
+            st["event"].set()
 
     # ------------- actor init -------------
 
@@ -319,10 +330,28 @@ class WorkerMain:
                     {"kind": "store", "node_addr": self.raylet_addr,
                      "size": size}
                 )
+            payload["addr"] = self.rt.addr  # ack channel back to us
             from .protocol import MSG_NOTIFY
 
             conn.send([MSG_NOTIFY, 0, "stream_item", payload])
             idx += 1
+            # producer backpressure (reference: generator_waiter.h:75):
+            # pause while too many items sit unconsumed
+            from .._config import config as _cfg
+
+            cap = _cfg.streaming_backpressure_items
+            if cap > 0:
+                st = self._stream_prod.setdefault(
+                    task_id, {"acked": 0, "event": asyncio.Event()}
+                )
+                waited = 0.0
+                while idx - st["acked"] >= cap and waited < 600.0:
+                    st["event"].clear()
+                    try:
+                        await asyncio.wait_for(st["event"].wait(), 1.0)
+                    except asyncio.TimeoutError:
+                        waited += 1.0
+        self._stream_prod.pop(task_id, None)
         return {"status": "ok", "streaming_done": idx}
 
     # ------------- actor calls -------------
