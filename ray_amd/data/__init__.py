@@ -8,6 +8,7 @@ in-flight window provides the same backpressure effect for the v1
 pipeline).
 """
 from .dataset import (  # noqa: F401
+    ActorPoolStrategy,
     Dataset,
     DataContext,
     from_arrow,

@@ -196,12 +196,27 @@ def _rows_to_table(rows: List[dict]) -> pa.Table:
     return _to_table({k: np.asarray([r[k] for r in rows]) for k in keys})
 
 
+class ActorPoolStrategy:
+    """compute= strategy for map_batches: a fixed pool of UDF actors
+    (reference: data ActorPoolMapOperator / ActorPoolStrategy)."""
+
+    def __init__(self, size: Optional[int] = None,
+                 min_size: Optional[int] = None,
+                 max_size: Optional[int] = None):
+        self.size = int(size or max_size or min_size or 2)
+
+
 def _apply_ops(table_or_ref, ops: List[tuple]) -> pa.Table:
     t = table_or_ref
     for op in ops:
         kind = op[0]
         if kind == "map_batches":
             t = _op_map_batches(t, op[1], op[2], op[3])
+        elif kind == "actor_map":
+            # outside an actor pool (e.g. downstream task re-exec):
+            # instantiate the UDF class per process as a fallback
+            inst = op[1]() if isinstance(op[1], type) else op[1]
+            t = _op_map_batches(t, inst, op[2], op[3])
         elif kind == "map":
             t = _op_map(t, op[1])
         elif kind == "flat_map":
@@ -274,6 +289,7 @@ class Dataset:
         self._materialized: Optional[List[Any]] = None
         self._task_opts: Optional[dict] = None
         self._concurrency: Optional[int] = None  # per-op in-flight cap
+        self._actor_pool_size: Optional[int] = None
 
     # ----- plan builders -----
 
@@ -281,6 +297,7 @@ class Dataset:
         ds = Dataset(self._inputs, self._ops + [op])
         ds._task_opts = self._task_opts
         ds._concurrency = self._concurrency
+        ds._actor_pool_size = self._actor_pool_size
         return ds
 
     def map(self, fn: Callable[[dict], dict], **kwargs) -> "Dataset":
@@ -305,19 +322,19 @@ class Dataset:
             kw = dict(fn_kwargs or {})
             fn = lambda b: base(b, *a, **kw)  # noqa: E731
         if isinstance(fn, type):
-            # actor-pool UDF class: instantiate per task (v1; actor pools
-            # proper when concurrency tuning matters)
-            cls = fn
-
-            def fn(batch, _cls=cls):
-                inst_attr = "_ray_amd_udf_inst"
-                import threading
-
-                tl = getattr(_cls, inst_attr, None)
-                if tl is None:
-                    tl = _cls()
-                    setattr(_cls, inst_attr, tl)
-                return tl(batch)
+            # class UDF -> real actor pool (reference:
+            # ActorPoolMapOperator): one instance per pool actor
+            if isinstance(compute, ActorPoolStrategy):
+                size = compute.size
+            elif isinstance(concurrency, int) and concurrency > 0:
+                size = concurrency
+            elif isinstance(concurrency, (tuple, list)):
+                size = int(concurrency[-1])
+            else:
+                size = 2
+            ds = self._with_op(("actor_map", fn, batch_format, batch_size))
+            ds._actor_pool_size = max(1, size)
+            return ds
 
         op = ("map_batches", fn, batch_format, batch_size)
         if num_gpus or kwargs.get("num_cpus") or concurrency:
@@ -413,6 +430,9 @@ class Dataset:
         window = DataContext.get_current().streaming_read_window
         if self._concurrency:
             window = min(window, self._concurrency)
+        if any(op[0] == "actor_map" for op in ops):
+            yield from self._iter_actor_pool(ops, window)
+            return
 
         @ray.remote
         def _exec_block(block, ops=ops):
@@ -444,6 +464,59 @@ class Dataset:
             ref = pending.pop(0)
             refill()
             yield ref
+
+    def _iter_actor_pool(self, ops, window):
+        """Blocks flow through a fixed pool of UDF actors (stateful,
+        init-once); ordered results, windowed in-flight."""
+        ray = _ray()
+
+        @ray.remote
+        class _PoolWorker:
+            def __init__(self, ops):
+                self._insts = {}
+                self._ops = [
+                    (
+                        ("map_batches", self._inst(i, op[1]), op[2], op[3])
+                        if op[0] == "actor_map"
+                        else op
+                    )
+                    for i, op in enumerate(ops)
+                ]
+
+            def _inst(self, i, cls):
+                if i not in self._insts:
+                    self._insts[i] = cls() if isinstance(cls, type) else cls
+                return self._insts[i]
+
+            def process(self, block):
+                t = block() if callable(block) else block
+                return _apply_ops(t, self._ops)
+
+        import builtins
+
+        size = self._actor_pool_size or 2
+        # NB: plain `range` is shadowed by the Dataset range() API here
+        workers = [_PoolWorker.remote(ops) for _ in builtins.range(size)]
+        try:
+            pending = []
+            i = 0
+            inputs = iter(self._inputs)
+            for inp in itertools.islice(inputs, max(window, size)):
+                pending.append(workers[i % size].process.remote(inp))
+                i += 1
+            while pending:
+                ref = pending.pop(0)
+                nxt = next(inputs, None)
+                if nxt is not None:
+                    pending.append(workers[i % size].process.remote(nxt))
+                    i += 1
+                yield ref
+        finally:
+            for w in workers:
+                try:
+                    ray.kill(w)
+                except Exception:
+                    pass
 
     def _iter_tables(self) -> Iterator[pa.Table]:
         ray = _ray()

@@ -210,3 +210,34 @@ def test_map_batches_concurrency_cap(ray_start_regular):
         lambda b: b, concurrency=2
     )
     assert ds.count() == 64  # capped window still processes everything
+
+
+def test_map_batches_actor_pool(ray_start_regular, tmp_path):
+    """Class UDFs run on a fixed actor pool: init once per actor,
+    state persists across blocks (reference: ActorPoolMapOperator)."""
+    from ray_amd import data as ray_data
+    from ray_amd.data.dataset import ActorPoolStrategy
+
+    init_dir = tmp_path
+
+    class AddModel:
+        def __init__(self):
+            import os
+            import time as _t
+
+            with open(init_dir / f"init_{os.getpid()}_{_t.time_ns()}",
+                      "w") as f:
+                f.write("x")
+            self.seen = 0
+
+        def __call__(self, batch):
+            self.seen += 1
+            return {"v": batch["id"] + 1000}
+
+    ds = ray_data.range(80, override_num_blocks=8).map_batches(
+        AddModel, compute=ActorPoolStrategy(size=3)
+    )
+    out = ds.take_all()
+    assert sorted(r["v"] for r in out) == list(range(1000, 1080))
+    inits = list(init_dir.glob("init_*"))
+    assert len(inits) == 3  # one instance per pool actor, not per block
