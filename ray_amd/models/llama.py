@@ -271,6 +271,110 @@ def _rope_one(x, cos, sin):
     return torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1).to(x.dtype)
 
 
+class BatchedDecoder:
+    """Single-token decode for a BATCH of independent sequences at
+    per-slot positions — the substrate for continuous batching
+    (reference: serve LLM's vLLM engine loop). Same static-buffer
+    design as GraphedDecoder but pos/mask are per-row; eager execution
+    (hipGraph capture of the batched step lands next round)."""
+
+    def __init__(self, model: "LlamaModel", batch_size: int, max_T: int,
+                 device):
+        self.m = model
+        cfg = model.cfg
+        self.B = batch_size
+        self.max_T = max_T
+        self.device = device
+        hd = cfg.hidden_size // cfg.num_heads
+        dt = model.dtype
+        self.cache_k = [
+            torch.zeros(batch_size, cfg.num_kv_heads, max_T, hd,
+                        device=device, dtype=dt)
+            for _ in range(cfg.num_layers)
+        ]
+        self.cache_v = [torch.zeros_like(k) for k in self.cache_k]
+        self.mask = torch.full((batch_size, 1, 1, max_T), float("-inf"),
+                               device=device, dtype=torch.float32)
+        self._rows = torch.arange(batch_size, device=device)
+
+    def set_slot_len(self, slot: int, length: int):
+        self.mask[slot, ..., :length] = 0.0
+        self.mask[slot, ..., length:] = float("-inf")
+
+    def prefill_slot(self, slot: int, toks: torch.Tensor):
+        """Run the prompt through the model, filling this slot's cache
+        rows; returns last-position logits [V]."""
+        caches = [
+            _SlotCacheView(self.cache_k[li], self.cache_v[li], slot)
+            for li in range(self.m.cfg.num_layers)
+        ]
+        logits = self.m(toks.view(1, -1), kv_caches=caches, pos0=0)[0, -1]
+        self.set_slot_len(slot, toks.numel())
+        return logits
+
+    @torch.no_grad()
+    def decode(self, toks: torch.Tensor, pos: torch.Tensor):
+        """toks [B] last tokens, pos [B] their positions; returns
+        logits [B, V] (inactive slots produce garbage rows)."""
+        m = self.m
+        cfg = m.cfg
+        B = self.B
+        hd = cfg.hidden_size // cfg.num_heads
+        x = m.embed(toks.view(B, 1))
+        cos = m.cosT.index_select(0, pos).view(B, 1, 1, hd // 2)
+        sin = m.sinT.index_select(0, pos).view(B, 1, 1, hd // 2)
+        for li, layer in enumerate(m.layers):
+            h = layer.attn_norm(x)
+            at = layer.attn
+            q = at.q_proj(h).view(B, 1, cfg.num_heads, hd)
+            k = at.k_proj(h).view(B, 1, cfg.num_kv_heads, hd)
+            v = at.v_proj(h).view(B, 1, cfg.num_kv_heads, hd)
+            q = _rope_one(q, cos, sin)
+            k = _rope_one(k, cos, sin)
+            # per-row cache write at each slot's own position
+            self.cache_k[li][self._rows, :, pos] = (
+                k[:, 0].to(self.cache_k[li].dtype)
+            )
+            self.cache_v[li][self._rows, :, pos] = (
+                v[:, 0].to(self.cache_v[li].dtype)
+            )
+            attn = F.scaled_dot_product_attention(
+                q.permute(0, 2, 1, 3),
+                self.cache_k[li],
+                self.cache_v[li],
+                attn_mask=self.mask.to(q.dtype),
+                enable_gqa=True,
+            )
+            x = x + at.o_proj(attn.permute(0, 2, 1, 3).reshape(B, 1, -1))
+            x = x + layer.mlp(layer.mlp_norm(x))
+        x = m.final_norm(x)
+        return m.lm_head(x)[:, 0]
+
+
+class _SlotCacheView:
+    """KV-cache adapter targeting one slot of a BatchedDecoder."""
+
+    def __init__(self, k_buf, v_buf, slot: int):
+        self.k_buf = k_buf
+        self.v_buf = v_buf
+        self.slot = slot
+        self.len = 0
+
+    def update(self, k, v, pos0):
+        T = k.size(2)
+        self.k_buf[self.slot : self.slot + 1, :, pos0 : pos0 + T] = k.to(
+            self.k_buf.dtype
+        )
+        self.v_buf[self.slot : self.slot + 1, :, pos0 : pos0 + T] = v.to(
+            self.v_buf.dtype
+        )
+        self.len = pos0 + T
+        return (
+            self.k_buf[self.slot : self.slot + 1, :, : self.len],
+            self.v_buf[self.slot : self.slot + 1, :, : self.len],
+        )
+
+
 class KVCache:
     """Per-layer KV cache for decode, preallocated in HBM."""
 
