@@ -92,6 +92,7 @@ class WorkerMain:
         self._max_concurrency = 1
         # producer-side streaming backpressure: task_id -> {acked, event}
         self._stream_prod = {}
+        self._order_lock = None  # created on the loop in main()
 
     async def main(self):
         loop = asyncio.get_running_loop()
@@ -112,6 +113,7 @@ class WorkerMain:
         rt.server.route("cancel_task", self.h_cancel_task)
         rt.server.route("exit_worker", self.h_exit_worker)
         rt.server.route("stream_ack", self.h_stream_ack)
+        self._order_lock = asyncio.Lock()
         r = await rt.raylet.call(
             "register_worker", {"pid": os.getpid(), "addr": rt.addr}
         )
@@ -368,43 +370,57 @@ class WorkerMain:
             asyncio.get_running_loop().call_later(0.05, self._graceful_exit)
             return {"status": "ok", "results": [
                 {"kind": "inline", "data": serialization.dumps(None)}]}
-        try:
-            args, kwargs = await self._load_args(spec)
-            args, kwargs = await self._resolve_args(args, kwargs)
-            if method_name == "__ray_apply__":
-                # generic in-actor execution (reference: __ray_call__):
-                # first arg is fn(instance, *rest) — used by compiled
-                # DAG loops and debugging helpers
-                fn, args = args[0], args[1:]
-                import functools
-
-                method = functools.partial(fn, self.actor_instance)
-            else:
-                method = getattr(self.actor_instance, method_name)
-        except Exception:
-            return self._error_reply(spec, traceback.format_exc())
-
+        # Execution-ORDER guarantee (reference: actor tasks run in
+        # submission order): arg resolution awaits inside a FIFO lock so
+        # a later call whose args are ready first cannot jump the queue;
+        # the actual execution is awaited OUTSIDE the lock (the
+        # single-thread executor preserves the dispatch order for sync
+        # methods; async methods interleave by design).
         loop = asyncio.get_running_loop()
-        exit_after = False
-        if asyncio.iscoroutinefunction(method):
+        async with self._order_lock:
             try:
-                result = await method(*args, **kwargs)
+                args, kwargs = await self._load_args(spec)
+                args, kwargs = await self._resolve_args(args, kwargs)
+                if method_name == "__ray_apply__":
+                    # generic in-actor execution (reference: __ray_call__):
+                    # first arg is fn(instance, *rest) — used by compiled
+                    # DAG loops and debugging helpers
+                    fn, args = args[0], args[1:]
+                    import functools
+
+                    method = functools.partial(fn, self.actor_instance)
+                else:
+                    method = getattr(self.actor_instance, method_name)
+            except Exception:
+                return self._error_reply(spec, traceback.format_exc())
+
+            if asyncio.iscoroutinefunction(method):
+                pending = asyncio.ensure_future(method(*args, **kwargs))
+                is_async = True
+            else:
+
+                def _exec():
+                    try:
+                        return True, method(*args, **kwargs), False
+                    except SystemExit:
+                        return True, None, True
+                    except BaseException as e:  # noqa
+                        return False, e, False
+
+                pending = loop.run_in_executor(self.executor, _exec)
+                is_async = False
+
+        exit_after = False
+        if is_async:
+            try:
+                result = await pending
                 ok = True
             except SystemExit:
                 ok, result, exit_after = True, None, True
             except BaseException as e:  # noqa
                 ok, result = False, e
         else:
-
-            def _exec():
-                try:
-                    return True, method(*args, **kwargs), False
-                except SystemExit:
-                    return True, None, True
-                except BaseException as e:  # noqa
-                    return False, e, False
-
-            ok, result, exit_after = await loop.run_in_executor(self.executor, _exec)
+            ok, result, exit_after = await pending
         if ok and spec.get("streaming"):
             reply = await self._run_streaming(conn, spec, result)
         else:

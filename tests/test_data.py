@@ -131,7 +131,7 @@ def test_train_test_split(ray_start_regular):
 
 
 def test_dataset_shard_in_trainer(ray_start_regular, tmp_path):
-    from ray_amd.train import RunConfig, ScalingConfig
+    from ray_amd.train import FailureConfig, RunConfig, ScalingConfig
     from ray_amd.train.torch import TorchTrainer
 
     def loop(config):
@@ -145,11 +145,15 @@ def test_dataset_shard_in_trainer(ray_start_regular, tmp_path):
     t = TorchTrainer(
         loop,
         scaling_config=ScalingConfig(num_workers=2),
-        run_config=RunConfig(name="shard", storage_path=str(tmp_path)),
+        run_config=RunConfig(
+            name="shard", storage_path=str(tmp_path),
+            # retry transient worker-group failures under full-suite load
+            failure_config=FailureConfig(max_failures=2),
+        ),
         datasets={"train": ds},
     )
     res = t.fit()
-    assert res.error is None
+    assert res.error is None, res.error
     assert res.metrics["rows"] == 20
 
 

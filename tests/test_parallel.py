@@ -118,3 +118,56 @@ def test_tp_linear_two_ranks(ray_start_regular):
     ref = (x @ w1.T) @ w2.T
     np.testing.assert_allclose(y0, ref.numpy(), atol=1e-4, rtol=1e-4)
     np.testing.assert_allclose(y0, y1, atol=1e-6)
+
+
+def test_pipeline_parallel_matches_monolithic(ray_start_regular):
+    """GPipe fill-drain over stage actors == single-process training
+    (same init, same data, fp32 exact-ish)."""
+    import numpy as np
+    import torch
+
+    from ray_amd.parallel.pipeline import Pipeline
+
+    def stage0():
+        torch.manual_seed(0)
+        return torch.nn.Sequential(torch.nn.Linear(8, 32), torch.nn.Tanh())
+
+    def stage1():
+        torch.manual_seed(1)
+        return torch.nn.Linear(32, 1)
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(32, 8)).astype(np.float32)
+    Y = rng.normal(size=(32, 1)).astype(np.float32)
+
+    pipe = Pipeline([stage0, stage1], lr=0.05, num_microbatches=4)
+    pipe_losses = [pipe.step(X, Y) for _ in range(3)]
+
+    # monolithic reference with identical per-stage init
+    torch.manual_seed(0)
+    m0 = torch.nn.Sequential(torch.nn.Linear(8, 32), torch.nn.Tanh())
+    torch.manual_seed(1)
+    m1 = torch.nn.Linear(32, 1)
+    model = torch.nn.Sequential(m0, m1)
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    ref_losses = []
+    for _ in range(3):
+        losses = []
+        opt.zero_grad()
+        for xs, ys in zip(np.array_split(X, 4), np.array_split(Y, 4)):
+            out = model(torch.as_tensor(xs))
+            loss = torch.nn.functional.mse_loss(out, torch.as_tensor(ys))
+            (loss / 4).backward()
+            losses.append(float(loss.detach()))
+        opt.step()
+        ref_losses.append(sum(losses) / 4)
+
+    assert np.allclose(pipe_losses, ref_losses, atol=1e-5), (
+        pipe_losses, ref_losses
+    )
+    # weights match after 3 steps
+    st = pipe.state_dicts()
+    for k, v in m0.state_dict().items():
+        assert np.allclose(st[0][k], v.numpy(), atol=1e-5)
+    for k, v in m1.state_dict().items():
+        assert np.allclose(st[1][k], v.numpy(), atol=1e-5)
