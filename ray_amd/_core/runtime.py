@@ -1091,6 +1091,17 @@ class CoreRuntime:
         if pool is None:
             pool = self._pools[key] = _LeasePool(key, res, pg_key)
         lease = await self._acquire_lease(pool, spec)
+        # cancels that landed while we waited for the lease win here
+        # (reference: queued tasks are cancellable until dispatch)
+        if spec["returns"] and spec["returns"][0] in self._cancelled_returns:
+            self._cancelled_returns.discard(spec["returns"][0])
+            self._release_or_reuse(pool, lease)
+            return {
+                "status": "error",
+                "error": serialization.dumps(
+                    exc.TaskCancelledError(spec.get("name", ""))
+                ),
+            }
         lease.busy = True
         try:
             reply = await lease.client.call("push_task", spec)
