@@ -162,6 +162,38 @@ def main(argv=None):
     s.add_argument("--duration", type=float, default=2.0)
     s.set_defaults(fn=cmd_microbenchmark)
 
+    # ray_amd serve run module:app (reference: `serve run`)
+    sv = sub.add_parser("serve")
+    svsub = sv.add_subparsers(dest="serve_cmd", required=True)
+    sr = svsub.add_parser("run")
+    sr.add_argument("--address", default=None)
+    sr.add_argument("--name", default="default")
+    sr.add_argument("--route-prefix", default="/")
+    sr.add_argument("--port", type=int, default=8000)
+    sr.add_argument("--blocking", action="store_true")
+    sr.add_argument("import_path", help="module:app, e.g. my_app:app")
+
+    def _serve_run(args):
+        import importlib
+
+        import ray_amd as ray
+        from ray_amd import serve as _serve
+
+        ray.init(address=args.address, ignore_reinit_error=True)
+        mod_name, _, attr = args.import_path.partition(":")
+        sys.path.insert(0, os.getcwd())
+        app = getattr(importlib.import_module(mod_name), attr or "app")
+        _serve.run(app, name=args.name, route_prefix=args.route_prefix,
+                   port=args.port)
+        print(f"serving {args.import_path} at "
+              f"http://127.0.0.1:{args.port}{args.route_prefix}")
+        if args.blocking:
+            while True:
+                time.sleep(3600)
+        return 0
+
+    sr.set_defaults(fn=_serve_run)
+
     # ray_amd job submit|status|logs|stop|list (reference: `ray job ...`)
     j = sub.add_parser("job")
     jsub = j.add_subparsers(dest="job_cmd", required=True)

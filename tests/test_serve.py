@@ -316,3 +316,27 @@ def test_long_poll_config_push(serve_session):
         time.sleep(0.2)
     assert len(h._replicas) == 3  # pushed, not pulled on failure
     assert h.version.remote().result(timeout_s=30) == "v2"
+
+
+def test_serve_run_cli(ray_start_regular, tmp_path, monkeypatch):
+    """`ray_amd serve run module:app` deploys an import path."""
+    import httpx
+
+    from ray_amd.scripts import main as cli
+
+    app_file = tmp_path / "cli_app.py"
+    app_file.write_text(
+        "from ray_amd import serve\n"
+        "@serve.deployment\n"
+        "class Hello:\n"
+        "    def __call__(self, request):\n"
+        "        return {'msg': 'hi-from-cli'}\n"
+        "app = Hello.bind()\n"
+    )
+    monkeypatch.chdir(tmp_path)
+    rc = cli(["serve", "run", "--name", "cliapp", "--port", "18434",
+              "cli_app:app"])
+    assert rc == 0
+    r = httpx.get("http://127.0.0.1:18434/", timeout=30)
+    assert r.status_code == 200 and r.json()["msg"] == "hi-from-cli"
+    serve.shutdown()
