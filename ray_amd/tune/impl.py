@@ -200,6 +200,41 @@ class ASHAScheduler:
 # ---------------- config ----------------
 
 
+class MedianStoppingRule:
+    """Stop trials whose running mean falls below the median of all
+    trials' running means at the same iteration (reference:
+    tune/schedulers/median_stopping_rule.py)."""
+
+    def __init__(self, *, metric=None, mode: str = "max",
+                 grace_period: int = 4, min_samples_required: int = 3):
+        self.metric = metric
+        self.mode = mode
+        self.grace_period = grace_period
+        self.min_samples = min_samples_required
+        self._history: Dict[str, list] = {}
+
+    def on_trial_result(self, trial_id, iteration, metric_value) -> str:
+        if metric_value is None:
+            return "CONTINUE"
+        h = self._history.setdefault(trial_id, [])
+        h.append(float(metric_value))
+        if iteration < self.grace_period:
+            return "CONTINUE"
+        if len(self._history) < self.min_samples:
+            return "CONTINUE"
+        import statistics
+
+        means = [
+            sum(v) / len(v) for k, v in self._history.items() if v
+        ]
+        if len(means) < self.min_samples:
+            return "CONTINUE"
+        med = statistics.median(means)
+        mine = sum(h) / len(h)
+        worse = mine < med if self.mode == "max" else mine > med
+        return "STOP" if worse else "CONTINUE"
+
+
 class PopulationBasedTraining:
     """PBT (reference: tune/schedulers/pbt.py PopulationBasedTraining).
 

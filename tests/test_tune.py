@@ -166,3 +166,33 @@ def test_pbt_exploit_and_explore(ray_start_regular, tmp_path):
     assert sched.num_perturbations >= 1
     best = rg.get_best_result()
     assert best.metrics["score"] > 5.0
+
+
+def test_median_stopping_rule(ray_start_regular):
+    from ray_amd import tune
+
+    def trainable(config):
+        import time as _tm
+
+        for i in range(25):
+            tune.report({"score": config["base"] + i * config["slope"]})
+            _tm.sleep(0.03)
+
+    sched = tune.MedianStoppingRule(metric="score", mode="max",
+                                    grace_period=5)
+    rg = tune.Tuner(
+        trainable,
+        param_space={
+            "base": tune.grid_search([0.0, 0.0, 100.0, 100.0]),
+            "slope": tune.sample_from(lambda _: 1.0),
+        },
+        tune_config=tune.TuneConfig(metric="score", mode="max",
+                                    scheduler=sched,
+                                    max_concurrent_trials=4),
+    ).fit()
+    iters = sorted(r.metrics["training_iteration"] for r in rg
+                   if r.metrics)
+    # the two low-base trials get median-stopped well before 25 iters
+    assert iters[0] < 25 and iters[-1] == 25
+    best = rg.get_best_result()
+    assert best.metrics["score"] >= 100
