@@ -94,6 +94,24 @@ def summarize_tasks(**kwargs) -> dict:
     return by_name
 
 
+def summarize_actors(**kwargs) -> dict:
+    """Counts by class and state (reference: state summary API)."""
+    out = {}
+    for a in list_actors():
+        e = out.setdefault(a.get("class_name", "Actor"), {})
+        e[a["state"]] = e.get(a["state"], 0) + 1
+    return out
+
+
+def summarize_objects(**kwargs) -> dict:
+    rows = list_objects()
+    return {
+        "total_objects": sum(r["num_objects_in_store"] for r in rows),
+        "total_store_bytes": sum(r["store_used_bytes"] for r in rows),
+        "nodes": len(rows),
+    }
+
+
 def get_log(filename: str = None, node_id: str = None, tail: int = 1000,
             **kwargs):
     """Yield log lines from the session's log directory (reference:
