@@ -259,9 +259,18 @@ class CoreRuntime:
         # pub/sub: channel -> local callbacks (invoked on the loop thread)
         self._pubsub_cbs: Dict[str, list] = {}
 
+        # borrow protocol (reference: reference_counter.h borrowers /
+        # WaitForRefRemoved): owner side counts live borrowers per oid
+        # and defers the free until they release; borrower side reports
+        # first-acquire / last-release of refs it does not own.
+        self._borrows: Dict[bytes, int] = {}
+        self._pending_free: set = set()
+
         self.server.route("fetch_object", self._h_fetch_object)
         self.server.route("locate_object", self._h_locate_object)
         self.server.route("reconstruct_object", self._h_reconstruct_object)
+        self.server.route("borrow_add", self._h_borrow_add)
+        self.server.route("borrow_del", self._h_borrow_del)
         self.server.route("ping", lambda c, p: "pong")
 
     # ------------- lifecycle -------------
@@ -464,12 +473,16 @@ class CoreRuntime:
     # ------------- reference counting -------------
 
     def _add_local_ref(self, oid: bytes, owner_addr: str):
+        new = False
         with self._refs_lock:
             e = self._refs.get(oid)
             if e is None:
                 self._refs[oid] = [1, 0, owner_addr]
+                new = True
             else:
                 e[0] += 1
+        if new and owner_addr and self.addr and owner_addr != self.addr:
+            self._notify_owner(owner_addr, "borrow_add", oid)
 
     def _remove_local_ref(self, oid: bytes):
         with self._refs_lock:
@@ -483,6 +496,36 @@ class CoreRuntime:
             owner = e[2]
         if owner == self.addr:
             self._free_owned(oid)
+        elif owner and not self._closed:
+            self._notify_owner(owner, "borrow_del", oid)
+
+    def _notify_owner(self, owner_addr: str, method: str, oid: bytes):
+        async def _send():
+            try:
+                c = await self._conn(owner_addr)
+                c.notify(method, {"id": oid})
+            except Exception:
+                pass
+
+        try:
+            self._run(_send())
+        except Exception:
+            pass
+
+    def _h_borrow_add(self, conn, p):
+        oid = bytes(p["id"])
+        self._borrows[oid] = self._borrows.get(oid, 0) + 1
+
+    def _h_borrow_del(self, conn, p):
+        oid = bytes(p["id"])
+        n = self._borrows.get(oid, 0) - 1
+        if n <= 0:
+            self._borrows.pop(oid, None)
+            if oid in self._pending_free:
+                self._pending_free.discard(oid)
+                self._free_owned(oid, _force=True)
+        else:
+            self._borrows[oid] = n
 
     def _add_submitted_ref(self, oid: bytes):
         with self._refs_lock:
@@ -503,7 +546,12 @@ class CoreRuntime:
         if owner == self.addr:
             self._free_owned(oid)
 
-    def _free_owned(self, oid: bytes):
+    def _free_owned(self, oid: bytes, _force: bool = False):
+        if not _force and self._borrows.get(oid, 0) > 0:
+            # a borrower still holds this ref: defer the free until the
+            # last borrow_del arrives (reference: WaitForRefRemoved)
+            self._pending_free.add(oid)
+            return
         ent = self.memory_store.pop(oid, None)
         self._events.pop(oid, None)
         mapped = self._mmaps.pop(oid, None)

@@ -598,3 +598,47 @@ def test_streaming_generator_backpressure(ray_start_regular, tmp_path):
     # drain: everything arrives, in order
     rest = [ray.get(r, timeout=60) for r in it]
     assert rest == list(range(1, n))
+
+
+def test_borrow_protocol_defers_free(ray_start_regular):
+    """The owner defers freeing an object while a borrower (an actor
+    holding the ref) is alive (reference: WaitForRefRemoved)."""
+    import gc
+
+    from ray_amd._core import runtime as rtmod
+
+    @ray.remote
+    class Holder:
+        def hold(self, refs):
+            self.ref = refs[0]  # nested -> stays a ref (borrowed)
+            return True
+
+        def read(self):
+            return int(ray.get(self.ref)[123])
+
+        def drop(self):
+            del self.ref
+            import gc as _gc
+
+            _gc.collect()
+            return True
+
+    h = Holder.remote()
+    big = ray.put(np.arange(300_000, dtype=np.int64))
+    oid = big.id
+    assert ray.get(h.hold.remote([big]), timeout=30)
+    time.sleep(0.3)  # let the borrow_add notify land
+
+    rt = rtmod.global_runtime()
+    del big
+    gc.collect()
+    time.sleep(0.3)
+    # owner count dropped, but the borrow defers the free
+    assert oid in rt._pending_free or rt._borrows.get(oid, 0) > 0
+    assert ray.get(h.read.remote(), timeout=30) == 123  # still readable
+
+    assert ray.get(h.drop.remote(), timeout=30)
+    deadline = time.time() + 10
+    while time.time() < deadline and oid in rt.memory_store:
+        time.sleep(0.1)
+    assert oid not in rt.memory_store  # freed after the last release
