@@ -26,6 +26,7 @@ from .api import (  # noqa: F401
     get,
     get_actor,
     get_gpu_ids,
+    timeline,
     get_runtime_context,
     init,
     is_initialized,

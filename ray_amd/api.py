@@ -275,6 +275,13 @@ def _stage_runtime_env(runtime_env: Optional[dict]) -> Optional[dict]:
 
 def _normalize_scheduling(opts: dict) -> dict:
     opts = dict(opts)
+    acc = opts.pop("accelerator_type", None)
+    if acc:
+        # reference: accelerator_type=X adds a tiny demand on the
+        # node's accelerator_type:X resource (util/accelerators)
+        res = dict(opts.get("resources") or {})
+        res[f"accelerator_type:{acc}"] = 0.001
+        opts["resources"] = res
     strat = opts.get("scheduling_strategy")
     if strat is not None and hasattr(strat, "placement_group"):
         pg = strat.placement_group
@@ -556,6 +563,28 @@ def available_resources() -> Dict[str, float]:
         for k, v in n["Available"].items():
             total[k] = total.get(k, 0) + v
     return total
+
+
+def timeline(filename: Optional[str] = None) -> Optional[list]:
+    """Chrome-trace task events (reference: ray.timeline); returns the
+    event list, and writes JSON when filename is given."""
+    import json as _json
+
+    from .util import state as _state
+
+    events = []
+    for t in _state.list_tasks():
+        events.append({
+            "cat": "task", "name": t["name"], "ph": "X",
+            "ts": t["start_time_ms"] * 1000,
+            "dur": (t["end_time_ms"] - t["start_time_ms"]) * 1000,
+            "pid": t["worker_pid"], "tid": t["worker_pid"],
+        })
+    if filename:
+        with open(filename, "w") as f:
+            _json.dump(events, f)
+        return None
+    return events
 
 
 def get_gpu_ids() -> List[int]:

@@ -642,3 +642,27 @@ def test_borrow_protocol_defers_free(ray_start_regular):
     while time.time() < deadline and oid in rt.memory_store:
         time.sleep(0.1)
     assert oid not in rt.memory_store  # freed after the last release
+
+
+def test_accelerator_type_and_timeline(ray_start_regular, tmp_path):
+    from ray_amd import experimental as exp
+    from ray_amd.util.accelerators import AMD_INSTINCT_MI355X
+
+    # accelerator_type maps to the node's accelerator resource
+    exp.set_resource(f"accelerator_type:{AMD_INSTINCT_MI355X}", 1)
+    time.sleep(0.3)
+
+    @ray.remote(accelerator_type=AMD_INSTINCT_MI355X)
+    def on_mi355x():
+        return "scheduled"
+
+    assert ray.get(on_mi355x.remote(), timeout=30) == "scheduled"
+
+    # ray.timeline: chrome-trace events incl. the task above
+    evs = ray.timeline()
+    assert any(e["name"] == "on_mi355x" for e in evs)
+    out = tmp_path / "tl.json"
+    ray.timeline(str(out))
+    import json
+
+    assert json.loads(out.read_text())
