@@ -658,8 +658,15 @@ def test_accelerator_type_and_timeline(ray_start_regular, tmp_path):
 
     assert ray.get(on_mi355x.remote(), timeout=30) == "scheduled"
 
-    # ray.timeline: chrome-trace events incl. the task above
-    evs = ray.timeline()
+    # ray.timeline: chrome-trace events incl. the task above (events
+    # batch through raylet -> GCS; poll briefly)
+    deadline = time.time() + 15
+    evs = []
+    while time.time() < deadline:
+        evs = ray.timeline()
+        if any(e["name"] == "on_mi355x" for e in evs):
+            break
+        time.sleep(0.3)
     assert any(e["name"] == "on_mi355x" for e in evs)
     out = tmp_path / "tl.json"
     ray.timeline(str(out))
