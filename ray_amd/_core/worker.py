@@ -114,6 +114,7 @@ class WorkerMain:
         rt.server.route("exit_worker", self.h_exit_worker)
         rt.server.route("stream_ack", self.h_stream_ack)
         self._order_lock = asyncio.Lock()
+        asyncio.ensure_future(self._event_flusher())
         r = await rt.raylet.call(
             "register_worker", {"pid": os.getpid(), "addr": rt.addr}
         )
@@ -499,6 +500,20 @@ class WorkerMain:
                     {"kind": "store", "node_addr": self.raylet_addr, "size": size}
                 )
         return {"status": "ok", "results": results}
+
+    async def _event_flusher(self):
+        # small batches flush on a timer so timeline/state views see
+        # short-lived tasks too (the 100-event flush alone starves them)
+        while True:
+            await asyncio.sleep(1.0)
+            if self._events:
+                try:
+                    self.rt.raylet.notify(
+                        "report_task_events", {"events": self._events}
+                    )
+                    self._events = []
+                except Exception:
+                    pass
 
     def _record_event(self, spec, t0, ok, method=None):
         self._events.append(
