@@ -11,7 +11,7 @@ from __future__ import annotations
 import asyncio
 import time
 import traceback
-from typing import Any, Dict, List, Optional
+from typing import Dict, List, Optional
 
 from . import ids
 from .protocol import RpcClient, RpcServer

@@ -20,7 +20,6 @@ from __future__ import annotations
 
 import os
 import threading
-import uuid
 from typing import Dict, Optional, Tuple
 
 import cloudpickle

@@ -4,7 +4,6 @@ release/perf_metrics/microbenchmark.json so results compare 1:1 with
 BASELINE.md."""
 from __future__ import annotations
 
-import asyncio
 import time
 from typing import Callable, List, Tuple
 

@@ -13,7 +13,7 @@ import asyncio
 import random
 import time
 from dataclasses import dataclass, field
-from typing import Any, Callable, Dict, List, Optional
+from typing import Callable, Dict, List, Optional
 
 # starlette State instances (FastAPI app.state) break pickle: their
 # __getattr__ recurses during reconstruction before _state exists.

@@ -7,7 +7,6 @@ import os
 import queue
 import shutil
 import threading
-import uuid
 from typing import Any, Dict, Optional
 
 from .checkpoint import Checkpoint
