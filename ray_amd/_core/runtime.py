@@ -263,6 +263,8 @@ class CoreRuntime:
         # WaitForRefRemoved): owner side counts live borrowers per oid
         # and defers the free until they release; borrower side reports
         # first-acquire / last-release of refs it does not own.
+        # Known limit (r2): a borrower that dies without releasing
+        # leaks its count — per-borrower tracking + death sweep needed.
         self._borrows: Dict[bytes, int] = {}
         self._pending_free: set = set()
 
