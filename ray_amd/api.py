@@ -254,6 +254,11 @@ def _stage_runtime_env(runtime_env: Optional[dict]) -> Optional[dict]:
     import shutil as _sh
 
     rt = _rt.global_runtime()
+    if getattr(rt, "is_client", False):
+        raise NotImplementedError(
+            "runtime_env working_dir/py_modules over Ray-Client mode "
+            "needs a file upload channel (next round); env_vars work"
+        )
     out = dict(runtime_env)
     base = os.path.join(rt.session_dir, "runtime_env")
     os.makedirs(base, exist_ok=True)
