@@ -424,6 +424,28 @@ class Tuner:
         self.param_space = param_space or {}
         self.tune_config = tune_config or TuneConfig()
         self.run_config = run_config or RunConfig(name=f"tune_{int(time.time())}")
+        self._restored: Optional[dict] = None  # Tuner.restore state
+
+    @classmethod
+    def restore(cls, path: str, trainable=None) -> "Tuner":
+        """Resume an interrupted experiment (reference: Tuner.restore):
+        finished trials load from disk, unfinished ones re-run."""
+        import cloudpickle
+
+        with open(os.path.join(path, "tuner_state.pkl"), "rb") as f:
+            state = cloudpickle.loads(f.read())
+        t = cls(
+            trainable if trainable is not None else
+            cloudpickle.loads(state["trainable"]),
+            param_space={},  # variants come from the saved state
+            tune_config=state["tune_config"],
+            run_config=RunConfig(
+                name=os.path.basename(path.rstrip("/")),
+                storage_path=os.path.dirname(path.rstrip("/")),
+            ),
+        )
+        t._restored = state
+        return t
 
     def fit(self) -> ResultGrid:
         import cloudpickle
@@ -440,14 +462,46 @@ class Tuner:
         if isinstance(self._trainable, DataParallelTrainer):
             return self._fit_trainer_trials(storage)
 
-        variants = generate_variants(self.param_space, tc.num_samples)
+        done_names = set()
+        restored_results: List[Result] = []
+        if self._restored is not None:
+            variants = self._restored["variants"]
+        else:
+            variants = generate_variants(self.param_space, tc.num_samples)
+        fn_bytes = cloudpickle.dumps(self._trainable)
+        # persist experiment state so Tuner.restore can resume it
+        with open(os.path.join(storage, "tuner_state.pkl"), "wb") as f:
+            f.write(cloudpickle.dumps({
+                "variants": variants,
+                "tune_config": tc,
+                "trainable": fn_bytes,
+            }))
+        if self._restored is not None:
+            import json as _json
+
+            for i in range(len(variants)):
+                rp = os.path.join(storage, f"trial_{i:05d}", "result.json")
+                if os.path.exists(rp):
+                    with open(rp) as f:
+                        row = _json.load(f)
+                    if row.get("error"):
+                        continue  # re-run failed trials
+                    done_names.add(i)
+                    restored_results.append(Result(
+                        metrics=row.get("metrics"),
+                        checkpoint=(Checkpoint(row["ckpt"])
+                                    if row.get("ckpt") else None),
+                        path=os.path.join(storage, f"trial_{i:05d}"),
+                        error=None,
+                    ))
         scheduler = tc.scheduler or FIFOScheduler()
         max_conc = tc.max_concurrent_trials or min(8, max(1, len(variants)))
-        fn_bytes = cloudpickle.dumps(self._trainable)
         Actor = ray.remote(_TrialActor)
 
         trials = []  # dicts: actor, config, rows, done, error, it
-        pending = list(enumerate(variants))
+        pending = [
+            (i, cfg) for i, cfg in enumerate(variants) if i not in done_names
+        ]
         running: List[dict] = []
         finished: List[dict] = []
         t_start = time.time()
@@ -533,7 +587,9 @@ class Tuner:
                 for t in running:
                     t["actor"].request_stop.remote()
 
-        results = []
+        results = list(restored_results)
+        import json as _json
+
         for t in finished:
             metrics = t["rows"][-1] if t["rows"] else None
             results.append(
@@ -544,6 +600,16 @@ class Tuner:
                     error=RuntimeError(t["error"]) if t["error"] else None,
                 )
             )
+            # per-trial completion record (drives Tuner.restore)
+            try:
+                with open(os.path.join(storage, t["name"],
+                                       "result.json"), "w") as f:
+                    _json.dump({
+                        "metrics": metrics, "ckpt": t["ckpt"],
+                        "error": t["error"],
+                    }, f, default=str)
+            except OSError:
+                pass
         return ResultGrid(results, tc.metric, tc.mode)
 
     def _pbt_exploit(self, ray, scheduler, t, donor, fn_bytes, storage):

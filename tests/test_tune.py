@@ -197,3 +197,45 @@ def test_median_stopping_rule(ray_start_regular):
     assert iters[0] < 50, iters
     best = rg.get_best_result()
     assert best.metrics["score"] >= 100
+
+
+def test_tuner_restore_resumes_unfinished(ray_start_regular, tmp_path):
+    """Tuner.restore re-runs only failed/missing trials; finished ones
+    load from disk (reference: Tuner.restore)."""
+    from ray_amd.train import RunConfig
+
+    flag = tmp_path / "fixed"
+    runs = tmp_path / "runs"
+    runs.mkdir()
+
+    def trainable(config, flag=str(flag), runs=str(runs)):
+        import os
+        import time as _tm
+
+        with open(os.path.join(runs, f"run_{config['x']}_{_tm.time_ns()}"),
+                  "w") as f:
+            f.write("x")
+        if config["x"] == 3 and not os.path.exists(flag):
+            raise RuntimeError("transient failure")
+        tune.report({"score": config["x"] * 10})
+
+    space = {"x": tune.grid_search([1, 2, 3, 4])}
+    exp_dir = str(tmp_path / "exp")
+    tuner = Tuner(
+        trainable, param_space=space,
+        tune_config=TuneConfig(metric="score", mode="max"),
+        run_config=RunConfig(name="exp", storage_path=str(tmp_path)),
+    )
+    rg = tuner.fit()
+    assert len(rg.errors) == 1
+    n_first = len(list(runs.iterdir()))
+    assert n_first == 4
+
+    flag.write_text("ok")
+    rg2 = Tuner.restore(exp_dir).fit()
+    assert len(rg2.errors) == 0
+    assert sorted(r.metrics["score"] for r in rg2 if r.metrics) == [
+        10, 20, 30, 40
+    ]
+    # only the failed trial re-ran
+    assert len(list(runs.iterdir())) == n_first + 1
