@@ -476,15 +476,35 @@ class ServeController:
                 elif desired < n and now - d["last_scale"] > asc.get(
                     "downscale_delay_s", 30.0
                 ):
-                    for r in d["replicas"][desired:]:
-                        try:
-                            ray.kill(r)
-                        except Exception:
-                            pass
+                    # graceful drain (reference: replica graceful
+                    # shutdown): pull the victims out of routing first
+                    # (version bump -> long-poll push), kill them only
+                    # once their in-flight requests hit zero
+                    victims = d["replicas"][desired:]
                     d["replicas"] = d["replicas"][:desired]
+                    d.setdefault("draining", []).extend(victims)
                     d["last_scale"] = now
                     self.version += 1
+                self._reap_drained(d)
         return self.version
+
+    def _reap_drained(self, d):
+        ray = _ray()
+        still = []
+        for r in d.get("draining", []):
+            try:
+                ongoing = ray.get(r.get_stats.remote(),
+                                  timeout=10)["ongoing"]
+            except Exception:
+                continue  # already dead
+            if ongoing <= 0:
+                try:
+                    ray.kill(r)
+                except Exception:
+                    pass
+            else:
+                still.append(r)
+        d["draining"] = still
 
     def ping(self):
         return "pong"

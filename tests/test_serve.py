@@ -340,3 +340,49 @@ def test_serve_run_cli(ray_start_regular, tmp_path, monkeypatch):
     r = httpx.get("http://127.0.0.1:18434/", timeout=30)
     assert r.status_code == 200 and r.json()["msg"] == "hi-from-cli"
     serve.shutdown()
+
+
+def test_autoscale_down_graceful_drain(serve_session):
+    """Downscale removes replicas from routing first and kills them
+    only once their in-flight requests drain."""
+    @serve.deployment(
+        autoscaling_config={"min_replicas": 1, "max_replicas": 3,
+                            "target_ongoing_requests": 1,
+                            "downscale_delay_s": 0.0},
+    )
+    class Slow:
+        async def __call__(self, delay=0.0):
+            import asyncio
+
+            await asyncio.sleep(delay)
+            return "ok"
+
+    h = serve.run(Slow.bind(), name="drain", http=False)
+    ctrl = ray.get_actor(
+        serve.api.SERVE_CONTROLLER_NAME, namespace=serve.api.SERVE_NAMESPACE
+    )
+    # scale up under load
+    resps = [h.remote(1.0) for _ in range(6)]
+    time.sleep(0.2)
+    ray.get(ctrl.autoscale_once.remote(), timeout=60)
+    for r in resps:
+        assert r.result(timeout_s=60) == "ok"
+
+    # keep ONE long request in flight on some replica, then downscale
+    slow = h.remote(3.0)
+    time.sleep(0.3)
+    ray.get(ctrl.autoscale_once.remote(), timeout=60)  # moves to draining
+    # the long request survives the downscale (drain, not kill)
+    assert slow.result(timeout_s=60) == "ok"
+    # subsequent passes reap the drained replicas
+    deadline = time.time() + 20
+    while time.time() < deadline:
+        ray.get(ctrl.autoscale_once.remote(), timeout=60)
+        st = serve.status()
+        n = st["drain"]["deployments"]["Slow"]["replica_states"]["RUNNING"]
+        if n == 1:
+            break
+        time.sleep(0.3)
+    assert n == 1
+    # service still healthy
+    assert h.remote(0.0).result(timeout_s=30) == "ok"
