@@ -98,3 +98,36 @@ def test_client_tasks_objects_actors(client_server):
         assert len(ray.nodes()) == 1
     finally:
         ray.shutdown()
+
+
+def test_client_mode_contracts(client_server):
+    """Client-mode edge contracts: streaming is explicitly unsupported,
+    runtime_env env_vars pass through, working_dir staging errors
+    clearly."""
+    ctx = ray.init(f"ray_amd://127.0.0.1:{client_server}")
+    try:
+        @ray.remote(num_returns="streaming")
+        def gen():
+            yield 1
+
+        with pytest.raises(NotImplementedError):
+            gen.remote()
+
+        @ray.remote
+        def read_env():
+            import os
+
+            return os.environ.get("CLIENT_FLAG")
+
+        # env_vars runtime_env works over the client
+        r = read_env.options(
+            runtime_env={"env_vars": {"CLIENT_FLAG": "on"}}
+        ).remote()
+        assert ray.get(r, timeout=60) == "on"
+
+        with pytest.raises(NotImplementedError):
+            read_env.options(
+                runtime_env={"working_dir": "."}
+            ).remote()
+    finally:
+        ray.shutdown()

@@ -673,3 +673,14 @@ def test_accelerator_type_and_timeline(ray_start_regular, tmp_path):
     import json
 
     assert json.loads(out.read_text())
+
+
+def test_pubsub_multi_subscriber_fanout(ray_start_regular):
+    from ray_amd.util import pubsub
+
+    with pubsub.Subscriber("fan") as a, pubsub.Subscriber("fan") as b:
+        # one GCS connection per process: publish counts conns, every
+        # local subscriber still receives
+        assert pubsub.publish("fan", 7) >= 1
+        assert a.poll(timeout=10) == 7
+        assert b.poll(timeout=10) == 7
