@@ -37,14 +37,16 @@ class WorkerProc:
         self.idle = True
         self.lease_id: Optional[int] = None
         self.gpu_ids: List[int] = []
+        self.gpu_alloc: List[tuple] = []  # [(device, fraction)]
 
 
 class Lease:
-    def __init__(self, lease_id, resources, worker, gpu_ids, pg=None):
+    def __init__(self, lease_id, resources, worker, gpu_alloc, pg=None):
         self.lease_id = lease_id
         self.resources = resources
         self.worker: WorkerProc = worker
-        self.gpu_ids = gpu_ids
+        self.gpu_alloc = gpu_alloc  # [(device, fraction)]
+        self.gpu_ids = [d for d, _ in gpu_alloc]
         self.pg = pg
 
 
@@ -83,9 +85,12 @@ class Raylet:
         self._worker_ready: Dict[int, asyncio.Future] = {}  # pid -> fut
         self._actor_start_futs: Dict[bytes, asyncio.Future] = {}
         self._proto_worker: Dict[int, WorkerProc] = {}
-        # GPU instance pool
+        # GPU instance pool with fractional occupancy per device
+        # (reference: NodeResourceInstanceSet per-instance GPU accounting,
+        # common/scheduling/resource_instance_set.h). free fraction in
+        # [0,1] per device id; num_gpus=0.5 shares one device.
         ngpus = int(resources.get("GPU", 0))
-        self._free_gpus = list(range(ngpus))
+        self._gpu_free: Dict[int, float] = {i: 1.0 for i in range(ngpus)}
         # placement-group bundles: (pg_id, idx) -> {"resources", "avail", "committed"}
         self.bundles: Dict[tuple, dict] = {}
         self._node_cache: List[dict] = []
@@ -222,8 +227,7 @@ class Raylet:
         if w.kind == "actor" and w.actor_id is not None:
             spec_res = getattr(w, "actor_resources", None) or {}
             bundle_key = getattr(w, "actor_bundle", None)
-            for g in w.gpu_ids:
-                self._free_gpus.append(g)
+            self._gpu_release(w.gpu_alloc)
             if bundle_key is not None:
                 b = self.bundles.get(bundle_key)
                 if b is not None:
@@ -299,6 +303,42 @@ class Raylet:
                 return key
         return None
 
+    # ---------------- GPU instance pool ----------------
+
+    _GPU_EPS = 1e-6
+
+    def _gpu_alloc(self, ngpu: float) -> Optional[List[tuple]]:
+        """Assign device ids for a GPU request. Integer requests take
+        whole free devices; a fractional request (0 < ngpu < 1) shares
+        one device, best-fit packed onto the fullest device that still
+        has room. Returns [(device, fraction)] or None if the request
+        cannot be satisfied right now (caller must queue, never
+        silently under-assign)."""
+        if ngpu <= 0:
+            return []
+        eps = self._GPU_EPS
+        if ngpu < 1.0 - eps:
+            fit = [d for d, f in self._gpu_free.items() if f >= ngpu - eps]
+            if not fit:
+                return None
+            # pack partially-used devices first to keep whole GPUs free
+            dev = min(fit, key=lambda d: self._gpu_free[d])
+            self._gpu_free[dev] -= ngpu
+            return [(dev, ngpu)]
+        n = int(round(ngpu))
+        whole = [d for d, f in self._gpu_free.items() if f >= 1.0 - eps]
+        if len(whole) < n:
+            return None
+        out = []
+        for d in whole[:n]:
+            self._gpu_free[d] = 0.0
+            out.append((d, 1.0))
+        return out
+
+    def _gpu_release(self, alloc: List[tuple]):
+        for d, frac in alloc:
+            self._gpu_free[d] = min(1.0, self._gpu_free.get(d, 0.0) + frac)
+
     def _try_grant(self):
         made_progress = True
         while made_progress and self._pending:
@@ -338,28 +378,30 @@ class Raylet:
             if w.proc.poll() is not None or w.proto is None:
                 made_progress = True
                 continue
+            ngpu = float(req.get("GPU", 0))
+            gpu_alloc = self._gpu_alloc(ngpu)
+            if gpu_alloc is None:
+                # resource accounting says it fits but no device has the
+                # fraction free (fragmentation) — keep the request queued
+                self._idle_task_workers.appendleft(w)
+                break
             self._pending.popleft()
-            gpu_ids = []
-            ngpu = int(req.get("GPU", 0))
             if bundle is None:
                 for k, v in req.items():
                     self.avail[k] = self.avail.get(k, 0) - v
-                if ngpu:
-                    gpu_ids = [self._free_gpus.pop() for _ in range(min(ngpu, len(self._free_gpus)))]
             else:
                 for k, v in req.items():
                     bundle["avail"][k] = bundle["avail"].get(k, 0) - v
-                if ngpu:
-                    gpu_ids = [self._free_gpus.pop() for _ in range(min(ngpu, len(self._free_gpus)))]
             self._lease_seq += 1
-            lease = Lease(self._lease_seq, req, w, gpu_ids, pg)
+            lease = Lease(self._lease_seq, req, w, gpu_alloc, pg)
             self.leases[lease.lease_id] = lease
             w.idle = False
             w.lease_id = lease.lease_id
-            w.gpu_ids = gpu_ids
+            w.gpu_ids = lease.gpu_ids
+            w.gpu_alloc = gpu_alloc
             fut.set_result(
-                {"addr": w.addr, "lease_id": lease.lease_id, "gpu_ids": gpu_ids,
-                 "raylet": self.addr}
+                {"addr": w.addr, "lease_id": lease.lease_id,
+                 "gpu_ids": lease.gpu_ids, "raylet": self.addr}
             )
             made_progress = True
 
@@ -373,8 +415,7 @@ class Raylet:
         else:
             for k, v in lease.resources.items():
                 self.avail[k] = self.avail.get(k, 0) + v
-        for g in lease.gpu_ids:
-            self._free_gpus.append(g)
+        self._gpu_release(lease.gpu_alloc)
 
     def h_return_lease(self, conn, p):
         lease = self.leases.pop(p["lease_id"], None)
@@ -433,8 +474,17 @@ class Raylet:
                 bundle["avail"][k] = bundle["avail"].get(k, 0) - v
         elif not self._fits(self.avail, res):
             raise RuntimeError(f"node {self.node_name}: insufficient resources {res}")
-        ngpu = int(res.get("GPU", 0))
-        gpu_ids = [self._free_gpus.pop() for _ in range(min(ngpu, len(self._free_gpus)))]
+        ngpu = float(res.get("GPU", 0))
+        gpu_alloc = self._gpu_alloc(ngpu)
+        if gpu_alloc is None:
+            if bundle_key is not None:
+                for k, v in res.items():
+                    bundle["avail"][k] = bundle["avail"].get(k, 0) + v
+            raise RuntimeError(
+                f"node {self.node_name}: no GPU device has {ngpu} free "
+                "(fragmented fractional occupancy)"
+            )
+        gpu_ids = [d for d, _ in gpu_alloc]
         if bundle_key is None:
             for k, v in res.items():
                 self.avail[k] = self.avail.get(k, 0) - v
@@ -445,6 +495,7 @@ class Raylet:
             spec["env_vars"]["RAY_AMD_GPU_IDS"] = ids_str
         w = self._spawn_worker(actor_spec=spec)
         w.gpu_ids = gpu_ids
+        w.gpu_alloc = gpu_alloc
         w.actor_resources = res
         w.actor_bundle = bundle_key
         fut = asyncio.get_running_loop().create_future()

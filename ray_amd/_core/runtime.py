@@ -171,13 +171,14 @@ class ObjectRefGenerator:
 
 
 class _Lease:
-    __slots__ = ("lease_id", "addr", "client", "busy")
+    __slots__ = ("lease_id", "addr", "client", "busy", "gpu_ids")
 
-    def __init__(self, lease_id, addr, client):
+    def __init__(self, lease_id, addr, client, gpu_ids=()):
         self.lease_id = lease_id
         self.addr = addr
         self.client = client
         self.busy = False
+        self.gpu_ids = list(gpu_ids)
 
 
 class _LeasePool:
@@ -933,8 +934,18 @@ class CoreRuntime:
             if ent is None:
                 if r.owner_addr == self.addr:
                     await self._store_wait(r.id, None)
-                else:
+                elif fetch_local:
                     await self._fetch_from_owner(r, None)
+                else:
+                    # fetch_local=False: only learn that the object
+                    # exists somewhere; poll the owner's location table
+                    # instead of resolving it into our memory store.
+                    c = await self._conn(r.owner_addr)
+                    while True:
+                        loc = await c.call("locate_object", {"id": r.id})
+                        if loc is not None:
+                            break
+                        await asyncio.sleep(0.02)
             return r
 
         tasks = {asyncio.ensure_future(_ready_one(r)): r for r in pending}
@@ -949,13 +960,35 @@ class CoreRuntime:
                     break
                 for t in done:
                     r = tasks.pop(t)
+                    e = t.exception()
+                    if e is None:
+                        ready.append(r)
+                        continue
+                    # A ref whose value IS an error object counts as
+                    # ready (reference semantics: ray.wait readies refs
+                    # holding exceptions); _ready_one stores error
+                    # objects and returns normally for those, so an
+                    # exception here is a transport/ownership failure —
+                    # surface it as a ready error object rather than
+                    # silently marking the ref ready with no value.
+                    self._store_put(
+                        r.id,
+                        ("err", serialization.dumps(
+                            exc.ObjectLostError(
+                                f"wait: failed to resolve {r.id.hex()}: {e}"
+                            )
+                        )),
+                    )
                     ready.append(r)
                 if deadline is not None and time.monotonic() >= deadline:
                     break
         finally:
             for t in tasks:
                 t.cancel()
-        not_ready = [r for r in refs if r not in ready]
+            # consume cancellation/exceptions so asyncio never logs
+            # "exception was never retrieved"
+            for t in tasks:
+                t.add_done_callback(lambda fut: fut.cancelled() or fut.exception())
         ready = ready[: max(num_returns, 0)]
         not_ready = [r for r in refs if r not in ready]
         return ready, not_ready
@@ -1153,6 +1186,8 @@ class CoreRuntime:
                 ),
             }
         lease.busy = True
+        if lease.gpu_ids:
+            spec["gpu_ids"] = lease.gpu_ids
         try:
             reply = await lease.client.call("push_task", spec)
             return reply
@@ -1197,7 +1232,8 @@ class CoreRuntime:
                 if r.get("error"):
                     raise exc.RaySystemError(r["error"])
                 client = await self._conn(r["addr"])
-                lease = _Lease((r["lease_id"], r.get("raylet", raylet.addr)), r["addr"], client)
+                lease = _Lease((r["lease_id"], r.get("raylet", raylet.addr)),
+                               r["addr"], client, r.get("gpu_ids") or ())
                 pool.leases.append(lease)
                 self._grant_to_queue(pool, lease)
                 return

@@ -239,12 +239,22 @@ class WorkerMain:
         if spec.get("env_vars"):
             os.environ.update({str(k): str(v) for k, v in spec["env_vars"].items()})
         _apply_code_env(spec)
+        lease_gpus = [int(g) for g in (spec.get("gpu_ids") or [])]
+        if lease_gpus and "RAY_AMD_GPU_IDS" not in os.environ:
+            # Like the reference (worker.py set_gpu_ids), device
+            # visibility is set per-lease in a generic pooled worker;
+            # effective only if HIP has not initialized here yet.
+            ids_str = ",".join(map(str, lease_gpus))
+            os.environ["RAY_AMD_GPU_IDS"] = ids_str
+            os.environ.setdefault("CUDA_VISIBLE_DEVICES", ids_str)
+            os.environ.setdefault("HIP_VISIBLE_DEVICES", ids_str)
 
         loop = asyncio.get_running_loop()
 
         def _exec():
             _task_ctx.task_id = task_id
             _task_ctx.task_name = spec.get("name")
+            _task_ctx.gpu_ids = lease_gpus
             try:
                 return True, fn(*args, **kwargs)
             except BaseException as e:  # noqa
@@ -318,7 +328,13 @@ class WorkerMain:
                 return {"status": "error", "error": data}
             if not more:
                 break
-            oid = task_id + struct.pack("<I", idx)
+            # Stream-item ids live in the same namespace as normal object
+            # ids (nonce8 + LE u64 seq). Tag the top bit of the final byte
+            # so item ids can never collide with a minted object id (a
+            # normal id would need obj_seq >= 2^63): task_id is
+            # nonce8+low4(task_seq), so without the tag item 0 of task N is
+            # byte-identical to the object with obj_seq == task_seq(N).
+            oid = task_id + struct.pack("<I", 0x80000000 | idx)
             meta, buffers, _ = self.rt._serialize_capture(item)
             size = serialization.serialized_size(meta, buffers)
             payload = {"task_id": task_id, "oid": oid}

@@ -311,10 +311,11 @@ def _normalize_scheduling(opts: dict) -> dict:
 
 
 class ActorMethod:
-    def __init__(self, handle: "ActorHandle", name: str, num_returns=1):
+    def __init__(self, handle: "ActorHandle", name: str, defaults=None):
         self._handle = handle
         self._name = name
-        self._num_returns = num_returns
+        # per-method defaults from @ray.method(...) on the class
+        self._defaults = dict(defaults or {})
 
     def remote(self, *args, **kwargs):
         return self._remote(args, kwargs, {})
@@ -330,12 +331,14 @@ class ActorMethod:
 
     def _remote(self, args, kwargs, opts):
         rt = _rt.global_runtime()
-        num_returns = opts.get("num_returns", self._num_returns)
+        merged = {**self._defaults, **opts}
+        merged.setdefault("num_returns", 1)
+        num_returns = merged["num_returns"]
         refs = rt.submit_actor_task(
             self._handle._actor_id,
             self._name,
             (args, kwargs),
-            {"num_returns": num_returns},
+            merged,
         )
         if num_returns == "streaming":
             return refs
@@ -345,22 +348,25 @@ class ActorMethod:
 
 
 class ActorHandle:
-    def __init__(self, actor_id: bytes, class_name: str = "Actor"):
+    def __init__(self, actor_id: bytes, class_name: str = "Actor",
+                 method_options=None):
         self._actor_id = actor_id
         self._class_name = class_name
+        self._method_options = method_options or {}
 
     def __getattr__(self, item):
         if item.startswith("_") and item not in (
             "__ray_apply__", "__ray_terminate__"
         ):
             raise AttributeError(item)
-        return ActorMethod(self, item)
+        return ActorMethod(self, item, self._method_options.get(item))
 
     def __repr__(self):
         return f"Actor({self._class_name}, {self._actor_id.hex()})"
 
     def __reduce__(self):
-        return (_deserialize_handle, (self._actor_id, self._class_name))
+        return (_deserialize_handle,
+                (self._actor_id, self._class_name, self._method_options))
 
     def __hash__(self):
         return hash(self._actor_id)
@@ -374,8 +380,8 @@ class ActorHandle:
         return self._actor_id
 
 
-def _deserialize_handle(actor_id, class_name):
-    return ActorHandle(actor_id, class_name)
+def _deserialize_handle(actor_id, class_name, method_options=None):
+    return ActorHandle(actor_id, class_name, method_options)
 
 
 class ActorClass:
@@ -417,7 +423,24 @@ class ActorClass:
         mc = opts.get("max_concurrency", 1)
         pickled, key = self._ensure_pickled(mc)
         actor_id = rt.create_actor(key, pickled, opts, (args, kwargs))
-        return ActorHandle(actor_id, self._cls.__name__)
+        return ActorHandle(actor_id, self._cls.__name__,
+                           self._method_options())
+
+    def _method_options(self) -> dict:
+        """Collect @ray.method(...) per-method option dicts off the class
+        (reference: actor.py __ray_method_options__ on ActorMethod)."""
+        out = {}
+        for name in dir(self._cls):
+            if name.startswith("__"):
+                continue
+            try:
+                attr = getattr(self._cls, name)
+            except Exception:
+                continue
+            mo = getattr(attr, "__ray_method_options__", None)
+            if mo:
+                out[name] = dict(mo)
+        return out
 
     def bind(self, *args, **kwargs):
         from .dag import ClassNode

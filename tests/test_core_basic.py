@@ -258,6 +258,28 @@ def test_streaming_actor_method(ray_start_regular):
     assert [ray.get(r) for r in it] == [0, 1, 2, 3]
 
 
+def test_streaming_ids_do_not_collide_with_object_ids(ray_start_regular):
+    """Regression (advisor, round 1): stream-item oids were
+    task_id(nonce8+low4(task_seq)) + idx4, byte-identical to the normal
+    object id with obj_seq == task_seq, so a streaming item silently
+    overwrote a live object's memory-store entry. Put refs while
+    streaming tasks run through the same seq range; every put must
+    still read back its own value."""
+
+    @ray.remote
+    def gen(n):
+        for i in range(n):
+            yield ("stream", i)
+
+    held = [ray.put(("kept", i)) for i in range(25)]
+    for _ in range(25):
+        g = gen.options(num_returns="streaming").remote(3)
+        out = [ray.get(r) for r in g]
+        assert out == [("stream", 0), ("stream", 1), ("stream", 2)]
+    for i, ref in enumerate(held):
+        assert ray.get(ref) == ("kept", i)
+
+
 def test_metrics_api(ray_start_regular):
     from ray_amd.util import metrics
 
