@@ -34,7 +34,12 @@ class TorchConfig:
 
 def get_device() -> torch.device:
     if torch.cuda.is_available():
-        return torch.device("cuda", 0)
+        import os
+
+        idx = int(os.environ.get("LOCAL_RANK", "0"))
+        if idx >= torch.cuda.device_count():
+            idx = 0
+        return torch.device("cuda", idx)
     return torch.device("cpu")
 
 

@@ -42,23 +42,73 @@ class TrainWorker:
         self.thread: Optional[threading.Thread] = None
         self.error: Optional[str] = None
         self.done = False
+        self.local_rank = 0
+        self.local_world_size = 1
 
-    def setup_dist(self, master_addr: str, master_port: int, backend: str):
+    def node_info(self):
+        """Topology probe for rendezvous wiring: which node this worker
+        landed on, which GPU device ids it was leased, and a reachable
+        IP for rank-0 to host the process-group store."""
+        import ray_amd as ray
+
+        return {
+            "node_id": ray.get_runtime_context().get_node_id(),
+            "gpu_ids": ray.get_gpu_ids(),
+            "ip": os.environ.get("RAY_AMD_NODE_IP") or "127.0.0.1",
+        }
+
+    def reserve_port(self):
+        """Pick a free TCP port on THIS worker's host (rank 0 hosts the
+        rendezvous store, so the port must be free here, not on the
+        driver's node)."""
+        s = socket.socket()
+        s.bind((os.environ.get("RAY_AMD_NODE_IP") or "127.0.0.1", 0))
+        p = s.getsockname()[1]
+        s.close()
+        return p
+
+    def setup_dist(self, master_addr: str, master_port: int, backend: str,
+                   local_rank: int = 0, local_world_size: int = 1,
+                   node_visible_gpus=None):
         import torch.distributed as dist
 
+        self.local_rank = local_rank
+        self.local_world_size = local_world_size
+        # Scrub inherited torch-elastic state: if the DRIVER ran under
+        # torchrun, workers inherit TORCHELASTIC_USE_AGENT_STORE and
+        # would try to join torchrun's agent store instead of hosting
+        # their own rendezvous (init then hangs until timeout).
+        for k in ("TORCHELASTIC_USE_AGENT_STORE", "TORCHELASTIC_RUN_ID",
+                  "TORCHELASTIC_RESTART_COUNT", "TORCHELASTIC_MAX_RESTARTS",
+                  "TORCHELASTIC_ERROR_FILE", "GROUP_RANK", "GROUP_WORLD_SIZE",
+                  "ROLE_RANK", "ROLE_WORLD_SIZE", "ROLE_NAME"):
+            os.environ.pop(k, None)
         os.environ["MASTER_ADDR"] = master_addr
         os.environ["MASTER_PORT"] = str(master_port)
         os.environ["RANK"] = str(self.rank)
         os.environ["WORLD_SIZE"] = str(self.world_size)
-        os.environ["LOCAL_RANK"] = "0" if not self.use_gpu else "0"
+        os.environ["LOCAL_RANK"] = str(local_rank)
+        os.environ["LOCAL_WORLD_SIZE"] = str(local_world_size)
+        if self.use_gpu and node_visible_gpus:
+            # Reference behavior (train/_internal/utils
+            # share_cuda_visible_devices): colocated workers see the
+            # UNION of their devices, ordered by local rank, so RCCL
+            # can open direct xGMI p2p between ranks instead of
+            # bouncing through host memory. Must happen before this
+            # process first touches HIP.
+            vis = ",".join(str(g) for g in node_visible_gpus)
+            os.environ["HIP_VISIBLE_DEVICES"] = vis
+            os.environ["CUDA_VISIBLE_DEVICES"] = vis
         if self.world_size > 1:
             dist.init_process_group(
-                backend=backend, rank=self.rank, world_size=self.world_size
+                backend=backend, rank=self.rank, world_size=self.world_size,
+                init_method=f"tcp://{master_addr}:{master_port}",
             )
         if self.use_gpu:
             import torch
 
-            torch.cuda.set_device(0)  # workers see only their own GPU
+            torch.cuda.set_device(min(local_rank,
+                                      torch.cuda.device_count() - 1))
         return True
 
     def start_training(self, fn_bytes: bytes, config: dict, storage_dir: str,
@@ -67,8 +117,8 @@ class TrainWorker:
         fn = cloudpickle.loads(fn_bytes)
         shards = cloudpickle.loads(shard_bytes) if shard_bytes else {}
         self.session = TrainSession(
-            self.rank, self.world_size, 0, self.world_size, storage_dir,
-            run_name,
+            self.rank, self.world_size, self.local_rank,
+            self.local_world_size, storage_dir, run_name,
             latest_checkpoint=Checkpoint(latest_ckpt_path)
             if latest_ckpt_path
             else None,
@@ -210,10 +260,30 @@ class DataParallelTrainer:
         ]
         try:
             try:
-                port = _free_port()
+                # topology: local ranks per node + shared device
+                # visibility, rendezvous hosted on rank-0's node
+                infos = ray.get([w.node_info.remote() for w in workers],
+                                timeout=120)
+                by_node: Dict[str, list] = {}
+                for i, inf in enumerate(infos):
+                    by_node.setdefault(inf["node_id"], []).append(i)
+                local_rank = {}
+                local_ws = {}
+                node_vis = {}
+                for idxs in by_node.values():
+                    vis = [g for i in idxs for g in infos[i]["gpu_ids"]]
+                    for lr, i in enumerate(idxs):
+                        local_rank[i] = lr
+                        local_ws[i] = len(idxs)
+                        node_vis[i] = vis
+                master_addr = infos[0]["ip"]
+                port = ray.get(workers[0].reserve_port.remote(), timeout=60)
                 ray.get([
-                    w.setup_dist.remote("127.0.0.1", port, self._backend())
-                    for w in workers
+                    w.setup_dist.remote(
+                        master_addr, port, self._backend(),
+                        local_rank[i], local_ws[i], node_vis[i],
+                    )
+                    for i, w in enumerate(workers)
                 ], timeout=180)
 
                 shard_payloads = self._make_shards(n)
