@@ -57,8 +57,52 @@ def build_shm(verbose: bool = True, force: bool = False) -> str:
     return OUT_SHM
 
 
+def _hipcc_torch_ext(src: str, out: str, name: str, extra_libs=(),
+                     verbose: bool = True):
+    hipcc = os.path.join(
+        os.environ.get("ROCM_PATH", "/opt/rocm"), "bin", "hipcc"
+    )
+    includes = cpp_extension.include_paths() + [_python_include(), CSRC]
+    lib_dirs = cpp_extension.library_paths()
+    abi = int(torch.compiled_with_cxx11_abi())
+    cmd = [
+        hipcc, "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
+        "-shared", src, "-o", out,
+        f"-D_GLIBCXX_USE_CXX11_ABI={abi}",
+        f"-DTORCH_EXTENSION_NAME={name}",
+        "-DTORCH_API_INCLUDE_EXTENSION_H",
+        "-DUSE_ROCM=1", "-fno-gpu-rdc", "-Wno-unused-result",
+    ]
+    for i in includes:
+        cmd.append(f"-I{i}")
+    for l in lib_dirs:
+        cmd.append(f"-L{l}")
+        cmd.append(f"-Wl,-rpath,{l}")
+    cmd += ["-ltorch", "-ltorch_python", "-lc10", "-ltorch_hip", "-lc10_hip",
+            "-lamdhip64"]
+    cmd += list(extra_libs)
+    if verbose:
+        print("[ray_amd build]", " ".join(cmd), file=sys.stderr)
+    subprocess.check_call(cmd)
+    return out
+
+
+OUT_RCCL = os.path.join(PKG, "_rccl_comm.so")
+
+
+def build_rccl(verbose: bool = True, force: bool = False) -> str:
+    """Native RCCL communicator extension (rccl_comm.hip)."""
+    src = os.path.join(CSRC, "rccl_comm.hip")
+    if (not force and os.path.exists(OUT_RCCL)
+            and os.path.getmtime(OUT_RCCL) >= os.path.getmtime(src)):
+        return OUT_RCCL
+    return _hipcc_torch_ext(src, OUT_RCCL, "_rccl_comm",
+                            extra_libs=["-lrccl"], verbose=verbose)
+
+
 def build(verbose: bool = True, force: bool = False) -> str:
     build_shm(verbose, force)
+    build_rccl(verbose, force)
     if not force and not needs_rebuild():
         return OUT
     hipcc = os.path.join(
