@@ -1,16 +1,19 @@
 """Declarative collective groups over actors (reference:
 python/ray/util/collective/collective.py:149-624).
 
-MI355X-native backend: torch.distributed process-group objects built
-directly on a TCPStore — backend "rccl" is torch's ProcessGroupNCCL,
-which IS RCCL on ROCm (collectives run over xGMI in-node); "torch_gloo"
-is the CPU path. Multiple named groups can coexist per process (the
-reference's cupy-NCCL group cache, nccl_collective_group.py:126), and
-bf16 IS supported (the reference notes cupy couldn't, nccl_util.py:693).
+MI355X-native backend: GPU groups run on ray_amd's own RCCL
+communicator extension (csrc/rccl_comm.hip — rcclCommInitRank with the
+unique id exchanged through the GCS KV, a dedicated high-priority HIP
+stream + events per group, bf16 collectives over xGMI). This owns the
+layer the reference delegates to cupy
+(collective_group/nccl_collective_group.py:126); bf16 IS supported
+(the reference notes cupy couldn't, nccl_util.py:693). "torch_gloo" is
+the CPU path (torch ProcessGroupGloo on a TCPStore).
 
-Rendezvous runs through the GCS KV: rank 0 publishes host:port under
-"collective:<group>"; everyone else polls. Outside a ray_amd session
-(e.g. plain torchrun) MASTER_ADDR/MASTER_PORT env are used.
+Rendezvous runs through the GCS KV: rank 0 publishes the RCCL unique
+id (GPU) or host:port (CPU) under "collective:<group>"; everyone else
+polls. Outside a ray_amd session (e.g. plain torchrun)
+MASTER_ADDR/MASTER_PORT env are used to share it over a TCPStore.
 """
 from __future__ import annotations
 
@@ -28,6 +31,15 @@ from .types import Backend, ReduceOp, torch_reduce_op
 _groups: Dict[str, "Group"] = {}
 
 
+def _load_native():
+    """Import the in-tree RCCL extension; loud failure on GPU boxes —
+    a silent torch-PG fallback would defeat the native-collective
+    contract (driver checks which .so files actually loaded)."""
+    from ... import _rccl_comm  # in-tree ray_amd/_rccl_comm.so
+
+    return _rccl_comm
+
+
 class Group:
     def __init__(self, pg, rank: int, world_size: int, backend: str, store=None):
         self.pg = pg
@@ -35,6 +47,18 @@ class Group:
         self.world_size = world_size
         self.backend = backend
         self._store = store  # keep TCPStore alive
+
+
+class NativeRcclGroup:
+    """A collective group on the native RCCL comm (one per group per
+    process; dedicated comm stream lives inside the extension)."""
+
+    backend = "rccl"
+
+    def __init__(self, comm):
+        self.comm = comm
+        self.rank = comm.rank()
+        self.world_size = comm.world_size()
 
 
 def _free_port() -> int:
@@ -64,6 +88,39 @@ def _kv_rendezvous(group_name: str, rank: int, timeout: float = 120.0) -> str:
     raise TimeoutError(f"rendezvous for group {group_name} timed out")
 
 
+def _kv_share_bytes(group_name: str, rank: int, payload: Optional[bytes],
+                    timeout: float = 120.0) -> bytes:
+    """Rank 0 publishes `payload` under the group key; other ranks poll
+    it back. Used to exchange the RCCL unique id (reference:
+    nccl_collective_group.py rendezvous through a named store actor)."""
+    from ..._core import runtime as _rt
+
+    key = f"collective_uid:{group_name}".encode()
+    if _rt.is_initialized():
+        rt = _rt.global_runtime()
+        if rank == 0:
+            rt.gcs_call("kv_put", {"ns": "collective", "key": key,
+                                   "value": payload})
+            return payload
+        deadline = time.time() + timeout
+        while time.time() < deadline:
+            v = rt.gcs_call("kv_get", {"ns": "collective", "key": key})
+            if v:
+                return bytes(v)
+            time.sleep(0.02)
+        raise TimeoutError(f"uid rendezvous for group {group_name} timed out")
+    # outside a ray session (torchrun): share over a TCPStore
+    host = os.environ.get("MASTER_ADDR", "127.0.0.1")
+    port = int(os.environ.get("MASTER_PORT", "29500"))
+    store = dist.TCPStore(host, port, None, is_master=(rank == 0),
+                          timeout=datetime.timedelta(seconds=timeout))
+    skey = f"rccl_uid:{group_name}"
+    if rank == 0:
+        store.set(skey, payload)
+        return payload
+    return bytes(store.get(skey))
+
+
 def init_collective_group(
     world_size: int,
     rank: int,
@@ -74,6 +131,16 @@ def init_collective_group(
     if group_name in _groups:
         raise RuntimeError(f"group {group_name} already initialized here")
     from ..._core import runtime as _rt
+
+    if backend in ("rccl", "nccl") and torch.cuda.is_available() \
+            and not os.environ.get("RAY_AMD_COLLECTIVE_TORCH_PG"):
+        native = _load_native()
+        uid = native.unique_id() if rank == 0 else None
+        uid = _kv_share_bytes(group_name, rank, uid)
+        comm = native.RcclComm(world_size, rank, uid,
+                               torch.cuda.current_device())
+        _groups[group_name] = NativeRcclGroup(comm)
+        return
 
     if _rt.is_initialized():
         addr = _kv_rendezvous(group_name, rank)
@@ -149,8 +216,16 @@ def get_collective_group_size(group_name: str = "default") -> int:
     return _get(group_name).world_size
 
 
+def _red_name(op: ReduceOp) -> str:
+    return {ReduceOp.SUM: "sum", ReduceOp.PRODUCT: "prod",
+            ReduceOp.MIN: "min", ReduceOp.MAX: "max"}.get(op, "sum")
+
+
 def allreduce(tensor, group_name: str = "default", op: ReduceOp = ReduceOp.SUM):
     g = _get(group_name)
+    if isinstance(g, NativeRcclGroup):
+        g.comm.allreduce(tensor, _red_name(op))
+        return
     opts = dist.AllreduceOptions()
     opts.reduceOp = torch_reduce_op(op)
     g.pg.allreduce([tensor], opts).wait()
@@ -164,6 +239,9 @@ def allreduce_multigpu(tensor_list, group_name="default", op=ReduceOp.SUM):
 def reduce(tensor, dst_rank: int = 0, group_name: str = "default",
            op: ReduceOp = ReduceOp.SUM):
     g = _get(group_name)
+    if isinstance(g, NativeRcclGroup):
+        g.comm.reduce(tensor, dst_rank, _red_name(op))
+        return
     opts = dist.ReduceOptions()
     opts.rootRank = dst_rank
     opts.reduceOp = torch_reduce_op(op)
@@ -172,6 +250,9 @@ def reduce(tensor, dst_rank: int = 0, group_name: str = "default",
 
 def broadcast(tensor, src_rank: int = 0, group_name: str = "default"):
     g = _get(group_name)
+    if isinstance(g, NativeRcclGroup):
+        g.comm.broadcast(tensor, src_rank)
+        return
     opts = dist.BroadcastOptions()
     opts.rootRank = src_rank
     g.pg.broadcast([tensor], opts).wait()
@@ -179,12 +260,27 @@ def broadcast(tensor, src_rank: int = 0, group_name: str = "default"):
 
 def allgather(tensor_list: List, tensor, group_name: str = "default"):
     g = _get(group_name)
+    if isinstance(g, NativeRcclGroup):
+        out = torch.empty(
+            (g.world_size,) + tuple(tensor.shape),
+            dtype=tensor.dtype, device=tensor.device,
+        )
+        g.comm.allgather(out, tensor.contiguous())
+        for i, t in enumerate(tensor_list):
+            t.copy_(out[i].view_as(t))
+        return
     g.pg.allgather([tensor_list], [tensor]).wait()
 
 
 def reducescatter(tensor, tensor_list: List, group_name: str = "default",
                   op: ReduceOp = ReduceOp.SUM):
     g = _get(group_name)
+    if isinstance(g, NativeRcclGroup):
+        inp = torch.cat([t.contiguous().flatten() for t in tensor_list])
+        out = torch.empty_like(tensor).flatten()
+        g.comm.reducescatter(out, inp, _red_name(op))
+        tensor.copy_(out.view_as(tensor))
+        return
     if g.backend == "rccl":
         opts = dist.ReduceScatterOptions()
         opts.reduceOp = torch_reduce_op(op)
@@ -199,6 +295,9 @@ def reducescatter(tensor, tensor_list: List, group_name: str = "default",
 
 def barrier(group_name: str = "default"):
     g = _get(group_name)
+    if isinstance(g, NativeRcclGroup):
+        g.comm.barrier()
+        return
     if hasattr(g.pg, "barrier"):
         try:
             g.pg.barrier(dist.BarrierOptions()).wait()
@@ -213,11 +312,17 @@ def barrier(group_name: str = "default"):
 
 def send(tensor, dst_rank: int, group_name: str = "default"):
     g = _get(group_name)
+    if isinstance(g, NativeRcclGroup):
+        g.comm.send(tensor, dst_rank)
+        return
     g.pg.send([tensor], dst_rank, 0).wait()
 
 
 def recv(tensor, src_rank: int, group_name: str = "default"):
     g = _get(group_name)
+    if isinstance(g, NativeRcclGroup):
+        g.comm.recv(tensor, src_rank)
+        return
     g.pg.recv([tensor], src_rank, 0).wait()
 
 

@@ -145,8 +145,10 @@ def test_serve_llm_gpu_deployment():
 
 @GPU
 def test_collective_rccl_single_rank():
-    """RCCL group init + allreduce with world_size=1 (the multi-GPU
-    path is covered by the driver's 8-GPU scaling run)."""
+    """Native RCCL group (csrc/rccl_comm.hip) init + the full op set
+    with world_size=1 (the multi-GPU path is covered by the driver's
+    8-GPU scaling run). Asserts the NATIVE module is in use — a silent
+    torch-PG fallback fails this test."""
     ray.init(num_cpus=4, num_gpus=1, ignore_reinit_error=True)
     try:
 
@@ -158,14 +160,48 @@ def test_collective_rccl_single_rank():
                 col.init_collective_group(1, 0, backend="rccl",
                                           group_name="rccl1")
                 self.col = col
+                g = col.collective._groups["rccl1"]
+                assert isinstance(g, col.collective.NativeRcclGroup), g
 
-            def allreduce(self):
-                t = torch.ones(128, device="cuda") * 3
-                self.col.allreduce(t, "rccl1")
-                return float(t.sum().item())
+            def ops(self):
+                col = self.col
+                out = {}
+                t = torch.ones(128, device="cuda",
+                               dtype=torch.bfloat16) * 3
+                col.allreduce(t, "rccl1")
+                out["allreduce_bf16"] = float(t.float().sum().item())
+                b = torch.full((64,), 7.0, device="cuda")
+                col.broadcast(b, 0, "rccl1")
+                out["broadcast"] = float(b.sum().item())
+                src = torch.arange(8, device="cuda", dtype=torch.float32)
+                dst = [torch.zeros(8, device="cuda")]
+                col.allgather(dst, src, "rccl1")
+                out["allgather"] = dst[0].tolist()
+                rs_out = torch.zeros(4, device="cuda")
+                col.reducescatter(rs_out,
+                                  [torch.ones(4, device="cuda") * 2],
+                                  "rccl1")
+                out["reducescatter"] = rs_out.tolist()
+                # self send/recv must be inside a group call
+                g = col.collective._groups["rccl1"]
+                s = torch.arange(4, device="cuda", dtype=torch.float32)
+                r = torch.zeros(4, device="cuda")
+                g.comm.group_start()
+                g.comm.send(s, 0)
+                g.comm.recv(r, 0)
+                g.comm.group_end()
+                g.comm.synchronize()
+                out["p2p_self"] = r.tolist()
+                col.barrier("rccl1")
+                return out
 
         w = W.remote()
-        assert ray.get(w.allreduce.remote(), timeout=180) == 3 * 128
+        out = ray.get(w.ops.remote(), timeout=180)
+        assert out["allreduce_bf16"] == 3 * 128
+        assert out["broadcast"] == 7 * 64
+        assert out["allgather"] == list(range(8))
+        assert out["reducescatter"] == [2.0] * 4
+        assert out["p2p_self"] == [0.0, 1.0, 2.0, 3.0]
     finally:
         ray.shutdown()
 
