@@ -506,7 +506,7 @@ class CoreRuntime:
         async def _send():
             try:
                 c = await self._conn(owner_addr)
-                c.notify(method, {"id": oid})
+                c.notify(method, {"id": oid, "addr": self.addr})
             except Exception:
                 pass
 
@@ -515,20 +515,84 @@ class CoreRuntime:
         except Exception:
             pass
 
+    # Per-borrower tracking (reference: reference_counter.h:44 borrower
+    # sets + WaitForRefRemoved). The owner keeps a SET of borrower
+    # addresses per oid: a borrower sends borrow_add once when it first
+    # holds the ref and borrow_del once when it fully drops it (ordered
+    # on its own connection). Because adds can also arrive out-of-band
+    # via the task-reply merge (see _ingest_result) while the
+    # borrower's own del races them on a different connection, a del
+    # for a not-yet-known borrower is kept as a tombstone that cancels
+    # the late add.
+
+    def _borrow_merge(self, oid: bytes, addr: str):
+        if not addr or addr == self.addr:
+            return
+        ts = self._borrow_tombstones.get(oid)
+        if ts and addr in ts:
+            ts.discard(addr)
+            if not ts:
+                self._borrow_tombstones.pop(oid, None)
+            return
+        self._borrows.setdefault(oid, set()).add(addr)
+
     def _h_borrow_add(self, conn, p):
-        oid = bytes(p["id"])
-        self._borrows[oid] = self._borrows.get(oid, 0) + 1
+        self._borrow_merge(bytes(p["id"]), p.get("addr") or "?")
 
     def _h_borrow_del(self, conn, p):
         oid = bytes(p["id"])
-        n = self._borrows.get(oid, 0) - 1
-        if n <= 0:
-            self._borrows.pop(oid, None)
+        addr = p.get("addr") or "?"
+        s = self._borrows.get(oid)
+        if s is not None and addr in s:
+            s.discard(addr)
+            if not s:
+                self._borrows.pop(oid, None)
+                if oid in self._pending_free:
+                    self._pending_free.discard(oid)
+                    self._free_owned(oid, _force=True)
+        else:
+            self._borrow_tombstones.setdefault(oid, set()).add(addr)
+
+    def _sweep_borrower(self, addr: str):
+        """Drop a dead borrower's holds everywhere; free objects whose
+        last borrower it was (reference: borrower death in
+        reference_counter.cc)."""
+        emptied = []
+        for oid, s in list(self._borrows.items()):
+            if addr in s:
+                s.discard(addr)
+                if not s:
+                    self._borrows.pop(oid, None)
+                    emptied.append(oid)
+        for oid in emptied:
             if oid in self._pending_free:
                 self._pending_free.discard(oid)
                 self._free_owned(oid, _force=True)
-        else:
-            self._borrows[oid] = n
+
+    def _ensure_borrow_sweeper(self):
+        if self._borrow_sweeper_running:
+            return
+        self._borrow_sweeper_running = True
+
+        async def _sweep_loop():
+            try:
+                while self._pending_free and not self._closed:
+                    await asyncio.sleep(1.0)
+                    addrs = set()
+                    for oid in list(self._pending_free):
+                        addrs |= self._borrows.get(oid, set())
+                    for addr in addrs:
+                        if addr == "?":
+                            continue
+                        try:
+                            c = await self._conn(addr)
+                            await asyncio.wait_for(c.call("ping", {}), 3.0)
+                        except Exception:
+                            self._sweep_borrower(addr)
+            finally:
+                self._borrow_sweeper_running = False
+
+        self._run(_sweep_loop())
 
     def _add_submitted_ref(self, oid: bytes):
         with self._refs_lock:

@@ -13,6 +13,7 @@
 #include "hip/flash_attn_v3.hip"
 #include "hip/flash_attn_v4.hip"
 #include "hip/flash_attn_v5.hip"
+#include "hip/flash_attn_v6.hip"
 #include "hip/fa_bwd.hip"
 
 #define CHECK_IN(x)                                                     \
@@ -263,9 +264,16 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
                        (const short*)k.data_ptr(), (const short*)v.data_ptr(),
                        (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
                        causal ? 1 : 0, (int)q_offset, scale);
-  else
+  else if (getenv("RAY_AMD_FA_V5") != nullptr || T % 256 != 0)
     hipLaunchKernelGGL(flash_attn_fwd_v5_bf16, dim3(T / 128, B * Hq),
                        dim3(256), 0, cur_stream(), (const short*)q.data_ptr(),
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
+                       causal ? 1 : 0, (int)q_offset, scale);
+  else
+    // v6 (default): 8-wave dbuf/async-stage/swizzled structure
+    hipLaunchKernelGGL(flash_attn_fwd_v6_bf16, dim3(T / 256, B * Hq),
+                       dim3(512), 0, cur_stream(), (const short*)q.data_ptr(),
                        (const short*)k.data_ptr(), (const short*)v.data_ptr(),
                        (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
                        causal ? 1 : 0, (int)q_offset, scale);

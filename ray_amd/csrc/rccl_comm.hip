@@ -91,9 +91,9 @@ class RcclComm {
     if (comm_ != nullptr) {
       ncclCommDestroy(comm_);
     }
-    if (stream_ != nullptr) hipStreamDestroy(stream_);
-    if (ev_in_ != nullptr) hipEventDestroy(ev_in_);
-    if (ev_out_ != nullptr) hipEventDestroy(ev_out_);
+    if (stream_ != nullptr) (void)hipStreamDestroy(stream_);
+    if (ev_in_ != nullptr) (void)hipEventDestroy(ev_in_);
+    if (ev_out_ != nullptr) (void)hipEventDestroy(ev_out_);
   }
 
   int rank() const { return rank_; }

@@ -52,7 +52,7 @@ def main():
 
     print(
         f"shape B{B} H{H} T{T} D{D} causal={causal}\n"
-        f"ray_amd flash v5: {t_ours*1e3:.3f} ms  "
+        f"ray_amd flash    : {t_ours*1e3:.3f} ms  "
         f"{flops/t_ours/1e12:.1f} TFLOP/s\n"
         f"torch SDPA      : {t_sdpa*1e3:.3f} ms  "
         f"{flops/t_sdpa/1e12:.1f} TFLOP/s\n"
