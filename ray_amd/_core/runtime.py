@@ -260,13 +260,14 @@ class CoreRuntime:
         # pub/sub: channel -> local callbacks (invoked on the loop thread)
         self._pubsub_cbs: Dict[str, list] = {}
 
-        # borrow protocol (reference: reference_counter.h borrowers /
-        # WaitForRefRemoved): owner side counts live borrowers per oid
-        # and defers the free until they release; borrower side reports
-        # first-acquire / last-release of refs it does not own.
-        # Known limit (r2): a borrower that dies without releasing
-        # leaks its count — per-borrower tracking + death sweep needed.
-        self._borrows: Dict[bytes, int] = {}
+        # borrow protocol (reference: reference_counter.h:44 borrower
+        # sets / WaitForRefRemoved): owner tracks the SET of borrower
+        # addresses per oid and defers frees until all release; a
+        # sweeper pings borrowers of pending-free objects and drops
+        # dead ones (borrower-crash no longer pins the object).
+        self._borrows: Dict[bytes, set] = {}
+        self._borrow_tombstones: Dict[bytes, set] = {}
+        self._borrow_sweeper_running = False
         self._pending_free: set = set()
 
         self.server.route("fetch_object", self._h_fetch_object)
@@ -614,11 +615,14 @@ class CoreRuntime:
             self._free_owned(oid)
 
     def _free_owned(self, oid: bytes, _force: bool = False):
-        if not _force and self._borrows.get(oid, 0) > 0:
+        if not _force and self._borrows.get(oid):
             # a borrower still holds this ref: defer the free until the
-            # last borrow_del arrives (reference: WaitForRefRemoved)
+            # last borrow_del arrives (reference: WaitForRefRemoved);
+            # the sweeper handles borrowers that die instead
             self._pending_free.add(oid)
+            self._ensure_borrow_sweeper()
             return
+        self._borrow_tombstones.pop(oid, None)
         ent = self.memory_store.pop(oid, None)
         self._events.pop(oid, None)
         mapped = self._mmaps.pop(oid, None)
@@ -1343,6 +1347,23 @@ class CoreRuntime:
 
     def _ingest_result(self, spec, reply):
         status = reply.get("status")
+        # Synchronous borrower merge (see worker._held_borrows): refs
+        # the executing worker still holds must be registered with
+        # their owner BEFORE this caller drops its submitted-refs.
+        waddr = reply.get("worker_addr")
+        for oid, owner in reply.get("borrows") or ():
+            oid = bytes(oid)
+            if owner == self.addr:
+                self._borrow_merge(oid, waddr)
+            elif owner and waddr:
+                async def _fwd(owner=owner, oid=oid):
+                    try:
+                        c = await self._conn(owner)
+                        c.notify("borrow_add", {"id": oid, "addr": waddr})
+                    except Exception:
+                        pass
+
+                self._run(_fwd())
         if spec.get("streaming"):
             self._stream_finish(
                 spec["task_id"],

@@ -515,7 +515,21 @@ class WorkerMain:
                 results.append(
                     {"kind": "store", "node_addr": self.raylet_addr, "size": size}
                 )
-        return {"status": "ok", "results": results}
+        return {"status": "ok", "results": results,
+                "borrows": self._held_borrows(), "worker_addr": self.rt.addr}
+
+    def _held_borrows(self):
+        """Borrowed refs this worker still holds at reply time; the
+        caller merges these into the owner's borrower set SYNCHRONOUSLY
+        before it may drop its submitted-ref and free (closes the
+        fire-and-forget borrow_add race; reference: the borrower-chain
+        merge carried on task replies, reference_counter.h:44)."""
+        out = []
+        with self.rt._refs_lock:
+            for oid, e in self.rt._refs.items():
+                if e[0] > 0 and e[2] and e[2] != self.rt.addr:
+                    out.append((oid, e[2]))
+        return out
 
     async def _event_flusher(self):
         # small batches flush on a timer so timeline/state views see
