@@ -108,7 +108,19 @@ class GcsServer:
             _os.unlink(self.sock_path)
         except OSError:
             pass
-        await self.server.start_unix(self.sock_path)
+        node_ip = _os.environ.get("RAY_AMD_NODE_IP")
+        if node_ip:
+            # multi-machine mode: listen on TCP, advertise the address
+            # through the sock_path as a regular file (node.start_gcs
+            # reads it back)
+            port = await self.server.start_tcp(node_ip, 0)
+            self.addr = f"tcp:{node_ip}:{port}"
+            with open(self.sock_path + ".tmp", "w") as f:
+                f.write(self.addr)
+            _os.rename(self.sock_path + ".tmp", self.sock_path)
+        else:
+            await self.server.start_unix(self.sock_path)
+            self.addr = "unix:" + self.sock_path
 
     # ---------- persistence ----------
     def _snapshot(self) -> bytes:

@@ -83,7 +83,13 @@ def start_gcs(session_dir: str, env=None) -> tuple:
         if time.time() > deadline:
             raise RuntimeError("GCS start timed out")
         time.sleep(0.01)
-    return proc, "unix:" + sock
+    import stat
+
+    if stat.S_ISSOCK(os.stat(sock).st_mode):
+        return proc, "unix:" + sock
+    # TCP mode (RAY_AMD_NODE_IP): the file carries the address
+    with open(sock) as f:
+        return proc, f.read().strip()
 
 
 def start_raylet(

@@ -203,6 +203,21 @@ class RpcServer:
                 pass
 
 
+async def bind_server(server: "RpcServer", session_dir: str, name: str) -> str:
+    """Bind a daemon server on the configured transport. With
+    RAY_AMD_NODE_IP set the server listens on TCP at that address
+    (multi-machine mode; reference: network-addressed raylet startup,
+    _private/node.py:1422); otherwise a unix socket under the session
+    dir (single-machine fast path)."""
+    node_ip = os.environ.get("RAY_AMD_NODE_IP")
+    if node_ip:
+        port = await server.start_tcp(node_ip, 0)
+        return f"tcp:{node_ip}:{port}"
+    sock = os.path.join(session_dir, "sock", name)
+    await server.start_unix(sock)
+    return "unix:" + sock
+
+
 class RpcClient:
     """Asyncio RPC client with auto seq correlation. Not thread-safe;
     use from the owning event loop."""

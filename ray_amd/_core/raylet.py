@@ -110,9 +110,12 @@ class Raylet:
     # ---------------- lifecycle ----------------
 
     async def start(self):
-        sock = os.path.join(self.session_dir, "sock", f"raylet_{self.node_id.hex()[:8]}")
-        await self.server.start_unix(sock)
-        self.addr = "unix:" + sock
+        from .protocol import bind_server
+
+        self.addr = await bind_server(
+            self.server, self.session_dir,
+            f"raylet_{self.node_id.hex()[:8]}",
+        )
         await self.gcs.connect(self.gcs_addr)
         await self.gcs.call(
             "register_node",

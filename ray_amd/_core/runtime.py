@@ -328,11 +328,12 @@ class CoreRuntime:
         raise ConnectionLost(f"GCS unreachable after restart wait: {last}")
 
     async def _async_start(self):
-        sock = os.path.join(
-            self.session_dir, "sock", f"rt_{os.getpid()}_{os.urandom(3).hex()}"
+        from .protocol import bind_server
+
+        self.addr = await bind_server(
+            self.server, self.session_dir,
+            f"rt_{os.getpid()}_{os.urandom(3).hex()}",
         )
-        await self.server.start_unix(sock)
-        self.addr = "unix:" + sock
         await self.gcs.connect(self.gcs_addr)
         self.gcs.on_notify = self._on_conn_notify
         await self.raylet.connect(self.raylet_addr)

@@ -28,11 +28,21 @@ class NodeHandle:
 
 class Cluster:
     def __init__(self, initialize_head: bool = True, head_node_args: Optional[dict] = None,
-                 connect: bool = False):
+                 connect: bool = False, tcp: bool = False):
+        """tcp=True runs every daemon on TCP (127.0.0.1) and gives each
+        added node its OWN shm directory — the two-"machine" harness:
+        object pulls must move bytes through the chunked raylet
+        transfer, never a shared filesystem path."""
+        self.tcp = tcp
         self.session_dir = _node.new_session_dir()
         os.environ["RAY_AMD_SHM_DIR"] = _node.session_shm_dir(self.session_dir)
         _node.export_driver_pythonpath()
-        self.gcs_proc, self.gcs_addr = _node.start_gcs(self.session_dir)
+        gcs_env = None
+        if tcp:
+            os.environ["RAY_AMD_NODE_IP"] = "127.0.0.1"
+            gcs_env = dict(os.environ)
+        self.gcs_proc, self.gcs_addr = _node.start_gcs(self.session_dir,
+                                                       env=gcs_env)
         self.nodes = []
         self.head_node: Optional[NodeHandle] = None
         if initialize_head:
@@ -53,9 +63,19 @@ class Cluster:
         res.setdefault("GPU", num_gpus)
         res.setdefault("memory", 16 * 2**30)
         name = f"node{len(self.nodes)}"
+        env_extra = dict(kwargs.pop("env_extra", None) or {})
+        if self.tcp:
+            env_extra.setdefault("RAY_AMD_NODE_IP", "127.0.0.1")
+            if self.nodes:  # non-head nodes get their own "machine" shm
+                shm = os.path.join(
+                    _node.session_shm_dir(self.session_dir) + f"_{name}"
+                )
+                os.makedirs(shm, exist_ok=True)
+                env_extra.setdefault("RAY_AMD_SHM_DIR", shm)
         proc, addr, node_id = _node.start_raylet(
             self.session_dir, self.gcs_addr, res, node_name=name,
             labels=labels, object_store_memory=object_store_memory,
+            env_extra=env_extra or None,
         )
         h = NodeHandle(proc, addr, node_id, res)
         self.nodes.append(h)
@@ -137,7 +157,11 @@ class Cluster:
         import shutil
 
         try:
-            shutil.rmtree(_node.session_shm_dir(self.session_dir), ignore_errors=True)
+            import glob
+
+            base = _node.session_shm_dir(self.session_dir)
+            for d in [base] + glob.glob(base + "_*"):
+                shutil.rmtree(d, ignore_errors=True)
         except Exception:
             pass
         shutil.rmtree(self.session_dir, ignore_errors=True)

@@ -15,6 +15,7 @@
 #include "hip/flash_attn_v5.hip"
 #include "hip/flash_attn_v6.hip"
 #include "hip/fa_bwd.hip"
+#include "hip/fa_bwd_v3.hip"
 
 #define CHECK_IN(x)                                                     \
   TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                       \
@@ -311,6 +312,28 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
                      dsum.data_ptr<float>(), (short*)dq.data_ptr(), B, Hq,
                      Hkv, T, causal ? 1 : 0, scale);
   static const bool split = getenv("RAY_AMD_FA_BWD_FUSED") == nullptr;
+  static const bool v3 = getenv("RAY_AMD_FA_BWD_V2") == nullptr;
+  if (split && v3) {
+    // v3: K/V register-resident, 64-row dbuf async-staged q/dO tiles,
+    // swizzled LDS reads, one barrier per tile
+    hipLaunchKernelGGL(fa_bwd_dv_v3_bf16, dim3(T / 128, B * Hkv),
+                       dim3(256), 0, cur_stream(),
+                       (const short*)q.data_ptr(),
+                       (const short*)k.data_ptr(),
+                       (const short*)v.data_ptr(),
+                       (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
+                       (short*)dv.data_ptr(), B, Hq, Hkv, T,
+                       causal ? 1 : 0, scale);
+    hipLaunchKernelGGL(fa_bwd_dk_v3_bf16, dim3(T / 128, B * Hkv),
+                       dim3(256), 0, cur_stream(),
+                       (const short*)q.data_ptr(),
+                       (const short*)k.data_ptr(),
+                       (const short*)v.data_ptr(),
+                       (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
+                       dsum.data_ptr<float>(), (short*)dk.data_ptr(), B,
+                       Hq, Hkv, T, causal ? 1 : 0, scale);
+    return {dq, dk, dv};
+  }
   if (split) {
     // dk/dv split: each kernel fits 2 waves/SIMD (the fused one is
     // register-bound to 1) at +25% mfma work — measured faster
