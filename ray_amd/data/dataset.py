@@ -421,49 +421,26 @@ class Dataset:
         return ds
 
     def _iter_block_refs(self) -> Iterator[Any]:
-        """Streaming execution: keep at most `window` blocks in flight."""
+        """Streaming execution through the operator-graph executor
+        (data/_executor.py): fused task segments / actor-pool segments
+        with per-op in-flight budgets, store-pressure backpressure on
+        the source, and pool autoscaling."""
         ray = _ray()
         if self._materialized is not None:
             yield from self._materialized
             return
-        ops = self._ops
         window = DataContext.get_current().streaming_read_window
-        if self._concurrency:
-            window = min(window, self._concurrency)
-        if any(op[0] == "actor_map" for op in ops):
-            yield from self._iter_actor_pool(ops, window)
-            return
+        from ._executor import build_chain
 
-        @ray.remote
-        def _exec_block(block, ops=ops):
-            t = block() if callable(block) else block
-            return _apply_ops(t, ops)
-
-        pending = []
-        inputs = iter(self._inputs)
-        sentinel = object()
-        held = None
-
-        def refill():
-            # top the window up; under store pressure stop producing and
-            # let in-flight blocks drain — but never starve (always
-            # launch when nothing is in flight)
-            nonlocal held
-            while len(pending) < window:
-                if held is None:
-                    held = next(inputs, sentinel)
-                if held is sentinel:
-                    return
-                if pending and _store_pressure(ray):
-                    return
-                pending.append(_exec_block.remote(held))
-                held = None
-
-        refill()
-        while pending:
-            ref = pending.pop(0)
-            refill()
-            yield ref
+        ex = build_chain(
+            ray, self._inputs, self._ops, _apply_ops,
+            window=window,
+            task_opts=self._task_opts,
+            concurrency=self._concurrency,
+            actor_pool_size=self._actor_pool_size,
+            store_pressure=lambda: _store_pressure(ray),
+        )
+        yield from ex.run()
 
     def _iter_actor_pool(self, ops, window):
         """Blocks flow through a fixed pool of UDF actors (stateful,
@@ -950,7 +927,11 @@ class Dataset:
         return sum(t.nbytes for t in self._iter_tables())
 
     def stats(self) -> str:
-        return f"Dataset(num_blocks={self.num_blocks()}, ops={len(self._ops)})"
+        from ._executor import StreamingExecutor
+
+        base = f"Dataset(num_blocks={self.num_blocks()}, ops={len(self._ops)})"
+        rep = StreamingExecutor.stats_report()
+        return base + ("\n" + rep if rep else "")
 
     def __repr__(self):
         return f"Dataset(num_blocks={self.num_blocks()})"

@@ -245,3 +245,47 @@ def test_map_batches_actor_pool(ray_start_regular, tmp_path):
     assert sorted(r["v"] for r in out) == list(range(1000, 1080))
     inits = list(init_dir.glob("init_*"))
     assert len(inits) == 3  # one instance per pool actor, not per block
+
+
+def test_streaming_executor_budgets_and_stats(ray_start_regular):
+    """Operator-graph executor: per-op in-flight stays under budget,
+    mixed task/actor chains split into segments, op-level stats
+    reported (reference: streaming_executor.py + resource_manager)."""
+    from ray_amd.data import DataContext
+    from ray_amd.data._executor import StreamingExecutor
+
+    ctx = DataContext.get_current()
+    prev = ctx.streaming_read_window
+    ctx.streaming_read_window = 3
+    try:
+        class AddOne:
+            def __call__(self, b):
+                return {"id": b["id"] + 1}
+
+        ds = (
+            rd.range(40, override_num_blocks=10)
+            .map_batches(lambda b: {"id": b["id"] * 2})
+            .map_batches(AddOne, concurrency=2)
+            .map_batches(lambda b: {"id": b["id"] + 100})
+        )
+        rows = sorted(r["id"] for r in ds.take_all())
+        assert rows == sorted(2 * i + 101 for i in range(40))
+        stats = StreamingExecutor.last_stats
+        assert stats is not None and len(stats) == 3  # task, actor, task
+        for s in stats:
+            assert s.completed == 10
+            assert s.peak_in_flight <= 3
+        actor_stats = stats[1]
+        assert actor_stats.actors >= 1
+        report = ds.stats()
+        assert "Streaming executor" in report
+    finally:
+        ctx.streaming_read_window = prev
+
+
+def test_streaming_executor_order_preserved(ray_start_regular):
+    ds = rd.range(100, override_num_blocks=20).map_batches(
+        lambda b: {"id": b["id"]}
+    )
+    got = [r["id"] for r in ds.take_all()]
+    assert got == list(range(100))
