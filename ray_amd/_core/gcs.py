@@ -299,12 +299,36 @@ class GcsServer:
         return True
 
     def _pick_node(self, req: Dict[str, float], strategy: str = "hybrid",
-                   exclude=(), soft_affinity: Optional[bytes] = None) -> Optional[NodeInfo]:
+                   exclude=(), soft_affinity: Optional[bytes] = None,
+                   node_affinity=None, label_selector=None) -> Optional[NodeInfo]:
         """Hybrid policy (reference: policy/hybrid_scheduling_policy.h:28):
         sort feasible nodes by load and pick uniformly among the top k
         (top-k randomization avoids herding every scheduler decision
-        onto one node; reference scheduler_top_k_fraction)."""
+        onto one node; reference scheduler_top_k_fraction). Node
+        affinity / label selectors constrain the candidate set first
+        (reference: node_affinity_scheduling_policy.cc, node-label
+        policy)."""
         cands = [n for n in self._alive_nodes() if n.node_id not in exclude and self._fits(n, req)]
+        if node_affinity is not None:
+            target_hex, soft = node_affinity
+            pinned = [n for n in cands if n.node_id.hex() == target_hex]
+            if pinned:
+                return pinned[0]
+            if not soft:
+                return None  # hard affinity: wait for that node or fail
+        if label_selector:
+            hard = dict(label_selector.get("hard") or {})
+            soft_l = dict(label_selector.get("soft") or {})
+            cands = [
+                n for n in cands
+                if all((n.labels or {}).get(k) == v for k, v in hard.items())
+            ]
+            soft_match = [
+                n for n in cands
+                if all((n.labels or {}).get(k) == v for k, v in soft_l.items())
+            ]
+            if soft_match:
+                cands = soft_match
         if not cands:
             return None
         if soft_affinity is not None:
@@ -372,6 +396,8 @@ class GcsServer:
                     req,
                     strategy=a.spec.get("scheduling_strategy", "hybrid"),
                     soft_affinity=a.spec.get("pg_node"),
+                    node_affinity=a.spec.get("node_affinity"),
+                    label_selector=a.spec.get("label_selector"),
                 )
             if node is None:
                 if time.time() > deadline:

@@ -81,7 +81,12 @@ class Raylet:
         self._starting = 0
         self._lease_seq = 0
         self.leases: Dict[int, Lease] = {}
-        self._pending: deque = deque()  # (req payload, future)
+        # fair dispatch across scheduling classes (reference: fairness
+        # by scheduling class in the local lease manager): one FIFO per
+        # resource-shape class, round-robin between classes — a large
+        # request at one class's head no longer blocks other classes
+        self._pending: Dict[tuple, deque] = {}
+        self._class_order: deque = deque()
         self._worker_ready: Dict[int, asyncio.Future] = {}  # pid -> fut
         self._actor_start_futs: Dict[bytes, asyncio.Future] = {}
         self._proto_worker: Dict[int, WorkerProc] = {}
@@ -142,7 +147,7 @@ class Raylet:
         while True:
             snap = dict(self.avail)
             now = time.time()
-            pending = len(self._pending)
+            pending = self._pending_total()
             if snap != last or now - last_time > 1.0:
                 try:
                     self.gcs.notify(
@@ -286,14 +291,26 @@ class Raylet:
             if bkey is None:
                 return {"error": "placement group bundle not found"}
         if pg is None and not self._feasible_total(req):
+            if p.get("no_spill"):
+                # hard node-affinity / label pin: never redirect
+                return {"error":
+                        f"infeasible resource request {req} on pinned node"}
             spill = await self._find_spill_target(req)
             if spill:
                 return {"spill": spill}
             return {"error": f"infeasible resource request {req} on this cluster"}
         fut = asyncio.get_running_loop().create_future()
-        self._pending.append((req, pg, fut))
+        key = (tuple(sorted(req.items())), pg)
+        q = self._pending.get(key)
+        if q is None:
+            q = self._pending[key] = deque()
+            self._class_order.append(key)
+        q.append((req, pg, fut))
         self._try_grant()
         return await fut
+
+    def _pending_total(self) -> int:
+        return sum(len(q) for q in self._pending.values())
 
     def _bundle_key(self, pg):
         pg_id, idx = pg
@@ -343,70 +360,88 @@ class Raylet:
             self._gpu_free[d] = min(1.0, self._gpu_free.get(d, 0.0) + frac)
 
     def _try_grant(self):
-        made_progress = True
-        while made_progress and self._pending:
-            made_progress = False
-            req, pg, fut = self._pending[0]
-            if fut.done():
-                self._pending.popleft()
-                made_progress = True
-                continue
-            pool_avail = self.avail
-            bundle = None
-            if pg is not None:
-                bkey = self._bundle_key(pg)
-                if bkey is None:
-                    self._pending.popleft()
-                    fut.set_result({"error": "placement group bundle lost"})
+        """Round-robin across scheduling classes; within a class, FIFO."""
+        progress = True
+        while progress and self._class_order:
+            progress = False
+            for _ in range(len(self._class_order)):
+                if not self._class_order:
+                    break
+                key = self._class_order[0]
+                self._class_order.rotate(-1)
+                q = self._pending.get(key)
+                if not q:
+                    self._pending.pop(key, None)
+                    try:
+                        self._class_order.remove(key)
+                    except ValueError:
+                        pass
                     continue
-                bundle = self.bundles[bkey]
-                pool_avail = bundle["avail"]
-            if not self._fits(pool_avail, req):
-                break
-            if not self._idle_task_workers:
-                # bounded pool (reference: worker_pool.h soft limit):
-                # never exceed what the CPU resource could run anyway
-                n_task_workers = sum(
-                    1 for w in self.workers.values() if w.kind == "task"
-                )
-                cap = int(max(self.resources_total.get("CPU", 1)
-                              * _cfg.worker_cap_factor, 8))
-                if (
-                    self._starting < min(len(self._pending), 4)
-                    and n_task_workers + self._starting < cap
-                ):
-                    self._spawn_worker()
-                break
-            w = self._idle_task_workers.popleft()
-            if w.proc.poll() is not None or w.proto is None:
-                made_progress = True
-                continue
-            ngpu = float(req.get("GPU", 0))
-            gpu_alloc = self._gpu_alloc(ngpu)
-            if gpu_alloc is None:
-                # resource accounting says it fits but no device has the
-                # fraction free (fragmentation) — keep the request queued
-                self._idle_task_workers.appendleft(w)
-                break
-            self._pending.popleft()
-            if bundle is None:
-                for k, v in req.items():
-                    self.avail[k] = self.avail.get(k, 0) - v
-            else:
-                for k, v in req.items():
-                    bundle["avail"][k] = bundle["avail"].get(k, 0) - v
-            self._lease_seq += 1
-            lease = Lease(self._lease_seq, req, w, gpu_alloc, pg)
-            self.leases[lease.lease_id] = lease
-            w.idle = False
-            w.lease_id = lease.lease_id
-            w.gpu_ids = lease.gpu_ids
-            w.gpu_alloc = gpu_alloc
-            fut.set_result(
-                {"addr": w.addr, "lease_id": lease.lease_id,
-                 "gpu_ids": lease.gpu_ids, "raylet": self.addr}
+                if self._grant_head(q):
+                    progress = True
+
+    def _grant_head(self, q: deque) -> bool:
+        """Try to grant the head of one class queue. Returns True if an
+        item was consumed (granted or dropped), False if blocked."""
+        req, pg, fut = q[0]
+        if fut.done():
+            q.popleft()
+            return True
+        pool_avail = self.avail
+        bundle = None
+        if pg is not None:
+            bkey = self._bundle_key(pg)
+            if bkey is None:
+                q.popleft()
+                fut.set_result({"error": "placement group bundle lost"})
+                return True
+            bundle = self.bundles[bkey]
+            pool_avail = bundle["avail"]
+        if not self._fits(pool_avail, req):
+            return False
+        if not self._idle_task_workers:
+            # bounded pool (reference: worker_pool.h soft limit):
+            # never exceed what the CPU resource could run anyway
+            n_task_workers = sum(
+                1 for w in self.workers.values() if w.kind == "task"
             )
-            made_progress = True
+            cap = int(max(self.resources_total.get("CPU", 1)
+                          * _cfg.worker_cap_factor, 8))
+            if (
+                self._starting < min(self._pending_total(), 4)
+                and n_task_workers + self._starting < cap
+            ):
+                self._spawn_worker()
+            return False
+        w = self._idle_task_workers.popleft()
+        if w.proc.poll() is not None or w.proto is None:
+            return True
+        ngpu = float(req.get("GPU", 0))
+        gpu_alloc = self._gpu_alloc(ngpu)
+        if gpu_alloc is None:
+            # resource accounting says it fits but no device has the
+            # fraction free (fragmentation) — keep the request queued
+            self._idle_task_workers.appendleft(w)
+            return False
+        q.popleft()
+        if bundle is None:
+            for k, v in req.items():
+                self.avail[k] = self.avail.get(k, 0) - v
+        else:
+            for k, v in req.items():
+                bundle["avail"][k] = bundle["avail"].get(k, 0) - v
+        self._lease_seq += 1
+        lease = Lease(self._lease_seq, req, w, gpu_alloc, pg)
+        self.leases[lease.lease_id] = lease
+        w.idle = False
+        w.lease_id = lease.lease_id
+        w.gpu_ids = lease.gpu_ids
+        w.gpu_alloc = gpu_alloc
+        fut.set_result(
+            {"addr": w.addr, "lease_id": lease.lease_id,
+             "gpu_ids": lease.gpu_ids, "raylet": self.addr}
+        )
+        return True
 
     def _release_resources(self, lease: Lease):
         if lease.pg is not None:
@@ -590,7 +625,7 @@ class Raylet:
             self.gcs.notify(
                 "report_resources",
                 {"node_id": self.node_id, "available": dict(self.avail),
-                 "pending": len(self._pending),
+                 "pending": self._pending_total(),
                  "total": dict(self.resources_total)},
             )
         except Exception:

@@ -182,10 +182,13 @@ class _Lease:
 
 
 class _LeasePool:
-    def __init__(self, key, resources, pg):
+    def __init__(self, key, resources, pg, node_affinity=None,
+                 label_selector=None):
         self.key = key
         self.resources = resources
         self.pg = pg  # (pg_id, bundle_index) or None
+        self.node_affinity = node_affinity  # (node_id_hex, soft) or None
+        self.label_selector = label_selector  # {"hard":…, "soft":…} or None
         self.leases: List[_Lease] = []
         self.queue: deque = deque()  # pending task dispatch callables
         self.requests_in_flight = 0
@@ -1172,7 +1175,15 @@ class CoreRuntime:
         pg_key = None
         if pg is not None:
             pg_key = (pg[0], pg[1])
-        return (tuple(sorted(res.items())), pg_key), res, pg_key
+        na = options.get("node_affinity")
+        ls = options.get("label_selector")
+        ls_key = None
+        if ls:
+            ls_key = (tuple(sorted(ls.get("hard", {}).items())),
+                      tuple(sorted(ls.get("soft", {}).items())))
+        key = (tuple(sorted(res.items())), pg_key,
+               tuple(na) if na else None, ls_key)
+        return key, res, pg_key
 
     async def _submit_with_retries(self, spec, options, retries, captured_ids):
         try:
@@ -1241,7 +1252,11 @@ class CoreRuntime:
         key, res, pg_key = self._pool_key(options)
         pool = self._pools.get(key)
         if pool is None:
-            pool = self._pools[key] = _LeasePool(key, res, pg_key)
+            pool = self._pools[key] = _LeasePool(
+                key, res, pg_key,
+                node_affinity=options.get("node_affinity"),
+                label_selector=options.get("label_selector"),
+            )
         lease = await self._acquire_lease(pool, spec)
         # cancels that landed while we waited for the lease win here
         # (reference: queued tasks are cancellable until dispatch)
@@ -1284,6 +1299,49 @@ class CoreRuntime:
             asyncio.ensure_future(self._request_lease(pool))
         return await fut
 
+    async def _target_raylet(self, pool: _LeasePool):
+        """Resolve the raylet to lease from for node-affinity / label
+        strategies (reference: node_affinity_scheduling_policy.cc,
+        node-label policy). Returns (client, no_spill)."""
+        na = pool.node_affinity
+        ls = pool.label_selector
+        if na is None and ls is None:
+            return self.raylet, False
+        nodes = await self._gcs_rpc("node_table", {})
+        alive = [n for n in nodes if n["alive"]]
+        if na is not None:
+            target_hex, soft = na
+            for n in alive:
+                if bytes(n["node_id"]).hex() == target_hex:
+                    return await self._conn(n["addr"]), not soft
+            if soft:
+                return self.raylet, False
+            raise exc.RaySystemError(
+                f"node affinity target {target_hex} is not alive"
+            )
+        hard = dict(ls.get("hard") or {})
+        soft_l = dict(ls.get("soft") or {})
+        cands = [
+            n for n in alive
+            if all(n["labels"].get(k) == v for k, v in hard.items())
+        ]
+        if not cands:
+            raise exc.RaySystemError(
+                f"no alive node matches label selector {hard}"
+            )
+        soft_match = [
+            n for n in cands
+            if all(n["labels"].get(k) == v for k, v in soft_l.items())
+        ]
+        pick_from = soft_match or cands
+
+        def load(n):
+            t = n["resources_total"].get("CPU", 1.0) or 1.0
+            return 1.0 - n["resources_available"].get("CPU", 0.0) / t
+
+        best = min(pick_from, key=load)
+        return await self._conn(best["addr"]), True
+
     async def _request_lease(self, pool: _LeasePool):
         try:
             req = {
@@ -1292,7 +1350,9 @@ class CoreRuntime:
             }
             if pool.pg is not None:
                 req["pg_id"], req["bundle_index"] = pool.pg
-            raylet = self.raylet
+            raylet, no_spill = await self._target_raylet(pool)
+            if no_spill:
+                req["no_spill"] = True
             for _hop in range(8):
                 r = await raylet.call("request_lease", req)
                 if r.get("spill"):
@@ -1430,6 +1490,8 @@ class CoreRuntime:
                 "env_vars": (options.get("runtime_env") or {}).get("env_vars"),
                 "working_dir": (options.get("runtime_env") or {}).get("working_dir"),
                 "py_modules": (options.get("runtime_env") or {}).get("py_modules"),
+                "node_affinity": options.get("node_affinity"),
+                "label_selector": options.get("label_selector"),
             }
             if pg is not None:
                 payload["pg_id"], payload["bundle_index"] = pg[0], pg[1]

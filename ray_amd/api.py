@@ -294,6 +294,16 @@ def _normalize_scheduling(opts: dict) -> dict:
             pg.id,
             getattr(strat, "placement_group_bundle_index", None),
         )
+    elif strat is not None and hasattr(strat, "node_id"):
+        # NodeAffinitySchedulingStrategy (reference:
+        # policy/node_affinity_scheduling_policy.cc)
+        opts["node_affinity"] = (str(strat.node_id), bool(strat.soft))
+    elif strat is not None and hasattr(strat, "hard"):
+        # NodeLabelSchedulingStrategy
+        opts["label_selector"] = {
+            "hard": dict(strat.hard or {}),
+            "soft": dict(strat.soft or {}),
+        }
     elif opts.get("placement_group") is not None and not isinstance(
         opts.get("placement_group"), tuple
     ):
