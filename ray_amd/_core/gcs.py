@@ -460,6 +460,7 @@ class GcsServer:
             "spec_kv_key": a.spec.get("spec_kv_key"),
             "working_dir": a.spec.get("working_dir"),
             "py_modules": a.spec.get("py_modules"),
+            "tensor_transport": a.spec.get("tensor_transport"),
         }
 
     async def h_actor_exit(self, conn, p):

@@ -1492,6 +1492,7 @@ class CoreRuntime:
                 "py_modules": (options.get("runtime_env") or {}).get("py_modules"),
                 "node_affinity": options.get("node_affinity"),
                 "label_selector": options.get("label_selector"),
+                "tensor_transport": options.get("tensor_transport"),
             }
             if pg is not None:
                 payload["pg_id"], payload["bundle_index"] = pg[0], pg[1]

@@ -154,6 +154,7 @@ class WorkerMain:
             )
             import cloudpickle
 
+            self.tensor_transport = r.get("tensor_transport")
             cls, self._max_concurrency = cloudpickle.loads(pickled_cls)
             if self._max_concurrency > 1:
                 self.executor = concurrent.futures.ThreadPoolExecutor(
@@ -498,6 +499,16 @@ class WorkerMain:
                 )
             values = list(result)
         results = []
+        if getattr(self, "tensor_transport", None):
+            # tensor_transport actors (reference: actor.py:621): GPU
+            # tensors in returns ship as auto-fetching GPU-store refs
+            # (zero-copy hipIpc same-node) instead of host-staged bytes
+            from ray_amd.experimental import rdt as _rdt
+
+            values = [
+                _rdt.offload_tensors(v)[0] if _rdt.has_cuda_tensors(v) else v
+                for v in values
+            ]
         for oid, v in zip(spec["returns"], values):
             try:
                 meta, buffers, _ = self.rt._serialize_capture(v)

@@ -344,6 +344,15 @@ class ActorMethod:
         merged = {**self._defaults, **opts}
         merged.setdefault("num_returns", 1)
         num_returns = merged["num_returns"]
+        if self._handle._tensor_transport:
+            # sender-side offload (reference: rdt_manager __ray_send__):
+            # GPU tensors in args ship as GPU-store refs
+            from .experimental import rdt as _rdt
+
+            if _rdt.has_cuda_tensors((args, kwargs)):
+                args, _ = _rdt.offload_tensors(list(args))
+                args = tuple(args)
+                kwargs, _ = _rdt.offload_tensors(kwargs)
         refs = rt.submit_actor_task(
             self._handle._actor_id,
             self._name,
@@ -359,10 +368,11 @@ class ActorMethod:
 
 class ActorHandle:
     def __init__(self, actor_id: bytes, class_name: str = "Actor",
-                 method_options=None):
+                 method_options=None, tensor_transport=None):
         self._actor_id = actor_id
         self._class_name = class_name
         self._method_options = method_options or {}
+        self._tensor_transport = tensor_transport
 
     def __getattr__(self, item):
         if item.startswith("_") and item not in (
@@ -376,7 +386,8 @@ class ActorHandle:
 
     def __reduce__(self):
         return (_deserialize_handle,
-                (self._actor_id, self._class_name, self._method_options))
+                (self._actor_id, self._class_name, self._method_options,
+                 self._tensor_transport))
 
     def __hash__(self):
         return hash(self._actor_id)
@@ -390,8 +401,10 @@ class ActorHandle:
         return self._actor_id
 
 
-def _deserialize_handle(actor_id, class_name, method_options=None):
-    return ActorHandle(actor_id, class_name, method_options)
+def _deserialize_handle(actor_id, class_name, method_options=None,
+                        tensor_transport=None):
+    return ActorHandle(actor_id, class_name, method_options,
+                       tensor_transport)
 
 
 class ActorClass:
@@ -434,7 +447,8 @@ class ActorClass:
         pickled, key = self._ensure_pickled(mc)
         actor_id = rt.create_actor(key, pickled, opts, (args, kwargs))
         return ActorHandle(actor_id, self._cls.__name__,
-                           self._method_options())
+                           self._method_options(),
+                           opts.get("tensor_transport"))
 
     def _method_options(self) -> dict:
         """Collect @ray.method(...) per-method option dicts off the class

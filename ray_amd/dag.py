@@ -164,6 +164,8 @@ def _dag_loop(instance, specs):
     one loop per participating actor)."""
     from ray_amd.experimental.channel import Channel, ChannelReader
 
+    from ray_amd.experimental import rdt as _rdt
+
     states = []
     for spec in specs:
         readers = {
@@ -171,7 +173,13 @@ def _dag_loop(instance, specs):
             for name, (path, slot) in spec["reads"].items()
         }
         out = Channel(spec["out_path"]) if spec.get("out_path") else None
-        states.append((spec, readers, out))
+        # per-out-channel history of offloaded GPU tensors: the channel
+        # ack protocol guarantees the reader finished computing with
+        # iteration N-1's tensors by the time we WRITE iteration N+1
+        # (it acked N before then), so a 2-deep history is the earliest
+        # safe free point (accelerator-channel lifetime rule; reference:
+        # torch_tensor_accelerator_channel.py buffer reuse)
+        states.append((spec, readers, out, []))
 
     def render(t, vals):
         kind = t[0]
@@ -188,9 +196,17 @@ def _dag_loop(instance, specs):
             return in_args[t[1]]
         return in_kwargs[t[1]]  # input_key
 
+    def _free_refs(refs):
+        store = _rdt.get_gpu_object_store()
+        for ref in refs:
+            try:
+                store.free(ref)
+            except Exception:
+                pass
+
     while True:
         stopping = False
-        for spec, readers, out in states:
+        for spec, readers, out, gpu_hist in states:
             vals = {}
             stop = False
             err = None
@@ -205,6 +221,9 @@ def _dag_loop(instance, specs):
             if stop:
                 if out is not None:
                     out.write_obj(("stop", None))
+                for refs in gpu_hist:
+                    _free_refs(refs)
+                gpu_hist.clear()
                 stopping = True
                 continue
             if err is not None:
@@ -218,6 +237,12 @@ def _dag_loop(instance, specs):
                 }
                 r = getattr(instance, spec["method"])(*args, **kwargs)
                 if out is not None:
+                    if _rdt.has_cuda_tensors(r):
+                        # GPU edge: ship hipIpc refs, not host bytes
+                        r, refs = _rdt.offload_tensors(r)
+                        gpu_hist.append(refs)
+                        if len(gpu_hist) > 2:
+                            _free_refs(gpu_hist.pop(0))
                     out.write_obj(("data", r))
             except BaseException as e:  # noqa
                 if out is not None:

@@ -62,6 +62,7 @@ class GpuObjectStore:
         self._rt = _rt.global_runtime()
         self._tensors: Dict[bytes, "torch.Tensor"] = {}
         self._cache: Dict[bytes, "torch.Tensor"] = {}
+        self._events: Dict[bytes, object] = {}
         self._rt.server.route("rdt_fetch", self._h_fetch)
         self._rt.server.route("rdt_free", self._h_free)
 
@@ -87,10 +88,23 @@ class GpuObjectStore:
         if t is None:
             return None
         if p.get("mode") == "ipc" and t.is_cuda:
-            # export hipIpc handle; sync so the consumer sees final data
-            torch.cuda.synchronize(t.device)
+            # export hipIpc handle; producer-stream ORDER via an
+            # interprocess hipEvent — the consumer's stream waits on it
+            # instead of a full-device synchronize here (reference:
+            # cuda_ipc_transport.py:71-101 IPC event sync)
+            ev_handle = None
+            try:
+                ev = torch.cuda.Event(interprocess=True)
+                ev.record(torch.cuda.current_stream(t.device))
+                ev_handle = ev.ipc_handle()
+                self._events[oid] = ev  # keep alive until freed
+            except Exception:
+                torch.cuda.synchronize(t.device)
             func, args = reductions.reduce_tensor(t)
-            return {"mode": "ipc", "payload": cloudpickle.dumps((func, args))}
+            return {"mode": "ipc",
+                    "payload": cloudpickle.dumps((func, args)),
+                    "event": ev_handle,
+                    "device": t.device.index or 0}
         # staged fallback: device -> host -> bytes
         cpu = t.detach().cpu().contiguous()
         return {
@@ -102,11 +116,13 @@ class GpuObjectStore:
         }
 
     def _h_free(self, conn, p):
-        self._tensors.pop(bytes(p["id"]), None)
+        oid = bytes(p["id"])
+        self._tensors.pop(oid, None)
+        self._events.pop(oid, None)
 
     # ---------------- consumer side ----------------
 
-    def get(self, ref: GpuObjectRef, device=None):
+    def get(self, ref: GpuObjectRef, device=None, consume: bool = False):
         import torch
 
         if ref.id in self._tensors:  # we are the owner
@@ -132,6 +148,12 @@ class GpuObjectStore:
         if r["mode"] == "ipc":
             func, args = cloudpickle.loads(r["payload"])
             t = func(*args)
+            if r.get("event") is not None:
+                # order after the producer's stream without host sync
+                ev = torch.cuda.Event.from_ipc_handle(
+                    r.get("device", 0), r["event"]
+                )
+                torch.cuda.current_stream(t.device).wait_event(ev)
         else:
             arr, dtype_str = cloudpickle.loads(r["payload"])
             t = torch.from_numpy(arr.copy())
@@ -142,6 +164,8 @@ class GpuObjectStore:
             if device is not None:
                 t = t.to(device)
         self._cache[ref.id] = t
+        if consume:
+            self.free(ref)
         return t
 
     def free(self, ref: GpuObjectRef):
@@ -169,3 +193,71 @@ def get_gpu_object_store() -> GpuObjectStore:
         if _store is None:
             _store = GpuObjectStore()
         return _store
+
+
+# --------------------------------------------------------------------------
+# tensor_transport integration (reference: actor.py:621 tensor_transport
+# option + rdt_manager.py __ray_send__/__ray_recv__ injection): GPU
+# tensors in actor-call args/returns (and compiled-DAG channel payloads)
+# are replaced with auto-fetching refs, so same-node consumers get a
+# zero-copy hipIpc view instead of a host-staged copy.
+# --------------------------------------------------------------------------
+
+
+def _fetch_on_load(oid, owner_addr, node_id, shape, dtype, device, consume):
+    ref = GpuObjectRef(oid, owner_addr, node_id, shape, dtype, device)
+    return get_gpu_object_store().get(ref, consume=consume)
+
+
+class _AutoFetchRef:
+    """Pickles as a thunk that fetches the tensor wherever it is
+    deserialized (driver or another actor)."""
+
+    __slots__ = ("ref", "consume")
+
+    def __init__(self, ref: GpuObjectRef, consume: bool = False):
+        self.ref = ref
+        self.consume = consume
+
+    def __reduce__(self):
+        r = self.ref
+        return (_fetch_on_load,
+                (r.id, r.owner_addr, r.node_id, r.shape, r.dtype, r.device,
+                 self.consume))
+
+
+def offload_tensors(value, consume: bool = False, _refs=None):
+    """Recursively replace CUDA torch tensors in value with
+    auto-fetching GPU-store refs. Returns (new_value, refs)."""
+    if _refs is None:
+        _refs = []
+    try:
+        import torch
+    except ImportError:
+        return value, _refs
+    if isinstance(value, torch.Tensor) and value.is_cuda:
+        ref = get_gpu_object_store().put(value)
+        _refs.append(ref)
+        return _AutoFetchRef(ref, consume), _refs
+    if isinstance(value, (list, tuple)):
+        out = [offload_tensors(v, consume, _refs)[0] for v in value]
+        return (type(value)(out) if not isinstance(value, tuple)
+                else tuple(out)), _refs
+    if isinstance(value, dict):
+        return {k: offload_tensors(v, consume, _refs)[0]
+                for k, v in value.items()}, _refs
+    return value, _refs
+
+
+def has_cuda_tensors(value) -> bool:
+    try:
+        import torch
+    except ImportError:
+        return False
+    if isinstance(value, torch.Tensor):
+        return value.is_cuda
+    if isinstance(value, (list, tuple)):
+        return any(has_cuda_tensors(v) for v in value)
+    if isinstance(value, dict):
+        return any(has_cuda_tensors(v) for v in value.values())
+    return False
