@@ -698,12 +698,12 @@ class CoreRuntime:
 
     # ------------- put / get / wait -------------
 
-    def put(self, value, _owner=None) -> ObjectRef:
+    def put(self, value, _owner=None, _force_store: bool = False) -> ObjectRef:
         oid = ids.new_object_id()
         meta, buffers, captured = self._serialize_capture(value)
         size = serialization.serialized_size(meta, buffers)
         ref = ObjectRef(oid, self.addr)
-        if size <= serialization.INLINE_MAX:
+        if size <= serialization.INLINE_MAX and not _force_store:
             blob = bytearray(size)
             n = serialization.write_to(memoryview(blob), meta, buffers)
             self._store_put_threadsafe(oid, ("val_ser", bytes(blob[:n])))
