@@ -96,6 +96,11 @@ class Raylet:
         # [0,1] per device id; num_gpus=0.5 shares one device.
         ngpus = int(resources.get("GPU", 0))
         self._gpu_free: Dict[int, float] = {i: 1.0 for i in range(ngpus)}
+        # reader pins: oid -> set(conn ids); per-conn reverse index so a
+        # dead worker's pins are swept on disconnect
+        self._pins: Dict[bytes, set] = {}
+        self._conn_pins: Dict[int, set] = {}
+        self._deferred_free: set = set()
         # placement-group bundles: (pg_id, idx) -> {"resources", "avail", "committed"}
         self.bundles: Dict[tuple, dict] = {}
         self._node_cache: List[dict] = []
@@ -106,7 +111,7 @@ class Raylet:
             "free_objects pull_object fetch_chunk object_stats start_actor "
             "actor_ready actor_failed kill_worker reserve_bundle commit_bundle "
             "rollback_bundle remove_bundle node_info ping prestart_workers "
-            "report_task_events"
+            "report_task_events pin_object unpin_object try_recycle"
         ).split():
             self.server.route(m, getattr(self, "h_" + m))
         self.server.on_conn_lost = self._conn_lost
@@ -219,6 +224,9 @@ class Raylet:
         return {"ok": True, "node_id": self.node_id}
 
     def _conn_lost(self, proto, exc_):
+        for oid in list(self._conn_pins.get(id(proto), ())):
+            self._unpin(oid, id(proto))
+        self._conn_pins.pop(id(proto), None)
         w = self._proto_worker.pop(id(proto), None)
         if w is None:
             return
@@ -649,7 +657,58 @@ class Raylet:
         return {"ok": ok, "size": self.store.table.get(oid, [0])[0]}
 
     def h_free_objects(self, conn, p):
-        self.store.free([bytes(i) for i in p["ids"]])
+        ids_ = [bytes(i) for i in p["ids"]]
+        now, deferred = [], []
+        for oid in ids_:
+            (deferred if self._pins.get(oid) else now).append(oid)
+        if now:
+            self.store.free(now)
+        for oid in deferred:
+            # pin-while-mapped (closes the round-1 recycle race,
+            # reference: object pinning in plasma's lifecycle manager):
+            # a reader still maps this segment; unlink when it unpins
+            self._deferred_free.add(oid)
+
+    # ---- mapping pins (readers register while they hold an mmap) ----
+
+    def h_pin_object(self, conn, p):
+        oid = bytes(p["id"])
+        self._pins.setdefault(oid, set()).add(id(conn))
+        self._conn_pins.setdefault(id(conn), set()).add(oid)
+
+    def h_unpin_object(self, conn, p):
+        self._unpin(bytes(p["id"]), id(conn))
+
+    def _unpin(self, oid: bytes, conn_id: int):
+        s = self._pins.get(oid)
+        if s is not None:
+            s.discard(conn_id)
+            if not s:
+                self._pins.pop(oid, None)
+                if oid in self._deferred_free:
+                    self._deferred_free.discard(oid)
+                    self.store.free([oid])
+        cp = self._conn_pins.get(conn_id)
+        if cp is not None:
+            cp.discard(oid)
+
+    def h_try_recycle(self, conn, p):
+        """Owner asks to take the freed segment into its hot pool.
+        Granted only when no reader pins the mapping; otherwise the
+        unlink is deferred to the last unpin and the pool gets
+        nothing (correctness over recycling)."""
+        oid = bytes(p["id"])
+        if self._pins.get(oid):
+            self._deferred_free.add(oid)
+            return {"ok": False}
+        ent = self.store.table.pop(oid, None)
+        if ent is None or ent[1]:
+            if ent is not None:
+                self.store.table[oid] = ent
+                self.store.free([oid])  # spilled: normal free
+            return {"ok": False}
+        self.store.used -= ent[0]
+        return {"ok": True}
 
     async def h_pull_object(self, conn, p):
         """Pull an object from a remote node's store into ours, chunked

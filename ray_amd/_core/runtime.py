@@ -505,6 +505,7 @@ class CoreRuntime:
         if owner == self.addr:
             self._free_owned(oid)
         elif owner and not self._closed:
+            self._mmaps.pop(oid, None)
             self._notify_owner(owner, "borrow_del", oid)
 
     def _notify_owner(self, owner_addr: str, method: str, oid: bytes):
@@ -632,15 +633,31 @@ class CoreRuntime:
         mapped = self._mmaps.pop(oid, None)
         self._drop_lineage_for(oid)
         if ent is not None and ent[0] == "store" and not self._closed:
-            # recycle same-node segments into the hot pool (plasma-arena
-            # equivalent) unless we still hold a mapping of it
-            if ent[1] == self.raylet_addr and mapped is None:
-                path = store.shm_path(self.shm_dir, oid)
+            if ent[1] == self.raylet_addr:
+                # recycle same-node segments into the hot pool
+                # (plasma-arena equivalent) — but ONLY with the
+                # raylet's blessing: try_recycle is refused while any
+                # reader still pins a mapping of the segment (the
+                # round-1 recycle-while-mapped race, now closed)
+                async def _recycle():
+                    try:
+                        r = await self.raylet.call("try_recycle",
+                                                   {"id": oid})
+                    except Exception:
+                        return
+                    if r and r.get("ok"):
+                        path = store.shm_path(self.shm_dir, oid)
+                        try:
+                            fsize = os.path.getsize(path)
+                            store._segment_pool.release(path, fsize)
+                        except OSError:
+                            pass
+
                 try:
-                    fsize = os.path.getsize(path)
-                    store._segment_pool.release(path, fsize)
-                except OSError:
+                    self._run(_recycle())
+                except Exception:
                     pass
+                return
 
             async def _free():
                 try:
@@ -796,6 +813,28 @@ class CoreRuntime:
         except OSError:
             raise exc.ObjectLostError(oid.hex())
         self._mmaps[oid] = mo
+        # pin-while-mapped (closes the recycle race): the raylet defers
+        # unlink/recycle of this segment until the MAPPING dies — the
+        # unpin fires from a GC finalizer, not at ref-drop, because
+        # zero-copy views handed to user code can outlive the ref
+        try:
+            self.raylet.notify("pin_object", {"id": oid})
+            import weakref as _weakref
+
+            rt = self
+
+            def _unpin(oid=oid, rt=rt):
+                try:
+                    rt.loop.call_soon_threadsafe(
+                        lambda: rt.raylet.notify("unpin_object",
+                                                 {"id": oid})
+                    )
+                except Exception:
+                    pass
+
+            _weakref.finalize(mo, _unpin)
+        except Exception:
+            pass
         value = serialization.loads_from(mo.view)
         self.memory_store[oid] = ("val", value)
         return value

@@ -706,3 +706,42 @@ def test_pubsub_multi_subscriber_fanout(ray_start_regular):
         assert pubsub.publish("fan", 7) >= 1
         assert a.poll(timeout=10) == 7
         assert b.poll(timeout=10) == 7
+
+
+def test_pin_while_mapped_blocks_recycle(ray_start_regular):
+    """Round-1 known limit, now closed: a consumer holding a ZERO-COPY
+    view of a sealed segment pins it at the raylet; the owner's free
+    must not recycle the segment into the hot pool (a later put of the
+    same size class would overwrite the consumer's live view)."""
+
+    @ray.remote
+    class Viewer:
+        def __init__(self):
+            self.view = None
+
+        def hold_view(self, ref_box):
+            # get() maps the segment; keep the zero-copy numpy view
+            self.view = ray.get(ref_box[0])
+            return float(self.view[:100].sum())
+
+        def re_read(self):
+            return float(self.view[:100].sum())
+
+    v = Viewer.remote()
+    size = 512 * 1024  # one pool size-class exactly
+    a = np.arange(size // 8, dtype=np.float64)
+    ref = ray.put(a)
+    want = float(a[:100].sum())
+    assert ray.get(v.hold_view.remote([ref]), timeout=60) == want
+    # owner drops its ref -> free path runs; the viewer actor dropped
+    # its REF (ref_box was transient) but still holds the mapped VIEW
+    del ref, a
+    import gc
+
+    gc.collect()
+    time.sleep(1.0)
+    # hammer same-size-class puts: with the race, one of these would
+    # recycle the viewer's segment and overwrite its bytes
+    spam = [ray.put(np.full(size // 8, 7.0)) for _ in range(8)]
+    assert ray.get(v.re_read.remote(), timeout=60) == want
+    del spam
