@@ -278,6 +278,30 @@ class ReplicaActor:
         finally:
             self._ongoing -= 1
 
+    async def handle_grpc(self, method: str, data: bytes) -> bytes:
+        """gRPC ingress entry (reference: replica gRPC user-method
+        dispatch): the named method (or __call__) gets a GrpcRequest;
+        bytes results pass through, str encodes, other values JSON."""
+        self._ongoing += 1
+        self._total += 1
+        try:
+            req = GrpcRequest(method, data)
+            target = getattr(self._callable, method, None)
+            if target is None:
+                target = self._callable
+            r = target(req)
+            if asyncio.iscoroutine(r):
+                r = await r
+            if isinstance(r, (bytes, bytearray)):
+                return bytes(r)
+            if isinstance(r, str):
+                return r.encode()
+            import json
+
+            return json.dumps(r, default=str).encode()
+        finally:
+            self._ongoing -= 1
+
     def get_stats(self):
         return {"ongoing": self._ongoing, "total": self._total}
 
@@ -872,19 +896,130 @@ async def _send_response(send, status, headers, body):
 
 
 # --------------------------------------------------------------------------
+# gRPC proxy (reference: _private/proxy.py:555 gRPCProxy)
+# --------------------------------------------------------------------------
+
+
+class GrpcRequest:
+    """What a deployment method receives from the gRPC ingress."""
+
+    def __init__(self, method: str, data: bytes):
+        self.method = method
+        self.data = data
+
+
+class GrpcProxyActor:
+    """gRPC ingress on a generic handler: any unary-unary call to
+    /<app_name>/<Method> with bytes request/response (no compiled proto
+    needed on either side — clients use bytes serializers). Routes the
+    call to the named application's deployment handle; the deployment's
+    method <Method> (or __call__) receives a GrpcRequest.
+
+    Reference: serve gRPCProxy (proxy.py:555), re-designed without the
+    user-servicer registration machinery: the generic handler covers
+    arbitrary service/method names."""
+
+    def __init__(self, port: int):
+        self.port = port
+        self._handles: Dict[str, DeploymentHandle] = {}
+        self._server = None
+
+    async def start_server(self):
+        import grpc
+
+        ray = _ray()
+        outer = self
+
+        class _Generic(grpc.GenericRpcHandler):
+            def service(self, hcd):
+                # method path "/<app>/<Method>"
+                parts = hcd.method.strip("/").split("/")
+                if len(parts) != 2:
+                    return None
+                app_name, method = parts
+
+                async def unary_unary(request: bytes, context):
+                    return await outer._route(app_name, method, request)
+
+                return grpc.unary_unary_rpc_method_handler(
+                    unary_unary,
+                    request_deserializer=None,
+                    response_serializer=None,
+                )
+
+        server = grpc.aio.server()
+        server.add_generic_rpc_handlers((_Generic(),))
+        self.port = server.add_insecure_port(f"127.0.0.1:{self.port}")
+        await server.start()
+        self._server = server
+        return self.port
+
+    async def _route(self, app_name: str, method: str, data: bytes) -> bytes:
+        from ray_amd._core import runtime as _rtmod
+
+        rt = _rtmod.global_runtime()
+        loop = asyncio.get_running_loop()
+        h = self._handles.get(app_name)
+        if h is None:
+            h = self._handles[app_name] = DeploymentHandle(app_name)
+            await loop.run_in_executor(None, h._refresh)
+        if not h._replicas:
+            await loop.run_in_executor(None, h._refresh)
+        last = None
+        for _ in range(2):
+            try:
+                idx = h._pick()
+                replica = h._replicas[idx]
+                ref = replica.handle_grpc.remote(method, data)
+                return (await rt.get_async([ref], 120))[0]
+            except Exception as e:
+                last = e
+                try:
+                    await loop.run_in_executor(None, h._refresh)
+                except Exception:
+                    break
+        raise RuntimeError(f"grpc route to {app_name} failed: {last}")
+
+
+# --------------------------------------------------------------------------
 # module-level API
 # --------------------------------------------------------------------------
 
 _http_port = 8000
+_grpc_port: Optional[int] = None
 _started = False
 
 
-def start(detached: bool = True, http_options: Optional[dict] = None, **kw):
-    global _http_port, _started
+def start(detached: bool = True, http_options: Optional[dict] = None,
+          grpc_options: Optional[dict] = None, **kw):
+    global _http_port, _grpc_port, _started
     if http_options:
         _http_port = http_options.get("port", _http_port)
+    if grpc_options:
+        _grpc_port = grpc_options.get("port", 9000)
     _ensure_controller()
+    if _grpc_port is not None:
+        _ensure_grpc_proxy(_grpc_port)
     _started = True
+
+
+SERVE_GRPC_PROXY_NAME = "SERVE_GRPC_PROXY"
+
+
+def _ensure_grpc_proxy(port: int):
+    ray = _ray()
+    try:
+        return ray.get_actor(SERVE_GRPC_PROXY_NAME, namespace=SERVE_NAMESPACE)
+    except ValueError:
+        P = ray.remote(GrpcProxyActor)
+        proxy = P.options(
+            name=SERVE_GRPC_PROXY_NAME, namespace=SERVE_NAMESPACE,
+            num_cpus=0.1, max_concurrency=64,
+        ).remote(port)
+        bound = ray.get(proxy.start_server.remote(), timeout=30)
+        if not bound:
+            raise RuntimeError("serve gRPC proxy failed to start")
+        return proxy
 
 
 def _ensure_controller():

@@ -386,3 +386,45 @@ def test_autoscale_down_graceful_drain(serve_session):
     assert n == 1
     # service still healthy
     assert h.remote(0.0).result(timeout_s=30) == "ok"
+
+
+def test_grpc_proxy_ingress(serve_session):
+    """gRPC ingress (reference: proxy.py:555 gRPCProxy): a generic
+    bytes-in/bytes-out unary call to /<app>/<Method> routes to the
+    deployment; named method preferred, __call__ fallback."""
+    import grpc
+
+    @serve.deployment
+    class Echo:
+        def Predict(self, req):
+            return b"pred:" + req.data
+
+        def __call__(self, req):
+            return {"method": req.method, "len": len(req.data)}
+
+    serve.start(grpc_options={"port": 0})
+    serve.run(Echo.bind(), name="echoapp", route_prefix="/echoapp",
+              http=False)
+    from ray_amd.serve.api import SERVE_GRPC_PROXY_NAME, SERVE_NAMESPACE
+
+    proxy = ray.get_actor(SERVE_GRPC_PROXY_NAME, namespace=SERVE_NAMESPACE)
+    port = ray.get(
+        proxy.__ray_apply__.remote(lambda self: self.port), timeout=30
+    )
+    channel = grpc.insecure_channel(f"127.0.0.1:{port}")
+    pred = channel.unary_unary(
+        "/echoapp/Predict",
+        request_serializer=None,
+        response_deserializer=None,
+    )
+    assert pred(b"abc", timeout=30) == b"pred:abc"
+    call = channel.unary_unary(
+        "/echoapp/Anything",
+        request_serializer=None,
+        response_deserializer=None,
+    )
+    import json
+
+    out = json.loads(call(b"xyz", timeout=30))
+    assert out == {"method": "Anything", "len": 3}
+    channel.close()
