@@ -92,7 +92,7 @@ class GcsServer:
             "report_resources register_actor resolve_actor actor_exit "
             "kill_actor list_actors next_job_id create_pg pg_wait_ready "
             "remove_pg pg_table ping timeline_events drain_node "
-            "subscribe unsubscribe publish"
+            "subscribe unsubscribe publish debug_state"
         ).split():
             self.server.route(m, getattr(self, "h_" + m))
         self.server.on_conn_lost = self._conn_lost
@@ -231,6 +231,14 @@ class GcsServer:
             n.pending = p.get("pending", 0)
             if p.get("total") is not None:  # dynamic resource change
                 n.resources_total = p["total"]
+
+    def h_debug_state(self, conn, p):
+        return {
+            "handler_stats": self.server.stats_table(),
+            "nodes": len(self.nodes),
+            "actors": len(self.actors),
+            "pgs": len(self.pgs),
+        }
 
     def h_ping(self, conn, p):
         return "pong"

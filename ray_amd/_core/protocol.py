@@ -124,6 +124,28 @@ class RpcServer:
         self._server: Optional[asyncio.AbstractServer] = None
         self._conns = set()
         self.on_conn_lost: Optional[Callable] = None
+        # per-handler latency stats (reference: instrumented_io_context /
+        # event_stats.cc per-handler counts): method -> [count, total_s,
+        # max_s]. Sync handlers measure inline; async handlers measure
+        # the awaited span.
+        self.handler_stats: Dict[str, list] = {}
+
+    def _record(self, method: str, dt: float):
+        st = self.handler_stats.get(method)
+        if st is None:
+            st = self.handler_stats[method] = [0, 0.0, 0.0]
+        st[0] += 1
+        st[1] += dt
+        if dt > st[2]:
+            st[2] = dt
+
+    def stats_table(self) -> Dict[str, dict]:
+        return {
+            m: {"count": c, "total_s": round(t, 6), "mean_us":
+                round(t / c * 1e6, 1) if c else 0.0,
+                "max_us": round(mx * 1e6, 1)}
+            for m, (c, t, mx) in sorted(self.handler_stats.items())
+        }
 
     def route(self, method: str, handler: Handler):
         self._handlers[method] = handler
@@ -152,6 +174,8 @@ class RpcServer:
                 traceback.print_exc()
 
     def _dispatch(self, msg, proto):
+        import time as _time
+
         mtype = msg[0]
         if mtype == MSG_REQUEST:
             _, seq, method, payload = msg
@@ -162,15 +186,18 @@ class RpcServer:
                 proto.send([MSG_ERROR, seq, f"no such method: {method}"])
                 return
             coro_or_val = None
+            t0 = _time.perf_counter()
             try:
                 coro_or_val = h(proto, payload)
             except Exception:
+                self._record(method, _time.perf_counter() - t0)
                 proto.send([MSG_ERROR, seq, traceback.format_exc()])
                 return
             if asyncio.iscoroutine(coro_or_val):
                 task = asyncio.ensure_future(coro_or_val)
 
-                def _done(t, seq=seq, proto=proto):
+                def _done(t, seq=seq, proto=proto, method=method, t0=t0):
+                    self._record(method, _time.perf_counter() - t0)
                     if t.cancelled():
                         return
                     e = t.exception()
@@ -183,12 +210,15 @@ class RpcServer:
 
                 task.add_done_callback(_done)
             else:
+                self._record(method, _time.perf_counter() - t0)
                 proto.send([MSG_REPLY, seq, coro_or_val])
         elif mtype == MSG_NOTIFY:
             _, _, method, payload = msg
             h = self._handlers.get(method)
             if h is not None:
+                t0 = _time.perf_counter()
                 r = h(proto, payload)
+                self._record(method, _time.perf_counter() - t0)
                 if asyncio.iscoroutine(r):
                     asyncio.ensure_future(r)
 

@@ -111,7 +111,8 @@ class Raylet:
             "free_objects pull_object fetch_chunk object_stats start_actor "
             "actor_ready actor_failed kill_worker reserve_bundle commit_bundle "
             "rollback_bundle remove_bundle node_info ping prestart_workers "
-            "report_task_events pin_object unpin_object try_recycle"
+            "report_task_events pin_object unpin_object try_recycle "
+            "debug_state"
         ).split():
             self.server.route(m, getattr(self, "h_" + m))
         self.server.on_conn_lost = self._conn_lost
@@ -655,6 +656,20 @@ class Raylet:
         if ok:
             self.store.ensure_local(oid)
         return {"ok": ok, "size": self.store.table.get(oid, [0])[0]}
+
+    def h_debug_state(self, conn, p):
+        """Per-handler event-loop stats + queue/pool gauges (reference:
+        event_stats.cc periodic DebugString)."""
+        return {
+            "handler_stats": self.server.stats_table(),
+            "pending_leases": self._pending_total(),
+            "workers": len(self.workers),
+            "idle_workers": len(self._idle_task_workers),
+            "leases": len(self.leases),
+            "pins": len(self._pins),
+            "store": dict(zip(("objects", "used", "capacity"),
+                              self.store.stats())),
+        }
 
     def h_free_objects(self, conn, p):
         ids_ = [bytes(i) for i in p["ids"]]

@@ -152,3 +152,16 @@ def _match(row: dict, filters) -> bool:
             if str(have) == str(val):
                 return False
     return True
+
+
+def node_debug_state() -> dict:
+    """Per-handler event-loop stats + queue gauges from the local
+    raylet and the GCS (reference: event_stats.cc DebugString /
+    `ray status -v`)."""
+    from ray_amd._core import runtime as _rtmod
+
+    rt = _rtmod.global_runtime()
+    out = {}
+    out["raylet"] = rt._call_sync(rt.raylet.call("debug_state", {}), 10)
+    out["gcs"] = rt.gcs_call("debug_state", {})
+    return out

@@ -745,3 +745,20 @@ def test_pin_while_mapped_blocks_recycle(ray_start_regular):
     spam = [ray.put(np.full(size // 8, 7.0)) for _ in range(8)]
     assert ray.get(v.re_read.remote(), timeout=60) == want
     del spam
+
+
+def test_handler_event_loop_stats(ray_start_regular):
+    """Per-handler event-loop stats (reference: event_stats.cc)."""
+    @ray.remote
+    def f():
+        return 1
+
+    ray.get([f.remote() for _ in range(5)])
+    from ray_amd.util.state import node_debug_state
+
+    st = node_debug_state()
+    hs = st["raylet"]["handler_stats"]
+    assert "request_lease" in hs and hs["request_lease"]["count"] >= 1
+    assert hs["request_lease"]["mean_us"] > 0
+    assert st["gcs"]["handler_stats"]["report_resources"]["count"] >= 1
+    assert st["raylet"]["store"]["capacity"] > 0
