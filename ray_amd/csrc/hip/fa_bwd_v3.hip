@@ -237,10 +237,12 @@ extern "C" __global__ __launch_bounds__(256) void fa_bwd_dk_v3_bf16(
   const int nq = (T - q_start + FB3_QT - 1) / FB3_QT;
   const int n_tiles = rep * nq;
 
-  short8 q_stage[4], do_stage[4];
-  float lse_stage = 0.f, dsum_stage = 0.f;
-
-#define FB3_DK_LOAD(t_idx)                                                \
+  // dk holds K AND V register-resident plus the dK accumulators —
+  // reg-staged async loads (like dv's) push it to 289 regs = 1
+  // wave/SIMD (measured: the whole-bwd regression). Stage DIRECTLY
+  // global->LDS into the double buffer instead: minimal register
+  // liveness, still one barrier per 64-row tile.
+#define FB3_DK_STAGE(t_idx, buf)                                          \
   do {                                                                    \
     const int g_ = (t_idx) / nq;                                          \
     const int q0_ = q_start + ((t_idx) % nq) * FB3_QT;                    \
@@ -251,6 +253,7 @@ extern "C" __global__ __launch_bounds__(256) void fa_bwd_dk_v3_bf16(
       int i = threadIdx.x + 256 * j;                                      \
       int r = i >> 4;                                                     \
       int c = (i & 15) * 8;                                               \
+      int csw = c ^ ((r & 7) << 3);                                       \
       int qrow = q0_ + r;                                                 \
       short8 qv{0, 0, 0, 0, 0, 0, 0, 0}, dv{0, 0, 0, 0, 0, 0, 0, 0};      \
       if (qrow < T) {                                                     \
@@ -259,40 +262,23 @@ extern "C" __global__ __launch_bounds__(256) void fa_bwd_dk_v3_bf16(
         dv = *reinterpret_cast<const short8*>(qb_ + dO +                  \
                                               (long long)qrow * FB3_D + c); \
       }                                                                   \
-      q_stage[j] = qv;                                                    \
-      do_stage[j] = dv;                                                   \
+      *reinterpret_cast<short8*>(&q_lds[buf][r][csw]) = qv;               \
+      *reinterpret_cast<short8*>(&do_lds[buf][r][csw]) = dv;              \
     }                                                                     \
     if (threadIdx.x < FB3_QT) {                                           \
       int qrow = q0_ + threadIdx.x;                                       \
-      lse_stage = (qrow < T) ? LSE[lb_ + qrow] : INFINITY;                \
-      dsum_stage = (qrow < T) ? Dsum[lb_ + qrow] : 0.f;                   \
+      lse_lds[buf][threadIdx.x] = (qrow < T) ? LSE[lb_ + qrow] : INFINITY; \
+      dsum_lds[buf][threadIdx.x] = (qrow < T) ? Dsum[lb_ + qrow] : 0.f;   \
     }                                                                     \
   } while (0)
 
-#define FB3_DK_WRITE(buf)                                                 \
-  do {                                                                    \
-    _Pragma("unroll") for (int j = 0; j < 4; ++j) {                       \
-      int i = threadIdx.x + 256 * j;                                      \
-      int r = i >> 4;                                                     \
-      int c = (i & 15) * 8;                                               \
-      int csw = c ^ ((r & 7) << 3);                                       \
-      *reinterpret_cast<short8*>(&q_lds[buf][r][csw]) = q_stage[j];       \
-      *reinterpret_cast<short8*>(&do_lds[buf][r][csw]) = do_stage[j];     \
-    }                                                                     \
-    if (threadIdx.x < FB3_QT) {                                           \
-      lse_lds[buf][threadIdx.x] = lse_stage;                              \
-      dsum_lds[buf][threadIdx.x] = dsum_stage;                            \
-    }                                                                     \
-  } while (0)
-
-  FB3_DK_LOAD(0);
-  FB3_DK_WRITE(0);
+  FB3_DK_STAGE(0, 0);
   __syncthreads();
 
   for (int t = 0; t < n_tiles; ++t) {
     const int cur = t & 1;
     const int q0s = q_start + (t % nq) * FB3_QT;
-    if (t + 1 < n_tiles) FB3_DK_LOAD(t + 1);
+    if (t + 1 < n_tiles) FB3_DK_STAGE(t + 1, cur ^ 1);
 
 #pragma unroll
     for (int qt = 0; qt < 2; ++qt) {
@@ -353,7 +339,6 @@ extern "C" __global__ __launch_bounds__(256) void fa_bwd_dk_v3_bf16(
       }
     }
 
-    if (t + 1 < n_tiles) FB3_DK_WRITE(cur ^ 1);
     __syncthreads();
   }
 
