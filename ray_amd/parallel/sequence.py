@@ -58,6 +58,123 @@ def _merge(out_a, lse_a, out_b, lse_b):
     return out, lse
 
 
+
+def _ring_exchange(group, send_to, recv_from, sends, recvs):
+    """Post simultaneous ring sends/recvs; returns waitable reqs."""
+    reqs = []
+    if group is None:
+        for t in sends:
+            reqs.append(dist.isend(t, send_to))
+        for t in recvs:
+            reqs.append(dist.irecv(t, recv_from))
+    else:
+        for i, t in enumerate(sends):
+            reqs.append(group.send([t], send_to, i))
+        for i, t in enumerate(recvs):
+            reqs.append(group.recv([t], recv_from, i))
+    return reqs
+
+
+class _RingAttnFn(torch.autograd.Function):
+    """Fused ring attention with LSE-merging backward.
+
+    Forward: rotate K/V blocks, per-block fused flash fwd (with LSE),
+    exact logsumexp merge. Backward (the standard ring-attention
+    gradient): with the GLOBAL merged LSE, each block's contribution is
+    dS_b = P_b ⊙ (dP_b − D) where P_b = exp(S_b − LSE_global) — exactly
+    what the HIP fa_bwd kernels compute when handed the global LSE and
+    the merged O (Dsum = rowsum(dO ⊙ O)). dK/dV partials rotate around
+    the ring with their blocks and arrive home after `world` hops.
+    """
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal, group, rank, world):
+        from ray_amd import ops
+
+        send_to = (rank + 1) % world
+        recv_from = (rank - 1) % world
+        qc = q.contiguous()
+        cur_k, cur_v = k.contiguous(), v.contiguous()
+        out = None
+        lse = None
+        with torch.no_grad():
+            for step in range(world):
+                src_block = (rank - step) % world
+                if step < world - 1:
+                    nxt_k = torch.empty_like(cur_k)
+                    nxt_v = torch.empty_like(cur_v)
+                    reqs = _ring_exchange(group, send_to, recv_from,
+                                          [cur_k, cur_v], [nxt_k, nxt_v])
+                mode = ("causal" if src_block == rank else
+                        "full" if (not causal or src_block < rank) else
+                        "none")
+                if mode != "none":
+                    o, l = ops.flash_attention(
+                        qc, cur_k, cur_v, causal=(mode == "causal"),
+                        return_lse=True,
+                    )
+                    o = o.float()
+                    l = l.unsqueeze(-1)
+                    if out is None:
+                        out, lse = o, l
+                    else:
+                        out, lse = _merge(out, lse, o, l)
+                if step < world - 1:
+                    for r_ in reqs:
+                        r_.wait()
+                    cur_k, cur_v = nxt_k, nxt_v
+        out_bf = out.to(q.dtype).contiguous()
+        ctx.save_for_backward(qc, k.contiguous(), v.contiguous(), out_bf,
+                              lse.squeeze(-1).contiguous())
+        ctx.causal = causal
+        ctx.ring = (group, rank, world)
+        return out_bf
+
+    @staticmethod
+    def backward(ctx, d_out):
+        from ray_amd.ops import _K
+
+        q, k_own, v_own, out, lse = ctx.saved_tensors
+        causal = ctx.causal
+        group, rank, world = ctx.ring
+        send_to = (rank + 1) % world
+        recv_from = (rank - 1) % world
+        do = d_out.contiguous()
+        dq_acc = torch.zeros_like(q, dtype=torch.float32)
+        # rotating parcel: (k, v, dk_acc, dv_acc)
+        cur_k, cur_v = k_own, v_own
+        cur_dk = torch.zeros_like(k_own, dtype=torch.float32)
+        cur_dv = torch.zeros_like(v_own, dtype=torch.float32)
+        for step in range(world):
+            src_block = (rank - step) % world
+            mode = ("causal" if src_block == rank else
+                    "full" if (not causal or src_block < rank) else
+                    "none")
+            if mode != "none":
+                dq_b, dk_b, dv_b = _K.flash_attn_bwd(
+                    q, cur_k, cur_v, out, do, lse, mode == "causal"
+                )
+                dq_acc += dq_b.float()
+                cur_dk += dk_b.float()
+                cur_dv += dv_b.float()
+            # rotate — including after the last compute step so every
+            # block's gradient parcel arrives back at its owner
+            nxt_k = torch.empty_like(cur_k)
+            nxt_v = torch.empty_like(cur_v)
+            nxt_dk = torch.empty_like(cur_dk)
+            nxt_dv = torch.empty_like(cur_dv)
+            reqs = _ring_exchange(
+                group, send_to, recv_from,
+                [cur_k, cur_v, cur_dk.contiguous(), cur_dv.contiguous()],
+                [nxt_k, nxt_v, nxt_dk, nxt_dv])
+            for r_ in reqs:
+                r_.wait()
+            cur_k, cur_v = nxt_k, nxt_v
+            cur_dk, cur_dv = nxt_dk, nxt_dv
+        return (dq_acc.to(q.dtype), cur_dk.to(k_own.dtype),
+                cur_dv.to(v_own.dtype), None, None, None, None)
+
+
 def ring_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                    group=None, causal: bool = True) -> torch.Tensor:
     """q/k/v: this rank's sequence shard [B, H, T_local, D]; the global
@@ -75,10 +192,25 @@ def ring_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     else:
         rank = group.rank()
         world = group.size()
+    fused_ok = (
+        q.is_cuda and q.shape[-1] == 128 and q.dtype == torch.bfloat16
+        and q.shape[2] % 128 == 0 and q.shape[2] == k.shape[2]
+        and q.shape[1] == k.shape[1]
+    )
+    needs_grad = torch.is_grad_enabled() and (
+        q.requires_grad or k.requires_grad or v.requires_grad
+    )
     if world == 1:
+        if fused_ok:
+            from ray_amd import ops
+
+            return ops.flash_attention(q, k, v, causal=causal)
         mode = "causal" if causal else "full"
         out, _ = _block_attn(q, k, v, mode)
         return out.to(q.dtype)
+    if fused_ok and needs_grad:
+        # LSE-merging fused backward (training path)
+        return _RingAttnFn.apply(q, k, v, causal, group, rank, world)
 
     cur_k, cur_v = k.contiguous(), v.contiguous()
     out = None
