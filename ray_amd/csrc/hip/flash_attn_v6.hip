@@ -147,29 +147,33 @@ flash_attn_fwd_v6_bf16(const short* __restrict__ Q,
 
     const bool wave_active = !causal || (k0 <= wave_gq_max);
     if (wave_active) {
+      // att[2]-style pipeline (T15): BOTH sub-tiles' S'=K·Q^T MFMAs
+      // issue up front, so sub-tile 1's matrix work fills the MFMA
+      // pipe while sub-tile 0's softmax runs on the VALU pipe (the
+      // pipes are independent; m114).
+      const bool do0 =
+          (k0 < k_end) && !(causal && k0 > wave_gq_max);
+      const bool do1 =
+          (k0 + 32 < k_end) && !(causal && k0 + 32 > wave_gq_max);
+
+      auto s_mfma = [&](int kt) {
+        fa6_f32x16 acc{};
+        const int rr = 32 * kt + (lane & 31);
+        const short* krow = &k_lds[cur][rr][0];
+        const int sw = (rr & 7) << 3;
 #pragma unroll
-      for (int kt = 0; kt < 2; ++kt) {
+        for (int c = 0; c < 8; ++c) {
+          fa6_bf16x8 kf = fa6_ld8(krow + ((16 * c + a_off) ^ sw));
+          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, q_frag[c],
+                                                        acc, 0, 0, 0);
+        }
+        return acc;
+      };
+
+      auto softmax_pv = [&](int kt, fa6_f32x16 s_acc) {
         const int k0s = k0 + 32 * kt;
-        if (k0s >= k_end) break;
-        if (causal && k0s > wave_gq_max) break;
         const bool full_tile =
             (k0s + 32 <= Tk) && (!causal || (k0s + 31 <= wave_gq_min));
-
-        // ---- S' = K Q^T (one 32x32 tile per wave) ----
-        fa6_f32x16 s_acc{};
-        {
-          const int rr = 32 * kt + (lane & 31);
-          const short* krow = &k_lds[cur][rr][0];
-          const int sw = (rr & 7) << 3;
-#pragma unroll
-          for (int c = 0; c < 8; ++c) {
-            fa6_bf16x8 kf = fa6_ld8(krow + ((16 * c + a_off) ^ sw));
-            s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, q_frag[c],
-                                                            s_acc, 0, 0, 0);
-          }
-        }
-
-        // ---- lane-local online softmax ----
         float p_reg[16];
         float rmax = -INFINITY;
         if (full_tile) {
@@ -218,18 +222,17 @@ flash_attn_fwd_v6_bf16(const short* __restrict__ Q,
             for (int r = 0; r < 16; ++r) o_acc[d][r] *= sc;
         }
 
-        // ---- P -> B-fragments via cvt_pk + permlane32_swap (T12),
-        //      then PV ----
+        // P -> B-fragments via cvt_pk + permlane32_swap (T12), then PV
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
           unsigned int a0 = fa6_cvt_pk(p_reg[8 * kc + 0], p_reg[8 * kc + 1]);
           unsigned int a1 = fa6_cvt_pk(p_reg[8 * kc + 2], p_reg[8 * kc + 3]);
           unsigned int b0 = fa6_cvt_pk(p_reg[8 * kc + 4], p_reg[8 * kc + 5]);
           unsigned int b1 = fa6_cvt_pk(p_reg[8 * kc + 6], p_reg[8 * kc + 7]);
-          auto s0 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
-          auto s1 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
-          fa6_u32x4 pw{(unsigned)s0[0], (unsigned)s1[0], (unsigned)s0[1],
-                       (unsigned)s1[1]};
+          auto sw0 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
+          auto sw1 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
+          fa6_u32x4 pw{(unsigned)sw0[0], (unsigned)sw1[0], (unsigned)sw0[1],
+                       (unsigned)sw1[1]};
           fa6_bf16x8 pb = __builtin_bit_cast(fa6_bf16x8, pw);
 #pragma unroll
           for (int dt = 0; dt < 4; ++dt) {
@@ -244,7 +247,13 @@ flash_attn_fwd_v6_bf16(const short* __restrict__ Q,
                 va, pb, o_acc[dt], 0, 0, 0);
           }
         }
-      }  // kt
+      };
+
+      fa6_f32x16 sA{}, sB{};
+      if (do0) sA = s_mfma(0);
+      if (do1) sB = s_mfma(1);  // in flight under sub-tile 0's softmax
+      if (do0) softmax_pv(0, sA);
+      if (do1) softmax_pv(1, sB);
     }    // wave_active
 
     if (t + 1 < n_tiles) {

@@ -305,14 +305,22 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
                      cur_stream(), (const short*)d_o.data_ptr(),
                      (const short*)o.data_ptr(), dsum.data_ptr<float>(),
                      rows);
-  hipLaunchKernelGGL(fa_bwd_dq_bf16, dim3(T / 128, B * Hq), dim3(256), 0,
-                     cur_stream(), (const short*)q.data_ptr(),
-                     (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-                     (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
-                     dsum.data_ptr<float>(), (short*)dq.data_ptr(), B, Hq,
-                     Hkv, T, causal ? 1 : 0, scale);
   static const bool split = getenv("RAY_AMD_FA_BWD_FUSED") == nullptr;
   static const bool v3 = getenv("RAY_AMD_FA_BWD_V2") == nullptr;
+  if (v3)
+    hipLaunchKernelGGL(fa_bwd_dq_v3_bf16, dim3(T / 128, B * Hq), dim3(256),
+                       0, cur_stream(), (const short*)q.data_ptr(),
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
+                       dsum.data_ptr<float>(), (short*)dq.data_ptr(), B, Hq,
+                       Hkv, T, causal ? 1 : 0, scale);
+  else
+    hipLaunchKernelGGL(fa_bwd_dq_bf16, dim3(T / 128, B * Hq), dim3(256), 0,
+                       cur_stream(), (const short*)q.data_ptr(),
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
+                       dsum.data_ptr<float>(), (short*)dq.data_ptr(), B, Hq,
+                       Hkv, T, causal ? 1 : 0, scale);
   if (split && v3) {
     // v3: K/V register-resident, 64-row dbuf async-staged q/dO tiles,
     // swizzled LDS reads, one barrier per tile
