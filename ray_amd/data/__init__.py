@@ -23,6 +23,11 @@ from .dataset import (  # noqa: F401
     read_csv,
     read_json,
     read_parquet,
+    read_numpy,
+    read_webdataset,
+    read_datasource,
+    Datasource,
+    Datasink,
 )
 from .preprocessors import (  # noqa: F401
     Concatenator,

@@ -910,6 +910,19 @@ class Dataset:
                 for r in _rows_of(t):
                     f.write(json.dumps({k: _jsonable(v) for k, v in r.items()}) + "\n")
 
+    def write_datasink(self, datasink, **kwargs):
+        """Run datasink.write(block) as tasks over all blocks, then
+        on_write_complete(results) on the driver."""
+        ray = _ray()
+
+        @ray.remote
+        def _w(t, sink=datasink):
+            return sink.write(t)
+
+        results = ray.get([_w.remote(r) for r in self._materialize_refs()])
+        datasink.on_write_complete(results)
+        return results
+
     def write_numpy(self, path: str, *, column: str, **kwargs):
         import os
 
@@ -1236,3 +1249,84 @@ def read_binary_files(paths, *, include_paths: bool = False, **kwargs) -> Datase
     ds = Dataset(refs, [])
     ds._materialized = refs
     return ds
+
+
+def read_numpy(paths, *, parallelism: int = -1, column: str = "data",
+               **kwargs) -> Dataset:
+    """.npy / .npz files, one block per file (reference:
+    data/_internal/datasource numpy datasource). Lazy: files decode
+    inside read tasks paced by the streaming executor."""
+    files = _expand_paths(paths, ".npy") or _expand_paths(paths, ".npz")
+
+    def _mk(f):
+        def _read(f=f):
+            arr = np.load(f, allow_pickle=False)
+            if hasattr(arr, "files"):  # npz: one column per array
+                return _to_table({k: arr[k] for k in arr.files})
+            return _to_table({column: arr})
+
+        return _read
+
+    return Dataset([_mk(f) for f in files], [])
+
+
+def read_webdataset(paths, *, parallelism: int = -1, **kwargs) -> Dataset:
+    """WebDataset tar shards: files grouped by key prefix, one row per
+    sample with a column per extension (reference:
+    data/_internal/datasource webdataset). Pure tarfile, lazy blocks."""
+    files = _expand_paths(paths, ".tar")
+
+    def _mk(f):
+        def _read(f=f):
+            import tarfile
+
+            samples = {}
+            with tarfile.open(f) as tf:
+                for m in tf.getmembers():
+                    if not m.isfile():
+                        continue
+                    base = m.name
+                    key, _, ext = base.partition(".")
+                    samples.setdefault(key, {"__key__": key})[ext] = (
+                        tf.extractfile(m).read()
+                    )
+            rows = sorted(samples.values(), key=lambda r: r["__key__"])
+            cols = sorted({c for r in rows for c in r})
+            return pa.table({c: [r.get(c) for r in rows] for c in cols})
+
+        return _read
+
+    return Dataset([_mk(f) for f in files], [])
+
+
+class Datasource:
+    """Plugin ABC for custom sources (reference:
+    data/datasource/datasource.py). Implement get_read_tasks() to
+    return zero-arg callables, each producing one pyarrow.Table
+    block."""
+
+    def get_read_tasks(self, parallelism: int):
+        raise NotImplementedError
+
+    def estimate_inmemory_data_size(self):
+        return None
+
+
+class Datasink:
+    """Plugin ABC for custom sinks (reference:
+    data/datasource/datasink.py). write() is called once per block
+    inside a task; on_write_complete() once on the driver."""
+
+    def write(self, block) -> Any:
+        raise NotImplementedError
+
+    def on_write_complete(self, results) -> None:
+        pass
+
+
+def read_datasource(datasource: Datasource, *, parallelism: int = -1,
+                    **kwargs) -> Dataset:
+    tasks = list(datasource.get_read_tasks(
+        parallelism if parallelism > 0 else 8
+    ))
+    return Dataset(list(tasks), [])

@@ -289,3 +289,64 @@ def test_streaming_executor_order_preserved(ray_start_regular):
     )
     got = [r["id"] for r in ds.take_all()]
     assert got == list(range(100))
+
+
+def test_read_numpy_and_npz(ray_start_regular, tmp_path):
+    a = np.arange(20, dtype=np.float32)
+    np.save(tmp_path / "a.npy", a)
+    ds = rd.read_numpy(str(tmp_path / "a.npy"))
+    rows = ds.take_all()
+    assert [float(r["data"]) for r in rows] == list(map(float, a))
+    np.savez(tmp_path / "b.npz", x=np.arange(4), y=np.arange(4) * 2)
+    ds2 = rd.read_numpy(str(tmp_path / "b.npz"))
+    rows = ds2.take_all()
+    assert [r["y"] for r in rows] == [0, 2, 4, 6]
+
+
+def test_read_webdataset(ray_start_regular, tmp_path):
+    import io
+    import tarfile
+
+    tarp = tmp_path / "shard-000.tar"
+    with tarfile.open(tarp, "w") as tf:
+        for key in ("s0", "s1"):
+            for ext, payload in (("txt", f"text-{key}".encode()),
+                                 ("cls", b"7")):
+                data = io.BytesIO(payload)
+                info = tarfile.TarInfo(f"{key}.{ext}")
+                info.size = len(payload)
+                tf.addfile(info, data)
+    ds = rd.read_webdataset(str(tarp))
+    rows = ds.take_all()
+    assert len(rows) == 2
+    assert rows[0]["__key__"] == "s0" and rows[0]["txt"] == b"text-s0"
+    assert rows[1]["cls"] == b"7"
+
+
+def test_datasource_datasink_plugins(ray_start_regular):
+    class SquaresSource(rd.Datasource):
+        def get_read_tasks(self, parallelism):
+            def mk(i):
+                def read(i=i):
+                    import pyarrow as pa
+
+                    return pa.table({"x": [i * i]})
+
+                return read
+
+            return [mk(i) for i in range(5)]
+
+    ds = rd.read_datasource(SquaresSource())
+    assert sorted(r["x"] for r in ds.take_all()) == [0, 1, 4, 9, 16]
+
+    collected = []
+
+    class ListSink(rd.Datasink):
+        def write(self, block):
+            return block.num_rows
+
+        def on_write_complete(self, results):
+            collected.extend(results)
+
+    ds.write_datasink(ListSink())
+    assert sum(collected) == 5
