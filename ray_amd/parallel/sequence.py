@@ -132,11 +132,23 @@ class _RingAttnFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, d_out):
-        from ray_amd.ops import _K
-
         q, k_own, v_own, out, lse = ctx.saved_tensors
-        causal = ctx.causal
         group, rank, world = ctx.ring
+        dq, dk, dv = _ring_attn_backward(
+            q, k_own, v_own, out, lse, d_out, ctx.causal, group, rank,
+            world,
+        )
+        return dq, dk, dv, None, None, None, None
+
+
+def _ring_attn_backward(q, k_own, v_own, out, lse, d_out, causal, group,
+                        rank, world):
+    """Ring-attention gradient with the global LSE (callable outside
+    the autograd engine — the engine serializes concurrent backwards on
+    one device thread, which would deadlock a blocking ring)."""
+    from ray_amd.ops import _K
+
+    if True:
         send_to = (rank + 1) % world
         recv_from = (rank - 1) % world
         do = d_out.contiguous()
@@ -172,7 +184,7 @@ class _RingAttnFn(torch.autograd.Function):
             cur_k, cur_v = nxt_k, nxt_v
             cur_dk, cur_dv = nxt_dk, nxt_dv
         return (dq_acc.to(q.dtype), cur_dk.to(k_own.dtype),
-                cur_dv.to(v_own.dtype), None, None, None, None)
+                cur_dv.to(v_own.dtype))
 
 
 def ring_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,

@@ -217,22 +217,45 @@ def test_ring_attention_fused_backward_two_rank_sim():
     seq._ring_exchange = fake_exchange
     results = {}
 
+    errors = {}
+
     def run_rank(r):
-        sl = slice(r * Tl, (r + 1) * Tl)
-        ql = q[:, :, sl].detach().clone().requires_grad_()
-        kl = k[:, :, sl].detach().clone().requires_grad_()
-        vl = v[:, :, sl].detach().clone().requires_grad_()
-        out = seq._RingAttnFn.apply(ql, kl, vl, True, None, r, world)
-        out.backward(g_out[:, :, sl])
-        results[r] = (out.detach(), ql.grad, kl.grad, vl.grad)
+        # the fwd/bwd cores are called directly (not through
+        # loss.backward()): the autograd engine serializes concurrent
+        # backwards onto one device thread, which would deadlock the
+        # blocking ring exchange between the two simulated ranks
+        try:
+            sl = slice(r * Tl, (r + 1) * Tl)
+            ql = q[:, :, sl].contiguous()
+            kl = k[:, :, sl].contiguous()
+            vl = v[:, :, sl].contiguous()
+
+            class _Ctx:
+                saved_tensors = ()
+
+                def save_for_backward(self, *ts):
+                    self.saved_tensors = ts
+
+            ctx = _Ctx()
+            out = seq._RingAttnFn.forward(ctx, ql, kl, vl, True, None,
+                                          r, world)
+            qs, ks, vs, outs, lses = ctx.saved_tensors
+            dq, dk, dv = seq._ring_attn_backward(
+                qs, ks, vs, outs, lses, g_out[:, :, sl], True, None, r,
+                world,
+            )
+            results[r] = (out.detach(), dq, dk, dv)
+        except BaseException as e:  # noqa
+            errors[r] = e
 
     try:
         ts = [threading.Thread(target=run_rank, args=(r,)) for r in range(world)]
         for t in ts:
             t.start()
         for t in ts:
-            t.join(timeout=300)
+            t.join(timeout=240)
             assert not t.is_alive(), "ring sim deadlocked"
+        assert not errors, errors
     finally:
         seq._ring_exchange = orig
 
