@@ -1,0 +1,334 @@
+// fa_bwd v4: 8-wave (256 kv rows) blocks — derived from v3; the
+// staged 64-row q/dO tile now feeds 8 waves instead of 4 (half the
+// LDS staging traffic and barriers per unit of mfma work).
+#include "common.h"
+
+// dV: grid (T/128, B*Hkv); 4 waves x 32 kv rows; K in registers,
+// dV accumulates in registers. dv[k][d] = sum_q P^T[k][q] dO[q][d]
+// ---------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(512) void fa_bwd_dv_v4_bf16(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, const short* __restrict__ dO,
+    const float* __restrict__ LSE, short* __restrict__ dV, int B, int Hq,
+    int Hkv, int T, int causal, float scale) {
+  __shared__ short q_lds[2][FB3_QT][FB3_D];
+  __shared__ short do_lds[2][FB3_QT][FB3_D];
+  __shared__ float lse_lds[2][FB3_QT];
+  __shared__ short scratch[8][32][40];
+
+  const int k0 = blockIdx.x * 256;
+  const int bh = blockIdx.y;
+  const int b = bh / Hkv;
+  const int hkv = bh % Hkv;
+  const int rep = Hq / Hkv;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int hi = lane >> 5;
+  const int a_off = 8 * hi;
+
+  const long long kbase = (((long long)b * Hkv + hkv) * T + k0) * FB3_D;
+  // K register-resident: only this wave's own 32 rows are ever read
+  fb3_bf16x8 k_frag[8];
+  {
+    const short* kp = K + kbase + ((long long)32 * wave + (lane & 31)) * FB3_D;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) k_frag[c] = fb3_ld8(kp + 16 * c + a_off);
+  }
+
+  fb3_f32x16 dv_acc[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) dv_acc[t] = fb3_f32x16{};
+  const float L2E = 1.4426950408889634f;
+
+  const int q_start = causal ? k0 : 0;
+  const int nq = (T - q_start + FB3_QT - 1) / FB3_QT;
+  const int n_tiles = rep * nq;
+
+  // staging regs: 64 rows x 16 chunks / 512 threads = 2 chunks/tensor
+  short8 q_stage[2], do_stage[2];
+  float lse_stage = 0.f;
+
+#define FB4_DV_LOAD(t_idx)                                                \
+  do {                                                                    \
+    const int g_ = (t_idx) / nq;                                          \
+    const int q0_ = q_start + ((t_idx) % nq) * FB3_QT;                    \
+    const int hq_ = hkv * rep + g_;                                       \
+    const long long qb_ = (((long long)b * Hq + hq_) * T) * FB3_D;        \
+    const long long lb_ = ((long long)b * Hq + hq_) * T;                  \
+    _Pragma("unroll") for (int j = 0; j < 2; ++j) {                       \
+      int i = threadIdx.x + 512 * j;                                      \
+      int r = i >> 4;                                                     \
+      int c = (i & 15) * 8;                                               \
+      int qrow = q0_ + r;                                                 \
+      short8 qv{0, 0, 0, 0, 0, 0, 0, 0}, dv{0, 0, 0, 0, 0, 0, 0, 0};      \
+      if (qrow < T) {                                                     \
+        qv = *reinterpret_cast<const short8*>(qb_ + Q +                   \
+                                              (long long)qrow * FB3_D + c); \
+        dv = *reinterpret_cast<const short8*>(qb_ + dO +                  \
+                                              (long long)qrow * FB3_D + c); \
+      }                                                                   \
+      q_stage[j] = qv;                                                    \
+      do_stage[j] = dv;                                                   \
+    }                                                                     \
+    if (threadIdx.x < FB3_QT) {                                           \
+      int qrow = q0_ + threadIdx.x;                                       \
+      lse_stage = (qrow < T) ? LSE[lb_ + qrow] : INFINITY;                \
+    }                                                                     \
+  } while (0)
+
+#define FB4_DV_WRITE(buf)                                                 \
+  do {                                                                    \
+    _Pragma("unroll") for (int j = 0; j < 2; ++j) {                       \
+      int i = threadIdx.x + 512 * j;                                      \
+      int r = i >> 4;                                                     \
+      int c = (i & 15) * 8;                                               \
+      int csw = c ^ ((r & 7) << 3);                                       \
+      *reinterpret_cast<short8*>(&q_lds[buf][r][csw]) = q_stage[j];       \
+      *reinterpret_cast<short8*>(&do_lds[buf][r][csw]) = do_stage[j];     \
+    }                                                                     \
+    if (threadIdx.x < FB3_QT) lse_lds[buf][threadIdx.x] = lse_stage;      \
+  } while (0)
+
+  FB4_DV_LOAD(0);
+  FB4_DV_WRITE(0);
+  __syncthreads();
+
+  for (int t = 0; t < n_tiles; ++t) {
+    const int cur = t & 1;
+    const int q0s = q_start + (t % nq) * FB3_QT;
+    if (t + 1 < n_tiles) FB4_DV_LOAD(t + 1);
+
+#pragma unroll
+    for (int qt = 0; qt < 2; ++qt) {
+      const int qts = q0s + 32 * qt;
+      if (qts >= T) break;
+      if (causal && k0 + 32 * wave > qts + 31) continue;
+
+      fb3_f32x16 s_acc{};
+      {
+        const int rr = 32 * qt + (lane & 31);
+        const short* qrow = &q_lds[cur][rr][0];
+        const int sw = (rr & 7) << 3;
+#pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          // S'^T tile: A = this wave's K rows (regs), B = Q^T from LDS
+          fb3_bf16x8 qf = fb3_ld8(qrow + ((16 * c + a_off) ^ sw));
+          s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(k_frag[c], qf,
+                                                          s_acc, 0, 0, 0);
+        }
+      }
+      // C layout: col = lane&31 = q (this lane's loaded q row),
+      // rows = this wave's k (same as v2)
+      const int gq = qts + (lane & 31);
+      const float lse_q = lse_lds[cur][32 * qt + (lane & 31)];
+      const bool full_tile = !causal || (k0 + 32 * wave + 31 <= qts);
+      short(*scr)[40] = scratch[wave];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int gk = k0 + 32 * wave + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float pv = __builtin_amdgcn_exp2f(
+            __builtin_fmaf(s_acc[r] * scale, L2E, -lse_q * L2E));
+        if (!full_tile && gk > gq) pv = 0.f;
+        scr[(r & 3) + 8 * (r >> 2) + 4 * hi][lane & 31] = f2bf(pv);
+      }
+      __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+      for (int qc = 0; qc < 2; ++qc) {
+        fb3_bf16x8 af = __builtin_bit_cast(
+            fb3_bf16x8, *reinterpret_cast<const short8*>(
+                            &scr[lane & 31][16 * qc + a_off]));
+#pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          short dtmp[8];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            int row = 32 * qt + 16 * qc + a_off + i;
+            int col = 32 * dt + (lane & 31);
+            dtmp[i] = do_lds[cur][row][col ^ ((row & 7) << 3)];
+          }
+          fb3_bf16x8 bf = __builtin_bit_cast(
+              fb3_bf16x8, *reinterpret_cast<short8*>(dtmp));
+          dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af, bf, dv_acc[dt], 0, 0, 0);
+        }
+      }
+    }
+
+    if (t + 1 < n_tiles) FB4_DV_WRITE(cur ^ 1);
+    __syncthreads();
+  }
+
+  short* outv = dV + kbase + (long long)wave * 32 * FB3_D;
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int krow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      outv[(long long)krow * FB3_D + 32 * dt + (lane & 31)] =
+          f2bf(dv_acc[dt][r]);
+    }
+}
+
+// ---------------------------------------------------------------------
+// dK: same structure; V also register-resident; needs Dsum and
+// dP^T = V·dO^T.  dk[k][d] = sum_q dS̃^T[k][q] Q[q][d]
+// ---------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(512) void fa_bwd_dk_v4_bf16(
+    const short* __restrict__ Q, const short* __restrict__ K,
+    const short* __restrict__ V, const short* __restrict__ dO,
+    const float* __restrict__ LSE, const float* __restrict__ Dsum,
+    short* __restrict__ dK, int B, int Hq, int Hkv, int T, int causal,
+    float scale) {
+  __shared__ short q_lds[2][FB3_QT][FB3_D];
+  __shared__ short do_lds[2][FB3_QT][FB3_D];
+  __shared__ float lse_lds[2][FB3_QT];
+  __shared__ float dsum_lds[2][FB3_QT];
+  __shared__ short scratch[8][32][40];
+
+  const int k0 = blockIdx.x * 256;
+  const int bh = blockIdx.y;
+  const int b = bh / Hkv;
+  const int hkv = bh % Hkv;
+  const int rep = Hq / Hkv;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int hi = lane >> 5;
+  const int a_off = 8 * hi;
+
+  const long long kbase = (((long long)b * Hkv + hkv) * T + k0) * FB3_D;
+  fb3_bf16x8 k_frag[8], v_frag[8];
+  {
+    const short* kp = K + kbase + ((long long)32 * wave + (lane & 31)) * FB3_D;
+    const short* vp = V + kbase + ((long long)32 * wave + (lane & 31)) * FB3_D;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      k_frag[c] = fb3_ld8(kp + 16 * c + a_off);
+      v_frag[c] = fb3_ld8(vp + 16 * c + a_off);
+    }
+  }
+
+  fb3_f32x16 dk_acc[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) dk_acc[t] = fb3_f32x16{};
+  const float L2E = 1.4426950408889634f;
+
+  const int q_start = causal ? k0 : 0;
+  const int nq = (T - q_start + FB3_QT - 1) / FB3_QT;
+  const int n_tiles = rep * nq;
+
+  // dk holds K AND V register-resident plus the dK accumulators —
+  // reg-staged async loads (like dv's) push it to 289 regs = 1
+  // wave/SIMD (measured: the whole-bwd regression). Stage DIRECTLY
+  // global->LDS into the double buffer instead: minimal register
+  // liveness, still one barrier per 64-row tile.
+#define FB4_DK_STAGE(t_idx, buf)                                          \
+  do {                                                                    \
+    const int g_ = (t_idx) / nq;                                          \
+    const int q0_ = q_start + ((t_idx) % nq) * FB3_QT;                    \
+    const int hq_ = hkv * rep + g_;                                       \
+    const long long qb_ = (((long long)b * Hq + hq_) * T) * FB3_D;        \
+    const long long lb_ = ((long long)b * Hq + hq_) * T;                  \
+    _Pragma("unroll") for (int j = 0; j < 2; ++j) {                       \
+      int i = threadIdx.x + 512 * j;                                      \
+      int r = i >> 4;                                                     \
+      int c = (i & 15) * 8;                                               \
+      int csw = c ^ ((r & 7) << 3);                                       \
+      int qrow = q0_ + r;                                                 \
+      short8 qv{0, 0, 0, 0, 0, 0, 0, 0}, dv{0, 0, 0, 0, 0, 0, 0, 0};      \
+      if (qrow < T) {                                                     \
+        qv = *reinterpret_cast<const short8*>(qb_ + Q +                   \
+                                              (long long)qrow * FB3_D + c); \
+        dv = *reinterpret_cast<const short8*>(qb_ + dO +                  \
+                                              (long long)qrow * FB3_D + c); \
+      }                                                                   \
+      *reinterpret_cast<short8*>(&q_lds[buf][r][csw]) = qv;               \
+      *reinterpret_cast<short8*>(&do_lds[buf][r][csw]) = dv;              \
+    }                                                                     \
+    if (threadIdx.x < FB3_QT) {                                           \
+      int qrow = q0_ + threadIdx.x;                                       \
+      lse_lds[buf][threadIdx.x] = (qrow < T) ? LSE[lb_ + qrow] : INFINITY; \
+      dsum_lds[buf][threadIdx.x] = (qrow < T) ? Dsum[lb_ + qrow] : 0.f;   \
+    }                                                                     \
+  } while (0)
+
+  FB4_DK_STAGE(0, 0);
+  __syncthreads();
+
+  for (int t = 0; t < n_tiles; ++t) {
+    const int cur = t & 1;
+    const int q0s = q_start + (t % nq) * FB3_QT;
+    if (t + 1 < n_tiles) FB4_DK_STAGE(t + 1, cur ^ 1);
+
+#pragma unroll
+    for (int qt = 0; qt < 2; ++qt) {
+      const int qts = q0s + 32 * qt;
+      if (qts >= T) break;
+      if (causal && k0 + 32 * wave > qts + 31) continue;
+
+      fb3_f32x16 s_acc{}, dp_acc{};
+      {
+        const int rr = 32 * qt + (lane & 31);
+        const short* qrow = &q_lds[cur][rr][0];
+        const short* dorow = &do_lds[cur][rr][0];
+        const int sw = (rr & 7) << 3;
+#pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          fb3_bf16x8 qf = fb3_ld8(qrow + ((16 * c + a_off) ^ sw));
+          fb3_bf16x8 dof = fb3_ld8(dorow + ((16 * c + a_off) ^ sw));
+          s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(k_frag[c], qf,
+                                                          s_acc, 0, 0, 0);
+          dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(v_frag[c], dof,
+                                                           dp_acc, 0, 0, 0);
+        }
+      }
+      const int gq = qts + (lane & 31);
+      const float lse_q = lse_lds[cur][32 * qt + (lane & 31)];
+      const float d_q = dsum_lds[cur][32 * qt + (lane & 31)];
+      const bool full_tile = !causal || (k0 + 32 * wave + 31 <= qts);
+      short(*scr)[40] = scratch[wave];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int gk = k0 + 32 * wave + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float pv = __builtin_amdgcn_exp2f(
+            __builtin_fmaf(s_acc[r] * scale, L2E, -lse_q * L2E));
+        if (!full_tile && gk > gq) pv = 0.f;
+        float dsv = pv * (dp_acc[r] - d_q) * scale;
+        scr[(r & 3) + 8 * (r >> 2) + 4 * hi][lane & 31] = f2bf(dsv);
+      }
+      __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+      for (int qc = 0; qc < 2; ++qc) {
+        fb3_bf16x8 af = __builtin_bit_cast(
+            fb3_bf16x8, *reinterpret_cast<const short8*>(
+                            &scr[lane & 31][16 * qc + a_off]));
+#pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          short qtmp[8];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            int row = 32 * qt + 16 * qc + a_off + i;
+            int col = 32 * dt + (lane & 31);
+            qtmp[i] = q_lds[cur][row][col ^ ((row & 7) << 3)];
+          }
+          fb3_bf16x8 bf = __builtin_bit_cast(
+              fb3_bf16x8, *reinterpret_cast<short8*>(qtmp));
+          dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af, bf, dk_acc[dt], 0, 0, 0);
+        }
+      }
+    }
+
+    __syncthreads();
+  }
+
+  short* outk = dK + kbase + (long long)wave * 32 * FB3_D;
+#pragma unroll
+  for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int krow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      outk[(long long)krow * FB3_D + 32 * dt + (lane & 31)] =
+          f2bf(dk_acc[dt][r]);
+    }
+}
+

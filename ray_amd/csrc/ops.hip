@@ -16,6 +16,7 @@
 #include "hip/flash_attn_v6.hip"
 #include "hip/fa_bwd.hip"
 #include "hip/fa_bwd_v3.hip"
+#include "hip/fa_bwd_v4.hip"
 
 #define CHECK_IN(x)                                                     \
   TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                       \
@@ -322,8 +323,29 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
                        dsum.data_ptr<float>(), (short*)dq.data_ptr(), B, Hq,
                        Hkv, T, causal ? 1 : 0, scale);
   if (split && v3) {
-    // v3: K/V register-resident, 64-row dbuf async-staged q/dO tiles,
-    // swizzled LDS reads, one barrier per tile
+    // v3/v4: K/V register-resident, 64-row dbuf staged q/dO tiles,
+    // swizzled LDS reads, one barrier per tile; v4 = 8-wave blocks
+    // (256 kv rows) when the shape allows
+    bool v4 = (T % 256 == 0) && getenv("RAY_AMD_FA_BWD_V3ONLY") == nullptr;
+    if (v4) {
+      hipLaunchKernelGGL(fa_bwd_dv_v4_bf16, dim3(T / 256, B * Hkv),
+                         dim3(512), 0, cur_stream(),
+                         (const short*)q.data_ptr(),
+                         (const short*)k.data_ptr(),
+                         (const short*)v.data_ptr(),
+                         (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
+                         (short*)dv.data_ptr(), B, Hq, Hkv, T,
+                         causal ? 1 : 0, scale);
+      hipLaunchKernelGGL(fa_bwd_dk_v4_bf16, dim3(T / 256, B * Hkv),
+                         dim3(512), 0, cur_stream(),
+                         (const short*)q.data_ptr(),
+                         (const short*)k.data_ptr(),
+                         (const short*)v.data_ptr(),
+                         (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
+                         dsum.data_ptr<float>(), (short*)dk.data_ptr(), B,
+                         Hq, Hkv, T, causal ? 1 : 0, scale);
+      return {dq, dk, dv};
+    }
     hipLaunchKernelGGL(fa_bwd_dv_v3_bf16, dim3(T / 128, B * Hkv),
                        dim3(256), 0, cur_stream(),
                        (const short*)q.data_ptr(),
