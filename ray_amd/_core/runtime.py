@@ -1591,7 +1591,10 @@ class CoreRuntime:
             self._store_put_threadsafe(aid, ("store", self.raylet_addr, size))
             with self._refs_lock:
                 self._refs[aid] = [1, 0, self.addr]
-        self._run(self._submit_actor_async(spec, captured_ids))
+        retries = int(options.get("max_task_retries") or 0)
+        if streaming:
+            retries = 0
+        self._run(self._submit_actor_async(spec, captured_ids, retries))
         if streaming:
             return ObjectRefGenerator(task_id, self)
         return refs
@@ -1627,10 +1630,10 @@ class CoreRuntime:
         finally:
             st["resolving"] = None
 
-    async def _submit_actor_async(self, spec, captured_ids):
+    async def _submit_actor_async(self, spec, captured_ids, retries=0):
         actor_id = spec["actor_id"]
         try:
-            for attempt in range(3):
+            for attempt in range(3 + retries):
                 st = await self._actor_state(actor_id)
                 if st["client"] is None or not st["client"].connected:
                     st["client"] = None
@@ -1652,10 +1655,19 @@ class CoreRuntime:
                     self._ingest_result(spec, reply)
                     return
                 except (ConnectionLost, ConnectionError):
+                    st["client"] = None
+                    if retries > 0:
+                        # max_task_retries > 0 (reference:
+                        # actor_task_submitter.cc:597 sequenced
+                        # resubmit): the call is resent to the
+                        # restarted instance — at-least-once, the task
+                        # may execute twice if the reply was lost
+                        retries -= 1
+                        await asyncio.sleep(0.2)
+                        continue
                     # at-most-once (reference default, max_task_retries=0):
                     # the call may have executed before the actor died,
                     # so it must NOT be resent to a restarted instance
-                    st["client"] = None
                     err = serialization.dumps(
                         exc.ActorUnavailableError(
                             f"actor {actor_id.hex()} died while this call "

@@ -343,6 +343,7 @@ class ActorMethod:
         rt = _rt.global_runtime()
         merged = {**self._defaults, **opts}
         merged.setdefault("num_returns", 1)
+        merged.setdefault("max_task_retries", self._handle._max_task_retries)
         num_returns = merged["num_returns"]
         if self._handle._tensor_transport:
             # sender-side offload (reference: rdt_manager __ray_send__):
@@ -368,11 +369,13 @@ class ActorMethod:
 
 class ActorHandle:
     def __init__(self, actor_id: bytes, class_name: str = "Actor",
-                 method_options=None, tensor_transport=None):
+                 method_options=None, tensor_transport=None,
+                 max_task_retries=0):
         self._actor_id = actor_id
         self._class_name = class_name
         self._method_options = method_options or {}
         self._tensor_transport = tensor_transport
+        self._max_task_retries = max_task_retries
 
     def __getattr__(self, item):
         if item.startswith("_") and item not in (
@@ -387,7 +390,7 @@ class ActorHandle:
     def __reduce__(self):
         return (_deserialize_handle,
                 (self._actor_id, self._class_name, self._method_options,
-                 self._tensor_transport))
+                 self._tensor_transport, self._max_task_retries))
 
     def __hash__(self):
         return hash(self._actor_id)
@@ -402,9 +405,9 @@ class ActorHandle:
 
 
 def _deserialize_handle(actor_id, class_name, method_options=None,
-                        tensor_transport=None):
+                        tensor_transport=None, max_task_retries=0):
     return ActorHandle(actor_id, class_name, method_options,
-                       tensor_transport)
+                       tensor_transport, max_task_retries)
 
 
 class ActorClass:
@@ -448,7 +451,8 @@ class ActorClass:
         actor_id = rt.create_actor(key, pickled, opts, (args, kwargs))
         return ActorHandle(actor_id, self._cls.__name__,
                            self._method_options(),
-                           opts.get("tensor_transport"))
+                           opts.get("tensor_transport"),
+                           opts.get("max_task_retries", 0))
 
     def _method_options(self) -> dict:
         """Collect @ray.method(...) per-method option dicts off the class

@@ -264,3 +264,42 @@ def test_ray_method_num_returns(ray_start_regular):
     p2 = cloudpickle.loads(cloudpickle.dumps(p))
     c, d = p2.two.remote()
     assert ray.get([c, d]) == [1, 2]
+
+
+def test_actor_max_task_retries_resubmits(ray_start_regular):
+    """max_task_retries > 0 (reference: actor_task_submitter.cc:597):
+    a call in flight when the actor dies is resubmitted to the
+    restarted instance instead of failing with ActorUnavailableError."""
+    import os as _os
+
+    @ray.remote(max_restarts=2, max_task_retries=2)
+    class Flaky:
+        def __init__(self):
+            self.n = 0
+
+        def die_then_answer(self, marker):
+            # first instance dies mid-call; the restarted one answers
+            if not _os.path.exists(marker):
+                open(marker, "w").write("x")
+                _os._exit(1)
+            return "answered"
+
+    import tempfile
+
+    marker = tempfile.mktemp()
+    a = Flaky.remote()
+    assert ray.get(a.die_then_answer.remote(marker), timeout=120) == "answered"
+
+    # default (max_task_retries=0) stays at-most-once
+    @ray.remote(max_restarts=2)
+    class Flaky0:
+        def die(self, marker):
+            if not _os.path.exists(marker):
+                open(marker, "w").write("x")
+                _os._exit(1)
+            return "no"
+
+    marker2 = tempfile.mktemp()
+    b = Flaky0.remote()
+    with pytest.raises(ray.exceptions.RayError):
+        ray.get(b.die.remote(marker2), timeout=60)
