@@ -1,7 +1,8 @@
 """Llama-3 architecture, MI355X-first.
 
 Hand-written HIP kernels (ray_amd.ops) for RMSNorm / RoPE / SwiGLU /
-fused cross-entropy; GQA attention via torch SDPA (CK flash backend on
+fused cross-entropy; GQA training attention on the hand-written CDNA4
+flash kernels (SDPA fallback for decode/odd shapes; formerly
 ROCm); plain GEMMs via nn.Linear (hipBLASLt). bf16 weights, fp32
 optimizer states via ray_amd.ops.FusedAdamW.
 
@@ -11,6 +12,8 @@ Config llama3-8b matches Meta-Llama-3-8B: hidden 4096, 32 layers,
 from __future__ import annotations
 
 from dataclasses import dataclass
+
+import os
 
 import torch
 import torch.nn as nn
@@ -80,9 +83,23 @@ class Attention(nn.Module):
         if kv_cache is not None:
             k, v = kv_cache.update(k, v, pos0)
         is_causal = kv_cache is None or T > 1
-        out = F.scaled_dot_product_attention(
-            q, k, v, is_causal=is_causal, enable_gqa=True
-        )
+        # training self-attention runs the hand-written CDNA4 flash
+        # kernels (fwd v6 + fa_bwd v4: measured at parity with AOTriton
+        # round-trip, profiles/flash_attn_v6_bench.log); decode/cache
+        # shapes keep SDPA
+        if (
+            kv_cache is None
+            and q.is_cuda
+            and q.dtype == torch.bfloat16
+            and self.head_dim == 128
+            and T % 128 == 0
+            and not os.environ.get("RAY_AMD_ATTN_SDPA")
+        ):
+            out = ops.flash_attention(q, k, v, causal=True)
+        else:
+            out = F.scaled_dot_product_attention(
+                q, k, v, is_causal=is_causal, enable_gqa=True
+            )
         out = out.transpose(1, 2).reshape(B, T, -1)
         return self.o_proj(out)
 
