@@ -37,7 +37,10 @@ DEV_INLINE fab_bf16x8 fab_ld8(const short* p) {
 // ---------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(256) void fa_bwd_prep_bf16(
     const short* __restrict__ dO, const short* __restrict__ O,
-    float* __restrict__ Dsum, long long total_rows) {
+    float* __restrict__ Dsum, long long total_rows, int Hq, int T,
+    int bthd) {
+  // Dsum is ALWAYS [B,Hq,T]-indexed; when O/dO storage is [B,T,H,D]
+  // the storage row (b,t,h) maps to dsum index (b*Hq+h)*T + t.
   const long long row = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
   if (row >= total_rows) return;
   const int lane = threadIdx.x & 63;
@@ -45,7 +48,16 @@ extern "C" __global__ __launch_bounds__(256) void fa_bwd_prep_bf16(
   const short* op = O + row * FAB_D + 2 * lane;
   float acc = bf2f(dop[0]) * bf2f(op[0]) + bf2f(dop[1]) * bf2f(op[1]);
   acc = wave_sum(acc);
-  if (lane == 0) Dsum[row] = acc;
+  if (lane == 0) {
+    long long idx = row;
+    if (bthd) {
+      long long b = row / ((long long)T * Hq);
+      long long rem = row % ((long long)T * Hq);
+      long long t = rem / Hq, h = rem % Hq;
+      idx = (b * Hq + h) * T + t;
+    }
+    Dsum[idx] = acc;
+  }
 }
 
 // ---------------------------------------------------------------------

@@ -371,7 +371,7 @@ extern "C" __global__ __launch_bounds__(256) void fa_bwd_dq_v3_bf16(
     const short* __restrict__ V, const short* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ Dsum,
     short* __restrict__ dQ, int B, int Hq, int Hkv, int T, int causal,
-    float scale) {
+    float scale, int bthd) {
   __shared__ short k_lds[2][32][FB3_D];
   __shared__ short v_lds[2][32][FB3_D];
 
@@ -385,14 +385,20 @@ extern "C" __global__ __launch_bounds__(256) void fa_bwd_dq_v3_bf16(
   const int hi = lane >> 5;
   const int a_off = 8 * hi;
 
-  const long long qbase = (((long long)b * Hq + hq) * T + q0) * FB3_D;
-  const long long kbase = (((long long)b * Hkv + hkv) * T) * FB3_D;
+  const long long q_rs = bthd ? (long long)Hq * FB3_D : FB3_D;
+  const long long kv_rs = bthd ? (long long)Hkv * FB3_D : FB3_D;
+  const long long qbase =
+      bthd ? (((long long)b * T + q0) * Hq + hq) * FB3_D
+           : (((long long)b * Hq + hq) * T + q0) * FB3_D;
+  const long long kbase =
+      bthd ? ((long long)b * T * Hkv + hkv) * FB3_D
+           : (((long long)b * Hkv + hkv) * T) * FB3_D;
   const int my_q = q0 + wave * 32 + (lane & 31);
 
   fb3_bf16x8 q_frag[8], do_frag[8];
   {
-    const short* qp = Q + qbase + ((long long)wave * 32 + (lane & 31)) * FB3_D;
-    const short* dp = dO + qbase + ((long long)wave * 32 + (lane & 31)) * FB3_D;
+    const short* qp = Q + qbase + ((long long)wave * 32 + (lane & 31)) * q_rs;
+    const short* dp = dO + qbase + ((long long)wave * 32 + (lane & 31)) * q_rs;
 #pragma unroll
     for (int c = 0; c < 8; ++c) {
       q_frag[c] = fb3_ld8(qp + 16 * c + a_off);
@@ -422,10 +428,10 @@ extern "C" __global__ __launch_bounds__(256) void fa_bwd_dq_v3_bf16(
       int csw = c ^ ((r & 7) << 3);                                       \
       *reinterpret_cast<short8*>(&k_lds[buf][r][csw]) =                   \
           *reinterpret_cast<const short8*>(                               \
-              K + kbase + (long long)(kt0 + r) * FB3_D + c);              \
+              K + kbase + (long long)(kt0 + r) * kv_rs + c);              \
       *reinterpret_cast<short8*>(&v_lds[buf][r][csw]) =                   \
           *reinterpret_cast<const short8*>(                               \
-              V + kbase + (long long)(kt0 + r) * FB3_D + c);              \
+              V + kbase + (long long)(kt0 + r) * kv_rs + c);              \
     }                                                                     \
   } while (0)
 
@@ -499,13 +505,13 @@ extern "C" __global__ __launch_bounds__(256) void fa_bwd_dq_v3_bf16(
     __syncthreads();
   }
 
-  short* out = dQ + qbase + (long long)wave * 32 * FB3_D;
+  short* out = dQ + qbase + (long long)wave * 32 * q_rs;
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
-      out[(long long)qrow * FB3_D + 32 * dt + (lane & 31)] =
+      out[(long long)qrow * q_rs + 32 * dt + (lane & 31)] =
           f2bf(dq_acc[dt][r]);
     }
 }

@@ -10,7 +10,7 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dv_v4_bf16(
     const short* __restrict__ Q, const short* __restrict__ K,
     const short* __restrict__ V, const short* __restrict__ dO,
     const float* __restrict__ LSE, short* __restrict__ dV, int B, int Hq,
-    int Hkv, int T, int causal, float scale) {
+    int Hkv, int T, int causal, float scale, int bthd) {
   __shared__ short q_lds[2][FB3_QT][FB3_D];
   __shared__ short do_lds[2][FB3_QT][FB3_D];
   __shared__ float lse_lds[2][FB3_QT];
@@ -26,11 +26,15 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dv_v4_bf16(
   const int hi = lane >> 5;
   const int a_off = 8 * hi;
 
-  const long long kbase = (((long long)b * Hkv + hkv) * T + k0) * FB3_D;
+  const long long q_rs = bthd ? (long long)Hq * FB3_D : FB3_D;
+  const long long kv_rs = bthd ? (long long)Hkv * FB3_D : FB3_D;
+  const long long kbase =
+      bthd ? (((long long)b * T + k0) * Hkv + hkv) * FB3_D
+           : (((long long)b * Hkv + hkv) * T + k0) * FB3_D;
   // K register-resident: only this wave's own 32 rows are ever read
   fb3_bf16x8 k_frag[8];
   {
-    const short* kp = K + kbase + ((long long)32 * wave + (lane & 31)) * FB3_D;
+    const short* kp = K + kbase + ((long long)32 * wave + (lane & 31)) * kv_rs;
 #pragma unroll
     for (int c = 0; c < 8; ++c) k_frag[c] = fb3_ld8(kp + 16 * c + a_off);
   }
@@ -53,7 +57,9 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dv_v4_bf16(
     const int g_ = (t_idx) / nq;                                          \
     const int q0_ = q_start + ((t_idx) % nq) * FB3_QT;                    \
     const int hq_ = hkv * rep + g_;                                       \
-    const long long qb_ = (((long long)b * Hq + hq_) * T) * FB3_D;        \
+    const long long qb_ =                                                 \
+        bthd ? ((long long)b * T * Hq + hq_) * FB3_D                      \
+             : (((long long)b * Hq + hq_) * T) * FB3_D;                   \
     const long long lb_ = ((long long)b * Hq + hq_) * T;                  \
     _Pragma("unroll") for (int j = 0; j < 2; ++j) {                       \
       int i = threadIdx.x + 512 * j;                                      \
@@ -63,9 +69,9 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dv_v4_bf16(
       short8 qv{0, 0, 0, 0, 0, 0, 0, 0}, dv{0, 0, 0, 0, 0, 0, 0, 0};      \
       if (qrow < T) {                                                     \
         qv = *reinterpret_cast<const short8*>(qb_ + Q +                   \
-                                              (long long)qrow * FB3_D + c); \
+                                              (long long)qrow * q_rs + c); \
         dv = *reinterpret_cast<const short8*>(qb_ + dO +                  \
-                                              (long long)qrow * FB3_D + c); \
+                                              (long long)qrow * q_rs + c); \
       }                                                                   \
       q_stage[j] = qv;                                                    \
       do_stage[j] = dv;                                                   \
@@ -158,13 +164,13 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dv_v4_bf16(
     __syncthreads();
   }
 
-  short* outv = dV + kbase + (long long)wave * 32 * FB3_D;
+  short* outv = dV + kbase + (long long)wave * 32 * kv_rs;
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       int krow = (r & 3) + 8 * (r >> 2) + 4 * hi;
-      outv[(long long)krow * FB3_D + 32 * dt + (lane & 31)] =
+      outv[(long long)krow * kv_rs + 32 * dt + (lane & 31)] =
           f2bf(dv_acc[dt][r]);
     }
 }
@@ -178,7 +184,7 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dk_v4_bf16(
     const short* __restrict__ V, const short* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ Dsum,
     short* __restrict__ dK, int B, int Hq, int Hkv, int T, int causal,
-    float scale) {
+    float scale, int bthd) {
   __shared__ short q_lds[2][FB3_QT][FB3_D];
   __shared__ short do_lds[2][FB3_QT][FB3_D];
   __shared__ float lse_lds[2][FB3_QT];
@@ -195,11 +201,15 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dk_v4_bf16(
   const int hi = lane >> 5;
   const int a_off = 8 * hi;
 
-  const long long kbase = (((long long)b * Hkv + hkv) * T + k0) * FB3_D;
+  const long long q_rs = bthd ? (long long)Hq * FB3_D : FB3_D;
+  const long long kv_rs = bthd ? (long long)Hkv * FB3_D : FB3_D;
+  const long long kbase =
+      bthd ? (((long long)b * T + k0) * Hkv + hkv) * FB3_D
+           : (((long long)b * Hkv + hkv) * T + k0) * FB3_D;
   fb3_bf16x8 k_frag[8], v_frag[8];
   {
-    const short* kp = K + kbase + ((long long)32 * wave + (lane & 31)) * FB3_D;
-    const short* vp = V + kbase + ((long long)32 * wave + (lane & 31)) * FB3_D;
+    const short* kp = K + kbase + ((long long)32 * wave + (lane & 31)) * kv_rs;
+    const short* vp = V + kbase + ((long long)32 * wave + (lane & 31)) * kv_rs;
 #pragma unroll
     for (int c = 0; c < 8; ++c) {
       k_frag[c] = fb3_ld8(kp + 16 * c + a_off);
@@ -226,7 +236,9 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dk_v4_bf16(
     const int g_ = (t_idx) / nq;                                          \
     const int q0_ = q_start + ((t_idx) % nq) * FB3_QT;                    \
     const int hq_ = hkv * rep + g_;                                       \
-    const long long qb_ = (((long long)b * Hq + hq_) * T) * FB3_D;        \
+    const long long qb_ =                                                 \
+        bthd ? ((long long)b * T * Hq + hq_) * FB3_D                      \
+             : (((long long)b * Hq + hq_) * T) * FB3_D;                   \
     const long long lb_ = ((long long)b * Hq + hq_) * T;                  \
     _Pragma("unroll") for (int j = 0; j < 2; ++j) {                       \
       int i = threadIdx.x + 512 * j;                                      \
@@ -237,9 +249,9 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dk_v4_bf16(
       short8 qv{0, 0, 0, 0, 0, 0, 0, 0}, dv{0, 0, 0, 0, 0, 0, 0, 0};      \
       if (qrow < T) {                                                     \
         qv = *reinterpret_cast<const short8*>(qb_ + Q +                   \
-                                              (long long)qrow * FB3_D + c); \
+                                              (long long)qrow * q_rs + c); \
         dv = *reinterpret_cast<const short8*>(qb_ + dO +                  \
-                                              (long long)qrow * FB3_D + c); \
+                                              (long long)qrow * q_rs + c); \
       }                                                                   \
       *reinterpret_cast<short8*>(&q_lds[buf][r][csw]) = qv;               \
       *reinterpret_cast<short8*>(&do_lds[buf][r][csw]) = dv;              \
@@ -321,13 +333,13 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dk_v4_bf16(
     __syncthreads();
   }
 
-  short* outk = dK + kbase + (long long)wave * 32 * FB3_D;
+  short* outk = dK + kbase + (long long)wave * 32 * kv_rs;
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       int krow = (r & 3) + 8 * (r >> 2) + 4 * hi;
-      outk[(long long)krow * FB3_D + 32 * dt + (lane & 31)] =
+      outk[(long long)krow * kv_rs + 32 * dt + (lane & 31)] =
           f2bf(dk_acc[dt][r]);
     }
 }

@@ -56,7 +56,9 @@ flash_attn_fwd_v6_bf16(const short* __restrict__ Q,
                        const short* __restrict__ V, short* __restrict__ O,
                        float* __restrict__ LSE, int B, int Hq, int Hkv,
                        int T, int Tk, int causal, int q_offset,
-                       float scale) {
+                       float scale, int bthd) {
+  // bthd=1: Q/K/V/O are [B,T,H,D] storage (the model's natural layout
+  // after RoPE — no transpose-contiguous copies); bthd=0: [B,H,T,D].
   __shared__ short k_lds[2][FA6_BN][FA6_D];
   __shared__ short v_lds[2][FA6_BN][FA6_D];
 
@@ -71,8 +73,14 @@ flash_attn_fwd_v6_bf16(const short* __restrict__ Q,
   const int hi = lane >> 5;
   const int a_off = 8 * hi;
 
-  const long long qbase = (((long long)b * Hq + hq) * T + q0) * FA6_D;
-  const long long kbase = (((long long)b * Hkv + hkv) * Tk) * FA6_D;
+  const long long q_rs = bthd ? (long long)Hq * FA6_D : FA6_D;
+  const long long kv_rs = bthd ? (long long)Hkv * FA6_D : FA6_D;
+  const long long qbase =
+      bthd ? (((long long)b * T + q0) * Hq + hq) * FA6_D
+           : (((long long)b * Hq + hq) * T + q0) * FA6_D;
+  const long long kbase =
+      bthd ? ((long long)b * Tk * Hkv + hkv) * FA6_D
+           : (((long long)b * Hkv + hkv) * Tk) * FA6_D;
   const int my_q = q0 + wave * FA6_QW + qcol;
   const int gq = q_offset + my_q;                  // causal-global index
   const int wave_gq_min = q_offset + q0 + wave * FA6_QW;
@@ -81,7 +89,7 @@ flash_attn_fwd_v6_bf16(const short* __restrict__ Q,
   // Q fragments (B operand): lane reads its own query row
   fa6_bf16x8 q_frag[8];
   {
-    const short* qp = Q + qbase + ((long long)wave * FA6_QW + qcol) * FA6_D;
+    const short* qp = Q + qbase + ((long long)wave * FA6_QW + qcol) * q_rs;
 #pragma unroll
     for (int c = 0; c < 8; ++c) q_frag[c] = fa6_ld8(qp + 16 * c + a_off);
   }
@@ -110,9 +118,9 @@ flash_attn_fwd_v6_bf16(const short* __restrict__ Q,
       short8 kv{0, 0, 0, 0, 0, 0, 0, 0}, vv{0, 0, 0, 0, 0, 0, 0, 0};      \
       if (krow < Tk) {                                                    \
         kv = *reinterpret_cast<const short8*>(                            \
-            K + kbase + (long long)krow * FA6_D + c);                     \
+            K + kbase + (long long)krow * kv_rs + c);                     \
         vv = *reinterpret_cast<const short8*>(                            \
-            V + kbase + (long long)krow * FA6_D + c);                     \
+            V + kbase + (long long)krow * kv_rs + c);                     \
       }                                                                   \
       k_stage[j] = kv;                                                    \
       v_stage[j] = vv;                                                    \
@@ -265,7 +273,7 @@ flash_attn_fwd_v6_bf16(const short* __restrict__ Q,
   // ---- epilogue: O[q][d] = O'[d][q] / l ----
   float inv = (l_run > 0.f) ? 1.f / l_run : 0.f;
   if (my_q < T) {
-    short* op = O + qbase + ((long long)wave * FA6_QW + qcol) * FA6_D;
+    short* op = O + qbase + ((long long)wave * FA6_QW + qcol) * q_rs;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt)
 #pragma unroll

@@ -8,10 +8,8 @@
 #include "hip/elementwise.hip"
 #include "hip/rl_scans.hip"
 #include "hip/cross_entropy.hip"
-#include "hip/flash_attn.hip"
-#include "hip/flash_attn_v2.hip"
-#include "hip/flash_attn_v3.hip"
-#include "hip/flash_attn_v4.hip"
+// Retired ladder rungs (flash_attn.hip v1-v4) stay in the tree as the
+// evidence trail but are no longer compiled into the shipped binary.
 #include "hip/flash_attn_v5.hip"
 #include "hip/flash_attn_v6.hip"
 #include "hip/fa_bwd.hip"
@@ -222,17 +220,41 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor mx,
   return dl;
 }
 
+
+// Layout of a [B,H,T,D] tensor: 0 = contiguous BHTD, 1 = a transpose
+// view of [B,T,H,D] storage (the model's natural layout), -1 = other.
+static inline int attn_layout(const at::Tensor& t) {
+  if (t.stride(3) != 1) return -1;
+  long long B = t.size(0), H = t.size(1), T = t.size(2), D = t.size(3);
+  if (t.stride(2) == D && t.stride(1) == T * D && t.stride(0) == H * T * D)
+    return 0;
+  if (t.stride(1) == D && t.stride(2) == H * D && t.stride(0) == H * T * D)
+    return 1;
+  return -1;
+}
+
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
                                         at::Tensor v, bool causal,
                                         int64_t q_offset, bool want_lse) {
-  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v);
+  TORCH_CHECK(q.is_cuda() && k.is_cuda() && v.is_cuda(),
+              "flash_attn tensors must be on GPU");
+  int bthd = attn_layout(q);
+  TORCH_CHECK(bthd >= 0 && attn_layout(k) == bthd && attn_layout(v) == bthd,
+              "q/k/v must share a BHTD-contiguous or BTHD-view layout");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "q must be bf16");
   TORCH_CHECK(q.size(3) == 128, "head_dim must be 128");
   TORCH_CHECK(q.size(2) % 128 == 0, "T must be a multiple of 128 (pad)");
   int B = (int)q.size(0), Hq = (int)q.size(1), T = (int)q.size(2);
   int Hkv = (int)k.size(1), Tk = (int)k.size(2);
   TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
-  auto o = at::empty_like(q);
+  TORCH_CHECK(!bthd || T % 256 == 0,
+              "BTHD layout needs the v6 kernel (T % 256 == 0)");
+  // O matches q's layout so the model's out.transpose().reshape stays
+  // a free view
+  auto o = bthd
+               ? at::empty({B, T, Hq, (int)q.size(3)}, q.options())
+                     .permute({0, 2, 1, 3})
+               : at::empty_like(q);
   at::Tensor lse;
   float* lse_ptr = nullptr;
   if (want_lse) {
@@ -240,33 +262,7 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
     lse_ptr = lse.data_ptr<float>();
   }
   float scale = 1.0f / sqrtf((float)q.size(3));
-  static const bool use_v1 = getenv("RAY_AMD_FA_V1") != nullptr;
-  static const bool use_v2 = getenv("RAY_AMD_FA_V2") != nullptr;
-  if (use_v1)
-    hipLaunchKernelGGL(flash_attn_fwd_bf16, dim3(T / 64, B * Hq), dim3(256),
-                       0, cur_stream(), (const short*)q.data_ptr(),
-                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-                       (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
-                       causal ? 1 : 0, (int)q_offset, scale);
-  else if (use_v2)
-    hipLaunchKernelGGL(flash_attn_fwd_v2_bf16, dim3(T / 64, B * Hq),
-                       dim3(256), 0, cur_stream(), (const short*)q.data_ptr(),
-                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-                       (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
-                       causal ? 1 : 0, (int)q_offset, scale);
-  else if (getenv("RAY_AMD_FA_V3") != nullptr)
-    hipLaunchKernelGGL(flash_attn_fwd_v3_bf16, dim3(T / 128, B * Hq),
-                       dim3(256), 0, cur_stream(), (const short*)q.data_ptr(),
-                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-                       (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
-                       causal ? 1 : 0, (int)q_offset, scale);
-  else if (getenv("RAY_AMD_FA_V4") != nullptr)
-    hipLaunchKernelGGL(flash_attn_fwd_v4_bf16, dim3(T / 128, B * Hq),
-                       dim3(256), 0, cur_stream(), (const short*)q.data_ptr(),
-                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
-                       (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
-                       causal ? 1 : 0, (int)q_offset, scale);
-  else if (getenv("RAY_AMD_FA_V5") != nullptr || T % 256 != 0)
+  if (!bthd && (getenv("RAY_AMD_FA_V5") != nullptr || T % 256 != 0))
     hipLaunchKernelGGL(flash_attn_fwd_v5_bf16, dim3(T / 128, B * Hq),
                        dim3(256), 0, cur_stream(), (const short*)q.data_ptr(),
                        (const short*)k.data_ptr(), (const short*)v.data_ptr(),
@@ -278,7 +274,7 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
                        dim3(512), 0, cur_stream(), (const short*)q.data_ptr(),
                        (const short*)k.data_ptr(), (const short*)v.data_ptr(),
                        (short*)o.data_ptr(), lse_ptr, B, Hq, Hkv, T, Tk,
-                       causal ? 1 : 0, (int)q_offset, scale);
+                       causal ? 1 : 0, (int)q_offset, scale, bthd);
   if (want_lse) return {o, lse};
   return {o};
 }
@@ -287,7 +283,11 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
                                         at::Tensor v, at::Tensor o,
                                         at::Tensor d_o, at::Tensor lse,
                                         bool causal) {
-  CHECK_IN(q); CHECK_IN(k); CHECK_IN(v); CHECK_IN(o); CHECK_IN(d_o);
+  TORCH_CHECK(q.is_cuda(), "flash_attn_bwd tensors must be on GPU");
+  int bthd = attn_layout(q);
+  TORCH_CHECK(bthd >= 0 && attn_layout(k) == bthd && attn_layout(v) == bthd
+                  && attn_layout(o) == bthd && attn_layout(d_o) == bthd,
+              "q/k/v/o/dO must share a BHTD or BTHD-view layout");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "q must be bf16");
   TORCH_CHECK(q.size(3) == 128, "head_dim must be 128");
   TORCH_CHECK(q.size(2) % 128 == 0, "T must be a multiple of 128");
@@ -296,25 +296,34 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
   int B = (int)q.size(0), Hq = (int)q.size(1), T = (int)q.size(2);
   int Hkv = (int)k.size(1);
   TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
-  auto dq = at::empty_like(q);
-  auto dk = at::empty_like(k);
-  auto dv = at::empty_like(v);
+  TORCH_CHECK(!bthd || T % 256 == 0,
+              "BTHD layout needs the v4 kernels (T % 256 == 0)");
+  int D = (int)q.size(3);
+  auto mk_like = [&](const at::Tensor& t, int H) {
+    return bthd ? at::empty({B, T, H, D}, t.options()).permute({0, 2, 1, 3})
+                : at::empty_like(t);
+  };
+  auto dq = mk_like(q, Hq);
+  auto dk = mk_like(k, Hkv);
+  auto dv = mk_like(v, Hkv);
   auto dsum = at::empty({B, Hq, T}, q.options().dtype(at::kFloat));
   float scale = 1.0f / sqrtf((float)q.size(3));
   long long rows = (long long)B * Hq * T;
   hipLaunchKernelGGL(fa_bwd_prep_bf16, dim3((rows + 3) / 4), dim3(256), 0,
                      cur_stream(), (const short*)d_o.data_ptr(),
                      (const short*)o.data_ptr(), dsum.data_ptr<float>(),
-                     rows);
+                     rows, Hq, T, bthd);
   static const bool split = getenv("RAY_AMD_FA_BWD_FUSED") == nullptr;
   static const bool v3 = getenv("RAY_AMD_FA_BWD_V2") == nullptr;
+  TORCH_CHECK(!bthd || (split && v3),
+              "BTHD layout requires the v3/v4 backward path");
   if (v3)
     hipLaunchKernelGGL(fa_bwd_dq_v3_bf16, dim3(T / 128, B * Hq), dim3(256),
                        0, cur_stream(), (const short*)q.data_ptr(),
                        (const short*)k.data_ptr(), (const short*)v.data_ptr(),
                        (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
                        dsum.data_ptr<float>(), (short*)dq.data_ptr(), B, Hq,
-                       Hkv, T, causal ? 1 : 0, scale);
+                       Hkv, T, causal ? 1 : 0, scale, bthd);
   else
     hipLaunchKernelGGL(fa_bwd_dq_bf16, dim3(T / 128, B * Hq), dim3(256), 0,
                        cur_stream(), (const short*)q.data_ptr(),
@@ -326,7 +335,8 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
     // v3/v4: K/V register-resident, 64-row dbuf staged q/dO tiles,
     // swizzled LDS reads, one barrier per tile; v4 = 8-wave blocks
     // (256 kv rows) when the shape allows
-    bool v4 = (T % 256 == 0) && getenv("RAY_AMD_FA_BWD_V3ONLY") == nullptr;
+    bool v4 = (T % 256 == 0) &&
+              (bthd || getenv("RAY_AMD_FA_BWD_V3ONLY") == nullptr);
     if (v4) {
       hipLaunchKernelGGL(fa_bwd_dv_v4_bf16, dim3(T / 256, B * Hkv),
                          dim3(512), 0, cur_stream(),
@@ -335,7 +345,7 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
                          (const short*)v.data_ptr(),
                          (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
                          (short*)dv.data_ptr(), B, Hq, Hkv, T,
-                         causal ? 1 : 0, scale);
+                         causal ? 1 : 0, scale, bthd);
       hipLaunchKernelGGL(fa_bwd_dk_v4_bf16, dim3(T / 256, B * Hkv),
                          dim3(512), 0, cur_stream(),
                          (const short*)q.data_ptr(),
@@ -343,7 +353,7 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
                          (const short*)v.data_ptr(),
                          (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
                          dsum.data_ptr<float>(), (short*)dk.data_ptr(), B,
-                         Hq, Hkv, T, causal ? 1 : 0, scale);
+                         Hq, Hkv, T, causal ? 1 : 0, scale, bthd);
       return {dq, dk, dv};
     }
     hipLaunchKernelGGL(fa_bwd_dv_v3_bf16, dim3(T / 128, B * Hkv),

@@ -374,6 +374,16 @@ def flash_attention_ref(q, k, v, causal=True):
     return out.to(q.dtype), lse
 
 
+def _is_bthd_view(t):
+    """[B,H,T,D] tensor that is a transpose view of [B,T,H,D] storage
+    (the model's natural post-RoPE layout) — the kernels read it
+    directly, no transpose-contiguous copy."""
+    B, H, T, D = t.shape
+    st = t.stride()
+    return (st[3] == 1 and st[1] == D and st[2] == H * D
+            and st[0] == H * T * D)
+
+
 class _FlashAttnFn(torch.autograd.Function):
     """Training flash attention: v5 forward + FA2-style two-kernel
     backward (csrc/hip/fa_bwd.hip). Requires T == Tk, T % 128 == 0,
@@ -381,17 +391,29 @@ class _FlashAttnFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, q, k, v, causal):
-        qc, kc, vc = q.contiguous(), k.contiguous(), v.contiguous()
+        T = q.shape[2]
+        bthd = (T % 256 == 0 and _is_bthd_view(q) and _is_bthd_view(k)
+                and _is_bthd_view(v))
+        if bthd:
+            qc, kc, vc = q, k, v
+        else:
+            qc, kc, vc = q.contiguous(), k.contiguous(), v.contiguous()
         out, lse = _K.flash_attn_fwd(qc, kc, vc, causal, 0, True)
         ctx.save_for_backward(qc, kc, vc, out, lse)
         ctx.causal = causal
+        ctx.bthd = bthd
         return out
 
     @staticmethod
     def backward(ctx, d_out):
         q, k, v, out, lse = ctx.saved_tensors
+        if ctx.bthd:
+            if not _is_bthd_view(d_out):
+                d_out = d_out.transpose(1, 2).contiguous().transpose(1, 2)
+        else:
+            d_out = d_out.contiguous()
         dq, dk, dv = _K.flash_attn_bwd(
-            q, k, v, out, d_out.contiguous(), lse, ctx.causal
+            q, k, v, out, d_out, lse, ctx.causal
         )
         return dq, dk, dv, None
 
@@ -426,8 +448,12 @@ def flash_attention(q, k, v, causal: bool = True, q_offset: int = 0,
     pad = (-T) % 128
     if pad:
         q = torch.nn.functional.pad(q, (0, 0, 0, pad))
-    r = _K.flash_attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(),
-                          causal, q_offset, return_lse)
+    if not pad and T % 256 == 0 and _is_bthd_view(q) and _is_bthd_view(k) \
+            and _is_bthd_view(v):
+        r = _K.flash_attn_fwd(q, k, v, causal, q_offset, return_lse)
+    else:
+        r = _K.flash_attn_fwd(q.contiguous(), k.contiguous(),
+                              v.contiguous(), causal, q_offset, return_lse)
     out = r[0][:, :, :T] if pad else r[0]
     if return_lse:
         lse = r[1][:, :, :T] if pad else r[1]

@@ -360,3 +360,31 @@ def test_flash_attention_backward_gpu():
             rel = (ours - refg).norm() / (refg.norm() + 1e-6)
             assert torch.isfinite(ours).all(), (name, B, Hq, T, causal)
             assert rel < 0.03, (name, B, Hq, Hkv, T, causal, float(rel))
+
+
+@pytest.mark.gpu
+def test_flash_attention_bthd_layout_matches():
+    """BTHD-view inputs (the model's natural post-RoPE layout) must
+    produce identical results and grads to the contiguous path — with
+    no transpose-contiguous copies."""
+    torch.manual_seed(0)
+    B, Hq, Hkv, T, D = 2, 8, 2, 512, 128
+    base_q = torch.randn(B, T, Hq, D, device="cuda", dtype=torch.bfloat16)
+    base_k = torch.randn(B, T, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    base_v = torch.randn(B, T, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    g = torch.randn(B, Hq, T, D, device="cuda", dtype=torch.bfloat16)
+
+    def run(make):
+        q = make(base_q).requires_grad_()
+        k = make(base_k).requires_grad_()
+        v = make(base_v).requires_grad_()
+        out = ops.flash_attention(q, k, v, causal=True)
+        out.backward(g)
+        return out.detach().float(), q.grad.float(), k.grad.float(), v.grad.float()
+
+    o1, dq1, dk1, dv1 = run(lambda t: t.detach().clone().transpose(1, 2))
+    o2, dq2, dk2, dv2 = run(
+        lambda t: t.detach().clone().transpose(1, 2).contiguous()
+    )
+    for a, b in ((o1, o2), (dq1, dq2), (dk1, dk2), (dv1, dv2)):
+        assert torch.equal(a, b) or (a - b).abs().max().item() < 1e-6
