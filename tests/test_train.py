@@ -280,3 +280,61 @@ def test_huggingface_model_in_trainer(ray_start_regular, tmp_path):
     res = t.fit()
     assert res.error is None
     assert res.metrics["last"] <= res.metrics["first"] + 0.1
+
+
+def test_hf_transformers_trainer_integration(ray_start_regular, tmp_path):
+    """transformers.Trainer inside a TorchTrainer worker: HF picks up
+    the worker's RANK/WORLD_SIZE env, metrics and checkpoints flow back
+    through ray_amd.train.report (reference:
+    train/huggingface/transformers prepare_trainer +
+    RayTrainReportCallback)."""
+    import ray_amd.train as train
+    from ray_amd.train import ScalingConfig, RunConfig
+    from ray_amd.train.torch import TorchTrainer
+
+    out_dir = str(tmp_path / "hf_out")
+
+    def loop(cfg):
+        import torch
+        import numpy as np
+        from transformers import (Trainer, TrainingArguments)
+        from ray_amd.train.huggingface import (RayTrainReportCallback,
+                                               prepare_trainer)
+
+        class TinyModel(torch.nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.lin = torch.nn.Linear(4, 2)
+
+            def forward(self, x=None, labels=None):
+                logits = self.lin(x)
+                loss = torch.nn.functional.cross_entropy(logits, labels)
+                return {"loss": loss, "logits": logits}
+
+        class DS(torch.utils.data.Dataset):
+            def __len__(self):
+                return 32
+
+            def __getitem__(self, i):
+                return {"x": torch.randn(4), "labels": i % 2}
+
+        args = TrainingArguments(
+            output_dir=cfg["out_dir"], num_train_epochs=1,
+            per_device_train_batch_size=8, logging_steps=1,
+            save_steps=2, save_total_limit=1, report_to=[],
+            use_cpu=True, disable_tqdm=True,
+        )
+        trainer = Trainer(model=TinyModel(), args=args,
+                          train_dataset=DS())
+        trainer.add_callback(RayTrainReportCallback().unwrap())
+        trainer = prepare_trainer(trainer)
+        trainer.train()
+
+    t = TorchTrainer(
+        loop, train_loop_config={"out_dir": out_dir},
+        scaling_config=ScalingConfig(num_workers=1),
+        run_config=RunConfig(name="hf_test"),
+    )
+    result = t.fit()
+    assert result.error is None, result.error
+    assert result.metrics and "loss" in result.metrics or "step" in (result.metrics or {})
