@@ -141,3 +141,36 @@ class BC(Algorithm):
 
     def set_weights(self, w):
         self.module.set_weights(w)
+
+
+def record_continuous_episodes(env: str, policy_fn=None,
+                               num_steps: int = 2000, num_envs: int = 8,
+                               seed: int = 0):
+    """Continuous-action transition recorder: rows carry next_obs too
+    (what CQL/IQL consume)."""
+    from .. import data as ray_data
+
+    vec = VectorEnv(env, num_envs, seed=seed)
+    rng = np.random.default_rng(seed)
+    act_dim = int(np.prod(vec.action_space.shape))
+    limit = float(np.max(np.abs(vec.action_space.high)))
+    obs = vec.reset()
+    rows = []
+    for _ in range(num_steps // num_envs):
+        if policy_fn is None:
+            acts = rng.uniform(-limit, limit,
+                               size=(num_envs, act_dim)).astype(np.float32)
+        else:
+            acts = np.asarray(policy_fn(obs), np.float32)
+        nobs, rew, term, trunc = vec.step(acts)
+        done = np.logical_or(term, trunc)
+        for i in range(num_envs):
+            rows.append({
+                "obs": obs[i].astype(np.float32),
+                "next_obs": nobs[i].astype(np.float32),
+                "action": acts[i],
+                "reward": float(rew[i]),
+                "done": bool(done[i]),
+            })
+        obs = nobs
+    return ray_data.from_items(rows)
