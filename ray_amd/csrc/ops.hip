@@ -15,6 +15,7 @@
 #include "hip/fa_bwd.hip"
 #include "hip/fa_bwd_v3.hip"
 #include "hip/fa_bwd_v4.hip"
+#include "hip/fa_bwd_dq_v4.hip"
 
 #define CHECK_IN(x)                                                     \
   TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                       \
@@ -317,7 +318,14 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
   static const bool v3 = getenv("RAY_AMD_FA_BWD_V2") == nullptr;
   TORCH_CHECK(!bthd || (split && v3),
               "BTHD layout requires the v3/v4 backward path");
-  if (v3)
+  if (v3 && T % 256 == 0)
+    hipLaunchKernelGGL(fa_bwd_dq_v4_bf16, dim3(T / 256, B * Hq), dim3(512),
+                       0, cur_stream(), (const short*)q.data_ptr(),
+                       (const short*)k.data_ptr(), (const short*)v.data_ptr(),
+                       (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
+                       dsum.data_ptr<float>(), (short*)dq.data_ptr(), B, Hq,
+                       Hkv, T, causal ? 1 : 0, scale, bthd);
+  else if (v3)
     hipLaunchKernelGGL(fa_bwd_dq_v3_bf16, dim3(T / 128, B * Hq), dim3(256),
                        0, cur_stream(), (const short*)q.data_ptr(),
                        (const short*)k.data_ptr(), (const short*)v.data_ptr(),

@@ -175,6 +175,14 @@ class LlamaModel(nn.Module):
                 x = layer(x, self.cosT, self.sinT, cache, pos0)
         x = self.final_norm(x)
         if targets is not None:
+            if x.is_cuda and not os.environ.get("RAY_AMD_NO_CHUNKED_CE"):
+                # chunked fused lm_head+CE: the 8.4 GB logits tensor
+                # (flagship shape) is never materialized whole
+                return ops.lm_head_cross_entropy(
+                    x.reshape(-1, self.cfg.hidden_size),
+                    self.lm_head.weight,
+                    targets.reshape(-1),
+                )
             logits = self.lm_head(x)
             return ops.cross_entropy(
                 logits.view(-1, self.cfg.vocab_size), targets.reshape(-1)

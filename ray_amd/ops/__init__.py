@@ -191,6 +191,60 @@ def cross_entropy(logits: torch.Tensor, target: torch.Tensor):
     return _FusedCE.apply(logits, target)
 
 
+class _ChunkedLmHeadCE(torch.autograd.Function):
+    """Fused lm_head + cross entropy, row-chunked: the [N, V] logits
+    are never materialized whole (N=B*T=32k, V=128k bf16 = 8.4 GB at
+    the flagship shape — plus its stored-activation copy). Forward
+    computes the loss per chunk; backward RECOMPUTES each chunk's
+    logits and uses the fused CE kernels for dlogits, folding straight
+    into dX and a fp32 dW accumulator. Costs one extra lm_head GEMM,
+    saves ~15 GB peak (headroom for larger micro-batches)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, target, chunk_rows):
+        _require_ext("cross_entropy")
+        x = x.contiguous()
+        t32 = target.to(torch.int32).contiguous()
+        N = x.shape[0]
+        nvalid = (t32 >= 0).sum().clamp(min=1)
+        loss_sum = torch.zeros((), dtype=torch.float32, device=x.device)
+        for s in range(0, N, chunk_rows):
+            e = min(s + chunk_rows, N)
+            logits = (x[s:e] @ weight.t()).contiguous()
+            l, mx, lse = _K.ce_fwd(logits, t32[s:e])
+            loss_sum += l.sum()
+        ctx.save_for_backward(x, weight, t32, nvalid)
+        ctx.chunk = chunk_rows
+        return loss_sum / nvalid.float()
+
+    @staticmethod
+    def backward(ctx, dloss):
+        x, w, t32, nvalid = ctx.saved_tensors
+        N = x.shape[0]
+        chunk = ctx.chunk
+        dx = torch.empty_like(x)
+        dw_acc = torch.zeros_like(w, dtype=torch.float32)
+        scale = dloss / nvalid.float()
+        for s in range(0, N, chunk):
+            e = min(s + chunk, N)
+            logits = (x[s:e] @ w.t()).contiguous()
+            _, mx, lse = _K.ce_fwd(logits, t32[s:e])
+            per_row = scale.expand(e - s).contiguous()
+            dl = _K.ce_bwd(logits, t32[s:e], mx, lse, per_row)
+            dx[s:e] = dl @ w
+            dw_acc += (dl.t() @ x[s:e]).float()
+        return dx, dw_acc.to(w.dtype), None, None
+
+
+def lm_head_cross_entropy(x: torch.Tensor, weight: torch.Tensor,
+                          target: torch.Tensor,
+                          chunk_rows: int = 8192) -> torch.Tensor:
+    """Chunked fused projection + CE; x [N, H] bf16, weight [V, H]."""
+    if not x.is_cuda:
+        return cross_entropy_ref(x.float() @ weight.t().float(), target)
+    return _ChunkedLmHeadCE.apply(x, weight, target, chunk_rows)
+
+
 # --------------------------------------------------------------------------
 # Fused AdamW
 # --------------------------------------------------------------------------

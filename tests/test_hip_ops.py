@@ -388,3 +388,31 @@ def test_flash_attention_bthd_layout_matches():
     )
     for a, b in ((o1, o2), (dq1, dq2), (dk1, dk2), (dv1, dv2)):
         assert torch.equal(a, b) or (a - b).abs().max().item() < 1e-6
+
+
+@pytest.mark.gpu
+def test_chunked_lm_head_ce_matches_plain():
+    """Chunked fused lm_head+CE (never materializes full logits) must
+    match the plain logits->cross_entropy path in loss and grads."""
+    torch.manual_seed(0)
+    N, H, V = 1024, 256, 1000
+    x0 = torch.randn(N, H, device="cuda", dtype=torch.bfloat16)
+    w0 = torch.randn(V, H, device="cuda", dtype=torch.bfloat16) * 0.02
+    t = torch.randint(0, V, (N,), device="cuda")
+    t[::7] = -100  # ignored rows
+
+    x1 = x0.clone().requires_grad_()
+    w1 = w0.clone().requires_grad_()
+    loss1 = ops.lm_head_cross_entropy(x1, w1, t, chunk_rows=300)
+    loss1.backward()
+
+    x2 = x0.clone().requires_grad_()
+    w2 = w0.clone().requires_grad_()
+    loss2 = ops.cross_entropy(x2 @ w2.t(), t)
+    loss2.backward()
+
+    assert abs(loss1.item() - loss2.item()) < 2e-3
+    def relerr(a, b):
+        return ((a.float() - b.float()).norm() / (b.float().norm() + 1e-6)).item()
+    assert relerr(x1.grad, x2.grad) < 0.03
+    assert relerr(w1.grad, w2.grad) < 0.03
