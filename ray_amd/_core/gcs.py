@@ -220,6 +220,7 @@ class GcsServer:
                 "resources_available": n.resources_available,
                 "labels": n.labels,
                 "pending": getattr(n, "pending", 0),
+                "pending_shapes": getattr(n, "pending_shapes", []),
             }
             for n in self.nodes.values()
         ]
@@ -229,6 +230,7 @@ class GcsServer:
         if n is not None:
             n.resources_available = p["available"]
             n.pending = p.get("pending", 0)
+            n.pending_shapes = p.get("pending_shapes") or []
             if p.get("total") is not None:  # dynamic resource change
                 n.resources_total = p["total"]
 

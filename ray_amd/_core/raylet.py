@@ -154,12 +154,16 @@ class Raylet:
             snap = dict(self.avail)
             now = time.time()
             pending = self._pending_total()
+            shapes = [
+                [dict(key[0]), len(q)]
+                for key, q in self._pending.items() if q
+            ]
             if snap != last or now - last_time > 1.0:
                 try:
                     self.gcs.notify(
                         "report_resources",
                         {"node_id": self.node_id, "available": snap,
-                         "pending": pending},
+                         "pending": pending, "pending_shapes": shapes},
                     )
                     last = snap
                     last_time = now

@@ -9,3 +9,9 @@ reported by raylets + explicit requests) and node lifecycle are real.
 """
 from . import sdk  # noqa: F401
 from .autoscaler import LocalAutoscaler  # noqa: F401
+from .v2 import (  # noqa: F401
+    AutoscalerV2,
+    LocalNodeProvider,
+    NodeProvider,
+    NodeType,
+)

@@ -175,9 +175,11 @@ class LlamaModel(nn.Module):
                 x = layer(x, self.cosT, self.sinT, cache, pos0)
         x = self.final_norm(x)
         if targets is not None:
-            if x.is_cuda and not os.environ.get("RAY_AMD_NO_CHUNKED_CE"):
-                # chunked fused lm_head+CE: the 8.4 GB logits tensor
-                # (flagship shape) is never materialized whole
+            if x.is_cuda and os.environ.get("RAY_AMD_CHUNKED_CE"):
+                # opt-in chunked fused lm_head+CE: saves ~15 GB peak
+                # (the full logits tensor is never materialized) at the
+                # cost of one extra lm_head GEMM (~2% step) — for
+                # memory-bound configs; measured slower at mb8
                 return ops.lm_head_cross_entropy(
                     x.reshape(-1, self.cfg.hidden_size),
                     self.lm_head.weight,

@@ -50,3 +50,46 @@ def test_request_resources_triggers_scale(ray_start_cluster):
         assert ray.cluster_resources()["CPU"] >= 10
     finally:
         asc.stop()
+
+
+def test_autoscaler_v2_scales_for_demand_shapes():
+    """v2: pending lease SHAPES drive bin-packed launches through the
+    instance-manager FSM; idle nodes terminate after the timeout
+    (reference: autoscaler/v2 scheduler + instance_manager)."""
+    import ray_amd as ray
+    from ray_amd.autoscaler import AutoscalerV2, LocalNodeProvider, NodeType
+    from ray_amd.cluster_utils import Cluster
+
+    cluster = Cluster(head_node_args={"num_cpus": 1})
+    try:
+        cluster.connect()
+        asc = AutoscalerV2(
+            LocalNodeProvider(cluster),
+            [NodeType("worker_2cpu", {"CPU": 2}, min_workers=0,
+                      max_workers=3)],
+            idle_timeout_s=2.0, poll_s=0.2, upscale_after_s=0.2,
+        ).start()
+        try:
+            @ray.remote(num_cpus=2)
+            def heavy(i):
+                time.sleep(1.0)
+                return i
+
+            # head has 1 CPU: every task needs a new 2-CPU worker shape
+            refs = [heavy.remote(i) for i in range(4)]
+            assert sorted(ray.get(refs, timeout=120)) == [0, 1, 2, 3]
+            summ = asc.summary()
+            assert summ.get("worker_2cpu", {}).get("RUNNING", 0) >= 1
+            # idle scale-down
+            deadline = time.time() + 40
+            while time.time() < deadline:
+                s = asc.summary().get("worker_2cpu", {})
+                if s.get("RUNNING", 0) == 0 and s.get("TERMINATED", 0) >= 1:
+                    break
+                time.sleep(0.3)
+            s = asc.summary().get("worker_2cpu", {})
+            assert s.get("TERMINATED", 0) >= 1, s
+        finally:
+            asc.stop()
+    finally:
+        cluster.shutdown()
