@@ -87,6 +87,9 @@ class Raylet:
         # request at one class's head no longer blocks other classes
         self._pending: Dict[tuple, deque] = {}
         self._class_order: deque = deque()
+        # cluster-infeasible requests waiting for new capacity
+        self._infeasible: List[tuple] = []
+        self._infeasible_running = False
         self._worker_ready: Dict[int, asyncio.Future] = {}  # pid -> fut
         self._actor_start_futs: Dict[bytes, asyncio.Future] = {}
         self._proto_worker: Dict[int, WorkerProc] = {}
@@ -158,6 +161,9 @@ class Raylet:
                 [dict(key[0]), len(q)]
                 for key, q in self._pending.items() if q
             ]
+            for req, fut in self._infeasible:
+                if not fut.done():
+                    shapes.append([dict(req), 1])
             if snap != last or now - last_time > 1.0:
                 try:
                     self.gcs.notify(
@@ -311,7 +317,15 @@ class Raylet:
             spill = await self._find_spill_target(req)
             if spill:
                 return {"spill": spill}
-            return {"error": f"infeasible resource request {req} on this cluster"}
+            # Cluster-infeasible RIGHT NOW: park the request and keep
+            # re-evaluating — an autoscaler may add a node that fits
+            # (reference: the infeasible lease queue in
+            # cluster_lease_manager; demand is advertised to the
+            # autoscaler through pending_shapes).
+            fut = asyncio.get_running_loop().create_future()
+            self._infeasible.append((req, fut))
+            self._ensure_infeasible_loop()
+            return await fut
         fut = asyncio.get_running_loop().create_future()
         key = (tuple(sorted(req.items())), pg)
         q = self._pending.get(key)
@@ -323,7 +337,42 @@ class Raylet:
         return await fut
 
     def _pending_total(self) -> int:
-        return sum(len(q) for q in self._pending.values())
+        return sum(len(q) for q in self._pending.values()) + len(
+            self._infeasible)
+
+    def _ensure_infeasible_loop(self):
+        if self._infeasible_running:
+            return
+        self._infeasible_running = True
+        asyncio.ensure_future(self._infeasible_loop())
+
+    async def _infeasible_loop(self):
+        try:
+            while self._infeasible:
+                await asyncio.sleep(0.25)
+                still = []
+                for req, fut in self._infeasible:
+                    if fut.done():
+                        continue
+                    if self._feasible_total(req):
+                        # this node can now host it: re-enter the
+                        # normal lease path
+                        key = (tuple(sorted(req.items())), None)
+                        q = self._pending.get(key)
+                        if q is None:
+                            q = self._pending[key] = deque()
+                            self._class_order.append(key)
+                        q.append((req, None, fut))
+                        self._try_grant()
+                        continue
+                    spill = await self._find_spill_target(req)
+                    if spill:
+                        fut.set_result({"spill": spill})
+                        continue
+                    still.append((req, fut))
+                self._infeasible = still
+        finally:
+            self._infeasible_running = False
 
     def _bundle_key(self, pg):
         pg_id, idx = pg
