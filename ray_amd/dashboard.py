@@ -37,6 +37,24 @@ def build_asgi_app():
                 from ray_amd.util import state as st
 
                 body = _json(st.list_placement_groups(limit=1000))
+            elif path == "/api/objects":
+                from ray_amd.util import state as st
+
+                body = _json(st.list_objects())
+            elif path == "/api/debug_state":
+                from ray_amd.util.state import node_debug_state
+
+                body = _json(node_debug_state())
+            elif path == "/api/serve":
+                body = _json(_serve_status())
+            elif path == "/api/timeline":
+                import ray_amd as ray
+
+                body = _json(ray.timeline() or [])
+            elif path == "/api/logs":
+                body = _json(_log_files())
+            elif path.startswith("/api/logs/"):
+                ctype, body = "text/plain", _log_tail(path[len("/api/logs/"):])
             elif path == "/metrics":
                 from ray_amd.util import metrics
 
@@ -85,6 +103,47 @@ def _jobs():
         return []
 
 
+def _serve_status():
+    """Serve app/deployment/replica view (reference:
+    dashboard/modules/serve)."""
+    import ray_amd as ray
+
+    try:
+        from ray_amd.serve.api import SERVE_CONTROLLER_NAME, SERVE_NAMESPACE
+
+        ctrl = ray.get_actor(SERVE_CONTROLLER_NAME, namespace=SERVE_NAMESPACE)
+        return ray.get(ctrl.status.remote(), timeout=10)
+    except Exception:
+        return {}
+
+
+def _log_files():
+    from ray_amd._core import runtime as rtmod
+    import os
+
+    rt = rtmod.global_runtime()
+    d = os.path.join(rt.session_dir, "logs")
+    try:
+        return sorted(os.listdir(d))
+    except OSError:
+        return []
+
+
+def _log_tail(name: str, n: int = 200) -> bytes:
+    from ray_amd._core import runtime as rtmod
+    import os
+
+    if "/" in name or ".." in name:
+        return b"bad log name"
+    rt = rtmod.global_runtime()
+    p = os.path.join(rt.session_dir, "logs", name)
+    try:
+        with open(p, "rb") as f:
+            return b"\n".join(f.read().splitlines()[-n:])
+    except OSError:
+        return b"no such log"
+
+
 def _cluster_status():
     import ray_amd as ray
     from ray_amd.util import state as st
@@ -115,7 +174,21 @@ async function refresh(){
 }
 refresh(); setInterval(refresh, 2000);
 </script>
-<p>APIs: /api/nodes /api/actors /api/tasks /api/jobs /api/placement_groups /metrics</p>
+<div id="tabs"></div>
+<pre id="detail"></pre>
+<script>
+const APIS=["nodes","actors","tasks","jobs","placement_groups","objects",
+            "serve","debug_state","logs"];
+const tabs=document.getElementById('tabs');
+APIS.forEach(a=>{
+  const b=document.createElement('button'); b.textContent=a;
+  b.onclick=async()=>{const r=await fetch('/api/'+a);
+    document.getElementById('detail').textContent=
+      JSON.stringify(await r.json(),null,2);};
+  tabs.appendChild(b);});
+</script>
+<p>APIs: /api/{nodes,actors,tasks,jobs,placement_groups,objects,serve,
+debug_state,timeline,logs,logs/&lt;file&gt;} /metrics</p>
 </body></html>"""
 
 

@@ -762,3 +762,51 @@ def test_handler_event_loop_stats(ray_start_regular):
     assert hs["request_lease"]["mean_us"] > 0
     assert st["gcs"]["handler_stats"]["report_resources"]["count"] >= 1
     assert st["raylet"]["store"]["capacity"] > 0
+
+
+def test_dashboard_routes(ray_start_regular):
+    """Dashboard REST surface (reference: dashboard/modules/*):
+    nodes/actors/objects/debug_state/logs all serve JSON."""
+    import asyncio
+    import json as _json
+
+    from ray_amd.dashboard import build_asgi_app
+
+    @ray.remote
+    def f():
+        return 1
+
+    ray.get(f.remote())
+    app = build_asgi_app()
+
+    async def hit(path):
+        out = {}
+
+        async def send(msg):
+            if msg["type"] == "http.response.start":
+                out["status"] = msg["status"]
+            else:
+                out.setdefault("body", b"")
+                out["body"] += msg.get("body", b"")
+
+        async def receive():
+            return {"type": "http.request"}
+
+        await app({"type": "http", "path": path, "headers": []},
+                  receive, send)
+        return out
+
+    loop = asyncio.new_event_loop()
+    try:
+        for p in ("/api/nodes", "/api/actors", "/api/objects",
+                  "/api/cluster_status", "/api/debug_state", "/api/logs",
+                  "/api/serve", "/api/placement_groups"):
+            r = loop.run_until_complete(hit(p))
+            assert r["status"] == 200, (p, r)
+            _json.loads(r["body"])
+        r = loop.run_until_complete(hit("/metrics"))
+        assert r["status"] == 200
+        r = loop.run_until_complete(hit("/"))
+        assert r["status"] == 200 and b"ray_amd" in r["body"]
+    finally:
+        loop.close()
