@@ -90,6 +90,12 @@ class Raylet:
         # cluster-infeasible requests waiting for new capacity
         self._infeasible: List[tuple] = []
         self._infeasible_running = False
+        # pull manager (dedup + priority) and push-side flow control
+        self._pulls: Dict[bytes, dict] = {}
+        self._pull_heap: list = []
+        self._pull_seq = 0
+        self._pulls_active = 0
+        self._pushes: Dict[tuple, dict] = {}
         self._worker_ready: Dict[int, asyncio.Future] = {}  # pid -> fut
         self._actor_start_futs: Dict[bytes, asyncio.Future] = {}
         self._proto_worker: Dict[int, WorkerProc] = {}
@@ -115,6 +121,7 @@ class Raylet:
             "actor_ready actor_failed kill_worker reserve_bundle commit_bundle "
             "rollback_bundle remove_bundle node_info ping prestart_workers "
             "report_task_events pin_object unpin_object try_recycle "
+            "request_push push_ack "
             "debug_state"
         ).split():
             self.server.route(m, getattr(self, "h_" + m))
@@ -778,40 +785,172 @@ class Raylet:
         self.store.used -= ent[0]
         return {"ok": True}
 
+    # ---------------- object transfer ----------------
+    #
+    # Pull manager (reference: object_manager/pull_manager.h:52 —
+    # prioritized bundles get > wait > task-args, dedup) + push-side
+    # streaming (push_manager.h:28 — the source streams chunks with a
+    # flow-control window instead of the round-trip-per-chunk fetch,
+    # so cross-node bandwidth is not RTT-bound).
+
+    PULL_PRIO_GET = 0
+    PULL_PRIO_WAIT = 1
+    PULL_PRIO_TASK_ARGS = 2
+    MAX_CONCURRENT_PULLS = 8
+    PUSH_WINDOW = 8  # chunks in flight before the source waits for acks
+
     async def h_pull_object(self, conn, p):
-        """Pull an object from a remote node's store into ours, chunked
-        (reference: PullManager / object_manager gRPC chunks)."""
+        """Pull an object from a remote node's store into ours.
+        Deduplicated (concurrent pulls of one oid share a transfer) and
+        prioritized; the transfer itself is source-paced push
+        streaming."""
         oid = bytes(p["id"])
         if self.store.contains(oid):
             self.store.ensure_local(oid)
             return {"ok": True}
-        src = p["src"]
-        c = RpcClient()
+        st = self._pulls.get(oid)
+        if st is None:
+            st = self._pulls[oid] = {
+                "fut": asyncio.get_running_loop().create_future(),
+                "prio": int(p.get("prio", self.PULL_PRIO_TASK_ARGS)),
+            }
+            import heapq
+
+            self._pull_seq += 1
+            heapq.heappush(self._pull_heap,
+                           (st["prio"], self._pull_seq, oid, p["src"]))
+            asyncio.ensure_future(self._pull_pump())
+        else:
+            # a higher-priority duplicate bumps nothing in-flight but
+            # is recorded for observability
+            st["prio"] = min(st["prio"], int(p.get("prio", 2)))
         try:
-            await c.connect(src, retries=5)
-            first = await c.call("fetch_chunk", {"id": oid, "off": 0, "len": CHUNK})
-            if not first.get("ok"):
-                return {"ok": False}
-            size = first["size"]
-            w = store.ObjectWriter(self.store.shm_dir, oid, size)
-            data = first["data"]
-            w.view[: len(data)] = data
-            off = len(data)
-            while off < size:
-                r = await c.call("fetch_chunk", {"id": oid, "off": off, "len": CHUNK})
-                if not r.get("ok"):
-                    return {"ok": False}
-                d = r["data"]
-                w.view[off : off + len(d)] = d
-                off += len(d)
-            w.seal()
-            self.store.seal(oid, size)
-            return {"ok": True}
+            ok = await asyncio.wait_for(
+                asyncio.shield(st["fut"]), p.get("timeout", 120.0)
+            )
+            return {"ok": ok}
+        except asyncio.TimeoutError:
+            return {"ok": False}
+
+    async def _pull_pump(self):
+        import heapq
+
+        while self._pull_heap and self._pulls_active < self.MAX_CONCURRENT_PULLS:
+            _, _, oid, src = heapq.heappop(self._pull_heap)
+            st = self._pulls.get(oid)
+            if st is None or st["fut"].done():
+                continue
+            self._pulls_active += 1
+            asyncio.ensure_future(self._do_pull(oid, src, st))
+
+    async def _do_pull(self, oid: bytes, src: str, st: dict):
+        ok = False
+        try:
+            ok = await self._pull_streamed(oid, src)
         except Exception:
             traceback.print_exc()
-            return {"ok": False}
+        finally:
+            self._pulls_active -= 1
+            self._pulls.pop(oid, None)
+            if not st["fut"].done():
+                st["fut"].set_result(ok)
+            asyncio.ensure_future(self._pull_pump())
+
+    async def _pull_streamed(self, oid: bytes, src: str) -> bool:
+        c = RpcClient()
+        recv = {"writer": None, "size": None, "received": 0,
+                "done": asyncio.get_running_loop().create_future()}
+
+        def on_notify(method, payload):
+            if method != "push_chunk" or bytes(payload["id"]) != oid:
+                return
+            try:
+                if recv["writer"] is None:
+                    recv["size"] = payload["size"]
+                    recv["writer"] = store.ObjectWriter(
+                        self.store.shm_dir, oid, max(payload["size"], 1)
+                    )
+                d = payload["data"]
+                off = payload["off"]
+                recv["writer"].view[off:off + len(d)] = d
+                recv["received"] += len(d)
+                c.notify("push_ack", {"id": oid, "received": recv["received"]})
+                if recv["received"] >= recv["size"]:
+                    if not recv["done"].done():
+                        recv["done"].set_result(True)
+            except Exception as e:
+                if not recv["done"].done():
+                    recv["done"].set_exception(e)
+
+        try:
+            await c.connect(src, retries=5)
+            c.on_notify = on_notify
+            r = await c.call("request_push", {"id": oid})
+            if not r.get("ok"):
+                return False
+            if r.get("size") == 0:
+                recv["writer"] = store.ObjectWriter(self.store.shm_dir, oid, 1)
+                recv["size"] = 0
+            else:
+                await asyncio.wait_for(recv["done"], 120.0)
+            recv["writer"].seal()
+            self.store.seal(oid, recv["size"])
+            return True
+        except Exception:
+            traceback.print_exc()
+            return False
         finally:
             c.close()
+
+    async def h_request_push(self, conn, p):
+        """Source side: stream the object to the requester over this
+        connection in CHUNK pieces, at most PUSH_WINDOW chunks ahead of
+        the receiver's acks; deduplicate per (conn, oid)."""
+        oid = bytes(p["id"])
+        ok = await self.store.wait_sealed(oid, 30.0)
+        if not ok:
+            return {"ok": False}
+        self.store.ensure_local(oid)
+        key = (id(conn), oid)
+        if key in self._pushes:
+            return {"ok": True, "dup": True}
+        size = self.store.table[oid][0]
+        if size == 0:
+            return {"ok": True, "size": 0}
+        st = self._pushes[key] = {"acked": 0,
+                                  "event": asyncio.Event()}
+
+        async def _stream():
+            try:
+                path = store.shm_path(self.store.shm_dir, oid)
+                with open(path, "rb") as f:
+                    off = 0
+                    while off < size:
+                        while (off - st["acked"]) >= self.PUSH_WINDOW * CHUNK:
+                            st["event"].clear()
+                            try:
+                                await asyncio.wait_for(st["event"].wait(),
+                                                       60.0)
+                            except asyncio.TimeoutError:
+                                return
+                        data = f.read(CHUNK)
+                        from .protocol import MSG_NOTIFY
+
+                        conn.send([MSG_NOTIFY, 0, "push_chunk",
+                                   {"id": oid, "off": off, "size": size,
+                                    "data": data}])
+                        off += len(data)
+            finally:
+                self._pushes.pop(key, None)
+
+        asyncio.ensure_future(_stream())
+        return {"ok": True, "size": size}
+
+    def h_push_ack(self, conn, p):
+        st = self._pushes.get((id(conn), bytes(p["id"])))
+        if st is not None:
+            st["acked"] = max(st["acked"], int(p.get("received", 0)))
+            st["event"].set()
 
     async def h_fetch_chunk(self, conn, p):
         oid = bytes(p["id"])

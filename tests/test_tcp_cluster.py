@@ -92,3 +92,28 @@ def test_tcp_node_death(tcp_cluster):
         ray.get(a.ping.remote(), timeout=30)
     alive = [n for n in ray.nodes() if n["Alive"]]
     assert len(alive) == 1
+
+
+def test_pull_dedup_and_priority(tcp_cluster):
+    """Pull manager: concurrent pulls of one object share a single
+    transfer (dedup, reference pull_manager.h), and the push-streamed
+    path moves large objects across separate shm dirs correctly."""
+    cluster = tcp_cluster
+    cluster.add_node(num_cpus=4, resources={"n2": 1})
+    cluster.connect()
+    cluster.wait_for_nodes()
+
+    @ray.remote(resources={"n2": 0.1})
+    def make_big():
+        return np.arange(2_000_000, dtype=np.float64)  # 16 MB, 4 chunks+
+
+    ref = make_big.remote()
+    ray.wait([ref], timeout=60)
+
+    # many concurrent consumers on the head node pulling the SAME oid
+    @ray.remote
+    def consume(x, i):
+        return float(x[i])
+
+    outs = ray.get([consume.remote(ref, i) for i in range(8)], timeout=120)
+    assert outs == [float(i) for i in range(8)]

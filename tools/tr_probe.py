@@ -6,6 +6,7 @@ import torch
 from torch.utils import cpp_extension
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+os.makedirs("/tmp/tr_probe_build", exist_ok=True)
 mod = cpp_extension.load(
     name="tr_probe_ext",
     sources=[os.path.join(os.path.dirname(__file__), "..", "ray_amd",
@@ -23,6 +24,5 @@ def run(label, addr_fn):
         print(f"lane {l:2d} addr {int(addrs[l]):4d} -> "
               + " ".join(f"{int(x):4d}" for x in out[l]))
 
-os.makedirs("/tmp/tr_probe_build", exist_ok=True)
 run("m156: (l&15) + (l>>4)*64", lambda l: (l & 15) + (l >> 4) * 64)
 run("linear: 4*l", lambda l: 4 * l)
