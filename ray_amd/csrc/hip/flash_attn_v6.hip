@@ -242,13 +242,27 @@ flash_attn_fwd_v6_bf16(const short* __restrict__ Q,
           fa6_u32x4 pw{(unsigned)sw0[0], (unsigned)sw1[0], (unsigned)sw0[1],
                        (unsigned)sw1[1]};
           fa6_bf16x8 pb = __builtin_bit_cast(fa6_bf16x8, pw);
+          // V^T A-fragments via ds_read_tr16_b64 (T10): each 16-lane
+          // group cooperatively transposes a 4x16 V tile — 2 tr reads
+          // replace 32 u16 gathers per (kc,dt). Probe-verified
+          // semantics (tools/tr_probe.py): supplier lane s provides
+          // the row (kq + 8*hi + 4*h + ((s&15)>>2)) at column block
+          // (dt*32 + 16*((s>>4)&1) + 4*(s&3)); consumer lane i
+          // receives V[kq+8*hi+4*h+j][dt*32 + (i&31)] in slot j.
+          const int gl = lane & 15;
+          const int grp16 = (lane >> 4) & 1;
+          const int vrow0 = 32 * kt + kc * 16 + a_off + (gl >> 2);
 #pragma unroll
           for (int dt = 0; dt < 4; ++dt) {
-            short vtmp[8];
-#pragma unroll
-            for (int i = 0; i < 8; ++i)
-              vtmp[i] = v_lds[cur][32 * kt + kc * 16 + a_off + i]
-                             [dt * 32 + (lane & 31)];
+            const int vcol = dt * 32 + 16 * grp16 + 4 * (gl & 3);
+            auto p0 = (__attribute__((address_space(3))) short4v*)
+                &v_lds[cur][vrow0][vcol];
+            auto p1 = (__attribute__((address_space(3))) short4v*)
+                &v_lds[cur][vrow0 + 4][vcol];
+            short4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(p0);
+            short4v hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(p1);
+            short vtmp[8] = {lo[0], lo[1], lo[2], lo[3],
+                             hi4[0], hi4[1], hi4[2], hi4[3]};
             fa6_bf16x8 va = __builtin_bit_cast(
                 fa6_bf16x8, *reinterpret_cast<short8*>(vtmp));
             o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
