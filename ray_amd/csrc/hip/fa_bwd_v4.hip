@@ -143,19 +143,32 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dv_v4_bf16(
         fb3_bf16x8 af = __builtin_bit_cast(
             fb3_bf16x8, *reinterpret_cast<const short8*>(
                             &scr[lane & 31][16 * qc + a_off]));
+        // B-fragments via ds_read_tr16_b64 (probe-verified; see
+        // flash_attn_v6): supplier lane provides row
+        // (base + 4h + (gl>>2)) at its 4-col block, XOR-swizzled
+        // per-row like the stores
+        {
+          const int gl = lane & 15;
+          const int grp16 = (lane >> 4) & 1;
 #pragma unroll
-        for (int dt = 0; dt < 4; ++dt) {
-          short dtmp[8];
+          for (int dt = 0; dt < 4; ++dt) {
+            short dtmp[8];
 #pragma unroll
-          for (int i = 0; i < 8; ++i) {
-            int row = 32 * qt + 16 * qc + a_off + i;
-            int col = 32 * dt + (lane & 31);
-            dtmp[i] = do_lds[cur][row][col ^ ((row & 7) << 3)];
+            for (int h = 0; h < 2; ++h) {
+              int row = 32 * qt + 16 * qc + a_off + 4 * h + (gl >> 2);
+              int col = (32 * dt + 16 * grp16 + 4 * (gl & 3))
+                        ^ ((row & 7) << 3);
+              auto p = (__attribute__((address_space(3))) short4v*)
+                  &do_lds[cur][row][col];
+              short4v r4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(p);
+#pragma unroll
+              for (int j = 0; j < 4; ++j) dtmp[4 * h + j] = r4[j];
+            }
+            fb3_bf16x8 bf = __builtin_bit_cast(
+                fb3_bf16x8, *reinterpret_cast<short8*>(dtmp));
+            dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                af, bf, dv_acc[dt], 0, 0, 0);
           }
-          fb3_bf16x8 bf = __builtin_bit_cast(
-              fb3_bf16x8, *reinterpret_cast<short8*>(dtmp));
-          dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              af, bf, dv_acc[dt], 0, 0, 0);
         }
       }
     }
@@ -313,19 +326,28 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dk_v4_bf16(
         fb3_bf16x8 af = __builtin_bit_cast(
             fb3_bf16x8, *reinterpret_cast<const short8*>(
                             &scr[lane & 31][16 * qc + a_off]));
+        {
+          const int gl = lane & 15;
+          const int grp16 = (lane >> 4) & 1;
 #pragma unroll
-        for (int dt = 0; dt < 4; ++dt) {
-          short qtmp[8];
+          for (int dt = 0; dt < 4; ++dt) {
+            short qtmp[8];
 #pragma unroll
-          for (int i = 0; i < 8; ++i) {
-            int row = 32 * qt + 16 * qc + a_off + i;
-            int col = 32 * dt + (lane & 31);
-            qtmp[i] = q_lds[cur][row][col ^ ((row & 7) << 3)];
+            for (int h = 0; h < 2; ++h) {
+              int row = 32 * qt + 16 * qc + a_off + 4 * h + (gl >> 2);
+              int col = (32 * dt + 16 * grp16 + 4 * (gl & 3))
+                        ^ ((row & 7) << 3);
+              auto p = (__attribute__((address_space(3))) short4v*)
+                  &q_lds[cur][row][col];
+              short4v r4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(p);
+#pragma unroll
+              for (int j = 0; j < 4; ++j) qtmp[4 * h + j] = r4[j];
+            }
+            fb3_bf16x8 bf = __builtin_bit_cast(
+                fb3_bf16x8, *reinterpret_cast<short8*>(qtmp));
+            dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                af, bf, dk_acc[dt], 0, 0, 0);
           }
-          fb3_bf16x8 bf = __builtin_bit_cast(
-              fb3_bf16x8, *reinterpret_cast<short8*>(qtmp));
-          dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              af, bf, dk_acc[dt], 0, 0, 0);
         }
       }
     }

@@ -9,6 +9,7 @@ from .impl import (  # noqa: F401
     Callback,
     FIFOScheduler,
     MedianStoppingRule,
+    PB2,
     PopulationBasedTraining,
     ResultGrid,
     TuneConfig,

@@ -306,6 +306,78 @@ class PopulationBasedTraining:
         return new
 
 
+class PB2(PopulationBasedTraining):
+    """PB2 (reference: tune/schedulers/pb2.py): PBT where explore()
+    picks new hyperparameters by GP-UCB over the observed
+    (config -> reward-improvement) data instead of random
+    perturbation — sample-efficient for small populations.
+    hyperparam_bounds: {name: [low, high]} (continuous)."""
+
+    def __init__(self, *, hyperparam_bounds=None, ucb_kappa=1.5, **kw):
+        kw.setdefault("hyperparam_mutations", {})
+        super().__init__(**kw)
+        self.hyperparam_bounds = hyperparam_bounds or {}
+        self.ucb_kappa = ucb_kappa
+        self._history = []        # (xvec, reward_delta)
+        self._prev_score: Dict[str, float] = {}
+        self._trial_cfgs: Dict[str, dict] = {}
+
+    def on_trial_result(self, trial_id, iteration, metric_value) -> str:
+        if metric_value is not None:
+            prev = self._prev_score.get(trial_id)
+            if prev is not None:
+                # record improvement for the trial's current config
+                cfg = self._trial_cfgs.get(trial_id)
+                if cfg is not None:
+                    x = self._vec(cfg)
+                    if x is not None:
+                        delta = metric_value - prev
+                        if self.mode == "min":
+                            delta = -delta
+                        self._history.append((x, delta))
+            self._prev_score[trial_id] = metric_value
+        return super().on_trial_result(trial_id, iteration, metric_value)
+
+    def observe_config(self, trial_id, config):
+        self._trial_cfgs[trial_id] = dict(config)
+
+    def _vec(self, cfg):
+        try:
+            return [float(cfg[k]) for k in sorted(self.hyperparam_bounds)]
+        except (KeyError, TypeError, ValueError):
+            return None
+
+    def explore(self, config: dict) -> dict:
+        keys = sorted(self.hyperparam_bounds)
+        if not keys or len(self._history) < 4:
+            return super().explore(config)
+        try:
+            import numpy as _np
+            from sklearn.gaussian_process import GaussianProcessRegressor
+            from sklearn.gaussian_process.kernels import Matern
+        except ImportError:
+            return super().explore(config)
+        X = _np.array([x for x, _ in self._history[-64:]])
+        y = _np.array([d for _, d in self._history[-64:]])
+        lo = _np.array([self.hyperparam_bounds[k][0] for k in keys])
+        hi = _np.array([self.hyperparam_bounds[k][1] for k in keys])
+        span = _np.where(hi > lo, hi - lo, 1.0)
+        gp = GaussianProcessRegressor(
+            kernel=Matern(nu=2.5), alpha=1e-4, normalize_y=True)
+        gp.fit((X - lo) / span, y)
+        cand = _np.random.default_rng(
+            self._rng.randrange(1 << 30)).uniform(0, 1, size=(256, len(keys)))
+        mu, sd = gp.predict(cand, return_std=True)
+        best = cand[int(_np.argmax(mu + self.ucb_kappa * sd))]
+        new = dict(config)
+        for k, v01, l, s in zip(keys, best, lo, span):
+            val = float(l + v01 * s)
+            if isinstance(config.get(k), int):
+                val = int(round(val))
+            new[k] = val
+        return new
+
+
 @dataclass
 class TuneConfig:
     metric: Optional[str] = None
@@ -516,6 +588,8 @@ class Tuner:
             t = {"actor": a, "config": cfg, "rows": [], "done": False,
                  "error": None, "it": 0, "name": name, "stopped": False,
                  "ckpt": None, "fetch_ref": None}
+            if scheduler is not None and hasattr(scheduler, "observe_config"):
+                scheduler.observe_config(name, cfg)
             running.append(t)
 
         while pending or running:
@@ -624,6 +698,8 @@ class Tuner:
         if ckpt is None:
             ckpt = donor.get("ckpt")
         new_cfg = scheduler.explore(donor["config"])
+        if hasattr(scheduler, "observe_config"):
+            scheduler.observe_config(t["name"], new_cfg)
         try:
             ray.kill(t["actor"])
         except Exception:

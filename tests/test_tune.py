@@ -239,3 +239,39 @@ def test_tuner_restore_resumes_unfinished(ray_start_regular, tmp_path):
     ]
     # only the failed trial re-ran
     assert len(list(runs.iterdir())) == n_first + 1
+
+
+def test_pb2_scheduler(ray_start_regular, tmp_path):
+    """PB2 (reference: tune/schedulers/pb2.py): GP-UCB explore over
+    hyperparam_bounds replaces random perturbation; bottom-quantile
+    trials restart from top-quantile checkpoints with GP-chosen
+    configs."""
+    from ray_amd import tune
+    from ray_amd.tune import PB2
+
+    def trainable(config):
+        import os
+
+        for it in range(8):
+            # score peaks at lr=0.5
+            score = -abs(config["lr"] - 0.5) * 10 + it * 0.1
+            ckpt_dir = str(tmp_path / f"ck_{os.getpid()}_{it}")
+            os.makedirs(ckpt_dir, exist_ok=True)
+            with open(os.path.join(ckpt_dir, "s.txt"), "w") as f:
+                f.write(str(it))
+            from ray_amd.train import Checkpoint
+
+            tune.report({"score": score},
+                        checkpoint=Checkpoint(ckpt_dir))
+
+    sched = PB2(metric="score", mode="max", perturbation_interval=2,
+                hyperparam_bounds={"lr": [0.0, 1.0]})
+    tuner = tune.Tuner(
+        trainable,
+        param_space={"lr": tune.uniform(0.0, 1.0)},
+        tune_config=tune.TuneConfig(metric="score", mode="max",
+                                    num_samples=4, scheduler=sched),
+    )
+    results = tuner.fit()
+    best = results.get_best_result()
+    assert best.metrics["score"] is not None
