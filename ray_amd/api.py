@@ -150,6 +150,13 @@ def shutdown(_exiting_interpreter: bool = False):
     global _cluster
     with _init_lock:
         try:
+            from .util.usage_stats import usage_stats_enabled, write_report
+
+            if _rt.is_initialized() and usage_stats_enabled():
+                write_report(_rt.global_runtime().session_dir)
+        except Exception:
+            pass
+        try:
             if _rt.is_initialized():
                 rt = _rt.global_runtime()
                 rt.shutdown()

@@ -36,3 +36,11 @@ from .preprocessors import (  # noqa: F401
     StandardScaler,
     TorchVisionNormalizer,
 )
+
+
+try:  # usage tagging (local-only; util/usage_stats.py)
+    from ray_amd.util.usage_stats import record_library_usage
+
+    record_library_usage("data")
+except Exception:  # pragma: no cover
+    pass

@@ -23,3 +23,11 @@ from .api import (  # noqa: F401
     start,
     status,
 )
+
+
+try:  # usage tagging (local-only; util/usage_stats.py)
+    from ray_amd.util.usage_stats import record_library_usage
+
+    record_library_usage("serve")
+except Exception:  # pragma: no cover
+    pass

@@ -30,3 +30,11 @@ def __getattr__(name):
 
         return _t
     raise AttributeError(name)
+
+
+try:  # usage tagging (local-only; util/usage_stats.py)
+    from ray_amd.util.usage_stats import record_library_usage
+
+    record_library_usage("train")
+except Exception:  # pragma: no cover
+    pass

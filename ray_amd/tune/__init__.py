@@ -29,3 +29,11 @@ from .impl import (  # noqa: F401
 )
 from ..train.checkpoint import Checkpoint  # noqa: F401
 from ..train.session import get_checkpoint, get_context  # noqa: F401
+
+
+try:  # usage tagging (local-only; util/usage_stats.py)
+    from ray_amd.util.usage_stats import record_library_usage
+
+    record_library_usage("tune")
+except Exception:  # pragma: no cover
+    pass
