@@ -982,6 +982,13 @@ class Raylet:
         return "pong"
 
     def h_report_task_events(self, conn, p):
+        spans = p.get("spans")
+        if spans:
+            try:
+                self.gcs.notify("timeline_events",
+                                {"events": [{"span": sp} for sp in spans]})
+            except Exception:
+                pass
         self._task_events.extend(p.get("events", []))
         if len(self._task_events) > 100000:
             del self._task_events[:50000]

@@ -1168,6 +1168,9 @@ class CoreRuntime:
             "working_dir": (options.get("runtime_env") or {}).get("working_dir"),
             "py_modules": (options.get("runtime_env") or {}).get("py_modules"),
         }
+        tctx = _trace_ctx()
+        if tctx:
+            spec["trace_ctx"] = tctx
         if streaming:
             import threading as _th
 
@@ -1571,6 +1574,9 @@ class CoreRuntime:
             "caller": self.addr,
             "num_returns": num_returns,
         }
+        tctx = _trace_ctx()
+        if tctx:
+            spec["trace_ctx"] = tctx
         if streaming:
             import threading as _th
 
@@ -1711,6 +1717,17 @@ class CoreRuntime:
 
     def raylet_call(self, method, payload, timeout=None):
         return self._call_sync(self.raylet.call(method, payload), timeout)
+
+
+def _trace_ctx():
+    """Caller-side span-context injection (reference:
+    tracing_helper.py:183); cheap no-op while tracing is off."""
+    try:
+        from ..util.tracing.tracing_helper import current_span_context
+
+        return current_span_context()
+    except Exception:
+        return None
 
 
 def fn_hash(pickled: bytes) -> bytes:

@@ -55,6 +55,32 @@ def _apply_code_env(spec: dict):
             _sys.path.insert(0, m)
 
 
+def _trace_activate(spec, name):
+    if not spec.get("trace_ctx") and not os.environ.get("RAY_AMD_TRACING"):
+        return None
+    try:
+        from ray_amd.util.tracing.tracing_helper import (
+            activate_remote_context,
+        )
+
+        return activate_remote_context(spec.get("trace_ctx"), name)
+    except Exception:
+        return None
+
+
+def _trace_finish(handle):
+    if handle is None:
+        return
+    try:
+        from ray_amd.util.tracing.tracing_helper import (
+            finish_remote_context,
+        )
+
+        finish_remote_context(handle)
+    except Exception:
+        pass
+
+
 class TaskContext:
     def __init__(self):
         self.task_id = None
@@ -251,6 +277,7 @@ class WorkerMain:
             os.environ.setdefault("HIP_VISIBLE_DEVICES", ids_str)
 
         loop = asyncio.get_running_loop()
+        _tr = _trace_activate(spec, "task:" + (spec.get("name") or "fn"))
 
         def _exec():
             _task_ctx.task_id = task_id
@@ -271,6 +298,7 @@ class WorkerMain:
                 ok, result = False, e
         else:
             ok, result = await loop.run_in_executor(self.executor, _exec)
+        _trace_finish(_tr)
         if ok and spec.get("streaming"):
             reply = await self._run_streaming(conn, spec, result)
         else:
@@ -395,6 +423,7 @@ class WorkerMain:
         # single-thread executor preserves the dispatch order for sync
         # methods; async methods interleave by design).
         loop = asyncio.get_running_loop()
+        _tr_a = _trace_activate(spec, "actor:" + method_name)
         async with self._order_lock:
             try:
                 args, kwargs = await self._load_args(spec)
@@ -439,6 +468,7 @@ class WorkerMain:
                 ok, result = False, e
         else:
             ok, result, exit_after = await pending
+        _trace_finish(_tr_a)
         if ok and spec.get("streaming"):
             reply = await self._run_streaming(conn, spec, result)
         else:

@@ -839,3 +839,47 @@ def test_usage_stats_local_report():
         assert not us.usage_stats_enabled()
     finally:
         _os.environ.pop("RAY_AMD_USAGE_STATS_ENABLED", None)
+
+
+def test_tracing_spans_cross_process():
+    """Spans cross process boundaries (reference:
+    util/tracing/tracing_helper.py:183-193): a driver span's trace_id
+    is inherited by the remote task's execution span, and a span
+    opened INSIDE the task nests under it."""
+    import os as _os
+
+    _os.environ["RAY_AMD_TRACING"] = "1"
+    try:
+        ray.init(num_cpus=2, ignore_reinit_error=True)
+        from ray_amd.util.tracing import get_trace_events, span
+
+        @ray.remote
+        def traced_task():
+            from ray_amd.util.tracing import span as span2
+
+            with span2("inner-work"):
+                return 42
+
+        with span("driver-root") as root:
+            assert ray.get(traced_task.remote()) == 42
+        import time as _t
+
+        deadline = _t.time() + 10
+        spans = []
+        while _t.time() < deadline:
+            spans = get_trace_events()
+            names = {s["name"] for s in spans
+                     if s["trace_id"] == root.trace_id}
+            if {"driver-root", "task:traced_task", "inner-work"} <= names:
+                break
+            _t.sleep(0.2)
+        by_name = {s["name"]: s for s in spans
+                   if s["trace_id"] == root.trace_id}
+        assert "task:traced_task" in by_name, by_name.keys()
+        assert by_name["task:traced_task"]["parent_id"] == root.span_id
+        assert "inner-work" in by_name
+        assert (by_name["inner-work"]["parent_id"]
+                == by_name["task:traced_task"]["span_id"])
+    finally:
+        _os.environ.pop("RAY_AMD_TRACING", None)
+        ray.shutdown()
