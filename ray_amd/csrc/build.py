@@ -38,11 +38,23 @@ def needs_rebuild() -> bool:
     return False
 
 
+def _sanitize_flags() -> list:
+    """RAY_AMD_SANITIZE=address|thread|undefined builds the native C++
+    with the matching sanitizer (reference: bazel --config=asan/tsan
+    presets; see docs/SANITIZERS.md). Run python with the sanitizer
+    runtime preloaded (LD_PRELOAD=$(g++ -print-file-name=libasan.so))."""
+    san = os.environ.get("RAY_AMD_SANITIZE")
+    if not san:
+        return []
+    return [f"-fsanitize={san}", "-fno-omit-frame-pointer", "-g", "-O1"]
+
+
 def build_shm(verbose: bool = True, force: bool = False) -> str:
     """Native shm store data path: plain C++ (g++), no HIP needed."""
     src = os.path.join(CSRC, "shm_store.cpp")
     if (not force and os.path.exists(OUT_SHM)
-            and os.path.getmtime(OUT_SHM) >= os.path.getmtime(src)):
+            and os.path.getmtime(OUT_SHM) >= os.path.getmtime(src)
+            and not os.environ.get("RAY_AMD_SANITIZE")):
         return OUT_SHM
     import pybind11
 
@@ -50,7 +62,7 @@ def build_shm(verbose: bool = True, force: bool = False) -> str:
         "g++", "-O3", "-std=c++17", "-fPIC", "-shared", src, "-o", OUT_SHM,
         f"-I{pybind11.get_include()}", f"-I{_python_include()}",
         "-pthread",
-    ]
+    ] + _sanitize_flags()
     if verbose:
         print("[ray_amd build]", " ".join(cmd), file=sys.stderr)
     subprocess.check_call(cmd)
