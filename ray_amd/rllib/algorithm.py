@@ -16,6 +16,8 @@ class AlgorithmConfig:
         self.env_config: dict = {}
         self.num_env_runners = 0
         self.num_envs_per_env_runner = 8
+        self.env_to_module_connector = None   # () -> ConnectorPipeline
+        self.learner_connector = None         # () -> ConnectorPipeline
         self.rollout_fragment_length = 200
         self.train_batch_size = 4000
         self.minibatch_size = 512
@@ -41,17 +43,30 @@ class AlgorithmConfig:
         return self
 
     def env_runners(self, *, num_env_runners=None, num_envs_per_env_runner=None,
-                    rollout_fragment_length=None, **kwargs):
+                    rollout_fragment_length=None,
+                    env_to_module_connector=None, **kwargs):
         if num_env_runners is not None:
             self.num_env_runners = num_env_runners
         if num_envs_per_env_runner is not None:
             self.num_envs_per_env_runner = num_envs_per_env_runner
         if rollout_fragment_length is not None:
             self.rollout_fragment_length = rollout_fragment_length
+        if env_to_module_connector is not None:
+            # factory () -> ConnectorPipeline (reference:
+            # config.env_runners(env_to_module_connector=...))
+            self.env_to_module_connector = env_to_module_connector
         return self
 
     def rollouts(self, **kwargs):  # old-stack alias
         return self.env_runners(**kwargs)
+
+    def learner_connections(self, *, learner_connector=None, **kw):
+        """reference: config.training(learner_connector=...) — factory
+        () -> ConnectorPipeline run on the train batch before the
+        loss."""
+        if learner_connector is not None:
+            self.learner_connector = learner_connector
+        return self
 
     def training(self, *, lr=None, gamma=None, train_batch_size=None,
                  minibatch_size=None, num_epochs=None, clip_param=None,

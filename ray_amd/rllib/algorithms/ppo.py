@@ -26,6 +26,8 @@ class PPO(Algorithm):
             config.num_env_runners,
             config.num_envs_per_env_runner,
             hidden=config.model_hidden,
+            env_to_module_connector=getattr(
+                config, "env_to_module_connector", None),
         )
         self.learner = PPOTorchLearner(
             obs_dim,
@@ -41,6 +43,7 @@ class PPO(Algorithm):
             num_epochs=config.num_epochs,
             minibatch_size=config.minibatch_size,
             use_gpu=config.num_gpus_per_learner > 0,
+            learner_connector=getattr(config, "learner_connector", None),
         )
         self._env_steps_total = 0
         self.env_runner_group.sync_weights(self.learner.get_weights())

@@ -16,10 +16,18 @@ class SingleAgentEnvRunner:
     """Runs vectorized envs with an inference-only module copy."""
 
     def __init__(self, env_name: str, num_envs: int = 8, seed: int = 0,
-                 hidden=(256, 256)):
+                 hidden=(256, 256), env_to_module=None):
         self.vec = VectorEnv(env_name, num_envs, seed)
         self.module = build_module_for_env(self.vec, hidden=hidden)
         self.obs = self.vec.reset()
+        # env->module connector pipeline (reference: rllib/connectors):
+        # transforms the observation batch before every inference
+        self.env_to_module = env_to_module
+
+    def _transform_obs(self, obs):
+        if self.env_to_module is None:
+            return obs
+        return self.env_to_module({"obs": obs})["obs"]
 
     def sample(self, num_steps: int) -> Dict[str, np.ndarray]:
         """Rollout num_steps per env; returns [T, B, ...] arrays."""
@@ -33,7 +41,8 @@ class SingleAgentEnvRunner:
         vf_buf = np.zeros((T + 1, B), np.float32)
         obs = self.obs
         for t in range(T):
-            a, logp, vf = self.module.forward_exploration(obs)
+            a, logp, vf = self.module.forward_exploration(
+                self._transform_obs(obs))
             obs_buf[t] = obs
             act_buf[t] = a
             logp_buf[t] = logp
@@ -42,7 +51,8 @@ class SingleAgentEnvRunner:
             rew_buf[t] = r
             done_buf[t] = np.logical_or(term, trunc).astype(np.float32)
         # bootstrap value
-        _, _, vf = self.module.forward_exploration(obs)
+        _, _, vf = self.module.forward_exploration(
+            self._transform_obs(obs))
         vf_buf[T] = vf
         self.obs = obs
         rets, lens = self.vec.pop_episode_stats()
@@ -71,21 +81,25 @@ class EnvRunnerGroup:
     """Manages N EnvRunner actors (+ a local runner when N==0)."""
 
     def __init__(self, env_name: str, num_runners: int, num_envs_per_runner: int,
-                 hidden=(256, 256)):
+                 hidden=(256, 256), env_to_module_connector=None):
         import ray_amd as ray
 
         self._ray = ray
         self.num_runners = num_runners
+        e2m = (env_to_module_connector()
+               if env_to_module_connector is not None else None)
         if num_runners == 0:
             self.local = SingleAgentEnvRunner(env_name, num_envs_per_runner,
-                                              hidden=hidden)
+                                              hidden=hidden,
+                                              env_to_module=e2m)
             self.remotes: List = []
         else:
             self.local = None
             cls = ray.remote(SingleAgentEnvRunner)
             self.remotes = [
                 cls.options(num_cpus=1).remote(
-                    env_name, num_envs_per_runner, seed=1000 * i, hidden=hidden
+                    env_name, num_envs_per_runner, seed=1000 * i,
+                    hidden=hidden, env_to_module=e2m,
                 )
                 for i in range(num_runners)
             ]

@@ -35,6 +35,7 @@ class PPOTorchLearner:
         minibatch_size: int = 512,
         use_gpu: bool = False,
         ddp: bool = False,
+        learner_connector=None,
     ):
         self.device = torch.device(
             "cuda:0" if use_gpu and torch.cuda.is_available() else "cpu"
@@ -58,13 +59,22 @@ class PPOTorchLearner:
         self.ent_coeff = entropy_coeff
         self.num_epochs = num_epochs
         self.minibatch_size = minibatch_size
+        if learner_connector is not None:
+            self.learner_connector = learner_connector()
+        else:
+            from .connectors import (ConnectorPipeline,
+                                     GeneralAdvantageEstimation)
+
+            self.learner_connector = ConnectorPipeline([
+                GeneralAdvantageEstimation(gamma, lambda_,
+                                           device=self.device),
+            ])
 
     def _advantages(self, batch: Dict[str, np.ndarray]):
-        rewards = torch.as_tensor(batch["rewards"], device=self.device)
-        values = torch.as_tensor(batch["vf"], device=self.device)
-        cont = 1.0 - torch.as_tensor(batch["dones"], device=self.device)
-        adv, vtarg = ops.gae(rewards, values, cont, self.gamma, self.lambda_)
-        return adv, vtarg
+        # learner-connector pipeline (reference: GAE as a learner
+        # connector, general_advantage_estimation.py:21)
+        out = self.learner_connector(dict(batch))
+        return out["advantages"], out["value_targets"]
 
     def update(self, samples: List[Dict[str, np.ndarray]]) -> Dict[str, float]:
         # concat along env axis
