@@ -473,6 +473,9 @@ class GcsServer:
             "tensor_transport": a.spec.get("tensor_transport"),
         }
 
+    # (profiler wrapper rides on the raylet start_actor spec, not the
+    # worker resolve path)
+
     async def h_actor_exit(self, conn, p):
         a = self.actors.get(p["actor_id"])
         if a is None:

@@ -198,6 +198,10 @@ class Raylet:
         env["RAY_AMD_RAYLET_ADDR"] = self.addr
         env["RAY_AMD_NODE_ID"] = self.node_id.hex()
         args = [sys.executable, "-m", "ray_amd._core.worker"]
+        if actor_spec is not None and actor_spec.get("profiler"):
+            # profiler runtime-env plugin: launch the worker UNDER the
+            # profiler (reference: runtime_env/rocprof_sys.py:17)
+            args = [str(x) for x in actor_spec["profiler"]] + args
         if actor_spec is not None:
             env["RAY_AMD_ACTOR_ID"] = actor_spec["actor_id"].hex()
             for k, v in (actor_spec.get("env_vars") or {}).items():

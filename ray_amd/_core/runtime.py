@@ -1535,6 +1535,7 @@ class CoreRuntime:
                 "node_affinity": options.get("node_affinity"),
                 "label_selector": options.get("label_selector"),
                 "tensor_transport": options.get("tensor_transport"),
+                "profiler": _profiler_cmd(options.get("runtime_env")),
             }
             if pg is not None:
                 payload["pg_id"], payload["bundle_index"] = pg[0], pg[1]
@@ -1717,6 +1718,27 @@ class CoreRuntime:
 
     def raylet_call(self, method, payload, timeout=None):
         return self._call_sync(self.raylet.call(method, payload), timeout)
+
+
+def _profiler_cmd(runtime_env):
+    """Profiler runtime-env plugins (reference:
+    _private/runtime_env/rocprof_sys.py:17, nsight.py): the worker
+    process is launched under the profiler. {"rocprof": {...}} expands
+    to a rocprofv3 wrapper; {"_wrapper_cmd": [...]} is the generic
+    escape hatch."""
+    if not runtime_env:
+        return None
+    if runtime_env.get("_wrapper_cmd"):
+        return list(runtime_env["_wrapper_cmd"])
+    rp = runtime_env.get("rocprof")
+    if rp is None:
+        return None
+    cmd = [rp.get("bin", "rocprofv3")]
+    cmd += list(rp.get("args", ["--kernel-trace", "--stats"]))
+    if rp.get("output_dir"):
+        cmd += ["-d", str(rp["output_dir"])]
+    cmd.append("--")
+    return cmd
 
 
 def _trace_ctx():

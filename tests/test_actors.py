@@ -303,3 +303,27 @@ def test_actor_max_task_retries_resubmits(ray_start_regular):
     b = Flaky0.remote()
     with pytest.raises(ray.exceptions.RayError):
         ray.get(b.die.remote(marker2), timeout=60)
+
+
+def test_profiler_runtime_env_wrapper(ray_start_regular):
+    """Profiler runtime-env plugin (reference:
+    runtime_env/rocprof_sys.py:17 — the worker launches UNDER the
+    profiler wrapper). Verified with the generic wrapper (`env VAR=x`)
+    so the test runs without a GPU; {'rocprof': {...}} expands to a
+    rocprofv3 wrapper the same way."""
+    import os as _os
+
+    @ray.remote(runtime_env={"_wrapper_cmd": ["env", "RAY_AMD_WRAPPED=yes"]})
+    class Probed:
+        def wrapped(self):
+            return _os.environ.get("RAY_AMD_WRAPPED")
+
+    a = Probed.remote()
+    assert ray.get(a.wrapped.remote(), timeout=60) == "yes"
+
+    # the rocprof sugar produces a rocprofv3 command line
+    from ray_amd._core.runtime import _profiler_cmd
+
+    cmd = _profiler_cmd({"rocprof": {"output_dir": "/tmp/p",
+                                     "args": ["--kernel-trace"]}})
+    assert cmd == ["rocprofv3", "--kernel-trace", "-d", "/tmp/p", "--"]
