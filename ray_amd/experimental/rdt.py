@@ -28,6 +28,65 @@ _store: Optional["GpuObjectStore"] = None
 _lock = threading.Lock()
 
 
+class _RdtIO:
+    """Dedicated I/O loop for RDT fetches: store.get() may be invoked
+    during DESERIALIZATION on the runtime's event-loop thread (auto-
+    fetching refs in task args), where blocking on the runtime loop
+    would deadlock. Fetch RPCs run on this loop instead."""
+
+    def __init__(self):
+        import asyncio
+
+        self.loop = asyncio.new_event_loop()
+        self._conns = {}
+        t = threading.Thread(target=self.loop.run_forever,
+                             name="ray_amd_rdt_io", daemon=True)
+        t.start()
+
+    async def _conn(self, addr):
+        from ray_amd._core.protocol import RpcClient
+
+        c = self._conns.get(addr)
+        if c is None or not c.connected:
+            c = RpcClient()
+            await c.connect(addr, retries=10)
+            self._conns[addr] = c
+        return c
+
+    def call(self, addr, method, payload, timeout=120.0):
+        import asyncio
+
+        async def _do():
+            c = await self._conn(addr)
+            return await c.call(method, payload)
+
+        return asyncio.run_coroutine_threadsafe(
+            _do(), self.loop).result(timeout)
+
+    def notify(self, addr, method, payload):
+        import asyncio
+
+        async def _do():
+            try:
+                c = await self._conn(addr)
+                c.notify(method, payload)
+            except Exception:
+                pass
+
+        asyncio.run_coroutine_threadsafe(_do(), self.loop)
+
+
+_io: Optional[_RdtIO] = None
+
+
+def _rdt_io() -> _RdtIO:
+    global _io
+    with _lock:
+        if _io is None:
+            _io = _RdtIO()
+        return _io
+
+
 class GpuObjectRef:
     """Tiny handle: (object id, owner runtime address, node id, meta)."""
 
@@ -138,11 +197,8 @@ class GpuObjectStore:
             else "staged"
         )
 
-        async def _fetch():
-            c = await self._rt._conn(ref.owner_addr)
-            return await c.call("rdt_fetch", {"id": ref.id, "mode": mode})
-
-        r = self._rt._call_sync(_fetch(), 120)
+        r = _rdt_io().call(ref.owner_addr, "rdt_fetch",
+                           {"id": ref.id, "mode": mode})
         if r is None:
             raise KeyError(f"GPU object {ref.id.hex()} not found at owner")
         if r["mode"] == "ipc":
@@ -174,14 +230,7 @@ class GpuObjectStore:
             del self._tensors[ref.id]
             return
 
-        async def _free():
-            c = await self._rt._conn(ref.owner_addr)
-            c.notify("rdt_free", {"id": ref.id})
-
-        try:
-            self._rt._call_sync(_free(), 10)
-        except Exception:
-            pass
+        _rdt_io().notify(ref.owner_addr, "rdt_free", {"id": ref.id})
 
     def num_objects(self) -> int:
         return len(self._tensors)
