@@ -400,6 +400,25 @@ class ServeController:
         opts = dict(spec.get("ray_actor_options") or {})
         opts.setdefault("num_cpus", 0.1)
         opts["max_concurrency"] = max(16, spec.get("max_ongoing_requests", 100))
+        # Deployment scheduler (reference: deployment_scheduler.py
+        # spread): round-robin replicas across alive nodes via soft
+        # node affinity so one node's failure doesn't take every
+        # replica down.
+        if "scheduling_strategy" not in opts:
+            try:
+                from ray_amd.util.scheduling_strategies import (
+                    NodeAffinitySchedulingStrategy,
+                )
+
+                alive = [n for n in ray.nodes() if n["Alive"]]
+                if len(alive) > 1:
+                    self._rr = getattr(self, "_rr", 0) + 1
+                    target = alive[self._rr % len(alive)]["NodeID"]
+                    opts["scheduling_strategy"] = (
+                        NodeAffinitySchedulingStrategy(target, soft=True)
+                    )
+            except Exception:
+                pass
         RA = ray.remote(ReplicaActor)
         return RA.options(**opts).remote(
             cls_or_fn, init_args, init_kwargs, spec.get("user_config")

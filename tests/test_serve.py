@@ -428,3 +428,27 @@ def test_grpc_proxy_ingress(serve_session):
     out = json.loads(call(b"xyz", timeout=30))
     assert out == {"method": "Anything", "len": 3}
     channel.close()
+
+
+def test_replica_spread_across_nodes():
+    """Deployment scheduler spread (reference: deployment_scheduler.py):
+    replicas round-robin across alive nodes via soft node affinity."""
+    from ray_amd.cluster_utils import Cluster
+
+    cluster = Cluster(head_node_args={"num_cpus": 4})
+    try:
+        cluster.add_node(num_cpus=4)
+        cluster.connect()
+        cluster.wait_for_nodes()
+
+        @serve.deployment(num_replicas=4)
+        class Where:
+            def __call__(self, _):
+                return ray.get_runtime_context().get_node_id()
+
+        h = serve.run(Where.bind(), http=False)
+        nodes = {h.remote(None).result(timeout_s=60) for _ in range(16)}
+        assert len(nodes) == 2, nodes  # both nodes host replicas
+        serve.shutdown()
+    finally:
+        cluster.shutdown()
