@@ -16,6 +16,7 @@
 #include "hip/fa_bwd_v3.hip"
 #include "hip/fa_bwd_v4.hip"
 #include "hip/fa_bwd_dq_v4.hip"
+#include "hip/fa_bwd_dkv_v5.hip"
 
 #define CHECK_IN(x)                                                     \
   TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                       \
@@ -345,6 +346,20 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
     // (256 kv rows) when the shape allows
     bool v4 = (T % 256 == 0) &&
               (bthd || getenv("RAY_AMD_FA_BWD_V3ONLY") == nullptr);
+    if (v4 && getenv("RAY_AMD_FA_NO_DKV5") == nullptr) {
+      // v5: fused dK+dV — S and dP computed once per tile, half the
+      // Q/dO staging; V comes from LDS to stay at 2 waves/SIMD
+      hipLaunchKernelGGL(fa_bwd_dkv_v5_bf16, dim3(T / 256, B * Hkv),
+                         dim3(512), 0, cur_stream(),
+                         (const short*)q.data_ptr(),
+                         (const short*)k.data_ptr(),
+                         (const short*)v.data_ptr(),
+                         (const short*)d_o.data_ptr(), lse.data_ptr<float>(),
+                         dsum.data_ptr<float>(), (short*)dk.data_ptr(),
+                         (short*)dv.data_ptr(), B, Hq, Hkv, T,
+                         causal ? 1 : 0, scale, bthd);
+      return {dq, dk, dv};
+    }
     if (v4) {
       hipLaunchKernelGGL(fa_bwd_dv_v4_bf16, dim3(T / 256, B * Hkv),
                          dim3(512), 0, cur_stream(),
