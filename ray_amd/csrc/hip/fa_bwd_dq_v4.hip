@@ -128,14 +128,23 @@ extern "C" __global__ __launch_bounds__(512) void fa_bwd_dq_v4_bf16(
         fb3_u32x4 pw{(unsigned)sw0[0], (unsigned)sw1[0], (unsigned)sw0[1],
                      (unsigned)sw1[1]};
         fb3_bf16x8 af = __builtin_bit_cast(fb3_bf16x8, pw);
+        // K B-fragments via ds_read_tr16_b64 (probe-verified; same
+        // cooperative transpose-read as dq_v3)
+        const int gl = lane & 15;
+        const int grp16 = (lane >> 4) & 1;
 #pragma unroll
         for (int dt = 0; dt < 4; ++dt) {
           short ktmp[8];
 #pragma unroll
-          for (int i = 0; i < 8; ++i) {
-            int row = 16 * kc + a_off + i;
-            int col = 32 * dt + (lane & 31);
-            ktmp[i] = k_lds[cur][row][col ^ ((row & 7) << 3)];
+          for (int h = 0; h < 2; ++h) {
+            int row = 16 * kc + a_off + 4 * h + (gl >> 2);
+            int col = (32 * dt + 16 * grp16 + 4 * (gl & 3))
+                      ^ ((row & 7) << 3);
+            auto p = (__attribute__((address_space(3))) short4v*)
+                &k_lds[cur][row][col];
+            short4v r4 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(p);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) ktmp[4 * h + j] = r4[j];
           }
           fb3_bf16x8 bf = __builtin_bit_cast(
               fb3_bf16x8, *reinterpret_cast<short8*>(ktmp));

@@ -319,7 +319,9 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
   static const bool v3 = getenv("RAY_AMD_FA_BWD_V2") == nullptr;
   TORCH_CHECK(!bthd || (split && v3),
               "BTHD layout requires the v3/v4 backward path");
-  if (v3 && T % 256 == 0 && getenv("RAY_AMD_FA_DQ_V4") != nullptr)
+  // dq v4 (8-wave + tr16 K-gathers) measured 12.68 vs v3 13.15 ms
+  // whole-bwd; the pre-tr16 v4 had regressed — tr16 flipped it
+  if (v3 && T % 256 == 0 && getenv("RAY_AMD_FA_NO_DQ_V4") == nullptr)
     hipLaunchKernelGGL(fa_bwd_dq_v4_bf16, dim3(T / 256, B * Hq), dim3(512),
                        0, cur_stream(), (const short*)q.data_ptr(),
                        (const short*)k.data_ptr(), (const short*)v.data_ptr(),
