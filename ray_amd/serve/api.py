@@ -836,17 +836,41 @@ class ProxyActor:
                 send, 500, {}, f'{{"error":"{last_exc}"}}'.encode()
             )
 
-        config = uvicorn.Config(
-            asgi, host="127.0.0.1", port=self.port, log_level="warning",
-            loop="asyncio",
-        )
-        self._server = uvicorn.Server(config)
-        self._server_task = asyncio.ensure_future(self._server.serve())
-        for _ in range(100):
-            if self._server.started:
-                return True
-            await asyncio.sleep(0.05)
+        # Try the configured port; on EADDRINUSE fall back to an
+        # ephemeral port (a second node's proxy in the single-machine
+        # multi-node sim — on a real cluster every node binds the same
+        # port on its own host). Returns the BOUND port (truthy).
+        async def _guarded_serve(server):
+            # uvicorn sys.exit()s on bind failure — a SystemExit escaping
+            # a task would tear down the actor's event loop
+            try:
+                await server.serve()
+            except (SystemExit, OSError):
+                pass
+
+        for try_port in (self.port, 0):
+            config = uvicorn.Config(
+                asgi, host="127.0.0.1", port=try_port, log_level="warning",
+                loop="asyncio",
+            )
+            self._server = uvicorn.Server(config)
+            self._server_task = asyncio.ensure_future(
+                _guarded_serve(self._server))
+            for _ in range(100):
+                if self._server.started:
+                    try:
+                        self.port = (self._server.servers[0].sockets[0]
+                                     .getsockname()[1])
+                    except Exception:
+                        pass
+                    return self.port
+                if self._server_task.done():
+                    break  # bind failed — retry ephemeral
+                await asyncio.sleep(0.05)
         return False
+
+    def bound_port(self):
+        return self.port
 
     async def _proxy_stream(self, send, replica, method, path, qs, headers,
                             body, rt, loop):
@@ -1071,6 +1095,60 @@ def _ensure_proxy(port: int):
         return proxy
 
 
+def _ensure_node_proxies(port: int) -> Dict[str, int]:
+    """One HTTP proxy per alive node (reference: ProxyStateManager in
+    _private/proxy_state.py starts a proxy actor on every node). The
+    head-node singleton keeps the legacy SERVE_PROXY_ACTOR name; the
+    others are SERVE_PROXY_ACTOR:<node_id> hard-pinned to their node.
+    Returns {node_id_hex: bound_port}."""
+    ray = _ray()
+    from ray_amd.util.scheduling_strategies import (
+        NodeAffinitySchedulingStrategy,
+    )
+
+    out: Dict[str, int] = {}
+    for n in ray.nodes():
+        if not n["Alive"]:
+            continue
+        nid = n["NodeID"]
+        aname = f"{SERVE_PROXY_NAME}:{nid}"
+        try:
+            proxy = ray.get_actor(aname, namespace=SERVE_NAMESPACE)
+            out[nid] = ray.get(proxy.bound_port.remote(), timeout=30)
+            continue
+        except ValueError:
+            pass
+        P = ray.remote(ProxyActor)
+        proxy = P.options(
+            name=aname, namespace=SERVE_NAMESPACE, num_cpus=0.1,
+            max_concurrency=64,
+            scheduling_strategy=NodeAffinitySchedulingStrategy(
+                nid, soft=False),
+        ).remote(port)
+        bound = ray.get(proxy.start_server.remote(), timeout=30)
+        if not bound:
+            raise RuntimeError(f"serve proxy failed to start on node {nid}")
+        out[nid] = bound
+    return out
+
+
+def proxy_ports() -> Dict[str, int]:
+    """Bound HTTP proxy port per node id (reference: serve.status()
+    proxies field)."""
+    ray = _ray()
+    out: Dict[str, int] = {}
+    for n in ray.nodes():
+        if not n["Alive"]:
+            continue
+        try:
+            p = ray.get_actor(f"{SERVE_PROXY_NAME}:{n['NodeID']}",
+                              namespace=SERVE_NAMESPACE)
+            out[n["NodeID"]] = ray.get(p.bound_port.remote(), timeout=30)
+        except Exception:
+            pass
+    return out
+
+
 def run(app: Application, *, name: str = "default", route_prefix: str = "/",
         blocking: bool = False, _local_testing_mode: bool = False,
         http: bool = True, port: Optional[int] = None) -> DeploymentHandle:
@@ -1130,6 +1208,12 @@ def run(app: Application, *, name: str = "default", route_prefix: str = "/",
     )
     if http:
         _ensure_proxy(port or _http_port)
+        try:
+            alive = [n for n in ray.nodes() if n["Alive"]]
+            if len(alive) > 1:
+                _ensure_node_proxies(port or _http_port)
+        except Exception:
+            pass  # single-node path keeps the head proxy only
     h = DeploymentHandle(name)
     h._refresh()
     return h
@@ -1162,7 +1246,12 @@ def delete(name: str, _blocking: bool = True):
 
 def shutdown():
     ray = _ray()
-    for n in (SERVE_PROXY_NAME, SERVE_CONTROLLER_NAME):
+    names = [SERVE_PROXY_NAME, SERVE_CONTROLLER_NAME]
+    try:
+        names += [f"{SERVE_PROXY_NAME}:{n['NodeID']}" for n in ray.nodes()]
+    except Exception:
+        pass
+    for n in names:
         try:
             a = ray.get_actor(n, namespace=SERVE_NAMESPACE)
             ray.kill(a)

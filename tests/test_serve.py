@@ -452,3 +452,35 @@ def test_replica_spread_across_nodes():
         serve.shutdown()
     finally:
         cluster.shutdown()
+
+
+def test_per_node_proxy_topology():
+    """One HTTP proxy per node (reference: ProxyStateManager): each
+    alive node gets a pinned proxy actor and every proxy routes."""
+    import urllib.request
+
+    from ray_amd.cluster_utils import Cluster
+    from ray_amd.serve.api import proxy_ports
+
+    cluster = Cluster(head_node_args={"num_cpus": 4})
+    try:
+        cluster.add_node(num_cpus=4)
+        cluster.connect()
+        cluster.wait_for_nodes()
+
+        @serve.deployment
+        class Hello:
+            def __call__(self, req):
+                return "hi"
+
+        serve.run(Hello.bind(), port=18431)
+        ports = proxy_ports()
+        assert len(ports) == 2, ports  # a proxy on every node
+        for p in ports.values():
+            body = urllib.request.urlopen(
+                f"http://127.0.0.1:{p}/", timeout=30
+            ).read()
+            assert b"hi" in body
+        serve.shutdown()
+    finally:
+        cluster.shutdown()
