@@ -41,13 +41,16 @@ class WorkerProc:
 
 
 class Lease:
-    def __init__(self, lease_id, resources, worker, gpu_alloc, pg=None):
+    def __init__(self, lease_id, resources, worker, gpu_alloc, pg=None,
+                 conn=None):
         self.lease_id = lease_id
         self.resources = resources
         self.worker: WorkerProc = worker
         self.gpu_alloc = gpu_alloc  # [(device, fraction)]
         self.gpu_ids = [d for d, _ in gpu_alloc]
         self.pg = pg
+        self.conn = conn  # granting client's connection (for revocation)
+        self.revoke_asked = 0.0
 
 
 class Raylet:
@@ -343,7 +346,7 @@ class Raylet:
         if q is None:
             q = self._pending[key] = deque()
             self._class_order.append(key)
-        q.append((req, pg, fut))
+        q.append((req, pg, fut, conn))
         self._try_grant()
         return await fut
 
@@ -373,7 +376,7 @@ class Raylet:
                         if q is None:
                             q = self._pending[key] = deque()
                             self._class_order.append(key)
-                        q.append((req, None, fut))
+                        q.append((req, None, fut, None))
                         self._try_grant()
                         continue
                     spill = await self._find_spill_target(req)
@@ -456,7 +459,7 @@ class Raylet:
     def _grant_head(self, q: deque) -> bool:
         """Try to grant the head of one class queue. Returns True if an
         item was consumed (granted or dropped), False if blocked."""
-        req, pg, fut = q[0]
+        req, pg, fut, rconn = q[0]
         if fut.done():
             q.popleft()
             return True
@@ -471,6 +474,8 @@ class Raylet:
             bundle = self.bundles[bkey]
             pool_avail = bundle["avail"]
         if not self._fits(pool_avail, req):
+            if pg is None:
+                self._ask_revocations(req, rconn)
             return False
         if not self._idle_task_workers:
             # bounded pool (reference: worker_pool.h soft limit):
@@ -504,7 +509,7 @@ class Raylet:
             for k, v in req.items():
                 bundle["avail"][k] = bundle["avail"].get(k, 0) - v
         self._lease_seq += 1
-        lease = Lease(self._lease_seq, req, w, gpu_alloc, pg)
+        lease = Lease(self._lease_seq, req, w, gpu_alloc, pg, conn=rconn)
         self.leases[lease.lease_id] = lease
         w.idle = False
         w.lease_id = lease.lease_id
@@ -515,6 +520,33 @@ class Raylet:
              "gpu_ids": lease.gpu_ids, "raylet": self.addr}
         )
         return True
+
+    def _ask_revocations(self, req: Dict[str, float], requester_conn):
+        """Queued lease blocked on resources held by granted leases:
+        ask OTHER holders to give back leases they are only caching
+        idle (reference: the lease holder returns workers on
+        ReleaseUnusedWorkers / idle-timeout; revocation makes that
+        demand-driven instead of timer-driven). Busy leases are left
+        alone — the holder returns them at task completion."""
+        import time as _t
+
+        from .protocol import MSG_NOTIFY
+
+        now = _t.monotonic()
+        for lease in self.leases.values():
+            c = lease.conn
+            if c is None or c is requester_conn:
+                continue
+            if now - lease.revoke_asked < 0.5:
+                continue
+            if not any(k in lease.resources for k in req):
+                continue  # disjoint resources — returning it won't help
+            lease.revoke_asked = now
+            try:
+                c.send([MSG_NOTIFY, 0, "revoke_lease",
+                        {"lease_id": lease.lease_id}])
+            except Exception:
+                pass
 
     def _release_resources(self, lease: Lease):
         if lease.pg is not None:

@@ -171,9 +171,10 @@ class ObjectRefGenerator:
 
 
 class _Lease:
-    __slots__ = ("lease_id", "addr", "client", "busy", "gpu_ids")
+    __slots__ = ("lease_id", "addr", "client", "busy", "gpu_ids", "revoked")
 
     def __init__(self, lease_id, addr, client, gpu_ids=()):
+        self.revoked = False
         self.lease_id = lease_id
         self.addr = addr
         self.client = client
@@ -340,6 +341,7 @@ class CoreRuntime:
         await self.gcs.connect(self.gcs_addr)
         self.gcs.on_notify = self._on_conn_notify
         await self.raylet.connect(self.raylet_addr)
+        self.raylet.on_notify = self._on_conn_notify
 
     def _run(self, coro):
         """Schedule a coroutine on the loop from any thread."""
@@ -398,6 +400,8 @@ class CoreRuntime:
     def _on_conn_notify(self, method, payload):
         if method == "stream_item":
             self._h_stream_item(payload)
+        elif method == "revoke_lease":
+            self._h_revoke_lease(payload)
         elif method == "pubsub":
             for cb in self._pubsub_cbs.get(payload["channel"], []):
                 try:
@@ -1423,10 +1427,32 @@ class CoreRuntime:
             if not fut.done():
                 fut.set_result(lease)
                 return
+        if lease.revoked:
+            # raylet asked for it back while we ran — no local waiter,
+            # so hand it over immediately instead of idling it out
+            if lease in pool.leases:
+                pool.leases.remove(lease)
+            asyncio.ensure_future(self._return_lease(pool, lease))
+            return
         # nobody waiting: keep lease idle; return after a short grace
         # period (long enough for submit->get->submit reuse)
         self.loop.call_later(_cfg.lease_idle_grace_s,
                              self._maybe_return_idle, pool, lease)
+
+    def _h_revoke_lease(self, p):
+        """Raylet-initiated lease revocation (contention): return the
+        lease now if it is idle, or flag it to be returned at task
+        completion when no local waiter needs it."""
+        lid = p.get("lease_id")
+        for pool in self._pools.values():
+            for lease in pool.leases:
+                if lease.lease_id[0] != lid:
+                    continue
+                lease.revoked = True
+                if not lease.busy and not pool.queue:
+                    pool.leases.remove(lease)
+                    asyncio.ensure_future(self._return_lease(pool, lease))
+                return
 
     def _release_or_reuse(self, pool: _LeasePool, lease: _Lease):
         self._grant_to_queue(pool, lease)

@@ -145,3 +145,50 @@ def test_fair_dispatch_across_scheduling_classes():
         assert ray.get(big_refs, timeout=60) == ["big"] * 3
     finally:
         ray.shutdown()
+
+
+def test_lease_revocation_under_contention(monkeypatch):
+    """A second driver's queued lease request revokes another driver's
+    idle cached lease instead of waiting out the idle grace period
+    (reference: ReleaseUnusedWorkers / lease-holder worker return,
+    made demand-driven)."""
+    import subprocess
+    import sys
+
+    # grace long enough that only revocation can explain a fast handoff
+    monkeypatch.setenv("RAY_AMD_LEASE_IDLE_GRACE_S", "30")
+    from ray_amd._config import config
+
+    config.reload()
+    try:
+        ray.init(num_cpus=1)
+
+        @ray.remote
+        def hold():
+            return "held"
+
+        # driver 1 runs a task, then CACHES the 1-CPU lease idle (30s)
+        assert ray.get(hold.remote(), timeout=30) == "held"
+        rt = ray.api._rt.global_runtime()
+        script = f"""
+import time, ray_amd as ray
+ray.init(address={rt.session_dir!r})
+
+@ray.remote
+def g():
+    return "ran"
+
+t0 = time.time()
+out = ray.get(g.remote(), timeout=20)
+print("RESULT", out, round(time.time() - t0, 1))
+ray.shutdown(_exiting_interpreter=True)
+"""
+        out = subprocess.run(
+            [sys.executable, "-c", script], capture_output=True, text=True,
+            timeout=60,
+        )
+        # without revocation driver 2 would block the full 30s grace
+        assert "RESULT ran" in out.stdout, out.stdout + out.stderr
+    finally:
+        ray.shutdown()
+        config.reload()
