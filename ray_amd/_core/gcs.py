@@ -75,6 +75,11 @@ class GcsServer:
     def __init__(self, sock_path: str, persist_path: str = ""):
         self.sock_path = sock_path
         self.persist_path = persist_path
+        self._storage = None
+        if persist_path:
+            from .gcs_storage import open_storage
+
+            self._storage = open_storage(persist_path)
         self._dirty = False
         self.server = RpcServer()
         self.kv: Dict[str, Dict[bytes, bytes]] = {}
@@ -139,19 +144,33 @@ class GcsServer:
         )
 
     def _restore(self):
-        import os as _os
         import pickle
 
-        if not _os.path.exists(self.persist_path):
-            return
-        try:
-            with open(self.persist_path, "rb") as f:
-                snap = pickle.load(f)
-        except Exception:
-            return
+        from .gcs_storage import decode_op
+
+        blob, ops = self._storage.load()
+        snap = {}
+        if blob is not None:
+            try:
+                snap = pickle.loads(blob)
+            except Exception:
+                snap = {}
         self.kv = snap.get("kv", {})
         self.named_actors = snap.get("named_actors", {})
         self.job_counter = snap.get("job_counter", 0)
+        # replay the synchronous KV journal on top of the snapshot
+        # (sqlite backend: acknowledged writes survive kill -9)
+        for raw in ops:
+            try:
+                op = decode_op(raw)
+            except Exception:
+                continue
+            if op[0] == "kv_put":
+                _, ns, key, value = op
+                self.kv.setdefault(ns, {})[key] = value
+            elif op[0] == "kv_del":
+                _, ns, key = op
+                self.kv.get(ns, {}).pop(key, None)
         # actors come back PENDING and are rescheduled once a raylet
         # re-registers (restart-based recovery)
         for aid, spec in snap.get("actor_specs", {}).items():
@@ -166,12 +185,7 @@ class GcsServer:
                 continue
             self._dirty = False
             try:
-                tmp = self.persist_path + ".tmp"
-                with open(tmp, "wb") as f:
-                    f.write(self._snapshot())
-                import os as _os
-
-                _os.replace(tmp, self.persist_path)
+                self._storage.save_snapshot(self._snapshot())
             except Exception:
                 pass
 
@@ -183,6 +197,11 @@ class GcsServer:
         if p.get("overwrite", True) or not exists:
             ns[key] = p["value"]
             self._dirty = True
+            if self._storage is not None:
+                from .gcs_storage import encode_op
+
+                self._storage.journal(
+                    encode_op("kv_put", p.get("ns", ""), key, p["value"]))
             return not exists
         return False
 
@@ -190,7 +209,13 @@ class GcsServer:
         return self.kv.get(p.get("ns", ""), {}).get(p["key"])
 
     def h_kv_del(self, conn, p):
-        return self.kv.get(p.get("ns", ""), {}).pop(p["key"], None) is not None
+        hit = self.kv.get(p.get("ns", ""), {}).pop(p["key"], None) is not None
+        if hit and self._storage is not None:
+            from .gcs_storage import encode_op
+
+            self._storage.journal(
+                encode_op("kv_del", p.get("ns", ""), p["key"]))
+        return hit
 
     def h_kv_exists(self, conn, p):
         return p["key"] in self.kv.get(p.get("ns", ""), {})
