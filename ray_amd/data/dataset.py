@@ -866,17 +866,48 @@ class Dataset:
         return self._agg("mean", on)
 
     def std(self, on: str):
-        import pandas as pd  # noqa
+        # pushdown: per-block (n, sum, sumsq) partials, driver combine
+        ray = _ray()
+        refs = self._materialize_refs()
 
-        vals = _col_to_numpy(
-            pa.concat_tables(list(self._iter_tables())).column(on)
-        )
-        return float(np.std(vals, ddof=1))
+        @ray.remote
+        def _part(t, on=on):
+            v = _col_to_numpy(t.column(on)) if t.num_rows else np.array([])
+            return (int(v.size), float(np.sum(v)), float(np.sum(v * v)))
+
+        parts = ray.get([_part.remote(r) for r in refs])
+        n = sum(p[0] for p in parts)
+        s = sum(p[1] for p in parts)
+        ss = sum(p[2] for p in parts)
+        return float(np.sqrt(max(ss - s * s / n, 0.0) / (n - 1)))
 
     def _agg(self, how: str, on: str):
-        t = pa.concat_tables(list(self._iter_tables()))
-        vals = _col_to_numpy(t.column(on))
-        return getattr(np, how)(vals).item()
+        """Global aggregate with pushdown (reference: AggregateFn map/
+        combine in _internal/planner/exchange): each block reduces
+        remotely; only scalars travel to the driver."""
+        ray = _ray()
+        refs = self._materialize_refs()
+        if len(refs) == 1:
+            t = pa.concat_tables(list(self._iter_tables()))
+            vals = _col_to_numpy(t.column(on))
+            return getattr(np, how)(vals).item()
+
+        @ray.remote
+        def _part(t, on=on, how=how):
+            if t.num_rows == 0:
+                return None
+            v = _col_to_numpy(t.column(on))
+            if how == "mean":
+                return (float(np.sum(v)), int(v.size))
+            return getattr(np, how)(v).item()
+
+        parts = [p for p in ray.get([_part.remote(r) for r in refs])
+                 if p is not None]
+        if how == "mean":
+            return sum(p[0] for p in parts) / sum(p[1] for p in parts)
+        if how == "sum":
+            return type(parts[0])(sum(parts))
+        return (min if how == "min" else max)(parts)
 
     def unique(self, column: str):
         t = pa.concat_tables(list(self._iter_tables()))
@@ -970,26 +1001,48 @@ class GroupedData:
         return t.group_by(self._key)
 
     def _agg_distributed(self, on: str, how: str) -> Optional[Dataset]:
-        """Multi-block: hash-exchange by key, per-partition aggregate
-        tasks (all rows of a key land in one partition), concat."""
+        """Multi-block: per-block PARTIAL aggregate (map-side combine —
+        reference: aggregate pushdown in _internal/planner/exchange),
+        hash-exchange only the combined partials, re-aggregate per
+        partition. Shuffle volume is O(distinct keys per block), not
+        O(rows)."""
         ray = _ray()
         refs = self._ds._materialize_refs()
         if len(refs) <= 1:
             return None
         key = self._key
-        parts = _shuffle_exchange(ray, refs, key, len(refs))
 
         @ray.remote
-        def _agg(t, on=on, how=how, key=key):
+        def _partial(t, on=on, how=how, key=key):
             if t.num_rows == 0:
                 return t
-            out = t.group_by(key).aggregate([(on, how)])
-            return out.rename_columns(
-                [f"{how}({on})" if c == f"{on}_{how}" else c
-                 for c in out.column_names]
-            )
+            if how == "mean":
+                return t.group_by(key).aggregate(
+                    [(on, "sum"), (on, "count")])
+            return t.group_by(key).aggregate([(on, how)])
 
-        out_refs = [_agg.remote(p) for p in parts]
+        partial_refs = [_partial.remote(r) for r in refs]
+        parts = _shuffle_exchange(ray, partial_refs, key, len(refs))
+
+        @ray.remote
+        def _final(t, on=on, how=how, key=key):
+            if t.num_rows == 0:
+                return t
+            if how == "mean":
+                g = t.group_by(key).aggregate(
+                    [(f"{on}_sum", "sum"), (f"{on}_count", "sum")])
+                s = _col_to_numpy(g.column(f"{on}_sum_sum"))
+                n = _col_to_numpy(g.column(f"{on}_count_sum"))
+                return pa.table({key: g.column(key),
+                                 f"mean({on})": pa.array(s / n)})
+            # sum of partial sums; min of mins; max of maxes
+            refn = "sum" if how == "sum" else how
+            g = t.group_by(key).aggregate([(f"{on}_{how}", refn)])
+            return g.rename_columns(
+                [f"{how}({on})" if c == f"{on}_{how}_{refn}" else c
+                 for c in g.column_names])
+
+        out_refs = [_final.remote(p) for p in parts]
         ds = Dataset(out_refs, [])
         ds._materialized = out_refs
         return ds
@@ -1002,6 +1055,34 @@ class GroupedData:
         return ds
 
     def count(self) -> Dataset:
+        ray = _ray()
+        refs = self._ds._materialize_refs()
+        key = self._key
+        if len(refs) > 1:
+            # pushdown: per-block counts, shuffle partials, sum
+
+            @ray.remote
+            def _partial(t, key=key):
+                if t.num_rows == 0:
+                    return t
+                return t.group_by(key).aggregate([(key, "count")])
+
+            parts = _shuffle_exchange(
+                ray, [_partial.remote(r) for r in refs], key, len(refs))
+
+            @ray.remote
+            def _final(t, key=key):
+                if t.num_rows == 0:
+                    return t
+                g = t.group_by(key).aggregate([(f"{key}_count", "sum")])
+                return g.rename_columns(
+                    ["count()" if c == f"{key}_count_sum" else c
+                     for c in g.column_names])
+
+            out_refs = [_final.remote(p) for p in parts]
+            ds = Dataset(out_refs, [])
+            ds._materialized = out_refs
+            return ds
         t = self._grouped().aggregate([(self._key, "count")])
         t = t.rename_columns([self._key, "count()"])
         return self._wrap(t)

@@ -177,6 +177,32 @@ def test_distributed_groupby(ray_start_regular):
     assert got == expect
 
 
+def test_aggregate_pushdown_multiblock(ray_start_regular):
+    """Map-side combine: grouped mean/count and global aggregates on a
+    multi-block dataset reduce per block before any shuffle/driver
+    transfer (reference: AggregateFn map/combine pushdown)."""
+    ds = rd.from_items([{"k": i % 4, "v": float(i)} for i in range(80)],
+                       parallelism=8)
+    mean = {int(r["k"]): r["mean(v)"]
+            for r in ds.groupby("k").mean("v").take_all()}
+    cnt = {int(r["k"]): r["count()"]
+           for r in ds.groupby("k").count().take_all()}
+    mn = {int(r["k"]): r["min(v)"]
+          for r in ds.groupby("k").min("v").take_all()}
+    import numpy as _np
+
+    for k in range(4):
+        vs = _np.array([float(i) for i in range(80) if i % 4 == k])
+        assert abs(mean[k] - vs.mean()) < 1e-9
+        assert cnt[k] == 20
+        assert mn[k] == vs.min()
+    # global aggregates on the same multi-block dataset
+    assert ds.sum("v") == sum(range(80))
+    assert abs(ds.mean("v") - 39.5) < 1e-9
+    assert ds.min("v") == 0.0 and ds.max("v") == 79.0
+    assert abs(ds.std("v") - _np.std(_np.arange(80.0), ddof=1)) < 1e-9
+
+
 def test_hash_join(ray_start_regular):
     left = rd.from_items(
         [{"k": i, "a": i * 10} for i in range(20)], parallelism=4
