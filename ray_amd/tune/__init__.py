@@ -27,6 +27,15 @@ from .impl import (  # noqa: F401
     with_parameters,
     with_resources,
 )
+from . import search  # noqa: F401
+from .search import (  # noqa: F401
+    BasicVariantGenerator,
+    BayesOptSearch,
+    ConcurrencyLimiter,
+    HyperOptSearch,
+    OptunaSearch,
+    Searcher,
+)
 from ..train.checkpoint import Checkpoint  # noqa: F401
 from ..train.session import get_checkpoint, get_context  # noqa: F401
 

@@ -536,10 +536,21 @@ class Tuner:
 
         done_names = set()
         restored_results: List[Result] = []
-        if self._restored is not None:
+        # search_alg drives sequential ask/tell config proposal
+        # (reference: tune/search/ Searcher protocol); without one,
+        # variants are pre-generated
+        searcher = tc.search_alg if self._restored is None else None
+        if searcher is not None:
+            searcher.set_search_properties(
+                tc.metric, tc.mode, self.param_space)
+            variants = []
+            n_target = tc.num_samples
+        elif self._restored is not None:
             variants = self._restored["variants"]
+            n_target = len(variants)
         else:
             variants = generate_variants(self.param_space, tc.num_samples)
+            n_target = len(variants)
         fn_bytes = cloudpickle.dumps(self._trainable)
         # persist experiment state so Tuner.restore can resume it
         with open(os.path.join(storage, "tuner_state.pkl"), "wb") as f:
@@ -567,7 +578,7 @@ class Tuner:
                         error=None,
                     ))
         scheduler = tc.scheduler or FIFOScheduler()
-        max_conc = tc.max_concurrent_trials or min(8, max(1, len(variants)))
+        max_conc = tc.max_concurrent_trials or min(8, max(1, n_target))
         Actor = ray.remote(_TrialActor)
 
         trials = []  # dicts: actor, config, rows, done, error, it
@@ -592,9 +603,19 @@ class Tuner:
                 scheduler.observe_config(name, cfg)
             running.append(t)
 
-        while pending or running:
+        drawn = 0
+        while pending or running or (searcher is not None
+                                     and drawn < n_target):
             while pending and len(running) < max_conc:
                 launch(pending.pop(0))
+            while (searcher is not None and drawn < n_target
+                   and len(running) < max_conc):
+                cfg = searcher.suggest(f"trial_{drawn:05d}")
+                if cfg is None:
+                    break  # searcher-imposed concurrency cap
+                variants.append(cfg)
+                launch((drawn, cfg))
+                drawn += 1
             # issue fetches and wait for any to become ready
             ref_to_trial = {}
             for t in running:
@@ -617,6 +638,9 @@ class Tuner:
                     t["done"] = True
                     running.remove(t)
                     finished.append(t)
+                    if searcher is not None:
+                        searcher.on_trial_complete(t["name"], None,
+                                                   error=True)
                     continue
                 for r in st["results"]:
                     t["it"] += 1
@@ -653,6 +677,12 @@ class Tuner:
                     t["done"] = True
                     running.remove(t)
                     finished.append(t)
+                    if searcher is not None:
+                        searcher.on_trial_complete(
+                            t["name"],
+                            t["rows"][-1] if t["rows"] else None,
+                            error=bool(t["error"]),
+                        )
                     try:
                         ray.kill(t["actor"])
                     except Exception:
