@@ -26,12 +26,17 @@
 namespace py = pybind11;
 
 static void parallel_memcpy(char* dst, const char* src, size_t n) {
-  constexpr size_t kParMin = 32u << 20;
+  constexpr size_t kParMin = 8u << 20;
   if (n < kParMin) {
     std::memcpy(dst, src, n);
     return;
   }
-  const int nt = 4;
+  // tmpfs copies scale with threads well past 4 on EPYC (measured:
+  // single thread ~1.6 GB/s in-container); one thread per 32 MiB up
+  // to the core count, capped at 16
+  int hw = (int)std::thread::hardware_concurrency();
+  int nt = (int)std::min<size_t>(std::max(hw, 2), n / (16u << 20));
+  nt = std::max(2, std::min(nt, 16));
   size_t chunk = (n + nt - 1) / nt;
   std::vector<std::thread> ts;
   for (int i = 0; i < nt; ++i) {
