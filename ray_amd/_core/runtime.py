@@ -765,10 +765,12 @@ class CoreRuntime:
                 ent = await self._store_wait(ref.id, timeout)
             else:
                 ent = await self._fetch_from_owner(ref, timeout)
-        return await self._materialize(ref.id, ent, timeout, ref.owner_addr)
+        # ray.get outranks wait/task-arg transfers at the pull manager
+        return await self._materialize(ref.id, ent, timeout, ref.owner_addr,
+                                       prio=0)
 
     async def _materialize(self, oid: bytes, ent: tuple, timeout=None,
-                           owner_addr=None):
+                           owner_addr=None, prio=2):
         kind = ent[0]
         if kind == "val":
             return ent[1]
@@ -781,7 +783,8 @@ class CoreRuntime:
         if kind == "store":
             while True:
                 try:
-                    return await self._materialize_store(oid, ent, timeout)
+                    return await self._materialize_store(oid, ent, timeout,
+                                                         prio=prio)
                 except exc.ObjectLostError:
                     # re-execute the producing task (lineage) or ask the
                     # owner to; _recover_entry raises ObjectLostError when
@@ -789,11 +792,12 @@ class CoreRuntime:
                     ent = await self._recover_entry(oid, owner_addr)
                     if ent[0] != "store":
                         return await self._materialize(
-                            oid, ent, timeout, owner_addr
+                            oid, ent, timeout, owner_addr, prio
                         )
         raise exc.RaySystemError(f"bad store entry {kind}")
 
-    async def _materialize_store(self, oid: bytes, ent: tuple, timeout=None):
+    async def _materialize_store(self, oid: bytes, ent: tuple, timeout=None,
+                                 prio=2):
         node_addr, size = ent[1], ent[2]
         path = store.shm_path(self.shm_dir, oid)
         if not os.path.exists(path):
@@ -808,7 +812,8 @@ class CoreRuntime:
             else:
                 r = await self.raylet.call(
                     "pull_object",
-                    {"id": oid, "src": node_addr, "timeout": timeout or 120.0},
+                    {"id": oid, "src": node_addr, "timeout": timeout or 120.0,
+                     "prio": prio},
                 )
                 if not r.get("ok"):
                     raise exc.ObjectLostError(oid.hex())
@@ -1048,9 +1053,19 @@ class CoreRuntime:
             ent = self.memory_store.get(r.id)
             if ent is None:
                 if r.owner_addr == self.addr:
-                    await self._store_wait(r.id, None)
+                    ent = await self._store_wait(r.id, None)
+                    if fetch_local and ent[0] == "store" and \
+                            ent[1] != self.raylet_addr:
+                        await self._materialize_store(r.id, ent, None,
+                                                      prio=1)
                 elif fetch_local:
-                    await self._fetch_from_owner(r, None)
+                    ent = await self._fetch_from_owner(r, None)
+                    # fetch_local semantics (reference: ray.wait pulls
+                    # the payload to this node): start the transfer at
+                    # WAIT priority (below get, above task-args)
+                    if ent[0] == "store" and ent[1] != self.raylet_addr:
+                        await self._materialize_store(r.id, ent, None,
+                                                      prio=1)
                 else:
                     # fetch_local=False: only learn that the object
                     # exists somewhere; poll the owner's location table

@@ -117,3 +117,30 @@ def test_pull_dedup_and_priority(tcp_cluster):
 
     outs = ray.get([consume.remote(ref, i) for i in range(8)], timeout=120)
     assert outs == [float(i) for i in range(8)]
+
+
+def test_wait_fetch_local_pulls_payload(tcp_cluster):
+    """ray.wait(fetch_local=True) starts the cross-node transfer at
+    WAIT priority: after wait readies the ref, the payload is already
+    in THIS node's shm dir (reference: ray.wait fetch_local
+    semantics)."""
+    cluster = tcp_cluster
+    cluster.add_node(num_cpus=2, resources={"n2": 1})
+    cluster.connect()
+    cluster.wait_for_nodes()
+
+    @ray.remote(resources={"n2": 0.5})
+    def big_on_n2():
+        return np.arange(700_000, dtype=np.float64)
+
+    ref = big_on_n2.remote()
+    ready, _ = ray.wait([ref], timeout=60, fetch_local=True)
+    assert ready == [ref]
+    rt = ray.api._rt.global_runtime()
+    path = os.path.join(rt.shm_dir, ref.id.hex())
+    deadline = time.time() + 10
+    while not os.path.exists(path) and time.time() < deadline:
+        time.sleep(0.05)
+    assert os.path.exists(path), "payload not pulled to the local node"
+    got = ray.get(ref, timeout=30)
+    assert got.shape == (700_000,)
