@@ -158,6 +158,16 @@ def main(argv=None):
         return 0
     s.set_defaults(fn=_dash)
 
+    s = sub.add_parser("debug")
+    s.add_argument("--address", default=None)
+
+    def _debug(args):
+        from ray_amd.util.rpdb import cmd_debug
+
+        return cmd_debug(args)
+
+    s.set_defaults(fn=_debug)
+
     s = sub.add_parser("microbenchmark")
     s.add_argument("--duration", type=float, default=2.0)
     s.set_defaults(fn=cmd_microbenchmark)
