@@ -52,6 +52,9 @@ class RunConfig:
     checkpoint_config: Optional[CheckpointConfig] = None
     verbose: int = 1
     log_to_file: bool = False
+    # experiment callbacks (reference: air.RunConfig.callbacks —
+    # tune.Callback hooks; wandb/mlflow loggers plug in here)
+    callbacks: Optional[list] = None
 
     def resolved_storage_path(self) -> str:
         base = self.storage_path or os.path.expanduser("~/ray_amd_results")

@@ -579,6 +579,18 @@ class Tuner:
                     ))
         scheduler = tc.scheduler or FIFOScheduler()
         max_conc = tc.max_concurrent_trials or min(8, max(1, n_target))
+        callbacks = list(getattr(self.run_config, "callbacks", None) or [])
+
+        def _cb(method, *a, **k):
+            for cb in callbacks:
+                try:
+                    getattr(cb, method)(*a, **k)
+                except Exception:
+                    import traceback as _tb
+
+                    _tb.print_exc()
+
+        _cb("setup")
         Actor = ray.remote(_TrialActor)
 
         trials = []  # dicts: actor, config, rows, done, error, it
@@ -602,6 +614,7 @@ class Tuner:
             if scheduler is not None and hasattr(scheduler, "observe_config"):
                 scheduler.observe_config(name, cfg)
             running.append(t)
+            _cb("on_trial_start", t["it"], running, t)
 
         drawn = 0
         while pending or running or (searcher is not None
@@ -650,6 +663,7 @@ class Tuner:
                         {f"config/{k}": v for k, v in t["config"].items()}
                     )
                     t["rows"].append(row)
+                    _cb("on_trial_result", t["it"], running, t, row)
                     if r.get("checkpoint_path"):
                         t["ckpt"] = r["checkpoint_path"]
                     mv = r["metrics"].get(tc.metric) if tc.metric else None
@@ -683,6 +697,8 @@ class Tuner:
                             t["rows"][-1] if t["rows"] else None,
                             error=bool(t["error"]),
                         )
+                    _cb("on_trial_error" if t["error"]
+                        else "on_trial_complete", t["it"], running, t)
                     try:
                         ray.kill(t["actor"])
                     except Exception:
@@ -691,6 +707,7 @@ class Tuner:
                 for t in running:
                     t["actor"].request_stop.remote()
 
+        _cb("on_experiment_end", finished)
         results = list(restored_results)
         import json as _json
 
@@ -792,7 +809,25 @@ def with_resources(fn, resources):
 
 
 class Callback:
-    def on_trial_result(self, *a, **k):
+    """Experiment-lifecycle hooks (reference: tune/callback.py) —
+    attach via RunConfig(callbacks=[...])."""
+
+    def setup(self, **info):
+        pass
+
+    def on_trial_start(self, iteration, trials, trial, **info):
+        pass
+
+    def on_trial_result(self, iteration, trials, trial, result, **info):
+        pass
+
+    def on_trial_complete(self, iteration, trials, trial, **info):
+        pass
+
+    def on_trial_error(self, iteration, trials, trial, **info):
+        pass
+
+    def on_experiment_end(self, trials, **info):
         pass
 
 
