@@ -125,7 +125,7 @@ class Raylet:
             "rollback_bundle remove_bundle node_info ping prestart_workers "
             "report_task_events pin_object unpin_object try_recycle "
             "request_push push_ack "
-            "debug_state"
+            "debug_state list_workers"
         ).split():
             self.server.route(m, getattr(self, "h_" + m))
         self.server.on_conn_lost = self._conn_lost
@@ -766,6 +766,15 @@ class Raylet:
             "store": dict(zip(("objects", "used", "capacity"),
                               self.store.stats())),
         }
+
+    def h_list_workers(self, conn, p):
+        """Registered workers on this node (drives `ray_amd stack`)."""
+        return [
+            {"pid": pid, "addr": w.addr, "kind": w.kind,
+             "idle": w.idle}
+            for pid, w in self.workers.items()
+            if w.addr and w.proc.poll() is None
+        ]
 
     def h_free_objects(self, conn, p):
         ids_ = [bytes(i) for i in p["ids"]]

@@ -139,6 +139,7 @@ class WorkerMain:
         rt.server.route("cancel_task", self.h_cancel_task)
         rt.server.route("exit_worker", self.h_exit_worker)
         rt.server.route("stream_ack", self.h_stream_ack)
+        rt.server.route("dump_stack", self.h_dump_stack)
         self._order_lock = asyncio.Lock()
         asyncio.ensure_future(self._event_flusher())
         r = await rt.raylet.call(
@@ -155,6 +156,19 @@ class WorkerMain:
         while rt.raylet.connected:
             await asyncio.sleep(0.5)
         sys.exit(0)
+
+    def h_dump_stack(self, conn, p):
+        """All-thread stack traces (reference: `ray stack` py-spy dump
+        — here served in-process, no ptrace needed)."""
+        import traceback as _tb
+
+        names = {t.ident: t.name for t in threading.enumerate()}
+        stacks = {}
+        for tid, fr in sys._current_frames().items():
+            stacks[f"{names.get(tid, 'thread')}-{tid}"] = "".join(
+                _tb.format_stack(fr))
+        return {"pid": os.getpid(), "actor": bool(self.actor_id),
+                "stacks": stacks}
 
     def h_stream_ack(self, conn, p):
         """Consumer progress for a streaming generator — wakes the

@@ -158,6 +158,57 @@ def main(argv=None):
         return 0
     s.set_defaults(fn=_dash)
 
+    s = sub.add_parser("stack")
+    s.add_argument("--address", default=None)
+
+    def _stack(args):
+        # reference: `ray stack` (py-spy dump of every worker); here
+        # each worker serves its own thread stacks over RPC
+        import ray_amd as ray
+
+        ray.init(address=args.address or "auto", ignore_reinit_error=True)
+        from ray_amd._core import runtime as _rt
+
+        rt = _rt.global_runtime()
+
+        async def _collect():
+            out = []
+            for n in ray.nodes():
+                if not n["Alive"]:
+                    continue
+                try:
+                    c = await rt._conn(n["Address"])
+                    workers = await c.call("list_workers", {})
+                except Exception as e:
+                    out.append((n["NodeID"], None, f"<unreachable: {e}>"))
+                    continue
+                for w in workers:
+                    try:
+                        wc = await rt._conn(w["addr"])
+                        d = await asyncio.wait_for(
+                            wc.call("dump_stack", {}), 10)
+                        out.append((n["NodeID"], w, d))
+                    except Exception as e:
+                        out.append((n["NodeID"], w, f"<no dump: {e}>"))
+            return out
+
+        import asyncio
+
+        for node_id, w, dump in rt._call_sync(_collect()):
+            hdr = (f"=== node {node_id[:12]} pid {w['pid']} "
+                   f"({w['kind']}) ===" if w else
+                   f"=== node {node_id[:12]} ===")
+            print(hdr)
+            if isinstance(dump, str):
+                print(f"  {dump}")
+                continue
+            for tname, stack in dump["stacks"].items():
+                print(f"--- {tname} ---")
+                print(stack)
+        return 0
+
+    s.set_defaults(fn=_stack)
+
     s = sub.add_parser("debug")
     s.add_argument("--address", default=None)
 

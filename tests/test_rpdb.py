@@ -105,3 +105,35 @@ def test_attach_helper_bridges_repl(ray_start_regular):
     rpdb.attach(rec["addr"], stdin=io.StringIO("c\n"), stdout=out)
     assert "(ray-pdb)" in out.getvalue()
     assert ray.get(ref, timeout=30) == "ok"
+
+
+def test_dump_stack_rpc(ray_start_regular):
+    """`ray_amd stack` plumbing: raylet lists workers, each worker
+    serves its thread stacks over RPC (reference: `ray stack`)."""
+    import asyncio
+
+    @ray.remote
+    class Sleeper:
+        def spin(self):
+            time.sleep(0.5)
+            return "ok"
+
+    a = Sleeper.remote()
+    assert ray.get(a.spin.remote(), timeout=30) == "ok"  # worker is up
+    ref = a.spin.remote()
+    rt = ray.api._rt.global_runtime()
+
+    async def collect():
+        workers = await rt.raylet.call("list_workers", {})
+        assert workers, "no workers listed"
+        dumps = []
+        for w in workers:
+            c = await rt._conn(w["addr"])
+            dumps.append(await asyncio.wait_for(c.call("dump_stack", {}), 10))
+        return dumps
+
+    dumps = rt._call_sync(collect())
+    assert all("stacks" in d and d["stacks"] for d in dumps)
+    joined = "\n".join("".join(d["stacks"].values()) for d in dumps)
+    assert "worker" in joined or "run" in joined  # real frames present
+    assert ray.get(ref, timeout=30) == "ok"
