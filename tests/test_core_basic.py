@@ -883,3 +883,23 @@ def test_tracing_spans_cross_process():
     finally:
         _os.environ.pop("RAY_AMD_TRACING", None)
         ray.shutdown()
+
+
+def test_experimental_shuffle(ray_start_regular):
+    """Push-based shuffle prototype (reference:
+    experimental/shuffle.py): mappers push partitions to reducer
+    actors; outputs partition by key."""
+    from ray_amd.experimental.shuffle import shuffle
+
+    blocks = [list(range(i * 10, (i + 1) * 10)) for i in range(4)]
+
+    def partition(block, n):
+        outs = [[] for _ in range(n)]
+        for x in block:
+            outs[x % n].append(x)
+        return outs
+
+    out = shuffle(blocks, 3, partition, lambda ps: sorted(sum(ps, [])))
+    assert sorted(sum(out, [])) == list(range(40))
+    for i, part in enumerate(out):
+        assert all(x % 3 == i for x in part)
