@@ -23,6 +23,7 @@ from .api import (  # noqa: F401
     available_resources,
     cancel,
     cluster_resources,
+    drain_node,
     get,
     get_actor,
     get_gpu_ids,

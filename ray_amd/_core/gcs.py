@@ -30,6 +30,7 @@ class NodeInfo:
         self.resources_available = dict(resources)
         self.labels = labels or {}
         self.alive = True
+        self.draining = False  # graceful drain: no new placements
         self.client: Optional[RpcClient] = None
         self.start_time = time.time()
 
@@ -241,6 +242,7 @@ class GcsServer:
                 "node_id": n.node_id,
                 "addr": n.addr,
                 "alive": n.alive,
+                "draining": n.draining,
                 "resources_total": n.resources_total,
                 "resources_available": n.resources_available,
                 "labels": n.labels,
@@ -311,7 +313,36 @@ class GcsServer:
         return n
 
     async def h_drain_node(self, conn, p):
-        await self._on_node_death(p["node_id"])
+        """Multi-phase node drain (reference: DrainNode RPC + the
+        raylet shutdown FSM): graceful mode marks the node DRAINING —
+        the scheduler stops placing work there immediately — then
+        waits for its leases to finish (resources fully free) or the
+        deadline before declaring it dead. Non-graceful keeps the old
+        immediate-death behavior."""
+        node_id = bytes(p["node_id"])
+        if not p.get("graceful"):
+            await self._on_node_death(node_id)
+            return True
+        n = self.nodes.get(node_id)
+        if n is None or not n.alive:
+            return False
+        n.draining = True
+        deadline = time.time() + float(p.get("deadline_s", 30.0))
+
+        async def _watch():
+            while time.time() < deadline:
+                cur = self.nodes.get(node_id)
+                if cur is None or not cur.alive:
+                    return
+                busy = any(
+                    cur.resources_available.get(k, 0) + 1e-9 < v
+                    for k, v in cur.resources_total.items())
+                if not busy and getattr(cur, "pending", 0) == 0:
+                    break
+                await asyncio.sleep(0.1)
+            await self._on_node_death(node_id)
+
+        asyncio.ensure_future(_watch())
         return True
 
     async def _on_node_death(self, node_id: bytes):
@@ -325,7 +356,9 @@ class GcsServer:
 
     # ---------- scheduling helpers ----------
     def _alive_nodes(self) -> List[NodeInfo]:
-        return [n for n in self.nodes.values() if n.alive]
+        # draining nodes take no NEW placements (existing work drains)
+        return [n for n in self.nodes.values()
+                if n.alive and not n.draining]
 
     def _fits(self, node: NodeInfo, req: Dict[str, float]) -> bool:
         for k, v in req.items():

@@ -599,6 +599,7 @@ def nodes() -> List[dict]:
             {
                 "NodeID": bytes(n["node_id"]).hex(),
                 "Alive": n["alive"],
+                "Draining": n.get("draining", False),
                 "Resources": n["resources_total"],
                 "Available": n["resources_available"],
                 "Labels": n.get("labels", {}),
@@ -606,6 +607,19 @@ def nodes() -> List[dict]:
             }
         )
     return out
+
+
+def drain_node(node_id: str, graceful: bool = True,
+               deadline_s: float = 30.0) -> bool:
+    """Drain a node (reference: DrainNode RPC / autoscaler scale-down).
+    Graceful: the scheduler stops placing work on it immediately, and
+    the node is declared dead once its leases finish or the deadline
+    passes. graceful=False kills it from the cluster view now."""
+    rt = _rt.global_runtime()
+    return bool(rt.gcs_call("drain_node", {
+        "node_id": bytes.fromhex(node_id), "graceful": graceful,
+        "deadline_s": deadline_s,
+    }))
 
 
 def cluster_resources() -> Dict[str, float]:

@@ -192,3 +192,41 @@ ray.shutdown(_exiting_interpreter=True)
     finally:
         ray.shutdown()
         config.reload()
+
+
+def test_graceful_node_drain(two_node_cluster):
+    """Multi-phase drain (reference: DrainNode): a DRAINING node takes
+    no new placements, finishes running work, then leaves the cluster."""
+    cluster = two_node_cluster
+    node_b_id = cluster.nodes[1].node_id.hex()
+
+    @ray.remote
+    def slow_on_b():
+        time.sleep(1.5)
+        return "finished"
+
+    pin = NodeAffinitySchedulingStrategy(node_b_id, soft=False)
+    running = slow_on_b.options(scheduling_strategy=pin).remote()
+    time.sleep(0.4)  # lease granted on b
+
+    assert ray.drain_node(node_b_id, graceful=True, deadline_s=30)
+    n = next(x for x in ray.nodes() if x["NodeID"] == node_b_id)
+    assert n["Draining"] and n["Alive"]
+    # the in-flight task still completes (drain waited for it)
+    assert ray.get(running, timeout=30) == "finished"
+    # node then leaves the cluster view; unpinned work keeps running
+    deadline = time.time() + 30
+    alive = [True]
+    while time.time() < deadline:
+        alive = [x for x in ray.nodes()
+                 if x["NodeID"] == node_b_id and x["Alive"]]
+        if not alive:
+            break
+        time.sleep(0.2)
+    assert not alive
+
+    @ray.remote
+    def on_head():
+        return "still scheduling"
+
+    assert ray.get(on_head.remote(), timeout=30) == "still scheduling"
