@@ -7,6 +7,8 @@ concurrency), with a windowed streaming iterator for consumption
 in-flight window provides the same backpressure effect for the v1
 pipeline).
 """
+from . import expressions  # noqa: F401
+from .expressions import col, lit  # noqa: F401
 from .dataset import (  # noqa: F401
     ActorPoolStrategy,
     Dataset,

@@ -353,6 +353,16 @@ class Dataset:
         return self._with_op(("flat_map", fn))
 
     def filter(self, fn=None, *, expr=None) -> "Dataset":
+        from .expressions import Expr
+
+        if fn is None and isinstance(expr, Expr):
+            # vectorized Expr filter (reference: data/expressions.py):
+            # evaluate the mask per batch, not a python UDF per row
+            def keep(batch, _e=expr):
+                mask = np.asarray(_e.eval(batch), dtype=bool)
+                return {k: np.asarray(v)[mask] for k, v in batch.items()}
+
+            return self._with_op(("map_batches", keep, "numpy", None))
         if fn is None and expr is not None:
             code = compile(expr, "<filter_expr>", "eval")
 
@@ -362,12 +372,26 @@ class Dataset:
         return self._with_op(("filter", fn))
 
     def add_column(self, name: str, fn) -> "Dataset":
+        from .expressions import Expr
+
         def add(batch):
             batch = dict(batch)
-            batch[name] = fn(batch)
+            batch[name] = fn.eval(batch) if isinstance(fn, Expr) \
+                else fn(batch)
             return batch
 
         return self._with_op(("map_batches", add, "numpy", None))
+
+    def with_column(self, name: str, expr) -> "Dataset":
+        """Expression-based column (reference: Dataset.with_columns +
+        data/expressions.py)."""
+        return self.add_column(name, expr)
+
+    def with_columns(self, exprs: Dict[str, "Any"]) -> "Dataset":
+        ds = self
+        for name, e in exprs.items():
+            ds = ds.add_column(name, e)
+        return ds
 
     def drop_columns(self, cols: List[str]) -> "Dataset":
         def drop(t: pa.Table):

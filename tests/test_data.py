@@ -376,3 +376,27 @@ def test_datasource_datasink_plugins(ray_start_regular):
 
     ds.write_datasink(ListSink())
     assert sum(collected) == 5
+
+
+def test_expressions_filter_and_with_columns(ray_start_regular):
+    """Expression API (reference: data/expressions.py col()/lit()):
+    vectorized filter masks and computed columns."""
+    from ray_amd.data.expressions import col, lit
+
+    ds = rd.from_items([{"x": i, "y": i * 2} for i in range(20)],
+                       parallelism=4)
+    out = ds.filter(expr=(col("x") > 5) & (col("y") < 30)).take_all()
+    assert [r["x"] for r in out] == [6, 7, 8, 9, 10, 11, 12, 13, 14]
+
+    ds2 = ds.with_columns({
+        "z": col("x") + col("y") * lit(10),
+        "neg": -col("x"),
+    })
+    rows = ds2.take(3)
+    assert rows[1]["z"] == 1 + 2 * 10 and rows[1]["neg"] == -1
+
+    # is_in + invert
+    out = ds.filter(expr=~col("x").is_in([0, 1, 2, 3, 4, 5, 6, 7, 8, 9,
+                                          10, 11, 12, 13, 14, 15, 16,
+                                          17])).take_all()
+    assert sorted(r["x"] for r in out) == [18, 19]
