@@ -338,3 +338,55 @@ def test_hf_transformers_trainer_integration(ray_start_regular, tmp_path):
     result = t.fit()
     assert result.error is None, result.error
     assert result.metrics and "loss" in result.metrics or "step" in (result.metrics or {})
+
+
+def test_sklearn_trainer(ray_start_regular, tmp_path):
+    """SklearnTrainer fits a real estimator in a ray worker and
+    checkpoints it (reference: train/sklearn/sklearn_trainer.py)."""
+    import numpy as np
+    from sklearn.ensemble import HistGradientBoostingClassifier
+
+    import ray_amd.data as rd
+    from ray_amd.train import RunConfig
+    from ray_amd.train.gbdt import SklearnTrainer
+
+    rng = np.random.default_rng(0)
+    rows = [{"a": float(x), "b": float(y),
+             "label": int(x + y > 1.0)}
+            for x, y in rng.random((200, 2))]
+    ds = rd.from_items(rows)
+    trainer = SklearnTrainer(
+        estimator=HistGradientBoostingClassifier(max_iter=20),
+        datasets={"train": ds, "valid": ds},
+        label_column="label",
+        run_config=RunConfig(name="skl", storage_path=str(tmp_path)),
+    )
+    res = trainer.fit()
+    assert res.error is None, res.error
+    assert res.metrics["train_score"] > 0.9
+    assert "valid_score" in res.metrics
+    model = SklearnTrainer.get_model(res.checkpoint)
+    import pandas as pd
+
+    pred = model.predict(pd.DataFrame([{"a": 0.9, "b": 0.9}]))
+    assert pred[0] == 1
+
+
+def test_xgboost_trainer_missing_lib(ray_start_regular, tmp_path):
+    import sys
+
+    if "xgboost" in sys.modules:
+        pytest.skip("xgboost installed")
+    import ray_amd.data as rd
+    from ray_amd.train import RunConfig
+    from ray_amd.train.gbdt import XGBoostTrainer
+
+    t = XGBoostTrainer(
+        params={"objective": "binary:logistic"},
+        datasets={"train": rd.from_items([{"x": 1.0, "label": 0}])},
+        label_column="label",
+        run_config=RunConfig(name="xgb", storage_path=str(tmp_path)),
+    )
+    res = t.fit()
+    assert res.error is not None
+    assert "xgboost" in str(res.error)
