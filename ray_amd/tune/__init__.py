@@ -8,6 +8,7 @@ from .impl import (  # noqa: F401
     ASHAScheduler,
     Callback,
     FIFOScheduler,
+    HyperBandScheduler,
     MedianStoppingRule,
     PB2,
     PopulationBasedTraining,

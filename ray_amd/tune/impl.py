@@ -200,6 +200,59 @@ class ASHAScheduler:
 # ---------------- config ----------------
 
 
+class HyperBandScheduler:
+    """HyperBand (reference: tune/schedulers/hyperband.py): trials are
+    assigned round-robin to brackets with geometrically increasing
+    grace periods (r, r*eta, ...); within a bracket, successive
+    halving keeps the top 1/eta at each rung. The bracket diversity is
+    what distinguishes this from plain ASHA — aggressive brackets
+    stop early, conservative ones let slow starters run."""
+
+    def __init__(self, *, metric=None, mode: str = "max",
+                 max_t: int = 81, reduction_factor: float = 3,
+                 time_attr="training_iteration"):
+        self.metric = metric
+        self.mode = mode
+        self.max_t = max_t
+        self.eta = reduction_factor
+        # bracket i has grace period eta^i (capped below max_t)
+        self.n_brackets = max(1, int(math.log(max_t, reduction_factor)))
+        self._bracket_of: Dict[str, int] = {}
+        self._next_bracket = 0
+        # (bracket, rung) -> recorded values
+        self._rungs: Dict[tuple, List[float]] = {}
+
+    def _bracket(self, trial_id: str) -> int:
+        b = self._bracket_of.get(trial_id)
+        if b is None:
+            b = self._next_bracket % self.n_brackets
+            self._next_bracket += 1
+            self._bracket_of[trial_id] = b
+        return b
+
+    def on_trial_result(self, trial_id, iteration, metric_value) -> str:
+        if metric_value is None:
+            return "CONTINUE"
+        if iteration >= self.max_t:
+            return "STOP"
+        v = float(metric_value) if self.mode == "max" else -float(
+            metric_value)
+        b = self._bracket(trial_id)
+        grace = self.eta ** b
+        rung = None
+        r = grace
+        while r <= iteration and r < self.max_t:
+            rung = r
+            r *= self.eta
+        if rung is None or iteration != int(rung):
+            return "CONTINUE"
+        rec = self._rungs.setdefault((b, int(rung)), [])
+        rec.append(v)
+        k = max(1, int(len(rec) / self.eta))
+        cutoff = sorted(rec, reverse=True)[k - 1]
+        return "CONTINUE" if v >= cutoff else "STOP"
+
+
 class MedianStoppingRule:
     """Stop trials whose running mean falls below the median of all
     trials' running means at the same iteration (reference:

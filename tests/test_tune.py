@@ -398,3 +398,33 @@ def test_optuna_missing_raises():
 
     with pytest.raises(ImportError, match="optuna"):
         OptunaSearch()
+
+
+def test_hyperband_scheduler(ray_start_regular, tmp_path):
+    """HyperBand brackets (reference: schedulers/hyperband.py): bad
+    trials stop at rung boundaries, good ones run to max_t; brackets
+    get different grace periods."""
+    from ray_amd.train import RunConfig
+    from ray_amd.tune import HyperBandScheduler
+
+    def trainable(config):
+        for i in range(9):
+            tune.report({"score": config["x"] + i * 0.01})
+
+    sched = HyperBandScheduler(metric="score", mode="max", max_t=9,
+                               reduction_factor=3)
+    tuner = Tuner(
+        trainable,
+        param_space={"x": tune.grid_search([0.0, 1.0, 2.0, 3.0, 4.0,
+                                            5.0])},
+        tune_config=TuneConfig(metric="score", mode="max",
+                               scheduler=sched,
+                               max_concurrent_trials=3),
+        run_config=RunConfig(name="hb", storage_path=str(tmp_path)),
+    )
+    grid = tuner.fit()
+    assert len(grid) == 6
+    best = grid.get_best_result()
+    assert best.metrics["config/x"] == 5.0
+    # brackets were assigned round-robin with distinct grace periods
+    assert len(set(sched._bracket_of.values())) > 1
