@@ -57,10 +57,15 @@ extern "C" __global__ __launch_bounds__(256) void swiglu_bwd_bf16(
 // cos/sin: [T, D/2] fp32 precomputed on host (guide §B: no on-device
 // trig), pos_of_row = (row / n_heads) % T.
 // Pairing: (x[d], x[d+D/2]) rotated by angle theta_d (LLaMA convention).
+// in_rs: input (b,t)-row stride in elements — n_heads*D when x is
+// contiguous, larger when x is a no-copy slice of a fused-QKV GEMM
+// output (the [H,D] tail of each row stays contiguous). y is always
+// written contiguous, so RoPE doubles as the gather.
 extern "C" __global__ __launch_bounds__(256) void rope_fwd_bf16(
     const short* __restrict__ x, short* __restrict__ y,
     const float* __restrict__ cosT, const float* __restrict__ sinT,
-    long long rows, int D, int n_heads, int T, int sign) {
+    long long rows, int D, int n_heads, int T, int sign,
+    long long in_rs) {
   int half = D / 2;
   long long total = rows * half;
   long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
@@ -68,14 +73,17 @@ extern "C" __global__ __launch_bounds__(256) void rope_fwd_bf16(
   for (; i < total; i += stride) {
     long long row = i / half;
     int d = (int)(i - row * half);
-    int t = (int)((row / n_heads) % T);
+    long long sr = row / n_heads;           // (b,t) super-row
+    int h = (int)(row - sr * n_heads);
+    int t = (int)(sr % T);
     float c = cosT[(long long)t * half + d];
     float s = sinT[(long long)t * half + d] * (float)sign;
-    long long b0 = row * D + d;
-    float x1 = bf2f(x[b0]);
-    float x2 = bf2f(x[b0 + half]);
-    y[b0] = f2bf(x1 * c - x2 * s);
-    y[b0 + half] = f2bf(x2 * c + x1 * s);
+    long long src = sr * in_rs + (long long)h * D + d;
+    long long dst = row * D + d;
+    float x1 = bf2f(x[src]);
+    float x2 = bf2f(x[src + half]);
+    y[dst] = f2bf(x1 * c - x2 * s);
+    y[dst + half] = f2bf(x2 * c + x1 * s);
   }
 }
 

@@ -98,16 +98,29 @@ std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor a, at::Tensor b) {
 
 at::Tensor rope_apply(at::Tensor x, at::Tensor cosT, at::Tensor sinT,
                       int64_t n_heads, int64_t T, int64_t sign) {
-  CHECK_IN(x); CHECK_IN(cosT); CHECK_IN(sinT);
+  CHECK_IN(cosT); CHECK_IN(sinT);
   int D = (int)x.size(-1);
   long long rows = x.numel() / D;
-  auto y = at::empty_like(x);
+  // accept a no-copy [B,T,H,D] slice of a fused-QKV row (contiguous
+  // [H,D] tail, arbitrary (b,t)-row stride); anything else must be
+  // contiguous
+  long long in_rs = (long long)n_heads * D;
+  if (!x.is_contiguous()) {
+    TORCH_CHECK(x.dim() == 4 && x.stride(3) == 1 && x.stride(2) == D
+                    && x.stride(0) == x.size(1) * x.stride(1),
+                "rope: input must be contiguous or a fused-QKV slice");
+    in_rs = (long long)x.stride(1);
+  }
+  auto y = x.is_contiguous()
+               ? at::empty_like(x)
+               : at::empty({x.size(0), x.size(1), x.size(2), x.size(3)},
+                           x.options());
   long long total = rows * (D / 2);
   hipLaunchKernelGGL(rope_fwd_bf16, dim3(grid_for(total)), dim3(256), 0,
                      cur_stream(), (const short*)x.data_ptr(),
                      (short*)y.data_ptr(), cosT.data_ptr<float>(),
                      sinT.data_ptr<float>(), rows, D, (int)n_heads, (int)T,
-                     (int)sign);
+                     (int)sign, in_rs);
   return y;
 }
 

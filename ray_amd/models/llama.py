@@ -56,25 +56,51 @@ class Attention(nn.Module):
         self.cfg = cfg
         self.head_dim = cfg.hidden_size // cfg.num_heads
         self.n_rep = cfg.num_heads // cfg.num_kv_heads
-        self.q_proj = nn.Linear(
-            cfg.hidden_size, cfg.num_heads * self.head_dim, bias=False, dtype=dtype
-        )
-        self.k_proj = nn.Linear(
-            cfg.hidden_size, cfg.num_kv_heads * self.head_dim, bias=False, dtype=dtype
-        )
-        self.v_proj = nn.Linear(
-            cfg.hidden_size, cfg.num_kv_heads * self.head_dim, bias=False, dtype=dtype
-        )
+        # Fused QKV projection (one GEMM instead of three: x streams
+        # from HBM once, and hipBLASLt's dgrad/wgrad run ~10% faster at
+        # the fatter N — measured in tools/gemm_fuse_ab.py). The q/k/v
+        # splits feed RoPE without a copy (strided-input rope kernel).
+        self._fused_qkv = not os.environ.get("RAY_AMD_NO_FUSED_QKV")
+        qd = cfg.num_heads * self.head_dim
+        kvd = cfg.num_kv_heads * self.head_dim
+        if self._fused_qkv:
+            self.qkv_proj = nn.Linear(
+                cfg.hidden_size, qd + 2 * kvd, bias=False, dtype=dtype
+            )
+        else:
+            self.q_proj = nn.Linear(
+                cfg.hidden_size, qd, bias=False, dtype=dtype)
+            self.k_proj = nn.Linear(
+                cfg.hidden_size, kvd, bias=False, dtype=dtype)
+            self.v_proj = nn.Linear(
+                cfg.hidden_size, kvd, bias=False, dtype=dtype)
         self.o_proj = nn.Linear(
             cfg.num_heads * self.head_dim, cfg.hidden_size, bias=False, dtype=dtype
         )
 
+    def qkv(self, x):
+        """[B,T,H] -> q [B,T,Hq,D], k/v [B,T,Hkv,D] (views when fused)."""
+        B, T, _ = x.shape
+        cfg = self.cfg
+        hd = self.head_dim
+        if self._fused_qkv:
+            qkv = self.qkv_proj(x)
+            q, k, v = qkv.split(
+                [cfg.num_heads * hd, cfg.num_kv_heads * hd,
+                 cfg.num_kv_heads * hd], dim=-1)
+            return (q.view(B, T, cfg.num_heads, hd),
+                    k.view(B, T, cfg.num_kv_heads, hd),
+                    v.view(B, T, cfg.num_kv_heads, hd))
+        return (self.q_proj(x).view(B, T, cfg.num_heads, hd),
+                self.k_proj(x).view(B, T, cfg.num_kv_heads, hd),
+                self.v_proj(x).view(B, T, cfg.num_kv_heads, hd))
+
     def forward(self, x, cosT, sinT, kv_cache=None, pos0: int = 0):
         B, T, H = x.shape
         cfg = self.cfg
-        q = self.q_proj(x).view(B, T, cfg.num_heads, self.head_dim)
-        k = self.k_proj(x).view(B, T, cfg.num_kv_heads, self.head_dim)
-        v = self.v_proj(x).view(B, T, cfg.num_kv_heads, self.head_dim)
+        q, k, v = self.qkv(x)
+        if self._fused_qkv:
+            v = v.contiguous()  # k/q gather happens inside rope
         q = ops.rope(q, cosT[pos0 : pos0 + T], sinT[pos0 : pos0 + T])
         k = ops.rope(k, cosT[pos0 : pos0 + T], sinT[pos0 : pos0 + T])
         q = q.transpose(1, 2)  # [B, Hq, T, D]
@@ -241,9 +267,7 @@ class GraphedDecoder:
         for li, layer in enumerate(m.layers):
             h = layer.attn_norm(x)
             at = layer.attn
-            q = at.q_proj(h).view(B, 1, cfg.num_heads, hd)
-            k = at.k_proj(h).view(B, 1, cfg.num_kv_heads, hd)
-            v = at.v_proj(h).view(B, 1, cfg.num_kv_heads, hd)
+            q, k, v = at.qkv(h)
             q = _rope_one(q, cos, sin)
             k = _rope_one(k, cos, sin)
             # static-address cache update
@@ -353,9 +377,7 @@ class BatchedDecoder:
         for li, layer in enumerate(m.layers):
             h = layer.attn_norm(x)
             at = layer.attn
-            q = at.q_proj(h).view(B, 1, cfg.num_heads, hd)
-            k = at.k_proj(h).view(B, 1, cfg.num_kv_heads, hd)
-            v = at.v_proj(h).view(B, 1, cfg.num_kv_heads, hd)
+            q, k, v = at.qkv(h)
             q = _rope_one(q, cos, sin)
             k = _rope_one(k, cos, sin)
             # per-row cache write at each slot's own position

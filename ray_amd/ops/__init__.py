@@ -137,7 +137,13 @@ class _Rope(torch.autograd.Function):
     def forward(ctx, x, cosT, sinT, n_heads, T):
         _require_ext("rope")
         ctx.meta = (cosT, sinT, n_heads, T)
-        return _K.rope_apply(x.contiguous(), cosT, sinT, n_heads, T, 1)
+        # fused-QKV slices ([H,D]-contiguous, strided (b,t) rows) go
+        # straight to the kernel — RoPE doubles as the gather
+        if not (x.dim() == 4 and x.stride(3) == 1
+                and x.stride(2) == x.size(3)
+                and x.stride(0) == x.size(1) * x.stride(1)):
+            x = x.contiguous()
+        return _K.rope_apply(x, cosT, sinT, n_heads, T, 1)
 
     @staticmethod
     def backward(ctx, dy):
