@@ -172,11 +172,17 @@ class GcsServer:
             elif op[0] == "kv_del":
                 _, ns, key = op
                 self.kv.get(ns, {}).pop(key, None)
+            elif op[0] == "actor_put":
+                snap.setdefault("actor_specs", {})[op[1]] = op[2]
+            elif op[0] == "actor_del":
+                snap.setdefault("actor_specs", {}).pop(op[1], None)
         # actors come back PENDING and are rescheduled once a raylet
         # re-registers (restart-based recovery)
         for aid, spec in snap.get("actor_specs", {}).items():
             a = ActorInfo(aid, spec)
             self.actors[aid] = a
+            if a.name:  # journaled actors re-register their name
+                self.named_actors.setdefault((a.namespace, a.name), aid)
             asyncio.ensure_future(self._schedule_actor(a))
 
     async def _persist_loop(self):
@@ -189,6 +195,16 @@ class GcsServer:
                 self._storage.save_snapshot(self._snapshot())
             except Exception:
                 pass
+
+    def _journal(self, *op):
+        if self._storage is None:
+            return
+        from .gcs_storage import encode_op
+
+        try:
+            self._storage.journal(encode_op(*op))
+        except Exception:
+            pass
 
     # ---------- KV ----------
     def h_kv_put(self, conn, p):
@@ -435,6 +451,7 @@ class GcsServer:
             self.named_actors[key] = actor_id
         self.actors[actor_id] = a
         self._dirty = True
+        self._journal("actor_put", actor_id, a.spec)
         asyncio.ensure_future(self._schedule_actor(a))
         return {"existing": None}
 
@@ -497,6 +514,7 @@ class GcsServer:
     async def _fail_actor(self, a: ActorInfo, cause: str):
         a.state = ACTOR_DEAD
         a.death_cause = cause
+        self._journal("actor_del", a.actor_id)
         for f in a.waiters:
             if not f.done():
                 f.set_result(None)
@@ -552,6 +570,7 @@ class GcsServer:
         else:
             a.state = ACTOR_DEAD
             a.death_cause = cause
+            self._journal("actor_del", a.actor_id)
             if a.name and self.named_actors.get((a.namespace, a.name)) == a.actor_id:
                 del self.named_actors[(a.namespace, a.name)]
             for f in a.waiters:

@@ -388,3 +388,62 @@ def test_gcs_sqlite_journal_survives_hard_kill(tmp_path, monkeypatch):
     finally:
         proc.kill()
         proc.wait(5)
+
+
+def test_gcs_sqlite_journal_actor_registration(tmp_path, monkeypatch):
+    """Actor registrations journal synchronously too: a named actor
+    registered right before kill -9 is rescheduled by the restarted
+    GCS (no snapshot flush needed)."""
+    import asyncio
+    import subprocess
+    import sys
+    import socket as _socket
+
+    monkeypatch.setenv("RAY_AMD_GCS_STORAGE", "sqlite")
+    sock = str(tmp_path / "gcs.sock")
+    persist = str(tmp_path / "gcs_state")
+    env = dict(os.environ)
+
+    def start():
+        p = subprocess.Popen(
+            [sys.executable, "-m", "ray_amd._core.gcs", sock, persist],
+            start_new_session=True, env=env,
+        )
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            if os.path.exists(sock):
+                try:
+                    s = _socket.socket(_socket.AF_UNIX)
+                    s.connect(sock)
+                    s.close()
+                    return p
+                except OSError:
+                    pass
+            time.sleep(0.05)
+        raise RuntimeError("gcs did not start")
+
+    from ray_amd._core.protocol import RpcClient
+
+    async def rpc(addr, method, payload):
+        c = RpcClient()
+        await c.connect("unix:" + addr)
+        r = await c.call(method, payload)
+        c.close()
+        return r
+
+    proc = start()
+    try:
+        asyncio.run(rpc(sock, "register_actor",
+                        {"actor_id": b"A" * 8, "name": "journaled_actor",
+                         "namespace": "default",
+                         "resources": {"CPU": 0}, "payload": b"x"}))
+        proc.kill()
+        proc.wait(5)
+        os.unlink(sock)
+        proc = start()
+        actors = asyncio.run(rpc(sock, "list_actors", {}))
+        names = [a.get("name") for a in actors]
+        assert "journaled_actor" in names, actors
+    finally:
+        proc.kill()
+        proc.wait(5)
