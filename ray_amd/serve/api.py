@@ -61,13 +61,21 @@ class Deployment:
     def __init__(self, func_or_class, name: str, *, num_replicas=1,
                  max_ongoing_requests=100, ray_actor_options=None,
                  user_config=None, autoscaling_config=None,
-                 health_check_period_s=10.0, **kwargs):
+                 health_check_period_s=10.0,
+                 placement_group_bundles=None,
+                 placement_group_strategy="PACK", **kwargs):
         self.func_or_class = func_or_class
         self.name = name
         self.num_replicas = num_replicas
         self.max_ongoing_requests = max_ongoing_requests
         self.ray_actor_options = ray_actor_options or {}
         self.user_config = user_config
+        # gang placement per replica (reference: serve deployment
+        # placement_group_bundles/placement_group_strategy): each
+        # replica reserves its bundles all-or-nothing and runs in
+        # bundle 0
+        self.placement_group_bundles = placement_group_bundles
+        self.placement_group_strategy = placement_group_strategy
         if isinstance(autoscaling_config, dict):
             autoscaling_config = AutoscalingConfig(**autoscaling_config)
         self.autoscaling_config = autoscaling_config
@@ -79,6 +87,8 @@ class Deployment:
             ray_actor_options=self.ray_actor_options,
             user_config=self.user_config,
             autoscaling_config=self.autoscaling_config,
+            placement_group_bundles=self.placement_group_bundles,
+            placement_group_strategy=self.placement_group_strategy,
         )
         name = kwargs.pop("name", self.name)
         merged.update(kwargs)
@@ -123,7 +133,9 @@ class Application:
 
 def deployment(_func_or_class=None, *, name=None, num_replicas=1,
                max_ongoing_requests=100, ray_actor_options=None,
-               user_config=None, autoscaling_config=None, **kwargs):
+               user_config=None, autoscaling_config=None,
+               placement_group_bundles=None,
+               placement_group_strategy="PACK", **kwargs):
     """@serve.deployment decorator (reference: serve/api.py:522)."""
 
     def make(fc):
@@ -138,6 +150,8 @@ def deployment(_func_or_class=None, *, name=None, num_replicas=1,
             fc, n, num_replicas=nr, max_ongoing_requests=max_ongoing_requests,
             ray_actor_options=ray_actor_options, user_config=user_config,
             autoscaling_config=asc,
+            placement_group_bundles=placement_group_bundles,
+            placement_group_strategy=placement_group_strategy,
         )
 
     if _func_or_class is not None:
@@ -400,6 +414,26 @@ class ServeController:
         opts = dict(spec.get("ray_actor_options") or {})
         opts.setdefault("num_cpus", 0.1)
         opts["max_concurrency"] = max(16, spec.get("max_ongoing_requests", 100))
+        if spec.get("pg_bundles"):
+            # gang placement (reference: deployment
+            # placement_group_bundles): reserve this replica's bundles
+            # all-or-nothing, run the replica actor in bundle 0
+            from ray_amd.util.placement_group import placement_group
+            from ray_amd.util.scheduling_strategies import (
+                PlacementGroupSchedulingStrategy,
+            )
+
+            pg = placement_group(spec["pg_bundles"],
+                                 strategy=spec.get("pg_strategy", "PACK"))
+            ray.get(pg.ready(), timeout=120)
+            self._replica_pgs = getattr(self, "_replica_pgs", {})
+            opts["scheduling_strategy"] = PlacementGroupSchedulingStrategy(
+                placement_group=pg, placement_group_bundle_index=0)
+            RA = ray.remote(ReplicaActor)
+            replica = RA.options(**opts).remote(
+                cls_or_fn, init_args, init_kwargs, spec.get("user_config"))
+            self._replica_pgs[replica] = pg
+            return replica
         # Deployment scheduler (reference: deployment_scheduler.py
         # spread): round-robin replicas across alive nodes via soft
         # node affinity so one node's failure doesn't take every
@@ -424,6 +458,18 @@ class ServeController:
             cls_or_fn, init_args, init_kwargs, spec.get("user_config")
         )
 
+    def _release_replica_pg(self, r):
+        pg = getattr(self, "_replica_pgs", {}).pop(r, None)
+        if pg is not None:
+            try:
+                from ray_amd.util.placement_group import (
+                    remove_placement_group,
+                )
+
+                remove_placement_group(pg)
+            except Exception:
+                pass
+
     def _teardown(self, app: dict):
         ray = _ray()
         for d in app["deployments"].values():
@@ -432,6 +478,7 @@ class ServeController:
                     ray.kill(r)
                 except Exception:
                     pass
+                self._release_replica_pg(r)
 
     def get_routing(self, app_name: str, deployment: Optional[str] = None):
         app = self.apps.get(app_name)
@@ -545,6 +592,7 @@ class ServeController:
                     ray.kill(r)
                 except Exception:
                     pass
+                self._release_replica_pg(r)
             else:
                 still.append(r)
         d["draining"] = still
@@ -1196,6 +1244,8 @@ def run(app: Application, *, name: str = "default", route_prefix: str = "/",
                 "ray_actor_options": d.ray_actor_options,
                 "user_config": d.user_config,
                 "autoscaling": asc,
+                "pg_bundles": d.placement_group_bundles,
+                "pg_strategy": d.placement_group_strategy,
                 "payload": cloudpickle.dumps(
                     (d.func_or_class, args, kwargs)
                 ),

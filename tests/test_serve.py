@@ -484,3 +484,24 @@ def test_per_node_proxy_topology():
         serve.shutdown()
     finally:
         cluster.shutdown()
+
+
+def test_deployment_placement_group_bundles(serve_session):
+    """Gang placement (reference: deployment placement_group_bundles):
+    each replica reserves its bundles all-or-nothing and runs in
+    bundle 0; teardown releases the groups."""
+    from ray_amd.util.placement_group import placement_group_table
+
+    @serve.deployment(num_replicas=2,
+                      placement_group_bundles=[{"CPU": 0.5}, {"CPU": 0.5}],
+                      placement_group_strategy="PACK")
+    class Gang:
+        def __call__(self, _):
+            return "gang"
+
+    h = serve.run(Gang.bind(), http=False)
+    assert h.remote(None).result(timeout_s=60) == "gang"
+    created = [g for g in placement_group_table().values()
+               if g["state"] == "CREATED"]
+    assert len(created) >= 2  # one group per replica
+    serve.shutdown()
