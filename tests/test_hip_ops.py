@@ -416,3 +416,27 @@ def test_chunked_lm_head_ce_matches_plain():
         return ((a.float() - b.float()).norm() / (b.float().norm() + 1e-6)).item()
     assert relerr(x1.grad, x2.grad) < 0.03
     assert relerr(w1.grad, w2.grad) < 0.03
+
+
+@GPU
+def test_rope_strided_fused_qkv_slice():
+    """The rope kernel accepts a no-copy slice of a fused-QKV GEMM
+    output (arbitrary (b,t)-row stride, contiguous [H,D] tail) and
+    matches the contiguous result."""
+    dev = _cuda()
+    torch.manual_seed(1)
+    B, T, Hq, Hkv, D = 2, 16, 4, 2, 64
+    cosT, sinT = ops.rope_tables(T, D, device=dev)
+    fused = torch.randn(B, T, (Hq + 2 * Hkv) * D, device=dev,
+                        dtype=torch.bfloat16)
+    q = fused[..., : Hq * D].view(B, T, Hq, D)           # strided slice
+    k = fused[..., Hq * D: (Hq + Hkv) * D].view(B, T, Hkv, D)
+    assert not q.is_contiguous() and not k.is_contiguous()
+    yq = ops.rope(q, cosT, sinT)
+    yk = ops.rope(k, cosT, sinT)
+    assert yq.is_contiguous() and yk.is_contiguous()
+    assert torch.allclose(yq.float(),
+                          ops.rope(q.contiguous(), cosT, sinT).float())
+    assert torch.allclose(yk.float(),
+                          ops.rope_ref(k.contiguous(), cosT, sinT).float(),
+                          atol=2e-2, rtol=2e-2)
