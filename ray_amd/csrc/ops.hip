@@ -17,6 +17,7 @@
 #include "hip/fa_bwd_v4.hip"
 #include "hip/fa_bwd_dq_v4.hip"
 #include "hip/fa_bwd_dkv_v5.hip"
+#include "hip/gemv.hip"
 
 #define CHECK_IN(x)                                                     \
   TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                       \
@@ -121,6 +122,26 @@ at::Tensor rope_apply(at::Tensor x, at::Tensor cosT, at::Tensor sinT,
                      (short*)y.data_ptr(), cosT.data_ptr<float>(),
                      sinT.data_ptr<float>(), rows, D, (int)n_heads, (int)T,
                      (int)sign, in_rs);
+  return y;
+}
+
+// ---------------- small-batch GEMV (decode linear) ----------------
+
+at::Tensor gemv(at::Tensor x, at::Tensor w) {
+  // x: [B, K] (B <= 8), w: [N, K] bf16 row-major -> y [B, N]
+  CHECK_IN(x); CHECK_IN(w);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+              w.scalar_type() == at::kBFloat16, "gemv: bf16 only");
+  int B = (int)x.size(0);
+  long long K = x.size(1);
+  int N = (int)w.size(0);
+  TORCH_CHECK(B >= 1 && B <= 8, "gemv: batch must be 1..8");
+  TORCH_CHECK(w.size(1) == K && K % 512 == 0, "gemv: K%512 != 0");
+  auto y = at::empty({B, N}, x.options());
+  int blocks = (int)std::min<long long>((N + 3) / 4, 4096);
+  hipLaunchKernelGGL(gemv_bf16, dim3(blocks), dim3(256), 0, cur_stream(),
+                     (const short*)w.data_ptr(), (const short*)x.data_ptr(),
+                     (short*)y.data_ptr(), N, B, K);
   return y;
 }
 
@@ -446,6 +467,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_bwd", &flash_attn_bwd);
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("gemv", &gemv);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);

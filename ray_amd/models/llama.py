@@ -78,22 +78,29 @@ class Attention(nn.Module):
             cfg.num_heads * self.head_dim, cfg.hidden_size, bias=False, dtype=dtype
         )
 
-    def qkv(self, x):
-        """[B,T,H] -> q [B,T,Hq,D], k/v [B,T,Hkv,D] (views when fused)."""
+    def qkv(self, x, lin=None):
+        """[B,T,H] -> q [B,T,Hq,D], k/v [B,T,Hkv,D] (views when fused).
+        `lin(x, weight)` overrides the matmul (decode paths pass the
+        GEMV fast path)."""
         B, T, _ = x.shape
         cfg = self.cfg
         hd = self.head_dim
         if self._fused_qkv:
-            qkv = self.qkv_proj(x)
+            qkv = (self.qkv_proj(x) if lin is None
+                   else lin(x, self.qkv_proj.weight))
             q, k, v = qkv.split(
                 [cfg.num_heads * hd, cfg.num_kv_heads * hd,
                  cfg.num_kv_heads * hd], dim=-1)
             return (q.view(B, T, cfg.num_heads, hd),
                     k.view(B, T, cfg.num_kv_heads, hd),
                     v.view(B, T, cfg.num_kv_heads, hd))
-        return (self.q_proj(x).view(B, T, cfg.num_heads, hd),
-                self.k_proj(x).view(B, T, cfg.num_kv_heads, hd),
-                self.v_proj(x).view(B, T, cfg.num_kv_heads, hd))
+        if lin is None:
+            return (self.q_proj(x).view(B, T, cfg.num_heads, hd),
+                    self.k_proj(x).view(B, T, cfg.num_kv_heads, hd),
+                    self.v_proj(x).view(B, T, cfg.num_kv_heads, hd))
+        return (lin(x, self.q_proj.weight).view(B, T, cfg.num_heads, hd),
+                lin(x, self.k_proj.weight).view(B, T, cfg.num_kv_heads, hd),
+                lin(x, self.v_proj.weight).view(B, T, cfg.num_kv_heads, hd))
 
     def forward(self, x, cosT, sinT, kv_cache=None, pos0: int = 0):
         B, T, H = x.shape
@@ -264,10 +271,11 @@ class GraphedDecoder:
         x = m.embed(self.in_tok)  # [B,1,H]
         cos = m.cosT.index_select(0, self.pos).view(1, 1, 1, hd // 2)
         sin = m.sinT.index_select(0, self.pos).view(1, 1, 1, hd // 2)
+        lin = ops.linear_sb  # GEMV fast path (hipBLASLt is 4x off at M<=8)
         for li, layer in enumerate(m.layers):
             h = layer.attn_norm(x)
             at = layer.attn
-            q, k, v = at.qkv(h)
+            q, k, v = at.qkv(h, lin=lin)
             q = _rope_one(q, cos, sin)
             k = _rope_one(k, cos, sin)
             # static-address cache update
@@ -284,11 +292,17 @@ class GraphedDecoder:
                 attn_mask=self.mask.to(q.dtype),
                 enable_gqa=True,
             )
-            x = x + at.o_proj(attn.permute(0, 2, 1, 3).reshape(B, 1, -1))
-            x = x + layer.mlp(layer.mlp_norm(x))
+            x = x + lin(attn.permute(0, 2, 1, 3).reshape(B, 1, -1),
+                        at.o_proj.weight)
+            hm = layer.mlp_norm(x)
+            mp = layer.mlp
+            x = x + lin(ops.swiglu(lin(hm, mp.gate_proj.weight),
+                                   lin(hm, mp.up_proj.weight)),
+                        mp.down_proj.weight)
         x = m.final_norm(x)
-        return m.lm_head(x)[:, 0]
+        return lin(x, m.lm_head.weight)[:, 0]
 
+    @torch.no_grad()  # grad mode gates the GEMV fast path (linear_sb)
     def capture(self):
         assert self.device.type == "cuda"
         s = torch.cuda.Stream()
@@ -374,10 +388,11 @@ class BatchedDecoder:
         x = m.embed(toks.view(B, 1))
         cos = m.cosT.index_select(0, pos).view(B, 1, 1, hd // 2)
         sin = m.sinT.index_select(0, pos).view(B, 1, 1, hd // 2)
+        lin = ops.linear_sb  # GEMV fast path for small decode batches
         for li, layer in enumerate(m.layers):
             h = layer.attn_norm(x)
             at = layer.attn
-            q, k, v = at.qkv(h)
+            q, k, v = at.qkv(h, lin=lin)
             q = _rope_one(q, cos, sin)
             k = _rope_one(k, cos, sin)
             # per-row cache write at each slot's own position
@@ -394,10 +409,15 @@ class BatchedDecoder:
                 attn_mask=self.mask.to(q.dtype),
                 enable_gqa=True,
             )
-            x = x + at.o_proj(attn.permute(0, 2, 1, 3).reshape(B, 1, -1))
-            x = x + layer.mlp(layer.mlp_norm(x))
+            x = x + lin(attn.permute(0, 2, 1, 3).reshape(B, 1, -1),
+                        at.o_proj.weight)
+            hm = layer.mlp_norm(x)
+            mp = layer.mlp
+            x = x + lin(ops.swiglu(lin(hm, mp.gate_proj.weight),
+                                   lin(hm, mp.up_proj.weight)),
+                        mp.down_proj.weight)
         x = m.final_norm(x)
-        return m.lm_head(x)[:, 0]
+        return lin(x, m.lm_head.weight)[:, 0]
 
 
 class _SlotCacheView:

@@ -160,6 +160,30 @@ def rope(x: torch.Tensor, cosT: torch.Tensor, sinT: torch.Tensor):
     return _Rope.apply(x, cosT[:T].contiguous(), sinT[:T].contiguous(), Hn, T)
 
 
+def linear_sb(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    """Small-batch linear for decode (csrc/hip/gemv.hip): one wave per
+    output row streams W once. Measured vs hipBLASLt
+    (tools/gemv_ab.py): wins 1.4-2x at rows<=2 with N<=8192 (the
+    qkv/o projections at decode batch 1-2); hipBLASLt already runs
+    near-bandwidth (5.7+ TB/s) at larger N/K, so those shapes fall
+    back. VALU-bound above rows=2 (each W element costs `rows` fmas)."""
+    K = x.shape[-1]
+    rows = x.numel() // K
+    if (
+        _K is not None
+        and x.is_cuda
+        and x.dtype == torch.bfloat16
+        and weight.dtype == torch.bfloat16
+        and rows <= 2
+        and weight.shape[0] <= 8192
+        and K % 512 == 0
+        and not torch.is_grad_enabled()
+    ):
+        y = _K.gemv(x.reshape(rows, K).contiguous(), weight)
+        return y.view(*x.shape[:-1], weight.shape[0])
+    return torch.nn.functional.linear(x, weight)
+
+
 # --------------------------------------------------------------------------
 # Cross entropy (fused, bf16 logits)
 # --------------------------------------------------------------------------

@@ -440,3 +440,17 @@ def test_rope_strided_fused_qkv_slice():
     assert torch.allclose(yk.float(),
                           ops.rope_ref(k.contiguous(), cosT, sinT).float(),
                           atol=2e-2, rtol=2e-2)
+
+
+@GPU
+def test_gemv_matches_linear():
+    """Decode GEMV kernel vs F.linear across the decode shapes."""
+    dev = _cuda()
+    torch.manual_seed(2)
+    for B, N, K in ((1, 4096, 4096), (4, 6144, 4096), (8, 4096, 14336),
+                    (1, 128256, 4096)):
+        x = torch.randn(B, K, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.02
+        y = ops.linear_sb(x, w)
+        yr = torch.nn.functional.linear(x.float(), w.float())
+        assert torch.allclose(y.float(), yr, atol=0.3, rtol=3e-2), (B, N, K)
