@@ -33,12 +33,14 @@ def cmd_start(args):
 def cmd_stop(args):
     from ray_amd._core.node import RAY_AMD_TMP
 
-    latest = os.path.join(RAY_AMD_TMP, "latest_session")
-    if not os.path.exists(latest):
-        print("no running session found")
-        return 0
-    with open(latest) as f:
-        session = f.read().strip()
+    session = getattr(args, "session", None)
+    if not session:
+        latest = os.path.join(RAY_AMD_TMP, "latest_session")
+        if not os.path.exists(latest):
+            print("no running session found")
+            return 0
+        with open(latest) as f:
+            session = f.read().strip()
     pid_file = os.path.join(session, "head_pids")
     if os.path.exists(pid_file):
         with open(pid_file) as f:
@@ -157,6 +159,64 @@ def main(argv=None):
         run_dashboard(port=args.port)
         return 0
     s.set_defaults(fn=_dash)
+
+    s = sub.add_parser("up")
+    s.add_argument("config", help="cluster YAML (local provider)")
+
+    def _up(args):
+        # reference: `ray up cluster.yaml` (autoscaler/commands.py) —
+        # local provider: start a head + worker raylets on this machine
+        import yaml
+
+        from ray_amd._core import node as _node
+
+        with open(args.config) as f:
+            cfg = yaml.safe_load(f) or {}
+        provider = (cfg.get("provider") or {}).get("type", "local")
+        if provider != "local":
+            print(f"provider {provider!r} needs cloud access; only "
+                  "'local' is supported offline", file=sys.stderr)
+            return 1
+        head = cfg.get("head_node") or {}
+        cluster = _node.start_local_cluster(
+            num_cpus=head.get("num_cpus"), num_gpus=head.get("num_gpus"),
+            resources=head.get("resources"),
+        )
+        pids = {"gcs": cluster.gcs_proc.pid,
+                "raylet": cluster.raylet_proc.pid}
+        wn = cfg.get("worker_nodes") or {}
+        n_workers = int(wn.get("count", cfg.get("max_workers", 0)))
+        for i in range(n_workers):
+            res = dict(wn.get("resources") or {})
+            res.setdefault("CPU", wn.get("num_cpus", 2))
+            res.setdefault("GPU", wn.get("num_gpus", 0))
+            res.setdefault("memory", 16 * 2**30)
+            proc, addr, node_id = _node.start_raylet(
+                cluster.session_dir, cluster.gcs_addr, res,
+                node_name=f"worker{i}",
+                labels=wn.get("labels"),
+            )
+            pids[f"worker{i}"] = proc.pid
+        with open(os.path.join(cluster.session_dir, "head_pids"),
+                  "w") as f:
+            json.dump(pids, f)
+        name = cfg.get("cluster_name", "local")
+        print(f"cluster {name!r} up: head + {n_workers} workers; "
+              f"session_dir={cluster.session_dir}")
+        print(f"connect with: ray_amd.init(address="
+              f"'{cluster.session_dir}')")
+        return 0
+
+    s.set_defaults(fn=_up)
+
+    s = sub.add_parser("down")
+    s.add_argument("session", nargs="?", default=None,
+                   help="session_dir (default: latest)")
+
+    def _down(args):
+        return cmd_stop(args)
+
+    s.set_defaults(fn=_down)
 
     s = sub.add_parser("stack")
     s.add_argument("--address", default=None)
