@@ -383,6 +383,37 @@ class ServeController:
         self.apps: Dict[str, dict] = {}
         self.routes: Dict[str, str] = {}  # route_prefix -> app
         self.version = 0
+        import threading
+
+        # reference: ServeController.run_control_loop — periodic
+        # autoscaling passes + proxy reconciliation (late-joining
+        # nodes get their per-node HTTP proxy without a redeploy)
+        self._stop_loop = threading.Event()
+        threading.Thread(target=self._control_loop, daemon=True,
+                         name="serve-control-loop").start()
+
+    def _control_loop(self):
+        while not self._stop_loop.wait(3.0):
+            try:
+                self.autoscale_once()
+            except Exception:
+                pass
+            try:
+                self._reconcile_proxies()
+            except Exception:
+                pass
+
+    def _reconcile_proxies(self):
+        if not self.routes:
+            return
+        ray = _ray()
+        try:  # only when HTTP was enabled (head proxy exists)
+            ray.get_actor(SERVE_PROXY_NAME, namespace=SERVE_NAMESPACE)
+        except ValueError:
+            return
+        alive = [n for n in ray.nodes() if n["Alive"]]
+        if len(alive) > 1:
+            _ensure_node_proxies(_http_port)
 
     def deploy_application(self, name: str, route_prefix: str,
                            specs: List[dict], ingress_name: str):

@@ -505,3 +505,34 @@ def test_deployment_placement_group_bundles(serve_session):
                if g["state"] == "CREATED"]
     assert len(created) >= 2  # one group per replica
     serve.shutdown()
+
+
+def test_proxy_reconcile_on_node_join():
+    """The controller's control loop starts a proxy on a node that
+    joins AFTER serve.run (reference: ProxyStateManager reconcile)."""
+    from ray_amd.cluster_utils import Cluster
+    from ray_amd.serve.api import proxy_ports
+
+    cluster = Cluster(head_node_args={"num_cpus": 4})
+    try:
+        cluster.connect()
+
+        @serve.deployment
+        class Hi:
+            def __call__(self, req):
+                return "hi"
+
+        serve.run(Hi.bind(), port=18433)  # single node at deploy time
+        cluster.add_node(num_cpus=2)
+        cluster.wait_for_nodes()
+        deadline = time.time() + 30
+        ports = {}
+        while time.time() < deadline:
+            ports = proxy_ports()
+            if len(ports) == 2:
+                break
+            time.sleep(0.5)
+        assert len(ports) == 2, ports
+        serve.shutdown()
+    finally:
+        cluster.shutdown()
