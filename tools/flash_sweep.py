@@ -47,5 +47,45 @@ def main():
               f"{t_sdpa / t_ours:5.2f}x {tf:8.1f}")
 
 
+
+def bwd_sweep():
+    torch.manual_seed(0)
+    D = 128
+    print(f"\nfwd+bwd: {'shape':26s} {'ours':>9s} {'SDPA':>9s} {'ratio':>6s}")
+    for B, Hq, Hkv, T in (
+        (16, 32, 8, 2048),
+        (8, 32, 8, 4096),
+        (2, 32, 8, 8192),
+        (8, 32, 32, 4096),
+    ):
+        def mk():
+            return (torch.randn(B, Hq, T, D, device="cuda",
+                                dtype=torch.bfloat16, requires_grad=True),
+                    torch.randn(B, Hkv, T, D, device="cuda",
+                                dtype=torch.bfloat16, requires_grad=True),
+                    torch.randn(B, Hkv, T, D, device="cuda",
+                                dtype=torch.bfloat16, requires_grad=True))
+        q, k, v = mk()
+        g = torch.randn(B, Hq, T, D, device="cuda", dtype=torch.bfloat16)
+
+        def ours():
+            out = ops.flash_attention(q, k, v, causal=True)
+            out.backward(g)
+            q.grad = k.grad = v.grad = None
+
+        def sdpa():
+            out = F.scaled_dot_product_attention(
+                q, k, v, is_causal=True, enable_gqa=True)
+            out.backward(g)
+            q.grad = k.grad = v.grad = None
+
+        t_ours = bench(ours, iters=15, warmup=5)
+        t_sdpa = bench(sdpa, iters=15, warmup=5)
+        name = f"B{B} Hq{Hq}/Hkv{Hkv} T{T}"
+        print(f"fwd+bwd: {name:26s} {t_ours * 1e3:8.2f}ms "
+              f"{t_sdpa * 1e3:8.2f}ms {t_sdpa / t_ours:5.2f}x")
+
+
 if __name__ == "__main__":
     main()
+    bwd_sweep()
