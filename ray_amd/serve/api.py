@@ -993,9 +993,23 @@ class ProxyActor:
                     "more_body": False})
 
     def _resolve_route(self, path):
+        # per-path TTL cache: a controller RPC per request would cap
+        # HTTP throughput (reference: proxies watch route tables via
+        # long-poll instead of resolving per request)
+        cache = getattr(self, "_route_cache", None)
+        if cache is None:
+            cache = self._route_cache = {}
+        hit = cache.get(path)
+        now = time.time()
+        if hit is not None and now - hit[1] < 2.0:
+            return hit[0]
         ray = _ray()
         ctrl = ray.get_actor(SERVE_CONTROLLER_NAME, namespace=SERVE_NAMESPACE)
-        return ray.get(ctrl.resolve_route.remote(path))
+        route = ray.get(ctrl.resolve_route.remote(path))
+        cache[path] = (route, now)
+        if len(cache) > 4096:
+            cache.clear()
+        return route
 
     def ping(self):
         return "pong"
