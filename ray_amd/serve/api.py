@@ -1330,7 +1330,16 @@ def get_deployment_handle(deployment_name: str, app_name: str = "default"
 def status():
     ray = _ray()
     ctrl = _ensure_controller()
-    return ray.get(ctrl.status.remote())
+    out = ray.get(ctrl.status.remote())
+    # reference serve.status(): proxies section (node -> state/port)
+    try:
+        out["proxies"] = {
+            nid: {"status": "HEALTHY", "port": port}
+            for nid, port in proxy_ports().items()
+        }
+    except Exception:
+        pass
+    return out
 
 
 def delete(name: str, _blocking: bool = True):
