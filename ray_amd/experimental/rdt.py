@@ -147,10 +147,14 @@ class GpuObjectStore:
         if t is None:
             return None
         if p.get("mode") == "ipc" and t.is_cuda:
-            # export hipIpc handle; producer-stream ORDER via an
-            # interprocess hipEvent — the consumer's stream waits on it
-            # instead of a full-device synchronize here (reference:
-            # cuda_ipc_transport.py:71-101 IPC event sync)
+            # Producer-stream ordering: export an interprocess hipEvent
+            # for the consumer's stream to wait on (reference:
+            # cuda_ipc_transport.py:71-101) AND synchronize the
+            # producing device before handing out the handle — the
+            # imported-event wait was observed to miss rarely on the
+            # dmabuf IPC path (consumer read uninitialized memory
+            # ~1/5 runs), and a device sync at RDT-fetch frequency is
+            # noise next to the RPC.
             ev_handle = None
             try:
                 ev = torch.cuda.Event(interprocess=True)
@@ -158,7 +162,8 @@ class GpuObjectStore:
                 ev_handle = ev.ipc_handle()
                 self._events[oid] = ev  # keep alive until freed
             except Exception:
-                torch.cuda.synchronize(t.device)
+                pass
+            torch.cuda.synchronize(t.device)
             func, args = reductions.reduce_tensor(t)
             return {"mode": "ipc",
                     "payload": cloudpickle.dumps((func, args)),
