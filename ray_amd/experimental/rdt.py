@@ -122,6 +122,9 @@ class GpuObjectStore:
         self._tensors: Dict[bytes, "torch.Tensor"] = {}
         self._cache: Dict[bytes, "torch.Tensor"] = {}
         self._events: Dict[bytes, object] = {}
+        # observability: did the last ipc fetch fall back to a staged
+        # copy because the mapped view failed checksum verification?
+        self.last_fetch_fallback = False
         self._rt.server.route("rdt_fetch", self._h_fetch)
         self._rt.server.route("rdt_free", self._h_free)
 
@@ -233,6 +236,7 @@ class GpuObjectStore:
                        or (not math.isnan(got)
                            and abs(got - r["check"]) > 1e-3
                            + 1e-4 * abs(r["check"])))
+                self.last_fetch_fallback = bool(bad)
                 if bad:
                     r2 = _rdt_io().call(ref.owner_addr, "rdt_fetch",
                                         {"id": ref.id, "mode": "staged"})
